@@ -1,0 +1,156 @@
+// CPU ORACLE — test infrastructure only (see field.hpp header).
+//
+// C API over the oracle, consumed exclusively by tests/ (ctypes) and by
+// bench.py's cpu_baseline leg.  All field elements cross this boundary as
+// 4 x u64 little-endian limbs; Montgomery form unless the name says
+// `canonical` — matching the reference's pinned limb layout
+// (plonk_proof_def.rs:22-52).
+#include <cstring>
+#include <vector>
+#include <omp.h>
+#include "field.hpp"
+#include "curve.hpp"
+#include "fq2.hpp"
+#include "ntt.hpp"
+#include "msm.hpp"
+#include "srs.hpp"
+#include "keccak.hpp"
+
+using namespace oracle;
+
+extern "C" {
+
+// ---- Fr/Fq ops (Montgomery-form in/out) ----
+#define FIELD_OPS(NAME, T)                                                      \
+    void orc_##NAME##_add(const u64* a, const u64* b, u64* out) {               \
+        T x, y; memcpy(x.l, a, 32); memcpy(y.l, b, 32);                         \
+        T r = x + y; memcpy(out, r.l, 32);                                      \
+    }                                                                           \
+    void orc_##NAME##_sub(const u64* a, const u64* b, u64* out) {               \
+        T x, y; memcpy(x.l, a, 32); memcpy(y.l, b, 32);                         \
+        T r = x - y; memcpy(out, r.l, 32);                                      \
+    }                                                                           \
+    void orc_##NAME##_mul(const u64* a, const u64* b, u64* out) {               \
+        T x, y; memcpy(x.l, a, 32); memcpy(y.l, b, 32);                         \
+        T r = x * y; memcpy(out, r.l, 32);                                      \
+    }                                                                           \
+    void orc_##NAME##_inv(const u64* a, u64* out) {                             \
+        T x; memcpy(x.l, a, 32);                                                \
+        T r = x.inverse(); memcpy(out, r.l, 32);                                \
+    }                                                                           \
+    void orc_##NAME##_from_canonical(const u64* a, u64* out) {                  \
+        T r = T::from_canonical(a); memcpy(out, r.l, 32);                       \
+    }                                                                           \
+    void orc_##NAME##_to_canonical(const u64* a, u64* out) {                    \
+        T x; memcpy(x.l, a, 32); x.to_canonical(out);                           \
+    }
+
+FIELD_OPS(fr, Fr)
+FIELD_OPS(fq, Fq)
+
+// ---- G1 (affine in/out: x, y Montgomery limbs + u64 infinity flag = 9 u64) ----
+static void store_affine(const G1Affine& p, u64* out) {
+    memcpy(out, p.x.l, 32);
+    memcpy(out + 4, p.y.l, 32);
+    out[8] = p.infinity ? 1 : 0;
+}
+static G1Affine load_affine(const u64* in) {
+    G1Affine p;
+    memcpy(p.x.l, in, 32);
+    memcpy(p.y.l, in + 4, 32);
+    p.infinity = in[8] != 0;
+    return p;
+}
+
+void orc_g1_generator(u64* out) { store_affine(G1Affine::generator(), out); }
+
+int orc_g1_is_on_curve(const u64* p) { return load_affine(p).is_on_curve() ? 1 : 0; }
+
+void orc_g1_add(const u64* a, const u64* b, u64* out) {
+    G1Proj r = G1Proj::from_affine(load_affine(a)).add_affine(load_affine(b));
+    store_affine(r.to_affine(), out);
+}
+
+void orc_g1_mul(const u64* p, const u64* scalar_canonical, u64* out) {
+    G1Proj r = G1Proj::from_affine(load_affine(p)).mul(scalar_canonical);
+    store_affine(r.to_affine(), out);
+}
+
+// ---- NTT (in-place over n Montgomery-form Fr elements) ----
+void orc_ntt(u64* data, u64 n, int inverse) {
+    Fr* a = reinterpret_cast<Fr*>(data);
+    if (inverse) ntt_inverse(a, n);
+    else ntt_forward(a, n);
+}
+
+void orc_coset_ntt(u64* data, u64 n, const u64* g_mont, int inverse) {
+    Fr* a = reinterpret_cast<Fr*>(data);
+    Fr g;
+    memcpy(g.l, g_mont, 32);
+    if (inverse) coset_ntt_inverse(a, n, g);
+    else coset_ntt_forward(a, n, g);
+}
+
+// ---- MSM ----
+// bases: n * 9 u64 (affine records as above); scalars: n * 4 u64 canonical.
+void orc_msm(const u64* bases, const u64* scalars, u64 n, u64* out, int window_c) {
+    std::vector<G1Affine> pts(n);
+    for (u64 i = 0; i < n; ++i) pts[i] = load_affine(bases + 9 * i);
+    G1Proj r = msm_pippenger(pts.data(), scalars, n, window_c > 0 ? window_c : 13);
+    store_affine(r.to_affine(), out);
+}
+
+void orc_msm_naive(const u64* bases, const u64* scalars, u64 n, u64* out) {
+    std::vector<G1Affine> pts(n);
+    for (u64 i = 0; i < n; ++i) pts[i] = load_affine(bases + 9 * i);
+    G1Proj r = msm_naive(pts.data(), scalars, n);
+    store_affine(r.to_affine(), out);
+}
+
+// ---- keccak256 ----
+void orc_keccak256(const uint8_t* data, u64 len, uint8_t* out32) {
+    keccak256(data, len, out32);
+}
+
+// ---- SRS ----
+// Returns required byte length for a ptau of `power`.
+u64 orc_srs_ptau_size(int power) {
+    size_t npoints = (size_t(1) << power) + 3;
+    return 12 + (12 + 44) + (12 + npoints * 64) + (12 + 256);
+}
+
+// Generate deterministic SRS and serialize to ptau bytes. Returns bytes written.
+u64 orc_srs_generate_ptau(int power, u64 seed, uint8_t* out, u64 cap) {
+    Srs srs = srs_generate(power, seed);
+    std::vector<uint8_t> bytes = srs_to_ptau(srs, power);
+    if (bytes.size() > cap) return 0;
+    memcpy(out, bytes.data(), bytes.size());
+    return bytes.size();
+}
+
+// Parse ptau bytes; writes (max_degree+1) G1 affine records; returns 0 on
+// success, negative on error.
+int orc_srs_parse(const uint8_t* bytes, u64 len, u64 max_degree, u64* g1_out,
+                  u64* h_out /*16 u64*/, u64* beta_h_out /*16 u64*/) {
+    try {
+        Srs srs = parse_ptau(bytes, len, max_degree);
+        for (size_t i = 0; i < srs.powers_of_g.size(); ++i)
+            store_affine(srs.powers_of_g[i], g1_out + 9 * i);
+        auto store_g2 = [](const G2Affine& g, u64* out) {
+            memcpy(out, g.x.c0.l, 32);
+            memcpy(out + 4, g.x.c1.l, 32);
+            memcpy(out + 8, g.y.c0.l, 32);
+            memcpy(out + 12, g.y.c1.l, 32);
+        };
+        store_g2(srs.h, h_out);
+        store_g2(srs.beta_h, beta_h_out);
+        return 0;
+    } catch (...) {
+        return -1;
+    }
+}
+
+int orc_num_threads() { return omp_get_max_threads(); }
+void orc_set_num_threads(int n) { omp_set_num_threads(n); }
+
+}  // extern "C"

@@ -1,0 +1,63 @@
+// CPU ORACLE — test infrastructure only (see field.hpp header).
+//
+// Restates: arkworks 0.4.2 ark-ec VariableBaseMSM (Pippenger) over BN254 G1
+// (SURVEY.md §8a row a5).  Scalars are canonical (non-Montgomery) 4 x u64
+// little-endian limbs, bases are affine Montgomery-form points — exactly the
+// shapes the KZG commit path hands to the extern MSM in the reference
+// (called from mpc-jellyfish via ark-ec; dep pinned Cargo.lock:933-1199).
+#pragma once
+#include <vector>
+#include <omp.h>
+#include "curve.hpp"
+
+namespace oracle {
+
+// Reference implementation: plain double-and-add sum (for small-n cross-checks).
+inline G1Proj msm_naive(const G1Affine* bases, const u64* scalars, size_t n) {
+    G1Proj acc = G1Proj::identity();
+    for (size_t i = 0; i < n; ++i) {
+        acc = acc.add(G1Proj::from_affine(bases[i]).mul(scalars + 4 * i));
+    }
+    return acc;
+}
+
+// Pippenger bucket method, window size c, unsigned digits, OpenMP over windows.
+inline G1Proj msm_pippenger(const G1Affine* bases, const u64* scalars, size_t n,
+                            int c = 13) {
+    const int num_windows = (256 + c - 1) / c;
+    const size_t num_buckets = (size_t(1) << c) - 1;  // digit 0 skipped
+    std::vector<G1Proj> window_sums(num_windows, G1Proj::identity());
+
+#pragma omp parallel for schedule(dynamic, 1)
+    for (int w = 0; w < num_windows; ++w) {
+        std::vector<G1Proj> buckets(num_buckets, G1Proj::identity());
+        const int bit0 = w * c;
+        for (size_t i = 0; i < n; ++i) {
+            const u64* s = scalars + 4 * i;
+            // extract c bits starting at bit0
+            u64 digit = 0;
+            int limb = bit0 / 64, off = bit0 % 64;
+            digit = s[limb] >> off;
+            if (off + c > 64 && limb + 1 < 4) digit |= s[limb + 1] << (64 - off);
+            digit &= (u64(1) << c) - 1;
+            if (digit != 0) buckets[digit - 1] = buckets[digit - 1].add_affine(bases[i]);
+        }
+        // running suffix sum: sum_d d * bucket[d]
+        G1Proj run = G1Proj::identity(), sum = G1Proj::identity();
+        for (size_t d = num_buckets; d >= 1; --d) {
+            run = run.add(buckets[d - 1]);
+            sum = sum.add(run);
+        }
+        window_sums[w] = sum;
+    }
+
+    // Horner over windows: acc = W_last; acc = acc*2^c + W_w
+    G1Proj acc = window_sums[num_windows - 1];
+    for (int w = num_windows - 2; w >= 0; --w) {
+        for (int k = 0; k < c; ++k) acc = acc.dbl();
+        acc = acc.add(window_sums[w]);
+    }
+    return acc;
+}
+
+}  // namespace oracle
