@@ -1,0 +1,159 @@
+/* renegade_amd — MI355X-native PlonK prover backend: C ABI.
+ *
+ * This is the drop-in boundary of the replacement prover.
+ * Each entry point mirrors an extern call the reference's circuits crate
+ * makes today (citations into /root/reference):
+ *
+ *   rng_prover_init      <- SRS load: parse_ptau_file semantics of
+ *                           crates/circuits/circuit-types/src/primitives/srs.rs:63-214
+ *                           (SYSTEM_SRS lazy static, srs.rs:30-42)
+ *   rng_preprocess       <- PlonkKzgSnark::preprocess(&SYSTEM_SRS, &cs),
+ *                           called at circuit-types/src/traits.rs:850
+ *   rng_prove            <- PlonkKzgSnark::prove_with_link_hint::<_,_,SolidityTranscript>,
+ *                           called at circuit-types/src/traits.rs:996
+ *   rng_link_proofs      <- PlonkKzgSnark::link_proofs::<SolidityTranscript>,
+ *                           called at circuits-core/src/zk_circuits/proof_linking/
+ *                           intent_and_balance.rs:66-73
+ *   rng_verify           <- PlonkKzgSnark::verify::<SolidityTranscript>,
+ *                           called at circuit-types/src/traits.rs:1012 (CPU pairing)
+ *   rng_msm_g1 / rng_ntt_fr <- the MSM/NTT primitives the north star names
+ *                           (arkworks ark-ec Pippenger / ark-poly radix-2 FFT,
+ *                           non-vendored deps)
+ *
+ * Conventions (pinned by the reference's rkyv shims,
+ * types-proofs/src/rkyv_impls/plonk_proof_def.rs):
+ *   - Fr/Fq elements: 4 x u64 little-endian Montgomery limbs (R = 2^256),
+ *     except scalars marked `canonical` (plain little-endian integers).
+ *   - G1 affine: x(4 u64), y(4 u64), infinity(u64 0/1)  => 9 u64 = 72 B.
+ *   - Proof buffer layout = the rkyv field order (plonk_proof_def.rs:197-222):
+ *     5 wire comms, 1 perm comm, 5 split-quotient comms, opening proof,
+ *     shifted opening proof (13 G1 affine records = 117 u64), then
+ *     evals: 5 wire evals, 4 sigma evals, 1 perm-next eval (10 Fr = 40 u64).
+ *     Total RNG_PROOF_U64S = 157 u64.
+ *   - Status codes map onto ProverError/PlonkError
+ *     (circuit-types/src/errors.rs): 0 ok, negative = error below.
+ *   - Thread-safety: context and pk handles are read-only after creation and
+ *     may be shared across threads (the reference proves from a rayon pool,
+ *     native_proof_manager.rs:143-148); each concurrent rng_prove call uses
+ *     its own HIP stream.
+ *
+ * A GPU box is REQUIRED for the compute entry points: they fail loudly
+ * (RNG_ERR_NO_GPU) rather than fall back to any CPU path.
+ */
+#ifndef RNG_PROVER_H
+#define RNG_PROVER_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define RNG_OK 0
+#define RNG_ERR_NO_GPU (-1)
+#define RNG_ERR_BAD_ARG (-2)
+#define RNG_ERR_HIP (-3)
+#define RNG_ERR_SRS (-4)          /* maps IoError from parse_ptau_file */
+#define RNG_ERR_UNSATISFIED (-5)  /* maps PlonkError::WrongProof / circuit unsat */
+#define RNG_ERR_VERIFY (-6)
+
+#define RNG_PROOF_G1S 13
+#define RNG_PROOF_EVALS 10
+#define RNG_PROOF_U64S (RNG_PROOF_G1S * 9 + RNG_PROOF_EVALS * 4) /* 157 */
+
+typedef struct RngCtx RngCtx;
+typedef struct RngProvingKey RngProvingKey;
+
+/* ---- lifecycle ---- */
+
+/* Parse a snarkjs-layout .ptau byte buffer (srs.rs:63-214 semantics) and
+ * upload the G1 powers to the GPU. max_degree + 1 G1 points are read. */
+RngCtx* rng_prover_init(const uint8_t* srs_ptau, size_t len, uint64_t max_degree);
+void rng_ctx_free(RngCtx* ctx);
+
+/* Library/device introspection */
+int rng_gpu_available(void);
+const char* rng_version(void);
+int rng_set_device(int device);
+
+/* Per-kernel HIP-event times (ms) of the LAST rng_msm_* / rng_ntt_* call on
+ * this thread, for roofline accounting: msm: [digits, sort, bucket_reduce,
+ * window_chunks, final]; ntt: [pass1(col/small), pass2(row)].  Returns number
+ * of entries written. */
+int rng_msm_last_times(double out_ms[5]);
+int rng_ntt_last_times(double out_ms[2]);
+
+/* ---- primitives (the two flagship kernels) ---- */
+
+/* In-place radix-2 NTT over BN254 Fr, natural order in/out,
+ * data = batch * n elements of 4 u64 Montgomery limbs (host buffer).
+ * n power of two, 2 <= n <= 2^26. */
+int rng_ntt_fr(RngCtx* ctx, uint64_t* data, uint64_t n, uint64_t batch, int inverse);
+
+/* MSM over BN254 G1: bases = n * 8 u64 (affine x,y Montgomery; no points at
+ * infinity), scalars = n * 4 u64 CANONICAL little-endian. out = 9 u64 affine
+ * record. window_c in [8,16], 0 = auto. */
+int rng_msm_g1(RngCtx* ctx, const uint64_t* bases, const uint64_t* scalars,
+               uint64_t n, uint64_t* out9, int window_c);
+
+/* Device-resident variants for benchmarking/pipelining (inputs already in
+ * HBM; see rng_dbuf_*). */
+int rng_ntt_fr_dev(RngCtx* ctx, void* dev_data, uint64_t n, uint64_t batch, int inverse);
+/* out-of-place: result lands in dev_out (dev_in preserved for n > 4096;
+ * for n <= 4096 the transform runs in dev_in then copies to dev_out). */
+int rng_ntt_fr_dev_oop(RngCtx* ctx, void* dev_in, void* dev_out, uint64_t n,
+                       uint64_t batch, int inverse);
+int rng_msm_g1_dev(RngCtx* ctx, const void* dev_bases, const void* dev_scalars,
+                   uint64_t n, uint64_t* out9, int window_c);
+
+/* Device buffer helpers */
+void* rng_dbuf_alloc(size_t bytes);
+void rng_dbuf_free(void* dbuf);
+int rng_dbuf_upload(void* dbuf, const void* host_src, size_t bytes);
+int rng_dbuf_download(const void* dbuf, void* host_dst, size_t bytes);
+int rng_device_sync(void);
+
+/* The SRS G1 powers already resident in HBM (bases for KZG commitments):
+ * returns device pointer (n*8 u64 packed affine) and count. */
+const void* rng_srs_dev_bases(RngCtx* ctx, uint64_t* count);
+
+/* ---- PlonK prover (populated as the plonk layer lands) ---- */
+
+/* Circuit description: the output of arithmetization, i.e. exactly what
+ * PlonkKzgSnark::preprocess consumes from a finalized PlonkCircuit
+ * (traits.rs:832-855): n gates (padded to a power of two), 13 selector
+ * columns, the wire permutation over 5*n wire slots, and public-input count.
+ * Layout documented in DESIGN.md §ABI. */
+typedef struct {
+    uint64_t n;             /* evaluation domain size (power of two) */
+    uint64_t num_public;    /* number of public inputs */
+    const uint64_t* selectors;  /* 13 * n Fr (Montgomery), column-major q[sel][gate] */
+    const uint64_t* sigma;      /* 5 * n u64 indices: wire permutation (values < 5n) */
+    uint64_t num_link_groups;
+    const uint64_t* link_offsets;  /* per group: (offset, stride=1?, count) triples */
+} RngCircuitDesc;
+
+RngProvingKey* rng_preprocess(RngCtx* ctx, const RngCircuitDesc* desc);
+void rng_pk_free(RngProvingKey* pk);
+
+/* witness: 5*n Fr wire values (Montgomery, column-major w[wire][gate]);
+ * public inputs: num_public Fr. seed drives the blinder (the reference uses
+ * thread_rng at traits.rs:994; a fixed seed pins proof bytes for parity).
+ * out_proof: RNG_PROOF_U64S u64. out_link_hint: optional (NULL to skip),
+ * n+? Fr coefficients + 9 u64 commitment; see DESIGN.md. */
+int rng_prove(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* wires,
+              const uint64_t* public_inputs, uint64_t seed,
+              uint64_t* out_proof, uint64_t* out_link_hint);
+
+int rng_verify(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* public_inputs,
+               const uint64_t* proof);
+
+int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a,
+                    const uint64_t* hint_b, uint64_t group_offset, uint64_t group_size,
+                    uint64_t* out_link_proof);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* RNG_PROVER_H */

@@ -1,0 +1,187 @@
+// PRODUCT PATH — MI355X-native BN254 field arithmetic (device + host).
+//
+// Replaces (new implementation, not a port): arkworks 0.4.2 ark-ff Montgomery
+// arithmetic for BN254 Fr/Fq as consumed by the reference prover through
+// mpc-jellyfish (limb layout pinned at
+// crates/relayer-types/types-proofs/src/rkyv_impls/plonk_proof_def.rs:22-52).
+// CIOS Montgomery multiplication with 4 x 64-bit limbs — deliberately a
+// different formulation from the CPU oracle's SOS reduce so the two paths
+// cross-check (oracle/field.hpp).
+#pragma once
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include "../../include/bn254_params.h"
+
+namespace rng {
+
+using u64 = uint64_t;
+using u128 = unsigned __int128;
+
+#define RNG_HD __host__ __device__ __forceinline__
+
+struct fr_params {
+    static constexpr u64 mod[4] = FR_MODULUS;
+    static constexpr u64 r[4] = FR_R;
+    static constexpr u64 r2[4] = FR_R2;
+    static constexpr u64 inv = FR_INV;
+};
+struct fq_params {
+    static constexpr u64 mod[4] = FQ_MODULUS;
+    static constexpr u64 r[4] = FQ_R;
+    static constexpr u64 r2[4] = FQ_R2;
+    static constexpr u64 inv = FQ_INV;
+};
+
+template <class P>
+struct Fp4 {
+    u64 l[4];
+
+    RNG_HD static Fp4 zero() { return Fp4{{0, 0, 0, 0}}; }
+    RNG_HD static Fp4 one() { return Fp4{{P::r[0], P::r[1], P::r[2], P::r[3]}}; }
+
+    RNG_HD bool is_zero() const { return (l[0] | l[1] | l[2] | l[3]) == 0; }
+    RNG_HD bool eq(const Fp4& o) const {
+        return l[0] == o.l[0] && l[1] == o.l[1] && l[2] == o.l[2] && l[3] == o.l[3];
+    }
+
+    RNG_HD static bool geq_mod(const u64 a[4]) {
+        for (int i = 3; i >= 0; --i) {
+            if (a[i] != P::mod[i]) return a[i] > P::mod[i];
+        }
+        return true;
+    }
+
+    RNG_HD Fp4 add(const Fp4& o) const {
+        Fp4 r;
+        u128 c = 0;
+        for (int i = 0; i < 4; ++i) {
+            c += (u128)l[i] + o.l[i];
+            r.l[i] = (u64)c;
+            c >>= 64;
+        }
+        if (geq_mod(r.l)) {
+            u128 b = 0;
+            for (int i = 0; i < 4; ++i) {
+                u128 d = (u128)r.l[i] - P::mod[i] - b;
+                r.l[i] = (u64)d;
+                b = (d >> 64) & 1;
+            }
+        }
+        return r;
+    }
+
+    RNG_HD Fp4 sub(const Fp4& o) const {
+        Fp4 r;
+        u128 b = 0;
+        for (int i = 0; i < 4; ++i) {
+            u128 d = (u128)l[i] - o.l[i] - b;
+            r.l[i] = (u64)d;
+            b = (d >> 64) & 1;
+        }
+        if (b) {
+            u128 c = 0;
+            for (int i = 0; i < 4; ++i) {
+                c += (u128)r.l[i] + P::mod[i];
+                r.l[i] = (u64)c;
+                c >>= 64;
+            }
+        }
+        return r;
+    }
+
+    RNG_HD Fp4 neg() const {
+        if (is_zero()) return *this;
+        Fp4 r;
+        u128 b = 0;
+        for (int i = 0; i < 4; ++i) {
+            u128 d = (u128)P::mod[i] - l[i] - b;
+            r.l[i] = (u64)d;
+            b = (d >> 64) & 1;
+        }
+        return r;
+    }
+
+    RNG_HD Fp4 dbl() const { return add(*this); }
+
+    // CIOS Montgomery multiplication (Acar), N = 4.
+    RNG_HD Fp4 mul(const Fp4& b) const {
+        u64 t[6] = {0, 0, 0, 0, 0, 0};
+        for (int i = 0; i < 4; ++i) {
+            // t += a * b[i]
+            u64 carry = 0;
+            for (int j = 0; j < 4; ++j) {
+                u128 cur = (u128)l[j] * b.l[i] + t[j] + carry;
+                t[j] = (u64)cur;
+                carry = (u64)(cur >> 64);
+            }
+            u128 cur = (u128)t[4] + carry;
+            t[4] = (u64)cur;
+            t[5] = (u64)(cur >> 64);
+            // reduce one limb
+            u64 m = t[0] * P::inv;
+            cur = (u128)m * P::mod[0] + t[0];
+            carry = (u64)(cur >> 64);
+            for (int j = 1; j < 4; ++j) {
+                cur = (u128)m * P::mod[j] + t[j] + carry;
+                t[j - 1] = (u64)cur;
+                carry = (u64)(cur >> 64);
+            }
+            cur = (u128)t[4] + carry;
+            t[3] = (u64)cur;
+            t[4] = t[5] + (u64)(cur >> 64);
+        }
+        Fp4 r{{t[0], t[1], t[2], t[3]}};
+        if (t[4] || geq_mod(r.l)) {
+            u128 bw = 0;
+            for (int i = 0; i < 4; ++i) {
+                u128 d = (u128)r.l[i] - P::mod[i] - bw;
+                r.l[i] = (u64)d;
+                bw = (d >> 64) & 1;
+            }
+        }
+        return r;
+    }
+
+    RNG_HD Fp4 sqr() const { return mul(*this); }
+
+    RNG_HD Fp4 pow(const u64 e[4]) const {
+        Fp4 acc = one();
+        Fp4 base = *this;
+        for (int i = 0; i < 256; ++i) {
+            if ((e[i >> 6] >> (i & 63)) & 1) acc = acc.mul(base);
+            base = base.sqr();
+        }
+        return acc;
+    }
+    RNG_HD Fp4 pow_u64(u64 e) const {
+        Fp4 acc = one();
+        Fp4 base = *this;
+        while (e) {
+            if (e & 1) acc = acc.mul(base);
+            base = base.sqr();
+            e >>= 1;
+        }
+        return acc;
+    }
+    RNG_HD Fp4 inverse() const {  // Fermat: a^(p-2); low limb of both moduli >= 2
+        u64 e[4] = {P::mod[0] - 2, P::mod[1], P::mod[2], P::mod[3]};
+        return pow(e);
+    }
+
+    RNG_HD static Fp4 from_canonical(const u64 c[4]) {
+        Fp4 a{{c[0], c[1], c[2], c[3]}};
+        Fp4 r2{{P::r2[0], P::r2[1], P::r2[2], P::r2[3]}};
+        return a.mul(r2);
+    }
+    RNG_HD void to_canonical(u64 out[4]) const {
+        // multiply by 1 (non-Montgomery) = Montgomery-reduce
+        Fp4 onev{{1, 0, 0, 0}};
+        Fp4 r = mul(onev);
+        out[0] = r.l[0]; out[1] = r.l[1]; out[2] = r.l[2]; out[3] = r.l[3];
+    }
+};
+
+using Fr = Fp4<fr_params>;
+using Fq = Fp4<fq_params>;
+
+}  // namespace rng

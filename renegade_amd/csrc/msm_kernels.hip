@@ -1,0 +1,172 @@
+// PRODUCT PATH — MI355X-native Pippenger MSM over BN254 G1.
+//
+// Replaces: arkworks ark-ec VariableBaseMSM as consumed through the
+// reference's KZG commitments (BASELINE config #2).
+//
+// Structure (sort-based, designed for gfx950):
+//  1. k_msm_digits: signed windowed digit decomposition (c bits, digits in
+//     [-2^(c-1), 2^(c-1)]), one (key = window<<16 | magnitude, val =
+//     sign<<31 | point index) pair per nonzero digit; zero digits get a
+//     sentinel key that sorts last.
+//  2. rocPRIM device radix sort on the 21-bit keys (groups all points of a
+//     bucket together — no atomics or EC critical sections anywhere).
+//  3. k_msm_bucket_reduce: one thread per segment head walks its bucket's
+//     points with Jacobian mixed-adds (bases gathered from HBM).
+//  4. k_msm_window_chunks: per (window, chunk of 2^(c-1)/CHUNK buckets):
+//     suffix running sums -> (sum, weighted-sum) partials.
+//  5. k_msm_final: combine chunk partials, fold windows (Horner with c
+//     doublings), single workgroup; result Jacobian to host.
+#include <hip/hip_runtime.h>
+#include <rocprim/device/device_radix_sort.hpp>
+#include "gpu_curve.hpp"
+
+namespace rng {
+
+constexpr uint32_t MSM_SENTINEL = 0x1FFFFFu;  // > any (w<<16|mag); 21 bits
+constexpr uint32_t MSM_CHUNK = 128;           // buckets per window-sum thread
+
+// ---- 1. digit decomposition ----
+// scalars: canonical LE 4xu64. keys/vals: n*W entries, window-major
+// (out[w*n + i]) so writes coalesce per window.
+__global__ void k_msm_digits(const uint64_t* scalars, uint32_t n, uint32_t c,
+                             uint32_t W, uint32_t* keys, uint32_t* vals) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t s[4];
+    s[0] = scalars[4 * i];
+    s[1] = scalars[4 * i + 1];
+    s[2] = scalars[4 * i + 2];
+    s[3] = scalars[4 * i + 3];
+    uint32_t carry = 0;
+    uint32_t half = 1u << (c - 1);
+    uint64_t cmask = (c == 64) ? ~0ull : ((1ull << c) - 1);
+    for (uint32_t w = 0; w < W; ++w) {
+        uint32_t bit0 = w * c;
+        uint32_t limb = bit0 >> 6, off = bit0 & 63;
+        uint64_t raw = s[limb] >> off;
+        if (off + c > 64 && limb + 1 < 4) raw |= s[limb + 1] << (64 - off);
+        raw = (raw & cmask) + carry;
+        uint32_t mag, sign;
+        if (raw >= half) {  // treat as negative digit raw - 2^c unless raw == half
+            if (raw > half) {
+                mag = (uint32_t)((1ull << c) - raw);
+                sign = 1;
+                carry = 1;
+            } else {  // raw == half: use +half, no carry (mag fits 16 bits for c<=16)
+                mag = half;
+                sign = 0;
+                carry = 0;
+            }
+        } else {
+            mag = (uint32_t)raw;
+            sign = 0;
+            carry = 0;
+        }
+        uint64_t o = (uint64_t)w * n + i;
+        keys[o] = mag == 0 ? MSM_SENTINEL : ((w << 16) | mag);
+        vals[o] = (sign << 31) | i;
+    }
+    // carry out of the top window must be zero for scalars < 2^(W*c-1)
+}
+
+// ---- 3. bucket segmented reduction ----
+// buckets: W * 2^(c-1) Jacobian points, zero-initialised (Z=0 == identity).
+__global__ void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
+                                    uint32_t total, const G1Aff* bases,
+                                    G1Jac* buckets, uint32_t c) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= total) return;
+    uint32_t key = keys[t];
+    if (key == MSM_SENTINEL) return;
+    if (t > 0 && keys[t - 1] == key) return;  // not a segment head
+    G1Jac acc = G1Jac::identity();
+    for (uint32_t j = t; j < total && keys[j] == key; ++j) {
+        uint32_t v = vals[j];
+        G1Aff p = bases[v & 0x7FFFFFFFu];
+        acc = acc.madd(p, (v >> 31) != 0);
+    }
+    uint32_t w = key >> 16;
+    uint32_t mag = key & 0xFFFFu;  // 1 .. 2^(c-1)
+    buckets[(uint64_t)w * (1u << (c - 1)) + (mag - 1)] = acc;
+}
+
+// ---- 4. per-window chunked suffix sums ----
+// grid: W * (2^(c-1) / MSM_CHUNK) threads total; partials: per thread
+// (T = plain sum, S = locally-weighted sum) -> 2 Jacobians.
+__global__ void k_msm_window_chunks(const G1Jac* buckets, uint32_t c, uint32_t W,
+                                    G1Jac* partials /* 2 per thread: T, S */) {
+    uint32_t nb = 1u << (c - 1);
+    uint32_t chunks_per_w = nb / MSM_CHUNK;
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= W * chunks_per_w) return;
+    uint32_t w = t / chunks_per_w;
+    uint32_t chunk = t % chunks_per_w;
+    const G1Jac* b = buckets + (uint64_t)w * nb + (uint64_t)chunk * MSM_CHUNK;
+    // local digits are base + j, j = 1..CHUNK, base = chunk*CHUNK
+    // suffix running sums over j descending: run = sum_{k>=j} S_k ; S += run
+    G1Jac run = G1Jac::identity(), S = G1Jac::identity();
+    for (int j = MSM_CHUNK - 1; j >= 0; --j) {
+        run = run.add(b[j]);
+        S = S.add(run);
+    }
+    partials[2 * t] = run;  // T = plain sum of chunk buckets
+    partials[2 * t + 1] = S;  // sum_j (j_local) * bucket, j_local = 1..CHUNK
+}
+
+// ---- 5. final combine ----
+// One workgroup (256 threads). For each (w, chunk): contribution =
+// S + base * T with base = chunk*CHUNK (scalar-mul by small integer).
+// Tree-reduce per window in LDS, then Horner across windows on thread 0.
+__global__ __launch_bounds__(256) void k_msm_final(const G1Jac* partials, uint32_t c,
+                                                   uint32_t W, G1Jac* out) {
+    __shared__ G1Jac red[256];
+    uint32_t nb = 1u << (c - 1);
+    uint32_t chunks_per_w = nb / MSM_CHUNK;
+    __shared__ G1Jac window_sum[32];
+
+    for (uint32_t w = 0; w < W; ++w) {
+        // each thread folds a strided subset of this window's chunks
+        G1Jac acc = G1Jac::identity();
+        for (uint32_t chunk = threadIdx.x; chunk < chunks_per_w; chunk += blockDim.x) {
+            uint32_t t = w * chunks_per_w + chunk;
+            G1Jac T = partials[2 * t];
+            G1Jac S = partials[2 * t + 1];
+            // base * T via double-and-add on the small integer base
+            uint32_t base = chunk * MSM_CHUNK;
+            G1Jac bT = G1Jac::identity();
+            G1Jac addend = T;
+            while (base) {
+                if (base & 1) bT = bT.add(addend);
+                addend = addend.dbl();
+                base >>= 1;
+            }
+            acc = acc.add(S).add(bT);
+        }
+        red[threadIdx.x] = acc;
+        __syncthreads();
+        for (uint32_t stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+            if (threadIdx.x < stride)
+                red[threadIdx.x] = red[threadIdx.x].add(red[threadIdx.x + stride]);
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) window_sum[w] = red[0];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        G1Jac acc = window_sum[W - 1];
+        for (int w = (int)W - 2; w >= 0; --w) {
+            for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
+            acc = acc.add(window_sum[w]);
+        }
+        *out = acc;
+    }
+}
+
+// ---- helpers ----
+__global__ void k_fr_to_canonical(const Fr* in, uint64_t* out, uint32_t n) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    in[i].to_canonical(out + 4 * i);
+}
+
+}  // namespace rng
