@@ -18,12 +18,14 @@
 //     doublings), single workgroup; result Jacobian to host.
 #include <hip/hip_runtime.h>
 #include <rocprim/device/device_radix_sort.hpp>
+#include <rocprim/device/device_select.hpp>
+#include <rocprim/iterator/counting_iterator.hpp>
 #include "gpu_curve.hpp"
 
 namespace rng {
 
 constexpr uint32_t MSM_SENTINEL = 0x1FFFFFu;  // > any (w<<16|mag); 21 bits
-constexpr uint32_t MSM_CHUNK = 128;           // buckets per window-sum thread
+constexpr uint32_t MSM_CHUNK = 32;            // buckets per window-sum thread
 
 // ---- 1. digit decomposition ----
 // scalars: canonical LE 4xu64. keys/vals: n*W entries, window-major
@@ -69,18 +71,53 @@ __global__ void k_msm_digits(const uint64_t* scalars, uint32_t n, uint32_t c,
     // carry out of the top window must be zero for scalars < 2^(W*c-1)
 }
 
-// ---- 3. bucket segmented reduction ----
-// buckets: W * 2^(c-1) Jacobian points, zero-initialised (Z=0 == identity).
-__global__ void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
-                                    uint32_t total, const G1Aff* bases,
-                                    G1Jac* buckets, uint32_t c) {
+// ---- 3a. segment-head flags (for stream compaction) ----
+__global__ void k_msm_head_flags(const uint32_t* keys, uint32_t total, uint8_t* flags) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= total) return;
     uint32_t key = keys[t];
-    if (key == MSM_SENTINEL) return;
-    if (t > 0 && keys[t - 1] == key) return;  // not a segment head
+    flags[t] = (key != MSM_SENTINEL && (t == 0 || keys[t - 1] != key)) ? 1 : 0;
+}
+
+// ---- 3b. segment lengths (for length-uniform wave scheduling) ----
+// heads are in increasing order (rocprim::select is stable); seg i spans
+// [heads[i], heads[i+1] or first sentinel/total).
+__global__ void k_msm_seg_lengths(const uint32_t* keys, const uint32_t* heads,
+                                  const uint32_t* head_count, uint32_t total,
+                                  uint32_t* lens) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    uint32_t hc = *head_count;
+    if (t >= hc) return;
+    uint32_t start = heads[t];
+    uint32_t end = (t + 1 < hc) ? heads[t + 1] : total;
+    // sentinel tail: the zero-digit entries sort after every real key, and
+    // the last real segment's end must exclude them
+    if (t + 1 == hc) {
+        uint32_t key = keys[start];
+        uint32_t e = start;
+        while (e < total && keys[e] == key) ++e;
+        end = e;
+    }
+    lens[t] = end - start;
+}
+
+// ---- 3c. bucket segmented reduction, one thread per segment ----
+// heads_sorted: head indices sorted by segment length so all 64 lanes of a
+// wave walk near-equal-length segments (kills divergence).
+// buckets: W * 2^(c-1) Jacobian points (fully overwritten for live buckets;
+// zero-initialised so untouched buckets read as identity).
+__global__ void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
+                                    const uint32_t* heads_sorted,
+                                    const uint32_t* lens_sorted,
+                                    const uint32_t* head_count, const G1Aff* bases,
+                                    G1Jac* buckets, uint32_t c) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= *head_count) return;
+    uint32_t start = heads_sorted[t];
+    uint32_t len = lens_sorted[t];
+    uint32_t key = keys[start];
     G1Jac acc = G1Jac::identity();
-    for (uint32_t j = t; j < total && keys[j] == key; ++j) {
+    for (uint32_t j = start; j < start + len; ++j) {
         uint32_t v = vals[j];
         G1Aff p = bases[v & 0x7FFFFFFFu];
         acc = acc.madd(p, (v >> 31) != 0);
@@ -113,50 +150,65 @@ __global__ void k_msm_window_chunks(const G1Jac* buckets, uint32_t c, uint32_t W
     partials[2 * t + 1] = S;  // sum_j (j_local) * bucket, j_local = 1..CHUNK
 }
 
-// ---- 5. final combine ----
-// One workgroup (256 threads). For each (w, chunk): contribution =
-// S + base * T with base = chunk*CHUNK (scalar-mul by small integer).
-// Tree-reduce per window in LDS, then Horner across windows on thread 0.
-__global__ __launch_bounds__(256) void k_msm_final(const G1Jac* partials, uint32_t c,
-                                                   uint32_t W, G1Jac* out) {
-    __shared__ G1Jac red[256];
+// ---- 5a. window combine: grid = W * MSM_SUBB blocks; block (w, sb) folds a
+// slice of window w's chunk partials (contribution = S + base*T with
+// base = chunk*CHUNK), LDS tree reduce -> window_partials[w*SUBB + sb].
+constexpr uint32_t MSM_SUBB = 16;
+
+__global__ __launch_bounds__(64) void k_msm_window_combine(const G1Jac* partials,
+                                                           uint32_t c,
+                                                           G1Jac* window_partials) {
+    __shared__ G1Jac red[64];
+    uint32_t w = blockIdx.x / MSM_SUBB;
+    uint32_t sb = blockIdx.x % MSM_SUBB;
     uint32_t nb = 1u << (c - 1);
     uint32_t chunks_per_w = nb / MSM_CHUNK;
-    __shared__ G1Jac window_sum[32];
-
-    for (uint32_t w = 0; w < W; ++w) {
-        // each thread folds a strided subset of this window's chunks
-        G1Jac acc = G1Jac::identity();
-        for (uint32_t chunk = threadIdx.x; chunk < chunks_per_w; chunk += blockDim.x) {
-            uint32_t t = w * chunks_per_w + chunk;
-            G1Jac T = partials[2 * t];
-            G1Jac S = partials[2 * t + 1];
-            // base * T via double-and-add on the small integer base
-            uint32_t base = chunk * MSM_CHUNK;
-            G1Jac bT = G1Jac::identity();
-            G1Jac addend = T;
-            while (base) {
-                if (base & 1) bT = bT.add(addend);
-                addend = addend.dbl();
-                base >>= 1;
-            }
-            acc = acc.add(S).add(bT);
+    uint32_t per_sb = (chunks_per_w + MSM_SUBB - 1) / MSM_SUBB;
+    uint32_t lo = sb * per_sb;
+    uint32_t hi = lo + per_sb < chunks_per_w ? lo + per_sb : chunks_per_w;
+    G1Jac acc = G1Jac::identity();
+    for (uint32_t chunk = lo + threadIdx.x; chunk < hi; chunk += blockDim.x) {
+        uint32_t t = w * chunks_per_w + chunk;
+        G1Jac T = partials[2 * t];
+        G1Jac S = partials[2 * t + 1];
+        uint32_t base = chunk * MSM_CHUNK;
+        G1Jac bT = G1Jac::identity();
+        G1Jac addend = T;
+        while (base) {
+            if (base & 1) bT = bT.add(addend);
+            addend = addend.dbl();
+            base >>= 1;
         }
-        red[threadIdx.x] = acc;
-        __syncthreads();
-        for (uint32_t stride = blockDim.x / 2; stride > 0; stride >>= 1) {
-            if (threadIdx.x < stride)
-                red[threadIdx.x] = red[threadIdx.x].add(red[threadIdx.x + stride]);
-            __syncthreads();
-        }
-        if (threadIdx.x == 0) window_sum[w] = red[0];
+        acc = acc.add(S).add(bT);
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (uint32_t stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if (threadIdx.x < stride)
+            red[threadIdx.x] = red[threadIdx.x].add(red[threadIdx.x + stride]);
         __syncthreads();
     }
-    if (threadIdx.x == 0) {
-        G1Jac acc = window_sum[W - 1];
-        for (int w = (int)W - 2; w >= 0; --w) {
+    if (threadIdx.x == 0) window_partials[w * MSM_SUBB + sb] = red[0];
+}
+
+// ---- 5b. reduce sub-partials per window + Horner fold across windows ----
+// One block of W threads: thread w sums its window's SUBB partials; thread 0
+// then Horner-folds the windows.
+__global__ void k_msm_horner(const G1Jac* window_partials, uint32_t c, uint32_t W,
+                             G1Jac* out) {
+    __shared__ G1Jac ws[32];
+    uint32_t w = threadIdx.x;
+    if (w < W) {
+        G1Jac s = window_partials[w * MSM_SUBB];
+        for (uint32_t i = 1; i < MSM_SUBB; ++i) s = s.add(window_partials[w * MSM_SUBB + i]);
+        ws[w] = s;
+    }
+    __syncthreads();
+    if (w == 0) {
+        G1Jac acc = ws[W - 1];
+        for (int i = (int)W - 2; i >= 0; --i) {
             for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
-            acc = acc.add(window_sum[w]);
+            acc = acc.add(ws[i]);
         }
         *out = acc;
     }

@@ -208,15 +208,27 @@ struct MsmScratch {
     uint32_t *keys_in = nullptr, *keys_out = nullptr, *vals_in = nullptr, *vals_out = nullptr;
     void* sort_temp = nullptr;
     size_t sort_temp_bytes = 0;
+    uint8_t* head_flags = nullptr;
+    uint32_t* heads = nullptr;
+    uint32_t* heads_sorted = nullptr;
+    uint32_t* lens = nullptr;
+    uint32_t* lens_sorted = nullptr;
+    uint32_t* head_count = nullptr;  // device u32
+    void* select_temp = nullptr;
+    size_t select_temp_bytes = 0;
     G1Jac* buckets = nullptr;
     G1Jac* partials = nullptr;
+    G1Jac* window_sums = nullptr;
     G1Jac* result = nullptr;
     uint64_t cap_entries = 0;
     uint32_t cap_c = 0;
 
     ~MsmScratch() {
         for (void* b : {(void*)keys_in, (void*)keys_out, (void*)vals_in, (void*)vals_out,
-                        sort_temp, (void*)buckets, (void*)partials, (void*)result})
+                        sort_temp, (void*)head_flags, (void*)heads, (void*)heads_sorted,
+                        (void*)lens, (void*)lens_sorted, (void*)head_count,
+                        select_temp, (void*)buckets, (void*)partials, (void*)window_sums,
+                        (void*)result})
             if (b) hipFree(b);
     }
 };
@@ -242,8 +254,22 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         rocprim::radix_sort_pairs(nullptr, s->sort_temp_bytes, s->keys_in, s->keys_out,
                                   s->vals_in, s->vals_out, total, 0, 21, stream);
         HIP_CHECK(hipMalloc(&s->sort_temp, s->sort_temp_bytes));
+        HIP_CHECK(hipMalloc(&s->head_flags, total));
+        uint64_t max_heads_cap = (nb < total ? nb : total) + 1;
+        HIP_CHECK(hipMalloc(&s->heads, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->heads_sorted, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->lens, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->lens_sorted, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->head_count, 4));
+        {
+            rocprim::counting_iterator<uint32_t> cit(0);
+            (void)rocprim::select(nullptr, s->select_temp_bytes, cit, s->head_flags,
+                                  s->heads, s->head_count, total, stream);
+            HIP_CHECK(hipMalloc(&s->select_temp, s->select_temp_bytes));
+        }
         HIP_CHECK(hipMalloc(&s->buckets, nb * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->partials, 2 * nchunks * sizeof(G1Jac)));
+        HIP_CHECK(hipMalloc(&s->window_sums, 32 * MSM_SUBB * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->result, sizeof(G1Jac)));
         s->cap_entries = total;
         s->cap_c = c;
@@ -259,17 +285,40 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     rocprim::radix_sort_pairs(s->sort_temp, s->sort_temp_bytes, s->keys_in, s->keys_out,
                               s->vals_in, s->vals_out, total, 0, 21, stream);
     HIP_CHECK(hipMemsetAsync(s->buckets, 0, nb * sizeof(G1Jac), stream));
+    hipLaunchKernelGGL(k_msm_head_flags, dim3((uint32_t)((total + tb - 1) / tb)), dim3(tb),
+                       0, stream, s->keys_out, (uint32_t)total, s->head_flags);
+    HIP_CHECK(hipGetLastError());
+    {
+        rocprim::counting_iterator<uint32_t> cit(0);
+        (void)rocprim::select(s->select_temp, s->select_temp_bytes, cit, s->head_flags,
+                              s->heads, s->head_count, total, stream);
+    }
+    uint64_t max_heads = nb < total ? nb : total;
+    hipLaunchKernelGGL(k_msm_seg_lengths, dim3((uint32_t)((max_heads + tb - 1) / tb)),
+                       dim3(tb), 0, stream, s->keys_out, s->heads, s->head_count,
+                       (uint32_t)total, s->lens);
+    HIP_CHECK(hipGetLastError());
+    // sort heads by segment length -> every wave walks near-equal segments
+    // (rocprim needs a host-side item count: 4-byte DtoH + sync, ~10us)
+    uint32_t hc = 0;
+    HIP_CHECK(hipMemcpyAsync(&hc, s->head_count, 4, hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    rocprim::radix_sort_pairs(s->sort_temp, s->sort_temp_bytes, s->lens, s->lens_sorted,
+                              s->heads, s->heads_sorted, hc, 0, 25, stream);
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((uint32_t)((total + tb - 1) / tb)),
-                       dim3(tb), 0, stream, s->keys_out, s->vals_out, (uint32_t)total,
-                       d_bases, s->buckets, c);
+    hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((uint32_t)((hc + tb - 1) / tb)),
+                       dim3(tb), 0, stream, s->keys_out, s->vals_out, s->heads_sorted,
+                       s->lens_sorted, s->head_count, d_bases, s->buckets, c);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
     hipLaunchKernelGGL(k_msm_window_chunks, dim3((uint32_t)((nchunks + tb - 1) / tb)),
                        dim3(tb), 0, stream, s->buckets, c, W, s->partials);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_final, dim3(1), dim3(256), 0, stream, s->partials, c, W,
+    hipLaunchKernelGGL(k_msm_window_combine, dim3(W * MSM_SUBB), dim3(64), 0, stream,
+                       s->partials, c, s->window_sums);
+    HIP_CHECK(hipGetLastError());
+    hipLaunchKernelGGL(k_msm_horner, dim3(1), dim3(64), 0, stream, s->window_sums, c, W,
                        s->result);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
