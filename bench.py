@@ -171,6 +171,38 @@ def main():
     breakdown = {k: round(float(np.mean([t[k] for t in kern_times])), 3)
                  for k in kern_times[0]}
 
+    # NTT side-measurement (BASELINE config #3): 2^22 forward+inverse,
+    # device-resident, out-of-place ping-pong (no copy-back)
+    ntt_extra = None
+    if rank == 0:
+        m = 1 << 22
+        nd = np.random.default_rng(5).integers(0, 1 << 61, size=4 * m, dtype=np.uint64)
+        da = ctx.dbuf_from(nd)
+        db = ctx.dbuf_alloc(nd.nbytes)
+        for _ in range(2):
+            ctx.ntt_dev_oop(da, db, m)
+            ctx.ntt_dev_oop(db, da, m, inverse=True)
+        ctx.sync()
+        t1 = time.perf_counter()
+        reps = 5
+        for _ in range(reps):
+            ctx.ntt_dev_oop(da, db, m)
+            ctx.ntt_dev_oop(db, da, m, inverse=True)
+        ctx.sync()
+        rt_ms = (time.perf_counter() - t1) / reps * 1e3
+        back = np.empty_like(nd)
+        ctx.dbuf_download(da, back)
+        assert np.array_equal(back, nd), "NTT round-trip mismatch in bench"
+        passes = plib.ntt_last_times()
+        alg_bytes_rt = 2 * 2 * 2 * 32 * m  # P=2 passes x r+w x 32 B, both directions
+        ntt_extra = {
+            "roundtrip_ms": round(rt_ms, 3),
+            "alg_gbs": round(alg_bytes_rt / (rt_ms / 1e3) / 1e9, 1),
+            "pass_ms": {k: round(v, 3) for k, v in passes.items()},
+        }
+        ctx.dbuf_free(da)
+        ctx.dbuf_free(db)
+
     if rank == 0:
         cores = os.cpu_count()
         cb = None
@@ -207,6 +239,7 @@ def main():
                 "breakdown_ms": breakdown,
             },
             "cpu_baseline": cb,
+            "extra": {"ntt_2^22": ntt_extra},
         }
         print(json.dumps(result), flush=True)
 

@@ -191,28 +191,9 @@ __global__ __launch_bounds__(64) void k_msm_window_combine(const G1Jac* partials
     if (threadIdx.x == 0) window_partials[w * MSM_SUBB + sb] = red[0];
 }
 
-// ---- 5b. reduce sub-partials per window + Horner fold across windows ----
-// One block of W threads: thread w sums its window's SUBB partials; thread 0
-// then Horner-folds the windows.
-__global__ void k_msm_horner(const G1Jac* window_partials, uint32_t c, uint32_t W,
-                             G1Jac* out) {
-    __shared__ G1Jac ws[32];
-    uint32_t w = threadIdx.x;
-    if (w < W) {
-        G1Jac s = window_partials[w * MSM_SUBB];
-        for (uint32_t i = 1; i < MSM_SUBB; ++i) s = s.add(window_partials[w * MSM_SUBB + i]);
-        ws[w] = s;
-    }
-    __syncthreads();
-    if (w == 0) {
-        G1Jac acc = ws[W - 1];
-        for (int i = (int)W - 2; i >= 0; --i) {
-            for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
-            acc = acc.add(ws[i]);
-        }
-        *out = acc;
-    }
-}
+// (The final fold across windows — W*SUBB <= 512 Jacobians, ~1.5 KB — is
+// done on the HOST: a single-lane dependent EC chain runs ~50x slower on a
+// GPU SIMT lane than on a host core, and the data is tiny.)
 
 // ---- helpers ----
 __global__ void k_fr_to_canonical(const Fr* in, uint64_t* out, uint32_t n) {

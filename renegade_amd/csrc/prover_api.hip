@@ -303,13 +303,19 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint32_t hc = 0;
     HIP_CHECK(hipMemcpyAsync(&hc, s->head_count, 4, hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
-    rocprim::radix_sort_pairs(s->sort_temp, s->sort_temp_bytes, s->lens, s->lens_sorted,
-                              s->heads, s->heads_sorted, hc, 0, 25, stream);
-    et.mark(stream);
-    hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((uint32_t)((hc + tb - 1) / tb)),
-                       dim3(tb), 0, stream, s->keys_out, s->vals_out, s->heads_sorted,
-                       s->lens_sorted, s->head_count, d_bases, s->buckets, c);
-    HIP_CHECK(hipGetLastError());
+    if (hc > 0) {
+        // descending: longest segments first so the tail waves pack short work
+        rocprim::radix_sort_pairs_desc(s->sort_temp, s->sort_temp_bytes, s->lens,
+                                       s->lens_sorted, s->heads, s->heads_sorted, hc, 0,
+                                       25, stream);
+        et.mark(stream);
+        hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((uint32_t)((hc + tb - 1) / tb)),
+                           dim3(tb), 0, stream, s->keys_out, s->vals_out, s->heads_sorted,
+                           s->lens_sorted, s->head_count, d_bases, s->buckets, c);
+        HIP_CHECK(hipGetLastError());
+    } else {
+        et.mark(stream);
+    }
     et.mark(stream);
     hipLaunchKernelGGL(k_msm_window_chunks, dim3((uint32_t)((nchunks + tb - 1) / tb)),
                        dim3(tb), 0, stream, s->buckets, c, W, s->partials);
@@ -318,14 +324,26 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     hipLaunchKernelGGL(k_msm_window_combine, dim3(W * MSM_SUBB), dim3(64), 0, stream,
                        s->partials, c, s->window_sums);
     HIP_CHECK(hipGetLastError());
-    hipLaunchKernelGGL(k_msm_horner, dim3(1), dim3(64), 0, stream, s->window_sums, c, W,
-                       s->result);
-    HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    HIP_CHECK(hipMemcpyAsync(h_result, s->result, sizeof(G1Jac), hipMemcpyDeviceToHost,
-                             stream));
+    // host-side fold: W*SUBB Jacobians (~1.5 KB); a single-lane dependent EC
+    // chain is far faster on a host core than on one GPU lane
+    G1Jac wsums[32 * MSM_SUBB];
+    HIP_CHECK(hipMemcpyAsync(wsums, s->window_sums, W * MSM_SUBB * sizeof(G1Jac),
+                             hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     et.collect(tls_msm_times, 5);
+    G1Jac ws[32];
+    for (uint32_t w = 0; w < W; ++w) {
+        G1Jac sum = wsums[w * MSM_SUBB];
+        for (uint32_t i = 1; i < MSM_SUBB; ++i) sum = sum.add(wsums[w * MSM_SUBB + i]);
+        ws[w] = sum;
+    }
+    G1Jac acc = ws[W - 1];
+    for (int w = (int)W - 2; w >= 0; --w) {
+        for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
+        acc = acc.add(ws[w]);
+    }
+    *h_result = acc;
     return RNG_OK;
 }
 
