@@ -30,7 +30,7 @@ constexpr uint32_t MSM_CHUNK = 32;            // buckets per window-sum thread
 // ---- 1. digit decomposition ----
 // scalars: canonical LE 4xu64. keys/vals: n*W entries, window-major
 // (out[w*n + i]) so writes coalesce per window.
-__global__ void k_msm_digits(const uint64_t* scalars, uint32_t n, uint32_t c,
+__global__ __launch_bounds__(256) void k_msm_digits(const uint64_t* scalars, uint32_t n, uint32_t c,
                              uint32_t W, uint32_t* keys, uint32_t* vals) {
     uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -106,7 +106,7 @@ __global__ void k_msm_seg_lengths(const uint32_t* keys, const uint32_t* heads,
 // wave walk near-equal-length segments (kills divergence).
 // buckets: W * 2^(c-1) Jacobian points (fully overwritten for live buckets;
 // zero-initialised so untouched buckets read as identity).
-__global__ void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
+__global__ __launch_bounds__(256) void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
                                     const uint32_t* heads_sorted,
                                     const uint32_t* lens_sorted,
                                     const uint32_t* head_count, const G1Aff* bases,
@@ -130,7 +130,7 @@ __global__ void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
 // ---- 4. per-window chunked suffix sums ----
 // grid: W * (2^(c-1) / MSM_CHUNK) threads total; partials: per thread
 // (T = plain sum, S = locally-weighted sum) -> 2 Jacobians.
-__global__ void k_msm_window_chunks(const G1Jac* buckets, uint32_t c, uint32_t W,
+__global__ __launch_bounds__(256) void k_msm_window_chunks(const G1Jac* buckets, uint32_t c, uint32_t W,
                                     G1Jac* partials /* 2 per thread: T, S */) {
     uint32_t nb = 1u << (c - 1);
     uint32_t chunks_per_w = nb / MSM_CHUNK;

@@ -42,7 +42,7 @@ __global__ void k_pow_table(Fr* out, Fr base, uint64_t step, uint64_t count) {
 }
 
 // ---- single-workgroup NTT (n <= 4096), one transform per block ----
-__global__ __launch_bounds__(256) void k_ntt_small(Fr* data, const Fr* wst,
+__global__ __launch_bounds__(512) void k_ntt_small(Fr* data, const Fr* wst,
                                                    uint32_t n, uint32_t logn,
                                                    Fr scale, int do_scale) {
     extern __shared__ Fr lds[];
@@ -74,7 +74,7 @@ __global__ __launch_bounds__(256) void k_ntt_small(Fr* data, const Fr* wst,
 
 // ---- pass 1: column-pair DFT of length N2 (stride N1) + outer twiddle ----
 // grid.x = N1/2 * batch; data viewed as [N2 rows][N1 cols].
-__global__ __launch_bounds__(256) void k_ntt_col(Fr* data, const Fr* wst2,
+__global__ __launch_bounds__(512) void k_ntt_col(Fr* data, const Fr* wst2,
                                                  const Fr* ta, const Fr* tb,
                                                  uint32_t N1, uint32_t N2,
                                                  uint32_t logN2, uint32_t split_log) {
@@ -116,7 +116,7 @@ __global__ __launch_bounds__(256) void k_ntt_col(Fr* data, const Fr* wst2,
 
 // ---- pass 2: row-pair DFT of length N1 (contiguous) + strided store ----
 // grid.x = N2/2 * batch.
-__global__ __launch_bounds__(256) void k_ntt_row(const Fr* in, Fr* out, const Fr* wst1,
+__global__ __launch_bounds__(512) void k_ntt_row(const Fr* in, Fr* out, const Fr* wst1,
                                                  uint32_t N1, uint32_t N2,
                                                  uint32_t logN1, Fr scale, int do_scale) {
     extern __shared__ Fr lds[];

@@ -173,7 +173,7 @@ static int ntt_dev_run(RngCtxImpl* ctx, Fr* data, Fr* out, uint32_t n, uint64_t 
     if (!p) return RNG_ERR_HIP;
     if (n <= 4096) {
         Fr* wst = inverse ? p->wst1_i : p->wst1_f;
-        hipLaunchKernelGGL(k_ntt_small, dim3((uint32_t)batch), dim3(256), n * sizeof(Fr),
+        hipLaunchKernelGGL(k_ntt_small, dim3((uint32_t)batch), dim3(512), n * sizeof(Fr),
                            stream, data, wst, n, p->logn, p->ninv, inverse ? 1 : 0);
         HIP_CHECK(hipGetLastError());
         if (out != data)
@@ -189,11 +189,11 @@ static int ntt_dev_run(RngCtxImpl* ctx, Fr* data, Fr* out, uint32_t n, uint64_t 
     uint32_t lds2 = 2 * p->N1 * sizeof(Fr);
     EvtTimer et;
     et.mark(stream);
-    hipLaunchKernelGGL(k_ntt_col, dim3((uint32_t)(p->N1 / 2 * batch)), dim3(256), lds1,
+    hipLaunchKernelGGL(k_ntt_col, dim3((uint32_t)(p->N1 / 2 * batch)), dim3(512), lds1,
                        stream, data, wst2, ta, tb, p->N1, p->N2, p->logN2, p->split_log);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    hipLaunchKernelGGL(k_ntt_row, dim3((uint32_t)(p->N2 / 2 * batch)), dim3(256), lds2,
+    hipLaunchKernelGGL(k_ntt_row, dim3((uint32_t)(p->N2 / 2 * batch)), dim3(512), lds2,
                        stream, data, out, wst1, p->N1, p->N2, p->logN1, p->ninv,
                        inverse ? 1 : 0);
     HIP_CHECK(hipGetLastError());
