@@ -15,6 +15,8 @@
 #include "msm.hpp"
 #include "srs.hpp"
 #include "keccak.hpp"
+#include "transcript.hpp"
+#include "plonk.hpp"
 
 using namespace oracle;
 
@@ -152,5 +154,88 @@ int orc_srs_parse(const uint8_t* bytes, u64 len, u64 max_degree, u64* g1_out,
 
 int orc_num_threads() { return omp_get_max_threads(); }
 void orc_set_num_threads(int n) { omp_set_num_threads(n); }
+
+// ---- TurboPlonk (oracle prover/verifier) ----
+// Proof buffer layout (matches include/rng_prover.h RNG_PROOF_U64S = 157):
+// 13 affine records (9 u64 each): wire comms 0..4, z comm, quot comms 0..4,
+// opening, shifted opening; then 10 Fr (4 u64 Montgomery): 5 wire evals,
+// 4 sigma evals, z_shift eval.
+
+static void proof_store(const OrcProof& pf, u64* out) {
+    const G1Affine* pts[13] = {&pf.wire_comms[0], &pf.wire_comms[1], &pf.wire_comms[2],
+                               &pf.wire_comms[3], &pf.wire_comms[4], &pf.z_comm,
+                               &pf.quot_comms[0], &pf.quot_comms[1], &pf.quot_comms[2],
+                               &pf.quot_comms[3], &pf.quot_comms[4], &pf.opening,
+                               &pf.shifted_opening};
+    for (int i = 0; i < 13; ++i) store_affine(*pts[i], out + 9 * i);
+    u64* e = out + 117;
+    for (int i = 0; i < 5; ++i) memcpy(e + 4 * i, pf.wire_evals[i].l, 32);
+    for (int i = 0; i < 4; ++i) memcpy(e + 20 + 4 * i, pf.sigma_evals[i].l, 32);
+    memcpy(e + 36, pf.z_shift_eval.l, 32);
+}
+
+static OrcProof proof_load(const u64* in) {
+    OrcProof pf;
+    G1Affine* pts[13] = {&pf.wire_comms[0], &pf.wire_comms[1], &pf.wire_comms[2],
+                         &pf.wire_comms[3], &pf.wire_comms[4], &pf.z_comm,
+                         &pf.quot_comms[0], &pf.quot_comms[1], &pf.quot_comms[2],
+                         &pf.quot_comms[3], &pf.quot_comms[4], &pf.opening,
+                         &pf.shifted_opening};
+    for (int i = 0; i < 13; ++i) *pts[i] = load_affine(in + 9 * i);
+    const u64* e = in + 117;
+    for (int i = 0; i < 5; ++i) memcpy(pf.wire_evals[i].l, e + 4 * i, 32);
+    for (int i = 0; i < 4; ++i) memcpy(pf.sigma_evals[i].l, e + 20 + 4 * i, 32);
+    memcpy(pf.z_shift_eval.l, e + 36, 32);
+    return pf;
+}
+
+void* orc_plonk_preprocess(u64 n, u64 num_public, const u64* selectors,
+                           const u64* sigma, const u64* srs_g1_records,
+                           u64 srs_points) {
+    OrcCircuitDesc d;
+    d.n = n;
+    d.num_public = num_public;
+    d.selectors = reinterpret_cast<const Fr*>(selectors);
+    d.sigma = sigma;
+    std::vector<G1Affine> srs(srs_points);
+    for (u64 i = 0; i < srs_points; ++i) srs[i] = load_affine(srs_g1_records + 9 * i);
+    try {
+        return orc_preprocess(d, srs);
+    } catch (...) {
+        return nullptr;
+    }
+}
+
+void orc_plonk_pk_free(void* pk) { delete static_cast<OrcProvingKey*>(pk); }
+
+int orc_plonk_prove(void* pk_, const u64* wires, const u64* pubs, u64 seed,
+                    u64* out157) {
+    auto* pk = static_cast<OrcProvingKey*>(pk_);
+    try {
+        OrcProof pf = orc_prove(*pk, reinterpret_cast<const Fr*>(wires),
+                                reinterpret_cast<const Fr*>(pubs), seed);
+        proof_store(pf, out157);
+        return 0;
+    } catch (...) {
+        return -1;
+    }
+}
+
+int orc_plonk_verify(void* pk_, const u64* pubs, const u64* proof157,
+                     const u64* tau_canonical) {
+    auto* pk = static_cast<OrcProvingKey*>(pk_);
+    Fr tau = Fr::from_canonical(tau_canonical);
+    try {
+        OrcProof pf = proof_load(proof157);
+        return orc_verify(*pk, reinterpret_cast<const Fr*>(pubs), pf, tau) ? 1 : 0;
+    } catch (...) {
+        return -1;
+    }
+}
+
+void orc_derive_tau(u64 seed, u64* out_canonical) {
+    Fr tau = derive_tau(seed);
+    tau.to_canonical(out_canonical);
+}
 
 }  // extern "C"

@@ -168,6 +168,10 @@ struct Fp4 {
         return pow(e);
     }
 
+    RNG_HD static Fp4 from_u64(u64 v) {
+        u64 c[4] = {v, 0, 0, 0};
+        return from_canonical(c);
+    }
     RNG_HD static Fp4 from_canonical(const u64 c[4]) {
         Fp4 a{{c[0], c[1], c[2], c[3]}};
         Fp4 r2{{P::r2[0], P::r2[1], P::r2[2], P::r2[3]}};

@@ -15,6 +15,8 @@
 #include "../../include/rng_prover.h"
 #include "ntt_kernels.hip"
 #include "msm_kernels.hip"
+#include "plonk_circuit.hpp"
+#include "test_circuits.hpp"
 
 namespace rng {
 
@@ -569,6 +571,43 @@ int rng_msm_g1(RngCtx* ctx, const uint64_t* bases, const uint64_t* scalars, uint
     hipFree(ds);
     return rc;
 }
+
+// ---- circuit construction (arithmetization front-end) ----
+// The tables handle wraps rng::CircuitTables; getters copy flat arrays out
+// so tests (and the oracle prover) consume identical inputs.
+
+void* rng_testcirc_build(uint64_t seed, uint64_t scale) {
+    try {
+        PlonkCircuit cs;
+        build_mixed_circuit(cs, seed, scale);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_testcirc_build: %s\n", why.c_str());
+            return nullptr;
+        }
+        auto* t = new CircuitTables(cs.finalize());
+        return t;
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_testcirc_build: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+uint64_t rng_circ_n(void* t) { return static_cast<CircuitTables*>(t)->n; }
+uint64_t rng_circ_npub(void* t) { return static_cast<CircuitTables*>(t)->num_public; }
+
+// selectors: 13*n*4 u64; sigma: 5*n u64; wires: 5*n*4 u64; pubs: npub*4 u64
+void rng_circ_get(void* t_, uint64_t* selectors, uint64_t* sigma, uint64_t* wires,
+                  uint64_t* pubs) {
+    auto* t = static_cast<CircuitTables*>(t_);
+    memcpy(selectors, t->selectors.data(), t->selectors.size() * sizeof(Fr));
+    memcpy(sigma, t->sigma.data(), t->sigma.size() * 8);
+    memcpy(wires, t->wires.data(), t->wires.size() * sizeof(Fr));
+    if (!t->public_inputs.empty())
+        memcpy(pubs, t->public_inputs.data(), t->public_inputs.size() * sizeof(Fr));
+}
+
+void rng_circ_free(void* t) { delete static_cast<CircuitTables*>(t); }
 
 // Plonk-layer entry points land with the plonk milestone (DESIGN.md roadmap);
 // fail loudly rather than silently succeed.
