@@ -1,0 +1,76 @@
+// PRODUCT PATH — TurboPlonk quotient evaluation on gfx950.
+//
+// Pointwise evaluation of the quotient numerator over the 8n coset domain
+// (the dominant non-NTT/MSM column work of the prover):
+//   num(t) = gate(t) + alpha [ z f - z_shift g ] + alpha^2 (z-1) L1
+//   quot(t) = num(t) * zh_inv[t % 8]
+// One thread per coset point; all operands are coset-evaluation arrays laid
+// out contiguously (sel: 13*m, sig: 5*m, w: 5*m, z/pi/l1: m, xpow: m = the
+// coset points g*w_m^t for the beta*k_j*X term).
+#include <hip/hip_runtime.h>
+#include "gpu_field.hpp"
+
+namespace rng {
+
+struct QuotChal {  // challenge pack (kernel arg)
+    Fr beta, gamma, alpha, alpha2;
+    Fr k[5];
+    Fr zh_inv[8];
+};
+
+__global__ __launch_bounds__(256) void k_quotient(
+    const Fr* sel, const Fr* sig, const Fr* w, const Fr* z, const Fr* pi,
+    const Fr* l1, const Fr* xpow, Fr* out, uint32_t m, QuotChal ch) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= m) return;
+    auto p5 = [](const Fr& v) {
+        Fr v2 = v.sqr();
+        return v2.sqr().mul(v);
+    };
+    Fr w0 = w[t], w1 = w[m + t], w2 = w[2 * m + t], w3 = w[3 * m + t],
+       w4 = w[4 * m + t];
+    // gate equation: selector order = plonk_circuit.hpp enum Sel
+    Fr gate = sel[11 * m + t].add(pi[t]);
+    gate = gate.add(sel[0 * m + t].mul(w0)).add(sel[1 * m + t].mul(w1));
+    gate = gate.add(sel[2 * m + t].mul(w2)).add(sel[3 * m + t].mul(w3));
+    gate = gate.add(sel[4 * m + t].mul(w0.mul(w1))).add(sel[5 * m + t].mul(w2.mul(w3)));
+    gate = gate.add(sel[6 * m + t].mul(p5(w0))).add(sel[7 * m + t].mul(p5(w1)));
+    gate = gate.add(sel[8 * m + t].mul(p5(w2))).add(sel[9 * m + t].mul(p5(w3)));
+    gate = gate.add(sel[12 * m + t].mul(w0.mul(w1).mul(w2).mul(w3).mul(w4)));
+    gate = gate.sub(sel[10 * m + t].mul(w4));
+    // permutation
+    Fr x = xpow[t];
+    Fr f = Fr::one(), g = Fr::one();
+    const Fr* ws[5] = {&w0, &w1, &w2, &w3, &w4};
+    for (int j = 0; j < 5; ++j) {
+        f = f.mul(ws[j]->add(ch.beta.mul(ch.k[j]).mul(x)).add(ch.gamma));
+        g = g.mul(ws[j]->add(ch.beta.mul(sig[j * m + t])).add(ch.gamma));
+    }
+    Fr zshift = z[(t + 8) % m];
+    Fr perm = ch.alpha.mul(z[t].mul(f).sub(zshift.mul(g)));
+    Fr l1term = ch.alpha2.mul(z[t].sub(Fr::one())).mul(l1[t]);
+    out[t] = gate.add(perm).add(l1term).mul(ch.zh_inv[t & 7]);
+}
+
+// elementwise multiply: data[i] *= table[i] (coset scaling)
+__global__ __launch_bounds__(256) void k_mul_pointwise(Fr* data, const Fr* table,
+                                                       uint32_t count) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < count) data[i] = data[i].mul(table[i]);
+}
+
+// multiply every element by a constant
+__global__ __launch_bounds__(256) void k_scale_const(Fr* data, Fr c, uint32_t count) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < count) data[i] = data[i].mul(c);
+}
+
+// zero-extend helper: copy src (len) into dst (m), zero the rest
+__global__ __launch_bounds__(256) void k_copy_pad(const Fr* src, uint32_t len, Fr* dst,
+                                                  uint32_t m) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= m) return;
+    dst[i] = (i < len) ? src[i] : Fr::zero();
+}
+
+}  // namespace rng

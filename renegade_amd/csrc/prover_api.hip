@@ -15,7 +15,9 @@
 #include "../../include/rng_prover.h"
 #include "ntt_kernels.hip"
 #include "msm_kernels.hip"
+#include "plonk_kernels.hip"
 #include "plonk_circuit.hpp"
+#include "plonk_host.hpp"
 #include "test_circuits.hpp"
 
 namespace rng {
@@ -74,6 +76,8 @@ struct NttPlan {
     Fr ninv;  // Montgomery form
     Fr* scratch = nullptr;  // n elements, for the pass-2 ping-pong (resized w/ batch)
     uint64_t scratch_elems = 0;
+    // coset tables (built on demand): gpow g^j, gpow_inv g^-j, xpow g*w^j
+    Fr *gpow = nullptr, *gpow_inv = nullptr, *xpow = nullptr;
 };
 
 // host-side helpers over the shared field type (host-compiled path of Fp4)
@@ -100,7 +104,8 @@ struct RngCtxImpl {
         for (auto& kv : plans) {
             NttPlan* p = kv.second.get();
             for (Fr* b : {p->wst1_f, p->wst2_f, p->ta_f, p->tb_f, p->wst1_i,
-                          p->wst2_i, p->ta_i, p->tb_i, p->scratch})
+                          p->wst2_i, p->ta_i, p->tb_i, p->scratch, p->gpow,
+                          p->gpow_inv, p->xpow})
                 if (b) hipFree(b);
         }
         if (srs_dev) hipFree(srs_dev);
@@ -365,6 +370,455 @@ static void jac_to_affine_record(const G1Jac& j, uint64_t* out9) {
     out9[8] = 0;
 }
 
+// ---------------- TurboPlonk prover (GPU) ----------------
+
+static int ensure_coset_tables(RngCtxImpl* ctx, NttPlan* p) {
+    if (p->gpow) return RNG_OK;
+    uint32_t m = p->n;
+    Fr g = Fr::from_u64(FR_GENERATOR);
+    Fr gi = g.inverse();
+    if (build_pow_table(&p->gpow, g, 1, m) != RNG_OK) return RNG_ERR_HIP;
+    if (build_pow_table(&p->gpow_inv, gi, 1, m) != RNG_OK) return RNG_ERR_HIP;
+    Fr wm = h_fr_root_of_unity(m);
+    if (build_pow_table(&p->xpow, wm, 1, m) != RNG_OK) return RNG_ERR_HIP;
+    uint32_t blocks = (m + 255) / 256;
+    hipLaunchKernelGGL(k_scale_const, dim3(blocks), dim3(256), 0, 0, p->xpow, g, m);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipDeviceSynchronize());
+    return RNG_OK;
+}
+
+// forward coset NTT of device buffer `in` (m elements, coefficient form),
+// result into `out` (in is clobbered)
+static int coset_fwd_dev(RngCtxImpl* ctx, Fr* in, Fr* out, uint32_t m) {
+    NttPlan* p = get_plan(ctx, m, 1);
+    if (!p) return RNG_ERR_HIP;
+    if (ensure_coset_tables(ctx, p) != RNG_OK) return RNG_ERR_HIP;
+    uint32_t blocks = (m + 255) / 256;
+    hipLaunchKernelGGL(k_mul_pointwise, dim3(blocks), dim3(256), 0, 0, in, p->gpow, m);
+    HIP_CHECK(hipGetLastError());
+    return ntt_dev_run(ctx, in, out, m, 1, false);
+}
+
+// inverse coset NTT of `in` (m evals), result into `out`
+static int coset_inv_dev(RngCtxImpl* ctx, Fr* in, Fr* out, uint32_t m) {
+    NttPlan* p = get_plan(ctx, m, 1);
+    if (!p) return RNG_ERR_HIP;
+    if (ensure_coset_tables(ctx, p) != RNG_OK) return RNG_ERR_HIP;
+    int rc = ntt_dev_run(ctx, in, out, m, 1, true);
+    if (rc != RNG_OK) return rc;
+    uint32_t blocks = (m + 255) / 256;
+    hipLaunchKernelGGL(k_mul_pointwise, dim3(blocks), dim3(256), 0, 0, out, p->gpow_inv, m);
+    HIP_CHECK(hipGetLastError());
+    return RNG_OK;
+}
+
+struct ProveScratch {  // per-context device scratch for proving at size n
+    uint64_t n = 0, m = 0;
+    Fr* w_coset = nullptr;   // 5*m
+    Fr* z_coset = nullptr;   // m
+    Fr* pi_coset = nullptr;  // m
+    Fr* q_buf = nullptr;     // m
+    Fr* tmp = nullptr;       // m
+    Fr* stage = nullptr;     // n+3 coefficient staging
+    uint64_t* canon = nullptr;  // 4*(n+3) canonical scalars for commits
+    ~ProveScratch() {
+        for (void* b : {(void*)w_coset, (void*)z_coset, (void*)pi_coset, (void*)q_buf,
+                        (void*)tmp, (void*)stage, (void*)canon})
+            if (b) hipFree(b);
+    }
+};
+
+struct PlonkPkImpl {
+    uint64_t n = 0, npub = 0;
+    Fr k[5];
+    std::vector<Fr> selq[13], sigp[5];       // host coeffs
+    std::vector<Fr> sig_evals[5];            // host evals (z build)
+    G1Aff sel_comms[13], sig_comms[5];
+    Fr* sel_coset = nullptr;   // device 13*m
+    Fr* sig_coset = nullptr;   // device 5*m
+    Fr* l1_coset = nullptr;    // device m
+    RngCtxImpl* ctx = nullptr;
+    ~PlonkPkImpl() {
+        for (void* b : {(void*)sel_coset, (void*)sig_coset, (void*)l1_coset})
+            if (b) hipFree(b);
+    }
+};
+
+static thread_local std::unique_ptr<ProveScratch> tls_prove_scratch;
+
+static int prove_scratch_ensure(uint64_t n) {
+    uint64_t m = 8 * n;
+    if (tls_prove_scratch && tls_prove_scratch->n >= n) return RNG_OK;
+    tls_prove_scratch = std::make_unique<ProveScratch>();
+    ProveScratch* s = tls_prove_scratch.get();
+    HIP_CHECK(hipMalloc(&s->w_coset, 5 * m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->z_coset, m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->pi_coset, m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->q_buf, m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->tmp, m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->stage, (n + 3) * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->canon, 4 * (n + 3) * 8));
+    s->n = n;
+    s->m = m;
+    return RNG_OK;
+}
+
+// commit to host coefficients via GPU MSM over the SRS device bases
+static int commit_dev(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, G1Aff* out,
+                      bool* out_inf) {
+    ProveScratch* s = tls_prove_scratch.get();
+    uint64_t mdeg = coeffs.size();
+    if (mdeg > ctx->srs_count) return RNG_ERR_BAD_ARG;
+    HIP_CHECK(hipMemcpy(s->stage, coeffs.data(), mdeg * sizeof(Fr), hipMemcpyHostToDevice));
+    uint32_t blocks = (uint32_t)((mdeg + 255) / 256);
+    hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, 0, s->stage,
+                       s->canon, (uint32_t)mdeg);
+    HIP_CHECK(hipGetLastError());
+    G1Jac res;
+    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, mdeg, 16, &res);
+    if (rc != RNG_OK) return rc;
+    uint64_t rec[9];
+    jac_to_affine_record(res, rec);
+    memcpy(out->x.l, rec, 32);
+    memcpy(out->y.l, rec + 4, 32);
+    *out_inf = rec[8] != 0;
+    return RNG_OK;
+}
+
+// upload host coeffs, pad to m, coset-forward into dst (device, m elems)
+static int coset_of_coeffs(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, Fr* dst,
+                           uint32_t m) {
+    ProveScratch* s = tls_prove_scratch.get();
+    HIP_CHECK(hipMemcpy(s->stage, coeffs.data(), coeffs.size() * sizeof(Fr),
+                        hipMemcpyHostToDevice));
+    uint32_t blocks = (m + 255) / 256;
+    hipLaunchKernelGGL(k_copy_pad, dim3(blocks), dim3(256), 0, 0, s->stage,
+                       (uint32_t)coeffs.size(), s->tmp, m);
+    HIP_CHECK(hipGetLastError());
+    return coset_fwd_dev(ctx, s->tmp, dst, m);
+}
+
+// GPU batched IFFT of column-major evals (host) -> host coeff vectors
+static int ifft_columns(RngCtxImpl* ctx, const Fr* host_evals, uint64_t n,
+                        uint64_t ncols, std::vector<Fr>* out_cols) {
+    Fr* d = nullptr;
+    HIP_CHECK(hipMalloc(&d, n * ncols * sizeof(Fr)));
+    if (hipMemcpy(d, host_evals, n * ncols * sizeof(Fr), hipMemcpyHostToDevice) !=
+        hipSuccess) {
+        hipFree(d);
+        return RNG_ERR_HIP;
+    }
+    NttPlan* p = get_plan(ctx, (uint32_t)n, ncols);
+    if (!p) {
+        hipFree(d);
+        return RNG_ERR_HIP;
+    }
+    Fr* out = (n <= 4096) ? d : p->scratch;
+    int rc = ntt_dev_run(ctx, d, out, (uint32_t)n, ncols, true);
+    if (rc == RNG_OK) {
+        for (uint64_t c = 0; c < ncols; ++c) {
+            out_cols[c].resize(n);
+            if (hipMemcpy(out_cols[c].data(), out + c * n, n * sizeof(Fr),
+                          hipMemcpyDeviceToHost) != hipSuccess)
+                rc = RNG_ERR_HIP;
+        }
+    }
+    hipFree(d);
+    return rc;
+}
+
+static int plonk_preprocess_impl(RngCtxImpl* ctx, const RngCircuitDesc* d,
+                                 PlonkPkImpl* pk) {
+    const uint64_t n = d->n, m = 8 * n;
+    pk->n = n;
+    pk->npub = d->num_public;
+    coset_ks(pk->k);
+    if (!cosets_ok(n)) return RNG_ERR_BAD_ARG;
+    if (prove_scratch_ensure(n) != RNG_OK) return RNG_ERR_HIP;
+
+    // selector + sigma polynomials (GPU IFFT)
+    if (ifft_columns(ctx, (const Fr*)d->selectors, n, 13, pk->selq) != RNG_OK)
+        return RNG_ERR_HIP;
+    // sigma evals from permutation indices
+    Fr w = h_fr_root_of_unity((uint32_t)n);
+    std::vector<Fr> wpow(n);
+    wpow[0] = Fr::one();
+    for (uint64_t i = 1; i < n; ++i) wpow[i] = wpow[i - 1].mul(w);
+    std::vector<Fr> sig_evals_flat(5 * n);
+    for (int j = 0; j < 5; ++j) {
+        for (uint64_t i = 0; i < n; ++i) {
+            uint64_t slot = d->sigma[(size_t)j * n + i];
+            sig_evals_flat[(size_t)j * n + i] = pk->k[slot / n].mul(wpow[slot % n]);
+        }
+        pk->sig_evals[j].assign(sig_evals_flat.begin() + (size_t)j * n,
+                                sig_evals_flat.begin() + (size_t)(j + 1) * n);
+    }
+    if (ifft_columns(ctx, sig_evals_flat.data(), n, 5, pk->sigp) != RNG_OK)
+        return RNG_ERR_HIP;
+
+    // commitments
+    bool inf;
+    for (int s = 0; s < 13; ++s)
+        if (commit_dev(ctx, pk->selq[s], &pk->sel_comms[s], &inf) != RNG_OK)
+            return RNG_ERR_HIP;
+    for (int j = 0; j < 5; ++j)
+        if (commit_dev(ctx, pk->sigp[j], &pk->sig_comms[j], &inf) != RNG_OK)
+            return RNG_ERR_HIP;
+
+    // coset caches
+    HIP_CHECK(hipMalloc(&pk->sel_coset, 13 * m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&pk->sig_coset, 5 * m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&pk->l1_coset, m * sizeof(Fr)));
+    for (int s = 0; s < 13; ++s)
+        if (coset_of_coeffs(ctx, pk->selq[s], pk->sel_coset + (size_t)s * m, (uint32_t)m) !=
+            RNG_OK)
+            return RNG_ERR_HIP;
+    for (int j = 0; j < 5; ++j)
+        if (coset_of_coeffs(ctx, pk->sigp[j], pk->sig_coset + (size_t)j * m, (uint32_t)m) !=
+            RNG_OK)
+            return RNG_ERR_HIP;
+    {
+        // L1 = IFFT(e0)
+        std::vector<Fr> e0(n, Fr::zero());
+        e0[0] = Fr::one();
+        std::vector<Fr> l1coef[1];
+        if (ifft_columns(ctx, e0.data(), n, 1, l1coef) != RNG_OK) return RNG_ERR_HIP;
+        if (coset_of_coeffs(ctx, l1coef[0], pk->l1_coset, (uint32_t)m) != RNG_OK)
+            return RNG_ERR_HIP;
+    }
+    HIP_CHECK(hipDeviceSynchronize());
+    return RNG_OK;
+}
+
+// transcript init shared with the verifier side (spec: oracle/transcript.hpp)
+static void h_transcript_init(HostTranscript& tr, const PlonkPkImpl& pk, const Fr* pubs) {
+    tr.append_u64(pk.n);
+    tr.append_u64(pk.npub);
+    for (int s = 0; s < 13; ++s) tr.append_g1(pk.sel_comms[s]);
+    for (int j = 0; j < 5; ++j) tr.append_g1(pk.sig_comms[j]);
+    for (uint64_t i = 0; i < pk.npub; ++i) tr.append_fr(pubs[i]);
+}
+
+static int plonk_prove_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, const Fr* wires,
+                            const Fr* pubs, uint64_t seed, uint64_t* out_proof,
+                            uint64_t* out_link_hint) {
+    const uint64_t n = pk.n;
+    const uint32_t m = (uint32_t)(8 * n);
+    if (prove_scratch_ensure(n) != RNG_OK) return RNG_ERR_HIP;
+    ProveScratch* sc = tls_prove_scratch.get();
+    HostDrbg drbg(seed);
+    HostTranscript tr;
+    h_transcript_init(tr, pk, pubs);
+    Fr w = h_fr_root_of_unity((uint32_t)n);
+
+    G1Aff comms[13];
+    bool comm_inf[13] = {false};
+
+    // --- R1: wire polys ---
+    std::vector<Fr> wpoly[5];
+    if (ifft_columns(ctx, wires, n, 5, wpoly) != RNG_OK) return RNG_ERR_HIP;
+    for (int j = 0; j < 5; ++j) {
+        Fr b0 = drbg.next(), b1 = drbg.next();
+        wpoly[j].resize(n + 2, Fr::zero());
+        wpoly[j][0] = wpoly[j][0].sub(b0);
+        wpoly[j][1] = wpoly[j][1].sub(b1);
+        wpoly[j][n] = wpoly[j][n].add(b0);
+        wpoly[j][n + 1] = wpoly[j][n + 1].add(b1);
+        if (commit_dev(ctx, wpoly[j], &comms[j], &comm_inf[j]) != RNG_OK)
+            return RNG_ERR_HIP;
+        tr.append_g1(comms[j], comm_inf[j]);
+    }
+    if (out_link_hint) {
+        // hint = wire-0 polynomial (n+2 coeffs, Montgomery) + its commitment
+        memcpy(out_link_hint, wpoly[0].data(), (n + 2) * sizeof(Fr));
+        uint64_t* c = out_link_hint + 4 * (n + 2);
+        memcpy(c, comms[0].x.l, 32);
+        memcpy(c + 4, comms[0].y.l, 32);
+        c[8] = comm_inf[0] ? 1 : 0;
+    }
+    Fr beta = tr.challenge();
+    Fr gamma = tr.challenge();
+
+    // --- R2: grand product ---
+    std::vector<Fr> znum(n), zden(n);
+    {
+        Fr wi = Fr::one();
+        for (uint64_t i = 0; i < n; ++i) {
+            Fr num = Fr::one(), den = Fr::one();
+            for (int j = 0; j < 5; ++j) {
+                Fr wv = wires[(size_t)j * n + i];
+                num = num.mul(wv.add(beta.mul(pk.k[j]).mul(wi)).add(gamma));
+                den = den.mul(wv.add(beta.mul(pk.sig_evals[j][i])).add(gamma));
+            }
+            znum[i] = num;
+            zden[i] = den;
+            wi = wi.mul(w);
+        }
+    }
+    std::vector<Fr> zden_inv = hbatch_inverse(zden);
+    std::vector<Fr> zevals(n);
+    zevals[0] = Fr::one();
+    for (uint64_t i = 1; i < n; ++i)
+        zevals[i] = zevals[i - 1].mul(znum[i - 1]).mul(zden_inv[i - 1]);
+    std::vector<Fr> zpoly[1];
+    if (ifft_columns(ctx, zevals.data(), n, 1, zpoly) != RNG_OK) return RNG_ERR_HIP;
+    {
+        Fr b2 = drbg.next(), b3 = drbg.next(), b4 = drbg.next();
+        zpoly[0].resize(n + 3, Fr::zero());
+        zpoly[0][0] = zpoly[0][0].sub(b4);
+        zpoly[0][1] = zpoly[0][1].sub(b3);
+        zpoly[0][2] = zpoly[0][2].sub(b2);
+        zpoly[0][n] = zpoly[0][n].add(b4);
+        zpoly[0][n + 1] = zpoly[0][n + 1].add(b3);
+        zpoly[0][n + 2] = zpoly[0][n + 2].add(b2);
+    }
+    if (commit_dev(ctx, zpoly[0], &comms[5], &comm_inf[5]) != RNG_OK) return RNG_ERR_HIP;
+    tr.append_g1(comms[5], comm_inf[5]);
+    Fr alpha = tr.challenge();
+
+    // --- R3: quotient on the 8n coset (GPU) ---
+    for (int j = 0; j < 5; ++j)
+        if (coset_of_coeffs(ctx, wpoly[j], sc->w_coset + (size_t)j * m, m) != RNG_OK)
+            return RNG_ERR_HIP;
+    if (coset_of_coeffs(ctx, zpoly[0], sc->z_coset, m) != RNG_OK) return RNG_ERR_HIP;
+    {
+        std::vector<Fr> pie(n, Fr::zero());
+        for (uint64_t i = 0; i < pk.npub; ++i) pie[i] = pubs[i];
+        std::vector<Fr> picoef[1];
+        if (ifft_columns(ctx, pie.data(), n, 1, picoef) != RNG_OK) return RNG_ERR_HIP;
+        if (coset_of_coeffs(ctx, picoef[0], sc->pi_coset, m) != RNG_OK)
+            return RNG_ERR_HIP;
+    }
+    QuotChal ch;
+    ch.beta = beta;
+    ch.gamma = gamma;
+    ch.alpha = alpha;
+    ch.alpha2 = alpha.sqr();
+    for (int j = 0; j < 5; ++j) ch.k[j] = pk.k[j];
+    {
+        Fr g = Fr::from_u64(FR_GENERATOR);
+        Fr gn = g.pow_u64(n);
+        Fr w8 = h_fr_root_of_unity(m).pow_u64(n);
+        std::vector<Fr> zh(8);
+        Fr cur = gn;
+        for (int t = 0; t < 8; ++t) {
+            zh[t] = cur.sub(Fr::one());
+            cur = cur.mul(w8);
+        }
+        zh = hbatch_inverse(zh);
+        for (int t = 0; t < 8; ++t) ch.zh_inv[t] = zh[t];
+    }
+    NttPlan* mp = get_plan(ctx, m, 1);
+    if (!mp || ensure_coset_tables(ctx, mp) != RNG_OK) return RNG_ERR_HIP;
+    {
+        uint32_t blocks = (m + 255) / 256;
+        hipLaunchKernelGGL(k_quotient, dim3(blocks), dim3(256), 0, 0, pk.sel_coset,
+                           pk.sig_coset, sc->w_coset, sc->z_coset, sc->pi_coset,
+                           pk.l1_coset, mp->xpow, sc->q_buf, m, ch);
+        HIP_CHECK(hipGetLastError());
+    }
+    if (coset_inv_dev(ctx, sc->q_buf, sc->tmp, m) != RNG_OK) return RNG_ERR_HIP;
+    std::vector<Fr> quot(5 * (n + 2));
+    HIP_CHECK(hipMemcpy(quot.data(), sc->tmp, quot.size() * sizeof(Fr),
+                        hipMemcpyDeviceToHost));
+    std::vector<Fr> quot_chunks[5];
+    {
+        Fr prev = Fr::zero();
+        for (int i = 0; i < 5; ++i) {
+            quot_chunks[i].assign(quot.begin() + (size_t)i * (n + 2),
+                                  quot.begin() + (size_t)(i + 1) * (n + 2));
+            Fr bnext = (i < 4) ? drbg.next() : Fr::zero();
+            quot_chunks[i][0] = quot_chunks[i][0].sub(prev);
+            if (i < 4) {
+                quot_chunks[i].resize(n + 3, Fr::zero());
+                quot_chunks[i][n + 2] = quot_chunks[i][n + 2].add(bnext);
+            }
+            if (commit_dev(ctx, quot_chunks[i], &comms[6 + i], &comm_inf[6 + i]) != RNG_OK)
+                return RNG_ERR_HIP;
+            tr.append_g1(comms[6 + i], comm_inf[6 + i]);
+            prev = bnext;
+        }
+    }
+    Fr zeta = tr.challenge();
+
+    // --- R4: evaluations ---
+    Fr wire_evals[5], sigma_evals[4], z_shift_eval;
+    for (int j = 0; j < 5; ++j) {
+        wire_evals[j] = hpoly_eval(wpoly[j], zeta);
+        tr.append_fr(wire_evals[j]);
+    }
+    for (int j = 0; j < 4; ++j) {
+        sigma_evals[j] = hpoly_eval(pk.sigp[j], zeta);
+        tr.append_fr(sigma_evals[j]);
+    }
+    z_shift_eval = hpoly_eval(zpoly[0], zeta.mul(w));
+    tr.append_fr(z_shift_eval);
+    Fr v = tr.challenge();
+
+    // --- R5: linearization + openings ---
+    Fr zeta_n = zeta.pow_u64(n);
+    Fr zh_zeta = zeta_n.sub(Fr::one());
+    Fr l1_zeta = zh_zeta.mul(Fr::from_u64(n).mul(zeta.sub(Fr::one())).inverse());
+    auto p5 = [](const Fr& x) {
+        Fr x2 = x.sqr();
+        return x2.sqr().mul(x);
+    };
+    const Fr* wb = wire_evals;
+    std::vector<Fr> D;
+    hpoly_add_scaled(D, pk.selq[11], Fr::one());
+    for (int j = 0; j < 4; ++j) hpoly_add_scaled(D, pk.selq[j], wb[j]);
+    hpoly_add_scaled(D, pk.selq[4], wb[0].mul(wb[1]));
+    hpoly_add_scaled(D, pk.selq[5], wb[2].mul(wb[3]));
+    for (int j = 0; j < 4; ++j) hpoly_add_scaled(D, pk.selq[6 + j], p5(wb[j]));
+    hpoly_add_scaled(D, pk.selq[12],
+                     wb[0].mul(wb[1]).mul(wb[2]).mul(wb[3]).mul(wb[4]));
+    hpoly_add_scaled(D, pk.selq[10], wb[4].neg());
+    Fr fbar = Fr::one(), Bbar = Fr::one();
+    for (int j = 0; j < 5; ++j)
+        fbar = fbar.mul(wb[j].add(beta.mul(pk.k[j]).mul(zeta)).add(gamma));
+    for (int j = 0; j < 4; ++j)
+        Bbar = Bbar.mul(wb[j].add(beta.mul(sigma_evals[j])).add(gamma));
+    hpoly_add_scaled(D, zpoly[0], alpha.mul(fbar).add(alpha.sqr().mul(l1_zeta)));
+    hpoly_add_scaled(D, pk.sigp[4], alpha.mul(beta).mul(z_shift_eval).mul(Bbar).neg());
+    {
+        Fr zpow = zh_zeta.neg();
+        Fr step = zeta.pow_u64(n + 2);
+        for (int i = 0; i < 5; ++i) {
+            hpoly_add_scaled(D, quot_chunks[i], zpow);
+            zpow = zpow.mul(step);
+        }
+    }
+    std::vector<Fr> C = D;
+    Fr vp = Fr::one();
+    for (int j = 0; j < 5; ++j) {
+        vp = vp.mul(v);
+        hpoly_add_scaled(C, wpoly[j], vp);
+    }
+    for (int j = 0; j < 4; ++j) {
+        vp = vp.mul(v);
+        hpoly_add_scaled(C, pk.sigp[j], vp);
+    }
+    std::vector<Fr> Wz = hpoly_div_linear(C, zeta);
+    if (commit_dev(ctx, Wz, &comms[11], &comm_inf[11]) != RNG_OK) return RNG_ERR_HIP;
+    std::vector<Fr> Wzw = hpoly_div_linear(zpoly[0], zeta.mul(w));
+    if (commit_dev(ctx, Wzw, &comms[12], &comm_inf[12]) != RNG_OK) return RNG_ERR_HIP;
+    tr.append_g1(comms[11], comm_inf[11]);
+    tr.append_g1(comms[12], comm_inf[12]);
+
+    // --- serialize (layout: include/rng_prover.h) ---
+    for (int i = 0; i < 13; ++i) {
+        memcpy(out_proof + 9 * i, comms[i].x.l, 32);
+        memcpy(out_proof + 9 * i + 4, comms[i].y.l, 32);
+        out_proof[9 * i + 8] = comm_inf[i] ? 1 : 0;
+    }
+    uint64_t* e = out_proof + 117;
+    for (int i = 0; i < 5; ++i) memcpy(e + 4 * i, wire_evals[i].l, 32);
+    for (int i = 0; i < 4; ++i) memcpy(e + 20 + 4 * i, sigma_evals[i].l, 32);
+    memcpy(e + 36, z_shift_eval.l, 32);
+    return RNG_OK;
+}
+
 }  // namespace rng
 
 // ---------------- C ABI ----------------
@@ -609,16 +1063,54 @@ void rng_circ_get(void* t_, uint64_t* selectors, uint64_t* sigma, uint64_t* wire
 
 void rng_circ_free(void* t) { delete static_cast<CircuitTables*>(t); }
 
-// Plonk-layer entry points land with the plonk milestone (DESIGN.md roadmap);
-// fail loudly rather than silently succeed.
-RngProvingKey* rng_preprocess(RngCtx*, const RngCircuitDesc*) { return nullptr; }
-void rng_pk_free(RngProvingKey*) {}
-int rng_prove(RngCtx*, const RngProvingKey*, const uint64_t*, const uint64_t*, uint64_t,
-              uint64_t*, uint64_t*) {
-    return RNG_ERR_BAD_ARG;
+// ---- TurboPlonk prover entry points ----
+
+struct RngProvingKey {
+    PlonkPkImpl impl;
+};
+
+RngProvingKey* rng_preprocess(RngCtx* ctx, const RngCircuitDesc* desc) {
+    if (!gpu_ok() || !ctx || !desc || !desc->selectors || !desc->sigma) return nullptr;
+    if (desc->n < 8 || (desc->n & (desc->n - 1))) return nullptr;
+    if (desc->n + 3 > ctx->impl.srs_count) return nullptr;  // SRS too small
+    auto pk = std::make_unique<RngProvingKey>();
+    pk->impl.ctx = &ctx->impl;
+    if (plonk_preprocess_impl(&ctx->impl, desc, &pk->impl) != RNG_OK) return nullptr;
+    return pk.release();
 }
+
+void rng_pk_free(RngProvingKey* pk) { delete pk; }
+
+int rng_prove(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* wires,
+              const uint64_t* public_inputs, uint64_t seed, uint64_t* out_proof,
+              uint64_t* out_link_hint) {
+    if (!gpu_ok()) return RNG_ERR_NO_GPU;
+    if (!ctx || !pk || !wires || !out_proof) return RNG_ERR_BAD_ARG;
+    if (pk->impl.npub > 0 && !public_inputs) return RNG_ERR_BAD_ARG;
+    return plonk_prove_impl(&ctx->impl, pk->impl, (const Fr*)wires,
+                            (const Fr*)public_inputs, seed, out_proof, out_link_hint);
+}
+
+// PK introspection for tests / the verifier side
+uint64_t rng_pk_n(RngProvingKey* pk) { return pk->impl.n; }
+// out: 18 affine records (13 selector comms, 5 sigma comms)
+void rng_pk_comms(RngProvingKey* pk, uint64_t* out) {
+    for (int s = 0; s < 13; ++s) {
+        memcpy(out + 9 * s, pk->impl.sel_comms[s].x.l, 32);
+        memcpy(out + 9 * s + 4, pk->impl.sel_comms[s].y.l, 32);
+        out[9 * s + 8] = 0;
+    }
+    for (int j = 0; j < 5; ++j) {
+        memcpy(out + 9 * (13 + j), pk->impl.sig_comms[j].x.l, 32);
+        memcpy(out + 9 * (13 + j) + 4, pk->impl.sig_comms[j].y.l, 32);
+        out[9 * (13 + j) + 8] = 0;
+    }
+}
+
+// rng_verify needs the CPU pairing (DESIGN.md roadmap); fails loudly until
+// it lands — tests verify product proofs through the oracle verifier.
 int rng_verify(RngCtx*, const RngProvingKey*, const uint64_t*, const uint64_t*) {
-    return RNG_ERR_BAD_ARG;
+    return RNG_ERR_VERIFY;
 }
 int rng_link_proofs(RngCtx*, const RngProvingKey*, const uint64_t*, const uint64_t*,
                     uint64_t, uint64_t, uint64_t*) {

@@ -1,0 +1,136 @@
+"""GPU prover parity: product rng_prove vs oracle prover, bit-exact
+(SURVEY.md §8c contract (i)), plus oracle-verifier acceptance of GPU proofs.
+"""
+import ctypes
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+U64P = ctypes.POINTER(ctypes.c_uint64)
+
+
+def ptr(a):
+    return a.ctypes.data_as(U64P)
+
+
+@pytest.fixture(scope="module")
+def setup(orc):
+    from renegade_amd import load_prover
+    plib = load_prover()
+    if not plib.gpu_available:
+        pytest.skip("no GPU")
+    lib = plib.lib
+    lib.rng_testcirc_build.restype = ctypes.c_void_p
+    lib.rng_testcirc_build.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    lib.rng_preprocess.restype = ctypes.c_void_p
+    lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                              ctypes.c_uint64, U64P, U64P]
+    lib.rng_pk_free.argtypes = [ctypes.c_void_p]
+    lib.rng_pk_n.restype = ctypes.c_uint64
+    lib.rng_pk_n.argtypes = [ctypes.c_void_p]
+    lib.rng_pk_comms.argtypes = [ctypes.c_void_p, U64P]
+
+    h = lib.rng_testcirc_build(777, 6)
+    assert h
+    n = lib.rng_circ_n(h)
+    npub = lib.rng_circ_npub(h)
+    sel = np.zeros(13 * n * 4, dtype=np.uint64)
+    sigma = np.zeros(5 * n, dtype=np.uint64)
+    wires = np.zeros(5 * n * 4, dtype=np.uint64)
+    pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+    lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+    lib.rng_circ_free(h)
+
+    power = max(4, int(n).bit_length())
+    ptau = orc.srs_generate_ptau(power, seed=42)
+    max_degree = (1 << power) + 2
+    g1, _, _ = orc.srs_parse(ptau, max_degree)
+    srs_records = np.ascontiguousarray(g1).reshape(-1)
+
+    ctx = plib.init(ptau, max_degree)
+
+    # product circuit-desc struct
+    class Desc(ctypes.Structure):
+        _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
+                    ("selectors", U64P), ("sigma", U64P),
+                    ("num_link_groups", ctypes.c_uint64), ("link_offsets", U64P)]
+
+    desc = Desc(n, npub, ptr(sel), ptr(sigma), 0, None)
+    pk = lib.rng_preprocess(ctx.h, ctypes.byref(desc))
+    assert pk, "rng_preprocess failed"
+
+    o = orc.lib
+    o.orc_plonk_preprocess.restype = ctypes.c_void_p
+    o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P, U64P,
+                                       U64P, ctypes.c_uint64]
+    o.orc_plonk_prove.argtypes = [ctypes.c_void_p, U64P, U64P, ctypes.c_uint64, U64P]
+    o.orc_plonk_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P]
+    o.orc_derive_tau.argtypes = [ctypes.c_uint64, U64P]
+    opk = o.orc_plonk_preprocess(n, npub, ptr(sel), ptr(sigma), ptr(srs_records),
+                                 max_degree + 1)
+    assert opk
+    tau = np.zeros(4, dtype=np.uint64)
+    o.orc_derive_tau(42, ptr(tau))
+    return dict(plib=plib, lib=lib, orc=orc, ctx=ctx, pk=pk, opk=opk, n=n, npub=npub,
+                wires=wires, pubs=pubs, tau=tau)
+
+
+def gpu_prove(s, seed=9, with_hint=False):
+    proof = np.zeros(157, dtype=np.uint64)
+    hint = np.zeros(4 * (s["n"] + 2) + 9, dtype=np.uint64) if with_hint else None
+    rc = s["lib"].rng_prove(s["ctx"].h, ctypes.c_void_p(s["pk"]), ptr(s["wires"]),
+                            ptr(s["pubs"]), seed, ptr(proof),
+                            ptr(hint) if with_hint else None)
+    assert rc == 0, f"rng_prove rc={rc}"
+    return (proof, hint) if with_hint else proof
+
+
+class TestGpuProver:
+    def test_bit_exact_vs_oracle(self, setup):
+        s = setup
+        gpu_proof = gpu_prove(s, seed=9)
+        orc_proof = np.zeros(157, dtype=np.uint64)
+        rc = s["orc"].lib.orc_plonk_prove(ctypes.c_void_p(s["opk"]), ptr(s["wires"]),
+                                          ptr(s["pubs"]), ctypes.c_uint64(9),
+                                          ptr(orc_proof))
+        assert rc == 0
+        assert np.array_equal(gpu_proof, orc_proof), \
+            f"proof mismatch at {np.nonzero(gpu_proof != orc_proof)[0][:8]}"
+
+    def test_verifies(self, setup):
+        s = setup
+        proof = gpu_prove(s, seed=11)
+        ok = s["orc"].lib.orc_plonk_verify(ctypes.c_void_p(s["opk"]), ptr(s["pubs"]),
+                                           ptr(proof), ptr(s["tau"]))
+        assert ok == 1
+
+    def test_pk_comms_match_oracle(self, setup):
+        s = setup
+        comms = np.zeros(18 * 9, dtype=np.uint64)
+        s["lib"].rng_pk_comms(ctypes.c_void_p(s["pk"]), ptr(comms))
+        # oracle recomputes the same commitments inside preprocess; compare via
+        # a proof transcript round trip instead: already covered by bit-exact
+        # proof equality (transcript binds the comms). Check non-degenerate:
+        assert np.any(comms != 0)
+
+    def test_link_hint(self, setup):
+        s = setup
+        proof, hint = gpu_prove(s, seed=13, with_hint=True)
+        # hint commitment equals the first wire commitment in the proof
+        assert np.array_equal(hint[-9:], proof[:9])
+
+    def test_seed_determinism(self, setup):
+        s = setup
+        p1 = gpu_prove(s, seed=21)
+        p2 = gpu_prove(s, seed=21)
+        p3 = gpu_prove(s, seed=22)
+        assert np.array_equal(p1, p2)
+        assert not np.array_equal(p1, p3)
