@@ -31,6 +31,7 @@ def setup(orc):
     lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
     lib.rng_circ_free.argtypes = [ctypes.c_void_p]
     lib.rng_preprocess.restype = ctypes.c_void_p
+    lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
     lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
                               ctypes.c_uint64, U64P, U64P]
     lib.rng_pk_free.argtypes = [ctypes.c_void_p]
