@@ -24,7 +24,10 @@ struct Keccak256 {
         buflen = 0;
     }
 
-    static inline uint64_t rotl(uint64_t x, int n) { return (x << n) | (x >> (64 - n)); }
+    // n == 0 must not shift by 64 (UB; clang -O3 miscompiles it)
+    static inline uint64_t rotl(uint64_t x, int n) {
+        return n == 0 ? x : (x << n) | (x >> (64 - n));
+    }
 
     void permute() {
         static const uint64_t RC[24] = {

@@ -25,7 +25,10 @@ struct Keccak {
     uint8_t buf[136];
     size_t fill = 0;
 
-    static uint64_t rol(uint64_t x, int n) { return (x << n) | (x >> (64 - n)); }
+    // n == 0 must not shift by 64 (UB clang -O3 miscompiles)
+    static uint64_t rol(uint64_t x, int n) {
+        return n == 0 ? x : (x << n) | (x >> (64 - n));
+    }
 
     void f1600() {
         static const uint64_t RC[24] = {

@@ -1063,6 +1063,21 @@ void rng_circ_get(void* t_, uint64_t* selectors, uint64_t* sigma, uint64_t* wire
 
 void rng_circ_free(void* t) { delete static_cast<CircuitTables*>(t); }
 
+// host keccak-256 (exported for known-answer tests of the transcript hash)
+void rng_keccak256(const uint8_t* data, size_t len, uint8_t* out32) {
+    keccak256_h(data, len, out32);
+}
+
+// blinder DRBG block i for a given seed (Montgomery limbs out) — lets tests
+// pin the oracle/product DRBG spec equality on CPU
+void rng_debug_drbg(uint64_t seed, uint32_t n_blocks, uint64_t* out) {
+    HostDrbg d(seed);
+    for (uint32_t i = 0; i < n_blocks; ++i) {
+        Fr b = d.next();
+        memcpy(out + 4 * i, b.l, 32);
+    }
+}
+
 // ---- TurboPlonk prover entry points ----
 
 struct RngProvingKey {
