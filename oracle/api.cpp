@@ -17,6 +17,7 @@
 #include "keccak.hpp"
 #include "transcript.hpp"
 #include "plonk.hpp"
+#include "poseidon2.hpp"
 
 using namespace oracle;
 
@@ -231,6 +232,14 @@ int orc_plonk_verify(void* pk_, const u64* pubs, const u64* proof157,
     } catch (...) {
         return -1;
     }
+}
+
+// Poseidon2 hash (Montgomery limbs in/out)
+void orc_poseidon2_hash(const u64* in_mont, u64 n, u64* out_mont) {
+    std::vector<Fr> in(n);
+    memcpy(in.data(), in_mont, n * 32);
+    Fr r = poseidon2_hash(in.data(), n);
+    memcpy(out_mont, r.l, 32);
 }
 
 void orc_derive_tau(u64 seed, u64* out_canonical) {

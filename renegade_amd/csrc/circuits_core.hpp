@@ -1,0 +1,374 @@
+// PRODUCT PATH — gadget library + darkpool types + circuits (host).
+//
+// New implementations of the reference's zk-gadget and type layers needed by
+// the circuit family (citations inline):
+//  - PoseidonHashGadget with fused external/internal sbox+MDS gates
+//    (zk_gadgets/primitives/poseidon/hash.rs:69-423, gates.rs:73-179)
+//  - CSPRNG / stream-cipher / commitment / recovery-id gadgets
+//    (zk_gadgets/state_primitives/{csprng,stream_cipher,commitment,
+//     recovery_id}.rs)
+//  - PoseidonCSPRNG + StateWrapper native semantics
+//    (darkpool-types/src/{csprng.rs,state_wrapper.rs})
+//  - darkpool value types' scalar layouts (darkpool-types/src/{balance.rs,
+//    deposit.rs}; circuit-types/src/primitives/schnorr.rs)
+//  - the `Valid Balance Create` circuit
+//    (zk_circuits/valid_balance_create.rs:44-133) and its fixed-seed
+//    witness/statement builder mirroring test_helpers (:225-283).
+#pragma once
+#include <vector>
+#include "plonk_circuit.hpp"
+#include "poseidon2.hpp"
+#include "test_circuits.hpp"  // Lcg
+
+namespace rng {
+
+// =============================== gadgets ===============================
+
+// In-circuit Poseidon2 sponge with fused gates (hash.rs:69-423).
+struct PoseidonHashGadget {
+    std::array<Var, 3> state;
+    int next_index = 0;
+    bool squeezing = false;
+
+    explicit PoseidonHashGadget(PlonkCircuit& cs) {
+        state = {cs.zero(), cs.zero(), cs.zero()};
+    }
+    void reset(PlonkCircuit& cs) {
+        state = {cs.zero(), cs.zero(), cs.zero()};
+        next_index = 0;
+        squeezing = false;
+    }
+
+    // fused external sbox+MDS gate for one output element (gates.rs:73-101):
+    // out = rc + sbox?(curr) + sbox?(s0) + sbox?(s1) + sbox?(s2)
+    static Var fused_external(PlonkCircuit& cs, bool sbox, const Fr& rc, Var curr,
+                              const std::array<Var, 3>& s) {
+        auto p5v = [&](Var x) { return Poseidon2::sbox(cs.witness(x)); };
+        Fr out = rc;
+        if (sbox) {
+            out = out.add(p5v(curr)).add(p5v(s[0])).add(p5v(s[1])).add(p5v(s[2]));
+        } else {
+            out = out.add(cs.witness(curr)).add(cs.witness(s[0])).add(cs.witness(s[1]))
+                      .add(cs.witness(s[2]));
+        }
+        Var o = cs.create_variable(out);
+        std::array<Fr, NUM_SELECTORS> q{};
+        for (auto& x : q) x = Fr::zero();
+        if (sbox) {
+            q[SEL_HASH0] = q[SEL_HASH1] = q[SEL_HASH2] = q[SEL_HASH3] = Fr::one();
+        } else {
+            q[SEL_LC0] = q[SEL_LC1] = q[SEL_LC2] = q[SEL_LC3] = Fr::one();
+        }
+        q[SEL_C] = rc;
+        q[SEL_O] = Fr::one();
+        cs.insert_gate({curr, s[0], s[1], s[2], o}, q);
+        return o;
+    }
+
+    // fused internal sbox+MDS gate (gates.rs:119-180):
+    // out = rc + coeff*sbox?(curr) + pow5(s0) + s1 + s2
+    static Var fused_internal(PlonkCircuit& cs, bool sbox, const Fr& rc, const Fr& coeff,
+                              Var curr, const std::array<Var, 3>& s) {
+        Fr cval = cs.witness(curr);
+        Fr celem = sbox ? Poseidon2::sbox(cval) : cval;
+        Fr out = rc.add(coeff.mul(celem)).add(Poseidon2::sbox(cs.witness(s[0])))
+                     .add(cs.witness(s[1])).add(cs.witness(s[2]));
+        Var o = cs.create_variable(out);
+        std::array<Fr, NUM_SELECTORS> q{};
+        for (auto& x : q) x = Fr::zero();
+        if (sbox) {
+            q[SEL_HASH0] = coeff;
+            q[SEL_HASH1] = Fr::one();
+            q[SEL_LC2] = q[SEL_LC3] = Fr::one();
+        } else {
+            q[SEL_HASH1] = Fr::one();
+            q[SEL_LC0] = coeff;
+            q[SEL_LC2] = q[SEL_LC3] = Fr::one();
+        }
+        q[SEL_C] = rc;
+        q[SEL_O] = Fr::one();
+        cs.insert_gate({curr, s[0], s[1], s[2], o}, q);
+        return o;
+    }
+
+    void external_round(PlonkCircuit& cs, bool sbox, const Fr rc[3]) {
+        std::array<Var, 3> in = state;
+        for (int i = 0; i < 3; ++i)
+            state[i] = fused_external(cs, sbox, rc[i], in[i], in);
+    }
+    void internal_round(PlonkCircuit& cs, const Fr rc[3]) {
+        std::array<Var, 3> in = state;
+        state[0] = fused_internal(cs, true, rc[0], Fr::one(), in[0], in);
+        state[1] = fused_internal(cs, false, rc[1], Fr::one(), in[1], in);
+        state[2] = fused_internal(cs, false, rc[2], Fr::from_u64(2), in[2], in);
+    }
+
+    // permutation with inter-round constant fusion (hash.rs:202-252)
+    void permute(PlonkCircuit& cs) {
+        constexpr int HALF = Poseidon2::R_F / 2;
+        Fr rc[3];
+        for (int w = 0; w < 3; ++w) rc[w] = Poseidon2::rc_full(0, w);
+        external_round(cs, false, rc);  // initial MDS + round-0 constants
+        for (int r = 0; r < HALF - 1; ++r) {
+            for (int w = 0; w < 3; ++w) rc[w] = Poseidon2::rc_full(r + 1, w);
+            external_round(cs, true, rc);
+        }
+        rc[0] = Poseidon2::rc_partial(0);
+        rc[1] = rc[2] = Fr::zero();
+        external_round(cs, true, rc);  // last ext round fused w/ 1st partial rc
+        for (int r = 0; r < Poseidon2::R_P - 1; ++r) {
+            rc[0] = Poseidon2::rc_partial(r + 1);
+            rc[1] = rc[2] = Fr::zero();
+            internal_round(cs, rc);
+        }
+        for (int w = 0; w < 3; ++w) rc[w] = Poseidon2::rc_full(HALF, w);
+        internal_round(cs, rc);  // last internal fused w/ next external rc
+        for (int r = HALF; r < Poseidon2::R_F - 1; ++r) {
+            for (int w = 0; w < 3; ++w) rc[w] = Poseidon2::rc_full(r + 1, w);
+            external_round(cs, true, rc);
+        }
+        rc[0] = rc[1] = rc[2] = Fr::zero();
+        external_round(cs, true, rc);  // final round, no trailing constant
+    }
+
+    void absorb(PlonkCircuit& cs, Var a) {
+        if (next_index == Poseidon2::RATE) {
+            permute(cs);
+            next_index = 0;
+        }
+        int at = next_index + Poseidon2::CAPACITY;
+        state[at] = cs.add(a, state[at]);
+        next_index++;
+    }
+    void batch_absorb(PlonkCircuit& cs, const std::vector<Var>& vs) {
+        for (Var v : vs) absorb(cs, v);
+    }
+    Var squeeze(PlonkCircuit& cs) {
+        if (!squeezing || next_index == Poseidon2::RATE) {
+            permute(cs);
+            next_index = 0;
+            squeezing = true;
+        }
+        return state[Poseidon2::CAPACITY + next_index++];
+    }
+    Var hash(PlonkCircuit& cs, const std::vector<Var>& input) {
+        batch_absorb(cs, input);
+        return squeeze(cs);
+    }
+};
+
+// CSPRNG state variable (darkpool-types/src/csprng.rs: {seed, index})
+struct CsprngVar {
+    Var seed, index;
+};
+
+// CSPRNGGadget (state_primitives/csprng.rs): next = H(seed, index); index++
+inline Var csprng_next(PlonkCircuit& cs, CsprngVar& st) {
+    PoseidonHashGadget h(cs);
+    Var out = h.hash(cs, {st.seed, st.index});
+    st.index = cs.add(st.index, cs.one());
+    return out;
+}
+inline std::vector<Var> csprng_next_k(PlonkCircuit& cs, CsprngVar& st, size_t k) {
+    std::vector<Var> out;
+    for (size_t i = 0; i < k; ++i) out.push_back(csprng_next(cs, st));
+    return out;
+}
+
+// StreamCipherGadget::encrypt (state_primitives/stream_cipher.rs:22-39):
+// pads = next_k, ciphertext_i = value_i - pad_i; returns (private, public)
+inline void stream_cipher_encrypt(PlonkCircuit& cs, const std::vector<Var>& values,
+                                  CsprngVar& st, std::vector<Var>& private_share,
+                                  std::vector<Var>& public_share) {
+    private_share = csprng_next_k(cs, st, values.size());
+    public_share.clear();
+    for (size_t i = 0; i < values.size(); ++i)
+        public_share.push_back(cs.sub(values[i], private_share[i]));
+}
+
+// resumable commitment (commitment.rs:433-446): comm = v0; comm = H(comm, v_i)
+inline Var resumable_commitment(PlonkCircuit& cs, const std::vector<Var>& vs) {
+    PoseidonHashGadget h(cs);
+    Var comm = vs[0];
+    for (size_t i = 1; i < vs.size(); ++i) {
+        comm = h.hash(cs, {comm, vs[i]});
+        h.reset(cs);
+    }
+    return comm;
+}
+
+// CommitmentGadget::compute_commitment (commitment.rs:72-135):
+//   private = H(private_shares || recovery_stream || share_stream)
+//   public  = resumable(public_shares)
+//   comm    = H(private, public)
+inline Var commitment_gadget(PlonkCircuit& cs, const std::vector<Var>& private_share,
+                             const CsprngVar& recovery, const CsprngVar& share,
+                             const std::vector<Var>& public_share) {
+    PoseidonHashGadget h(cs);
+    std::vector<Var> inputs = private_share;
+    inputs.push_back(recovery.seed);
+    inputs.push_back(recovery.index);
+    inputs.push_back(share.seed);
+    inputs.push_back(share.index);
+    Var priv = h.hash(cs, inputs);
+    Var pub = resumable_commitment(cs, public_share);
+    PoseidonHashGadget h2(cs);
+    return h2.hash(cs, {priv, pub});
+}
+
+// ============================ native types ============================
+
+// PoseidonCSPRNG (csprng.rs): next = H(seed, index); index++
+struct Csprng {
+    Fr seed;
+    uint64_t index = 0;
+    Fr next() {
+        Fr in[2] = {seed, Fr::from_u64(index)};
+        index++;
+        return poseidon_hash(in, 2);
+    }
+};
+
+struct Balance {  // DarkpoolBalance (balance.rs): 8 scalars in field order
+    Fr mint, owner, relayer_fee_recipient, authority_x, authority_y,
+        relayer_fee_balance, protocol_fee_balance, amount;
+    std::vector<Fr> to_scalars() const {
+        return {mint, owner, relayer_fee_recipient, authority_x, authority_y,
+                relayer_fee_balance, protocol_fee_balance, amount};
+    }
+    static constexpr size_t NUM_SCALARS = 8;
+};
+
+struct Deposit {  // deposit.rs: 3 scalars
+    Fr from, token, amount;
+    std::vector<Fr> to_scalars() const { return {from, token, amount}; }
+};
+
+// native commitment over a state-wrapped element (state_wrapper.rs:125-163)
+inline Fr native_commitment(const std::vector<Fr>& private_shares,
+                            const Csprng& recovery, const Csprng& share,
+                            const std::vector<Fr>& public_shares) {
+    std::vector<Fr> in = private_shares;
+    in.push_back(recovery.seed);
+    in.push_back(Fr::from_u64(recovery.index));
+    in.push_back(share.seed);
+    in.push_back(Fr::from_u64(share.index));
+    Fr priv = poseidon_hash(in.data(), in.size());
+    Fr pub = public_shares[0];
+    for (size_t i = 1; i < public_shares.size(); ++i) {
+        Fr two[2] = {pub, public_shares[i]};
+        pub = poseidon_hash(two, 2);
+    }
+    Fr fin[2] = {priv, pub};
+    return poseidon_hash(fin, 2);
+}
+
+// ====================== Valid Balance Create ======================
+
+struct VbcWitness {  // valid_balance_create.rs:143-150 (12 scalars)
+    Csprng initial_share_stream;
+    Csprng initial_recovery_stream;
+    Balance balance;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = {initial_share_stream.seed, Fr::from_u64(initial_share_stream.index),
+                             initial_recovery_stream.seed,
+                             Fr::from_u64(initial_recovery_stream.index)};
+        auto b = balance.to_scalars();
+        v.insert(v.end(), b.begin(), b.end());
+        return v;
+    }
+};
+
+struct VbcStatement {  // valid_balance_create.rs:159-170 (13 scalars)
+    Deposit deposit;
+    Fr balance_commitment;
+    Fr recovery_id;
+    std::vector<Fr> new_balance_share;  // 8
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = deposit.to_scalars();
+        v.push_back(balance_commitment);
+        v.push_back(recovery_id);
+        v.insert(v.end(), new_balance_share.begin(), new_balance_share.end());
+        return v;
+    }
+};
+
+// fixed-seed witness/statement builder mirroring test_helpers
+// (valid_balance_create.rs:225-283 + darkpool-types/src/fuzzing.rs:58-71,
+//  324-344): addresses are 160-bit, amounts < 2^100 (AMOUNT_BITS=100,
+//  circuit-types/src/lib.rs:59), stream states random with random indices.
+inline void vbc_build_witness_statement(uint64_t seed, VbcWitness& w, VbcStatement& st) {
+    Lcg rng(seed);
+    auto addr = [&]() {  // 160-bit address scalar (fuzzing.rs:66-71)
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    Fr amount = Fr::from_u64(rng.next());  // < 2^53 < 2^100
+
+    st.deposit = {addr(), addr(), amount};
+    w.balance = {st.deposit.token, st.deposit.from, addr(), rng.fr(), rng.fr(),
+                 Fr::zero(), Fr::zero(), amount};
+    w.initial_share_stream = {rng.fr(), rng.next()};
+    w.initial_recovery_stream = {rng.fr(), rng.next()};
+
+    // encrypt the balance: pads from the share stream, public = value - pad
+    Csprng share = w.initial_share_stream;
+    auto values = w.balance.to_scalars();
+    std::vector<Fr> pads, pub;
+    for (auto& v : values) {
+        Fr p = share.next();
+        pads.push_back(p);
+        pub.push_back(v.sub(p));
+    }
+    st.new_balance_share = pub;
+
+    // recovery id + commitment over the UPDATED streams
+    Csprng recovery = w.initial_recovery_stream;
+    st.recovery_id = recovery.next();
+    st.balance_commitment = native_commitment(pads, recovery, share, pub);
+}
+
+// apply_constraints (valid_balance_create.rs:44-133).  Returns nothing; all
+// witness/statement vars are allocated in order (create_witness then
+// create_public_var, traits.rs:984-991).
+inline void vbc_apply_constraints(PlonkCircuit& cs, const VbcWitness& w,
+                                  const VbcStatement& st) {
+    // --- allocate witness vars (field order) ---
+    CsprngVar share{cs.create_variable(w.initial_share_stream.seed),
+                    cs.create_variable(Fr::from_u64(w.initial_share_stream.index))};
+    CsprngVar recovery{cs.create_variable(w.initial_recovery_stream.seed),
+                       cs.create_variable(Fr::from_u64(w.initial_recovery_stream.index))};
+    auto bscal = w.balance.to_scalars();
+    std::vector<Var> balance;
+    for (auto& s : bscal) balance.push_back(cs.create_variable(s));
+
+    // --- allocate statement vars as public inputs ---
+    auto sscal = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : sscal) pub.push_back(cs.create_public_variable(s));
+    Var dep_from = pub[0], dep_token = pub[1], dep_amount = pub[2];
+    Var st_commitment = pub[3], st_recovery_id = pub[4];
+    std::vector<Var> st_share(pub.begin() + 5, pub.begin() + 13);
+
+    // 1. validate deposit (:71-87)
+    cs.enforce_in_range(dep_amount, 100);  // AmountGadget, AMOUNT_BITS=100
+    cs.enforce_equal(dep_token, balance[0]);  // token == mint
+    cs.enforce_equal(dep_from, balance[1]);   // from == owner
+    // 2. validate new balance (:90-104)
+    cs.enforce_equal(balance[7], dep_amount);  // amount
+    cs.enforce_equal(balance[5], cs.zero());   // relayer fee
+    cs.enforce_equal(balance[6], cs.zero());   // protocol fee
+    // 3. encryption (:110-133)
+    std::vector<Var> private_share, public_share;
+    stream_cipher_encrypt(cs, balance, share, private_share, public_share);
+    for (int i = 0; i < 8; ++i) cs.enforce_equal(public_share[i], st_share[i]);
+    // 4. recovery id (:58-59)
+    Var rid = csprng_next(cs, recovery);
+    cs.enforce_equal(rid, st_recovery_id);
+    // 5. commitment (:64-65) — uses the UPDATED stream states
+    Var comm = commitment_gadget(cs, private_share, recovery, share, public_share);
+    cs.enforce_equal(comm, st_commitment);
+}
+
+}  // namespace rng

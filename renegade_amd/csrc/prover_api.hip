@@ -19,6 +19,7 @@
 #include "plonk_circuit.hpp"
 #include "plonk_host.hpp"
 #include "test_circuits.hpp"
+#include "circuits_core.hpp"
 
 namespace rng {
 
@@ -1045,6 +1046,35 @@ void* rng_testcirc_build(uint64_t seed, uint64_t scale) {
         fprintf(stderr, "rng_testcirc_build: %s\n", e.what());
         return nullptr;
     }
+}
+
+// `Valid Balance Create` circuit builder (zk_circuits/valid_balance_create.rs)
+// with fixed-seed witness/statement; returns finalized tables.
+void* rng_circ_build_vbc(uint64_t seed) {
+    try {
+        VbcWitness w;
+        VbcStatement st;
+        vbc_build_witness_statement(seed, w, st);
+        PlonkCircuit cs;
+        vbc_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_vbc: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_vbc: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// native Poseidon2 hash (for cross-checks vs the oracle's restatement)
+void rng_poseidon_hash(const uint64_t* inputs_mont, uint64_t n, uint64_t* out_mont) {
+    std::vector<Fr> in(n);
+    memcpy(in.data(), inputs_mont, n * sizeof(Fr));
+    Fr r = poseidon_hash(in.data(), n);
+    memcpy(out_mont, r.l, 32);
 }
 
 uint64_t rng_circ_n(void* t) { return static_cast<CircuitTables*>(t)->n; }
