@@ -4,24 +4,28 @@
 Driver contract: `python bench.py --gpus N --steps K --warmup W` — one rank
 per GPU (torchrun for N>1), rank 0 prints ONE JSON line.
 
-Current workload (BASELINE.json configs[1], the config the MSM-GB/s leg of
-the metric is quoted on): one step = one 2^20-point BN254 G1 Pippenger MSM
-with bases+scalars already resident in HBM.  When the full PlonK prover
-lands, the workload switches to configs[3] (private-settlement proofs/sec).
+Workload (BASELINE.json configs[3], the config the proofs/sec leg of the
+metric is quoted on): one step = one full `Intent And Balance Private
+Settlement` TurboPlonk proof (the VALID MATCH MPC successor)
+on the GPU prover — synthetic fixed-seed witness, deterministic generated
+SRS, measured domain n reported in config.  Proof jobs are embarrassingly
+parallel across GPUs (exactly how the reference's rayon pool treats them,
+native_proof_manager.rs:193-198) => weak scaling, no data-path collective.
 
-Roofline accounting: bytes_alg per MSM = W*64*N + 2*32*N
-(per-window base re-read + one scalar read + one digit write/read);
-c=16 => W=16 => 1088 B/point = 1.140 GB per 2^20 MSM.  The dominant kernel
-(bucket_reduce) is timed with HIP events on the launch stream inside the
-library (rng_msm_last_times).
+The roofline object reports the flagship MSM kernel (BASELINE configs[1]:
+2^20-point BN254 G1 Pippenger) measured in the same run with HIP events:
+bytes_alg per MSM = W*64*N + 2*32*N = 1088 B/point, the
+bucket_reduce kernel's share = W*(64+8)*N.  `extra` carries the NTT 2^22
+round-trip leg (configs[2]) and the per-kernel MSM breakdown.
 
-cpu_baseline: the CPU oracle's Pippenger MSM (oracle/msm.hpp — a restatement,
-kind="port") timed on the host cores of the same box, on a bounded sample.
+cpu_baseline: the CPU oracle prover (oracle/plonk.hpp, kind="port") on the
+same circuit tables, all host cores, bounded sample.
 """
 import argparse
 import ctypes
 import json
 import os
+import subprocess
 import sys
 import time
 from pathlib import Path
@@ -32,68 +36,95 @@ REPO = Path(__file__).resolve().parent
 sys.path.insert(0, str(REPO))
 
 HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (8 TB/s), MI355X_MICROARCH.md
-LOG2N = 20
-WINDOW_C = 16
+MSM_LOG2N = 20
+MSM_WINDOW_C = 16
+U64P = ctypes.POINTER(ctypes.c_uint64)
+
+
+def ptr(a):
+    return a.ctypes.data_as(U64P)
 
 
 def log(msg):
     print(msg, file=sys.stderr, flush=True)
 
 
-def mk_inputs(orc, n, rank):
-    """Synthetic inputs: bases = tau-power points from the deterministic SRS
-    (valid curve points), scalars = seeded uniform (numpy, rejection-free via
-    top-limb mask below r)."""
-    power = 12  # 4k distinct bases, tiled to n (bucket pattern depends only on
-    # scalars, so tiled bases are perf-equivalent; keeps setup fast)
-    ptau = orc.srs_generate_ptau(power, seed=42)
-    g1, _, _ = orc.srs_parse(ptau, (1 << power) + 2)
-    npts = g1.shape[0]
-    reps = (n + npts - 1) // npts
-    bases8 = np.tile(np.ascontiguousarray(g1[:, :8]), (reps, 1))[:n].reshape(-1)
-    bases8 = np.ascontiguousarray(bases8)
-    rng = np.random.default_rng(12345 + rank)
-    scalars = rng.integers(0, 1 << 64, size=4 * n, dtype=np.uint64)
-    scalars[3::4] &= (1 << 61) - 1  # < 2^253 < r: canonical scalars
-    return ptau, bases8, scalars
+def build_settlement_tables(lib):
+    lib.rng_circ_build_settlement.restype = ctypes.c_void_p
+    lib.rng_circ_build_settlement.argtypes = [ctypes.c_uint64]
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    h = lib.rng_circ_build_settlement(42)
+    assert h, "settlement circuit build failed"
+    n = lib.rng_circ_n(h)
+    npub = lib.rng_circ_npub(h)
+    sel = np.zeros(13 * n * 4, dtype=np.uint64)
+    sigma = np.zeros(5 * n, dtype=np.uint64)
+    wires = np.zeros(5 * n * 4, dtype=np.uint64)
+    pubs = np.zeros(npub * 4, dtype=np.uint64)
+    lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+    lib.rng_circ_free(h)
+    return n, npub, sel, sigma, wires, pubs
 
 
-def cpu_baseline(orc, bases8_tiled, n_sample, cores):
-    """Oracle Pippenger on n_sample points, all host cores."""
-    from tests import py_ref as ref
-    rng = np.random.default_rng(999)
-    scalars = rng.integers(0, 1 << 64, size=4 * n_sample, dtype=np.uint64)
-    scalars[3::4] &= (1 << 61) - 1
-    # build 9-u64 records from 8-u64 packed
-    b8 = bases8_tiled[:8 * n_sample].reshape(n_sample, 8)
-    b9 = np.zeros((n_sample, 9), dtype=np.uint64)
-    b9[:, :8] = b8
-    b9 = np.ascontiguousarray(b9.reshape(-1))
-    orc.lib.orc_set_num_threads(cores)
+class Desc(ctypes.Structure):
+    _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
+                ("selectors", U64P), ("sigma", U64P),
+                ("num_link_groups", ctypes.c_uint64), ("link_offsets", U64P)]
+
+
+def cpu_baseline_proofs(orc, n, npub, sel, sigma, wires, pubs, srs_records, max_degree,
+                        cores):
+    o = orc.lib
+    o.orc_plonk_preprocess.restype = ctypes.c_void_p
+    o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P, U64P,
+                                       U64P, ctypes.c_uint64]
+    o.orc_plonk_prove.argtypes = [ctypes.c_void_p, U64P, U64P, ctypes.c_uint64, U64P]
+    pk = o.orc_plonk_preprocess(n, npub, ptr(sel), ptr(sigma), ptr(srs_records),
+                                max_degree + 1)
+    assert pk
+    from concurrent.futures import ThreadPoolExecutor
+
+    def one(seed):
+        proof = np.zeros(157, dtype=np.uint64)
+        rc = o.orc_plonk_prove(ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
+                               ctypes.c_uint64(seed), ptr(proof))
+        assert rc == 0
+        return proof
+
+    one(0)  # warm
     t0 = time.perf_counter()
-    reps = 0
-    while time.perf_counter() - t0 < 10.0:
-        orc.msm(b9, scalars, n_sample, window_c=WINDOW_C)
-        reps += 1
-        if reps >= 8:
-            break
-    dt = (time.perf_counter() - t0) / reps
-    bytes_per_point = (16 * 64 + 2 * 32)  # same accounting as GPU value
+    done = 0
+    with ThreadPoolExecutor(max_workers=cores) as ex:
+        futs = []
+        # bound the sample: ~15 s wall target, chunked submission
+        while time.perf_counter() - t0 < 15.0 and done < 4 * cores:
+            futs.append(ex.submit(one, 1000 + done))
+            done += 1
+        for f in futs:
+            f.result()
+    dt = time.perf_counter() - t0
     return {
-        "value": round(n_sample * bytes_per_point / dt / 1e9, 3),
-        "unit": "GB/s",
+        "value": round(done / dt, 3),
+        "unit": "proofs/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"{reps}x 2^{n_sample.bit_length()-1} MSM, same accounting (1088 B/pt)",
+        "sample": f"{done} settlement proofs (n={n}) on {cores} threads, oracle prover",
     }
 
 
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-kernel-legs", action="store_true",
+                    help="skip the MSM/NTT kernel side-measurements")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -103,13 +134,17 @@ def main():
 
     from renegade_amd import load_prover
     from tests.orc_bindings import OracleLib
-    import subprocess
     if not (REPO / "oracle" / "liborc.so").exists():
         subprocess.run(["make", "-C", str(REPO / "oracle")], check=True)
     orc = OracleLib(str(REPO / "oracle" / "liborc.so"))
     plib = load_prover()
     plib.require_gpu()
     plib.set_device(local_rank)
+    lib = plib.lib
+    lib.rng_preprocess.restype = ctypes.c_void_p
+    lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                              ctypes.c_uint64, U64P, U64P]
 
     dist = None
     if world > 1:
@@ -120,34 +155,39 @@ def main():
         torch.cuda.set_device(local_rank)
         dist = td
 
-    n = 1 << LOG2N
-    ptau, bases8, scalars = mk_inputs(orc, n, rank)
-    ctx = plib.init(ptau, (1 << 12) + 2)
+    # --- setup: circuit tables + SRS + PK ---
+    n, npub, sel, sigma, wires, pubs = build_settlement_tables(lib)
+    power = max(4, int(n).bit_length())
+    ptau = orc.srs_generate_ptau(power, seed=42)
+    max_degree = (1 << power) + 2
+    ctx = plib.init(ptau, max_degree)
+    desc = Desc(n, npub, ptr(sel), ptr(sigma), 0, None)
+    pk = lib.rng_preprocess(ctx.h, ctypes.byref(desc))
+    assert pk, "rng_preprocess failed"
+    proof = np.zeros(157, dtype=np.uint64)
 
-    dbases = ctx.dbuf_from(bases8)
-    dscalars = ctx.dbuf_from(scalars)
+    def step(seed):
+        rc = lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
+                           ctypes.c_uint64(seed), ptr(proof), None)
+        assert rc == 0, f"rng_prove rc={rc}"
 
-    def step():
-        return ctx.msm_dev(dbases, dscalars, n, window_c=WINDOW_C)
-
-    # warmup (also JIT-allocates MSM scratch)
-    for _ in range(args.warmup):
-        out = step()
+    for i in range(args.warmup):
+        step(i)
     ctx.sync()
 
-    # one oracle spot-check of the first result (tiny n would be cheating;
-    # instead verify determinism across steps)
-    out2 = step()
-    assert np.array_equal(out, out2), "nondeterministic MSM result"
+    # determinism spot check
+    p1 = None
+    step(12345)
+    p1 = proof.copy()
+    step(12345)
+    assert np.array_equal(p1, proof), "nondeterministic proof"
 
     if dist:
         dist.barrier()
     ctx.sync()
     t0 = time.perf_counter()
-    kern_times = []
-    for _ in range(args.steps):
-        step()
-        kern_times.append(plib.msm_last_times())
+    for i in range(args.steps):
+        step(10_000 + rank * 100_000 + i)
     ctx.sync()
     if dist:
         import torch
@@ -157,24 +197,54 @@ def main():
     else:
         elapsed = time.perf_counter() - t0
 
-    bytes_per_point = WINDOW_C * 64 + 2 * 32
-    bytes_alg = n * bytes_per_point  # per MSM
-    total_bytes = bytes_alg * args.steps * n_gpus
-    value = total_bytes / elapsed / 1e9  # GB/s whole-job
+    proofs_per_s = args.steps * n_gpus / elapsed
 
-    # roofline: dominant kernel = bucket_reduce; its algorithmic bytes =
-    # per-window base gathers + sorted digit-pair reads = W*(64+8)*N
-    avg_bucket_ms = float(np.mean([k["bucket_reduce"] for k in kern_times]))
-    bucket_bytes = WINDOW_C * (64 + 8) * n
-    achieved = bucket_bytes / (avg_bucket_ms / 1e3) / 1e9
-
-    breakdown = {k: round(float(np.mean([t[k] for t in kern_times])), 3)
-                 for k in kern_times[0]}
-
-    # NTT side-measurement (BASELINE config #3): 2^22 forward+inverse,
-    # device-resident, out-of-place ping-pong (no copy-back)
+    # --- kernel legs (rank 0): MSM 2^20 roofline + NTT 2^22 ---
+    roofline = None
     ntt_extra = None
-    if rank == 0:
+    msm_extra = None
+    if rank == 0 and not args.no_kernel_legs:
+        g1, _, _ = orc.srs_parse(ptau, max_degree)
+        npts = g1.shape[0]
+        nb = 1 << MSM_LOG2N
+        reps = (nb + npts - 1) // npts
+        bases8 = np.ascontiguousarray(
+            np.tile(np.ascontiguousarray(g1[:, :8]), (reps, 1))[:nb].reshape(-1))
+        rng = np.random.default_rng(12345)
+        scalars = rng.integers(0, 1 << 64, size=4 * nb, dtype=np.uint64)
+        scalars[3::4] &= (1 << 61) - 1
+        dbases = ctx.dbuf_from(bases8)
+        dscalars = ctx.dbuf_from(scalars)
+        for _ in range(2):
+            ctx.msm_dev(dbases, dscalars, nb, window_c=MSM_WINDOW_C)
+        t1 = time.perf_counter()
+        kt = []
+        for _ in range(5):
+            ctx.msm_dev(dbases, dscalars, nb, window_c=MSM_WINDOW_C)
+            kt.append(plib.msm_last_times())
+        msm_ms = (time.perf_counter() - t1) / 5 * 1e3
+        avg_bucket_ms = float(np.mean([k["bucket_reduce"] for k in kt]))
+        bucket_bytes = MSM_WINDOW_C * (64 + 8) * nb
+        achieved = bucket_bytes / (avg_bucket_ms / 1e3) / 1e9
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved, 1),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved / HBM_PEAK_GBS, 4),
+            "traffic": None,
+            "kernel": "k_msm_bucket_reduce (msm 2^20 leg)",
+            "kernel_ms": round(avg_bucket_ms, 3),
+        }
+        msm_extra = {
+            "msm_2^20_ms": round(msm_ms, 3),
+            "alg_gbs": round(nb * (MSM_WINDOW_C * 64 + 64) / (msm_ms / 1e3) / 1e9, 1),
+            "breakdown_ms": {k: round(float(np.mean([t[k] for t in kt])), 3)
+                             for k in kt[0]},
+        }
+        ctx.dbuf_free(dbases)
+        ctx.dbuf_free(dscalars)
+
         m = 1 << 22
         nd = np.random.default_rng(5).integers(0, 1 << 61, size=4 * m, dtype=np.uint64)
         da = ctx.dbuf_from(nd)
@@ -184,21 +254,18 @@ def main():
             ctx.ntt_dev_oop(db, da, m, inverse=True)
         ctx.sync()
         t1 = time.perf_counter()
-        reps = 5
-        for _ in range(reps):
+        for _ in range(5):
             ctx.ntt_dev_oop(da, db, m)
             ctx.ntt_dev_oop(db, da, m, inverse=True)
         ctx.sync()
-        rt_ms = (time.perf_counter() - t1) / reps * 1e3
+        rt_ms = (time.perf_counter() - t1) / 5 * 1e3
         back = np.empty_like(nd)
         ctx.dbuf_download(da, back)
         assert np.array_equal(back, nd), "NTT round-trip mismatch in bench"
-        passes = plib.ntt_last_times()
-        alg_bytes_rt = 2 * 2 * 2 * 32 * m  # P=2 passes x r+w x 32 B, both directions
         ntt_extra = {
             "roundtrip_ms": round(rt_ms, 3),
-            "alg_gbs": round(alg_bytes_rt / (rt_ms / 1e3) / 1e9, 1),
-            "pass_ms": {k: round(v, 3) for k, v in passes.items()},
+            "alg_gbs": round(2 * 2 * 2 * 32 * m / (rt_ms / 1e3) / 1e9, 1),
+            "pass_ms": {k: round(v, 3) for k, v in plib.ntt_last_times().items()},
         }
         ctx.dbuf_free(da)
         ctx.dbuf_free(db)
@@ -207,11 +274,14 @@ def main():
         cores = os.cpu_count()
         cb = None
         if not args.no_cpu_baseline:
-            cb = cpu_baseline(orc, bases8, 1 << 17, cores)
+            g1, _, _ = orc.srs_parse(ptau, max_degree)
+            srs_records = np.ascontiguousarray(g1).reshape(-1)
+            cb = cpu_baseline_proofs(orc, n, npub, sel, sigma, wires, pubs, srs_records,
+                                     max_degree, cores)
         result = {
-            "metric": "BN254 G1 MSM throughput (algorithmic GB/s, VALID MATCH MPC commit primitive)",
-            "value": round(value, 2),
-            "unit": "GB/s",
+            "metric": "VALID MATCH MPC proofs/sec (Intent And Balance Private Settlement)",
+            "value": round(proofs_per_s, 3),
+            "unit": "proofs/s",
             "n_gpus": n_gpus,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -222,24 +292,15 @@ def main():
             "dtype": "u256",
             "data": "synthetic",
             "config": {
-                "workload": "msm_2^20_bn254_g1",
-                "points": n,
-                "window_c": WINDOW_C,
-                "parallelism": f"replicated x{n_gpus}",
+                "workload": "intent_and_balance_private_settlement_proof",
+                "domain_n": int(n),
+                "num_public": int(npub),
+                "srs_power": power,
+                "parallelism": f"independent proofs x{n_gpus}",
             },
-            "roofline": {
-                "bound": "hbm",
-                "achieved": round(achieved, 1),
-                "peak": HBM_PEAK_GBS,
-                "unit": "GB/s",
-                "frac": round(achieved / HBM_PEAK_GBS, 4),
-                "traffic": None,
-                "kernel": "k_msm_bucket_reduce",
-                "kernel_ms": round(avg_bucket_ms, 3),
-                "breakdown_ms": breakdown,
-            },
+            "roofline": roofline,
             "cpu_baseline": cb,
-            "extra": {"ntt_2^22": ntt_extra},
+            "extra": {"msm_2^20": msm_extra, "ntt_2^22": ntt_extra},
         }
         print(json.dumps(result), flush=True)
 

@@ -371,4 +371,297 @@ inline void vbc_apply_constraints(PlonkCircuit& cs, const VbcWitness& w,
     cs.enforce_equal(comm, st_commitment);
 }
 
+// ================== Intent And Balance Private Settlement ==================
+// (zk_circuits/settlement/intent_and_balance_private_settlement.rs — the
+//  VALID MATCH MPC successor; BASELINE config #4)
+
+constexpr int AMOUNT_BITS = 100;       // circuit-types/src/lib.rs:59
+constexpr int FP_PRECISION = 63;       // fixed_point.rs:41 (repr = x * 2^63)
+
+// GreaterThanEqGadget (comparators.rs:259-267): range-check a-b in `bits`
+inline void gte_gadget(PlonkCircuit& cs, Var a, Var b, int bits) {
+    Var diff = cs.sub(a, b);
+    cs.enforce_in_range(diff, bits);
+}
+
+// FixedPointGadget::floor (fixed_point.rs gadget:87-133):
+// allocate floor = repr >> 63 and constrain repr - 2^63*floor in [0, 2^63)
+inline Var fp_floor_gadget(PlonkCircuit& cs, Var repr) {
+    u64 limbs[4];
+    cs.witness(repr).to_canonical(limbs);
+    // shift right by 63 (value < 2^227 in valid uses)
+    u64 fl[4];
+    for (int i = 0; i < 4; ++i) {
+        u64 lo = limbs[i] >> 63;
+        u64 hi = (i + 1 < 4) ? (limbs[i + 1] << 1) : 0;
+        fl[i] = lo | hi;
+    }
+    Var floor_v = cs.create_variable(Fr::from_canonical(fl));
+    Fr two63 = Fr::from_u64(1ull << 62).dbl();
+    std::array<Fr, 4> cf{Fr::one(), two63.neg(), Fr::zero(), Fr::zero()};
+    Var diff = cs.lc({repr, floor_v, cs.zero(), cs.zero()}, cf);
+    cs.enforce_in_range(diff, FP_PRECISION);
+    return floor_v;
+}
+
+struct FeeTakeVars {
+    Var relayer_fee, protocol_fee;
+};
+
+// FeeGadget::compute_fee_take (state_gadgets/fee.rs:21-37)
+inline FeeTakeVars fee_take_gadget(PlonkCircuit& cs, Var amount, Var relayer_rate_repr,
+                                   Var protocol_rate_repr) {
+    Var rf_fp = cs.mul(relayer_rate_repr, amount);   // mul_integer
+    Var pf_fp = cs.mul(protocol_rate_repr, amount);
+    return {fp_floor_gadget(cs, rf_fp), fp_floor_gadget(cs, pf_fp)};
+}
+
+struct Obligation {  // settlement_obligation.rs:36-47 (4 scalars)
+    Fr input_token, output_token, amount_in, amount_out;
+    std::vector<Fr> to_scalars() const {
+        return {input_token, output_token, amount_in, amount_out};
+    }
+};
+struct Intent {  // intent.rs:49-70 (5 scalars; min_price = FixedPoint repr)
+    Fr in_token, out_token, owner, min_price_repr, amount_in;
+    std::vector<Fr> to_scalars() const {
+        return {in_token, out_token, owner, min_price_repr, amount_in};
+    }
+};
+struct PostMatchShare {  // balance.rs PostMatchBalanceShare (3 scalars)
+    Fr amount, relayer_fee_balance, protocol_fee_balance;
+    std::vector<Fr> to_scalars() const {
+        return {amount, relayer_fee_balance, protocol_fee_balance};
+    }
+};
+
+struct SettlementParty {
+    Obligation obligation;
+    Intent intent;
+    Fr pre_amount_share;
+    Balance input_balance;
+    PostMatchShare pre_in_shares;
+    Balance output_balance;
+    PostMatchShare pre_out_shares;
+};
+
+struct SettlementWitness {  // 2 * 32 = 64 scalars, field order per the struct
+    SettlementParty p[2];
+};
+struct SettlementStatement {  // 17 scalars
+    Fr new_amount_share[2];
+    PostMatchShare new_in_shares[2];
+    PostMatchShare new_out_shares[2];
+    Fr relayer_fee_repr[2];
+    Fr protocol_fee_repr;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v;
+        for (int i = 0; i < 2; ++i) {
+            v.push_back(new_amount_share[i]);
+            auto a = new_in_shares[i].to_scalars();
+            v.insert(v.end(), a.begin(), a.end());
+            auto b = new_out_shares[i].to_scalars();
+            v.insert(v.end(), b.begin(), b.end());
+        }
+        v.push_back(relayer_fee_repr[0]);
+        v.push_back(relayer_fee_repr[1]);
+        v.push_back(protocol_fee_repr);
+        return v;
+    }
+};
+
+// fixed-seed witness/statement (test_helpers create_witness_statement,
+// intent_and_balance_private_settlement.rs:334-420; f64 price sampling
+// replaced by exact integer fixed-point construction)
+inline void settlement_build_witness_statement(uint64_t seed, SettlementWitness& w,
+                                               SettlementStatement& st) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    auto amount_u64 = [&]() { return (rng.next() & ((1ull << 52) - 1)) + 2; };
+    Fr token0 = addr(), token1 = addr();
+    uint64_t t0_amount = amount_u64(), t1_amount = amount_u64();
+    uint64_t t0_traded = t0_amount / 2 + 1, t1_traded = t1_amount / 2 + 1;
+
+    // min_price = half the trade price, as exact 2^63 fixed point:
+    // repr = floor((traded_out << 63) / traded_in / 2)
+    auto half_price_repr = [&](uint64_t out_amt, uint64_t in_amt) {
+        unsigned __int128 r = ((unsigned __int128)out_amt << 63) / in_amt / 2;
+        u64 l[4] = {(u64)r, (u64)(r >> 64), 0, 0};
+        return Fr::from_canonical(l);
+    };
+
+    Obligation ob0{token0, token1, Fr::from_u64(t0_traded), Fr::from_u64(t1_traded)};
+    Obligation ob1{token1, token0, Fr::from_u64(t1_traded), Fr::from_u64(t0_traded)};
+    Fr owner0 = addr(), owner1 = addr();
+    Intent in0{token0, token1, owner0, half_price_repr(t1_traded, t0_traded),
+               Fr::from_u64(t0_amount)};
+    Intent in1{token1, token0, owner1, half_price_repr(t0_traded, t1_traded),
+               Fr::from_u64(t1_amount)};
+
+    auto send_balance = [&](const Fr& owner, const Obligation& ob, uint64_t amt) {
+        return Balance{ob.input_token, owner, addr(), rng.fr(), rng.fr(),
+                       Fr::from_u64(rng.next() & 0xFFFF), Fr::from_u64(rng.next() & 0xFFFF),
+                       Fr::from_u64(amt)};
+    };
+    auto recv_balance = [&](const Fr& owner, const Obligation& ob) {
+        return Balance{ob.output_token, owner, addr(), rng.fr(), rng.fr(),
+                       Fr::from_u64(rng.next() & 0xFFFF), Fr::from_u64(rng.next() & 0xFFFF),
+                       Fr::from_u64(rng.next() & ((1ull << 40) - 1))};
+    };
+    w.p[0] = {ob0, in0, rng.fr(), send_balance(owner0, ob0, t0_amount), {rng.fr(), rng.fr(), rng.fr()},
+              recv_balance(owner0, ob0), {rng.fr(), rng.fr(), rng.fr()}};
+    w.p[1] = {ob1, in1, rng.fr(), send_balance(owner1, ob1, t1_amount), {rng.fr(), rng.fr(), rng.fr()},
+              recv_balance(owner1, ob1), {rng.fr(), rng.fr(), rng.fr()}};
+
+    // fee rates: ~0.1% as exact fixed point
+    auto fee_repr = [&]() {
+        unsigned __int128 r = ((unsigned __int128)1 << 63) / (1000 + (rng.next() & 1023));
+        u64 l[4] = {(u64)r, (u64)(r >> 64), 0, 0};
+        return Fr::from_canonical(l);
+    };
+    st.relayer_fee_repr[0] = fee_repr();
+    st.relayer_fee_repr[1] = fee_repr();
+    st.protocol_fee_repr = fee_repr();
+
+    // native share updates (test_helpers :356-390)
+    auto floor_mul = [&](const Fr& rate_repr, const Fr& amt) {
+        // floor(rate * amt / 2^63) over canonical integers (fits 128 bits here)
+        u64 rl[4], al[4];
+        rate_repr.to_canonical(rl);
+        amt.to_canonical(al);
+        unsigned __int128 rate = ((unsigned __int128)rl[1] << 64) | rl[0];
+        unsigned __int128 prod = rate * al[0];  // amounts < 2^64 here
+        unsigned __int128 fl = prod >> 63;
+        u64 l[4] = {(u64)fl, (u64)(fl >> 64), 0, 0};
+        return Fr::from_canonical(l);
+    };
+    for (int i = 0; i < 2; ++i) {
+        const auto& ob = w.p[i].obligation;
+        Fr rf = floor_mul(st.relayer_fee_repr[i], ob.amount_out);
+        Fr pf = floor_mul(st.protocol_fee_repr, ob.amount_out);
+        Fr net = ob.amount_out.sub(rf).sub(pf);
+        st.new_amount_share[i] = w.p[i].pre_amount_share.sub(ob.amount_in);
+        st.new_in_shares[i] = {w.p[i].pre_in_shares.amount.sub(ob.amount_in),
+                               w.p[i].pre_in_shares.relayer_fee_balance,
+                               w.p[i].pre_in_shares.protocol_fee_balance};
+        st.new_out_shares[i] = {w.p[i].pre_out_shares.amount.add(net),
+                                w.p[i].pre_out_shares.relayer_fee_balance.add(rf),
+                                w.p[i].pre_out_shares.protocol_fee_balance.add(pf)};
+    }
+}
+
+struct SettlementVars {
+    std::array<Var, 4> ob;
+    std::array<Var, 5> intent;
+    Var pre_amount;
+    std::array<Var, 8> in_bal;
+    std::array<Var, 3> pre_in;
+    std::array<Var, 8> out_bal;
+    std::array<Var, 3> pre_out;
+};
+
+// apply_constraints (intent_and_balance_private_settlement.rs:46-158 +
+// settlement_lib.rs:30-199).  Creates the four proof-linking groups.
+inline void settlement_apply_constraints(PlonkCircuit& cs, const SettlementWitness& w,
+                                         const SettlementStatement& st) {
+    const char* party_group[2] = {"intent_and_balance_settlement_party0",
+                                  "intent_and_balance_settlement_party1"};
+    const char* out_group[2] = {"output_balance_settlement_party0",
+                                "output_balance_settlement_party1"};
+    for (int i = 0; i < 2; ++i) {
+        cs.create_link_group(party_group[i]);
+        cs.create_link_group(out_group[i]);
+    }
+
+    // --- witness allocation (struct field order; link groups per the
+    //     #[link_groups] annotations) ---
+    SettlementVars pv[2];
+    for (int i = 0; i < 2; ++i) {
+        auto alloc_list = [&](const std::vector<Fr>& vals, const char* group) {
+            std::vector<Var> out;
+            for (auto& v : vals) {
+                Var x = cs.create_variable(v);
+                if (group) cs.add_to_link_group(x, group);
+                out.push_back(x);
+            }
+            return out;
+        };
+        auto obv = alloc_list(w.p[i].obligation.to_scalars(), nullptr);
+        auto inv = alloc_list(w.p[i].intent.to_scalars(), party_group[i]);
+        auto pam = alloc_list({w.p[i].pre_amount_share}, party_group[i]);
+        auto ibv = alloc_list(w.p[i].input_balance.to_scalars(), party_group[i]);
+        auto piv = alloc_list(w.p[i].pre_in_shares.to_scalars(), party_group[i]);
+        auto obal = alloc_list(w.p[i].output_balance.to_scalars(), out_group[i]);
+        auto pov = alloc_list(w.p[i].pre_out_shares.to_scalars(), out_group[i]);
+        std::copy(obv.begin(), obv.end(), pv[i].ob.begin());
+        std::copy(inv.begin(), inv.end(), pv[i].intent.begin());
+        pv[i].pre_amount = pam[0];
+        std::copy(ibv.begin(), ibv.end(), pv[i].in_bal.begin());
+        std::copy(piv.begin(), piv.end(), pv[i].pre_in.begin());
+        std::copy(obal.begin(), obal.end(), pv[i].out_bal.begin());
+        std::copy(pov.begin(), pov.end(), pv[i].pre_out.begin());
+    }
+    // --- statement allocation (public inputs, field order) ---
+    auto sscal = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : sscal) pub.push_back(cs.create_public_variable(s));
+    Var new_amount[2] = {pub[0], pub[7]};
+    std::array<Var, 3> new_in[2] = {{pub[1], pub[2], pub[3]}, {pub[8], pub[9], pub[10]}};
+    std::array<Var, 3> new_out[2] = {{pub[4], pub[5], pub[6]}, {pub[11], pub[12], pub[13]}};
+    Var relayer_fee[2] = {pub[14], pub[15]};
+    Var protocol_fee = pub[16];
+
+    // --- 1. obligation compatibility (:135-158) ---
+    for (int i = 0; i < 2; ++i) {
+        cs.enforce_in_range(pv[i].ob[2], AMOUNT_BITS);
+        cs.enforce_in_range(pv[i].ob[3], AMOUNT_BITS);
+    }
+    cs.enforce_equal(pv[0].ob[0], pv[1].ob[1]);  // in0 == out1 token
+    cs.enforce_equal(pv[0].ob[1], pv[1].ob[0]);
+    cs.enforce_equal(pv[0].ob[2], pv[1].ob[3]);  // amount_in0 == amount_out1
+    cs.enforce_equal(pv[0].ob[3], pv[1].ob[2]);
+
+    for (int i = 0; i < 2; ++i) {
+        // --- 2. fee take (:109-133) ---
+        FeeTakeVars ft = fee_take_gadget(cs, pv[i].ob[3], relayer_fee[i], protocol_fee);
+        // --- 3. intent constraints (settlement_lib.rs:46-79) ---
+        cs.enforce_equal(pv[i].ob[0], pv[i].intent[0]);  // input_token == in_token
+        cs.enforce_equal(pv[i].ob[1], pv[i].intent[1]);
+        gte_gadget(cs, pv[i].intent[4], pv[i].ob[2], AMOUNT_BITS);
+        Var min_out_fp = cs.mul(pv[i].intent[3], pv[i].ob[2]);  // mul_integer
+        Var min_out = fp_floor_gadget(cs, min_out_fp);
+        gte_gadget(cs, pv[i].ob[3], min_out, AMOUNT_BITS);
+        // --- in-balance (:86-99) ---
+        gte_gadget(cs, pv[i].in_bal[7], pv[i].ob[2], AMOUNT_BITS);
+        // --- out-balance (:102-129) ---
+        cs.enforce_equal(pv[i].out_bal[0], pv[i].ob[1]);     // mint == output token
+        cs.enforce_equal(pv[i].out_bal[1], pv[i].intent[2]); // owner
+        Var total_fee = cs.add(ft.relayer_fee, ft.protocol_fee);
+        Var net_receive = cs.sub(pv[i].ob[3], total_fee);
+        Var new_bal_amount = cs.add(pv[i].out_bal[7], net_receive);
+        cs.enforce_in_range(new_bal_amount, AMOUNT_BITS);
+        Var new_rfb = cs.add(pv[i].out_bal[5], ft.relayer_fee);
+        Var new_pfb = cs.add(pv[i].out_bal[6], ft.protocol_fee);
+        cs.enforce_in_range(new_rfb, AMOUNT_BITS);
+        cs.enforce_in_range(new_pfb, AMOUNT_BITS);
+        // --- 4. state updates (settlement_lib.rs:141-199) ---
+        Var exp_amount = cs.sub(pv[i].pre_amount, pv[i].ob[2]);
+        cs.enforce_equal(exp_amount, new_amount[i]);
+        Var exp_in_amt = cs.sub(pv[i].pre_in[0], pv[i].ob[2]);
+        cs.enforce_equal(exp_in_amt, new_in[i][0]);
+        cs.enforce_equal(pv[i].pre_in[1], new_in[i][1]);
+        cs.enforce_equal(pv[i].pre_in[2], new_in[i][2]);
+        Var exp_out_amt = cs.add(pv[i].pre_out[0], net_receive);
+        cs.enforce_equal(exp_out_amt, new_out[i][0]);
+        Var exp_out_rfb = cs.add(pv[i].pre_out[1], ft.relayer_fee);
+        cs.enforce_equal(exp_out_rfb, new_out[i][1]);
+        Var exp_out_pfb = cs.add(pv[i].pre_out[2], ft.protocol_fee);
+        cs.enforce_equal(exp_out_pfb, new_out[i][2]);
+    }
+}
+
 }  // namespace rng

@@ -1069,6 +1069,40 @@ void* rng_circ_build_vbc(uint64_t seed) {
     }
 }
 
+// `Intent And Balance Private Settlement` circuit builder (the VALID MATCH
+// MPC successor; zk_circuits/settlement/intent_and_balance_private_settlement.rs)
+void* rng_circ_build_settlement(uint64_t seed) {
+    try {
+        SettlementWitness w;
+        SettlementStatement st;
+        settlement_build_witness_statement(seed, w, st);
+        PlonkCircuit cs;
+        settlement_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_settlement: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_settlement: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+uint64_t rng_circ_num_link_groups(void* t) {
+    return static_cast<CircuitTables*>(t)->link_groups.size();
+}
+// out: per group 3 u64 (offset, stride, count), in creation order
+void rng_circ_link_groups(void* t_, uint64_t* out) {
+    auto* t = static_cast<CircuitTables*>(t_);
+    for (size_t i = 0; i < t->link_groups.size(); ++i) {
+        out[3 * i] = t->link_groups[i].offset;
+        out[3 * i + 1] = t->link_groups[i].stride;
+        out[3 * i + 2] = t->link_groups[i].count;
+    }
+}
+
 // native Poseidon2 hash (for cross-checks vs the oracle's restatement)
 void rng_poseidon_hash(const uint64_t* inputs_mont, uint64_t n, uint64_t* out_mont) {
     std::vector<Fr> in(n);

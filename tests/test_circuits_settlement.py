@@ -1,0 +1,99 @@
+"""`Intent And Balance Private Settlement` circuit (the VALID MATCH MPC
+successor; BASELINE config #4 input) — build, satisfiability, oracle
+prove/verify, link groups (CPU)."""
+import ctypes
+
+import numpy as np
+import pytest
+
+U64P = ctypes.POINTER(ctypes.c_uint64)
+
+
+def ptr(a):
+    return a.ctypes.data_as(U64P)
+
+
+@pytest.fixture(scope="module")
+def plib():
+    from renegade_amd import load_prover
+    return load_prover()
+
+
+@pytest.fixture(scope="module")
+def stl(plib):
+    lib = plib.lib
+    lib.rng_circ_build_settlement.restype = ctypes.c_void_p
+    lib.rng_circ_build_settlement.argtypes = [ctypes.c_uint64]
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_num_link_groups.restype = ctypes.c_uint64
+    lib.rng_circ_num_link_groups.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_link_groups.argtypes = [ctypes.c_void_p, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    h = lib.rng_circ_build_settlement(42)
+    assert h, "settlement circuit build failed (unsatisfied?)"
+    n = lib.rng_circ_n(h)
+    npub = lib.rng_circ_npub(h)
+    nlg = lib.rng_circ_num_link_groups(h)
+    lg = np.zeros(3 * max(1, nlg), dtype=np.uint64)
+    lib.rng_circ_link_groups(h, ptr(lg))
+    sel = np.zeros(13 * n * 4, dtype=np.uint64)
+    sigma = np.zeros(5 * n, dtype=np.uint64)
+    wires = np.zeros(5 * n * 4, dtype=np.uint64)
+    pubs = np.zeros(npub * 4, dtype=np.uint64)
+    lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+    lib.rng_circ_free(h)
+    return dict(n=n, npub=npub, nlg=nlg, lg=lg.reshape(-1, 3), sel=sel, sigma=sigma,
+                wires=wires, pubs=pubs)
+
+
+class TestSettlementCircuit:
+    def test_shape(self, stl):
+        # statement = 17 scalars (intent_and_balance_private_settlement.rs:217-243)
+        assert stl["npub"] == 17
+        # 4 proof-linking groups (rs:271-279): party0/1 (17 vars), out0/1 (11)
+        assert stl["nlg"] == 4
+        counts = sorted(int(c) for c in stl["lg"][:, 2])
+        assert counts == [11, 11, 17, 17]
+        print("settlement n =", stl["n"])
+
+    def test_oracle_prove_verify(self, stl, orc):
+        n, npub = stl["n"], stl["npub"]
+        power = max(4, int(n).bit_length())
+        ptau = orc.srs_generate_ptau(power, seed=42)
+        max_degree = (1 << power) + 2
+        g1, _, _ = orc.srs_parse(ptau, max_degree)
+        srs_records = np.ascontiguousarray(g1).reshape(-1)
+        o = orc.lib
+        o.orc_plonk_preprocess.restype = ctypes.c_void_p
+        o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P, U64P,
+                                           U64P, ctypes.c_uint64]
+        o.orc_plonk_prove.argtypes = [ctypes.c_void_p, U64P, U64P, ctypes.c_uint64, U64P]
+        o.orc_plonk_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P]
+        o.orc_derive_tau.argtypes = [ctypes.c_uint64, U64P]
+        pk = o.orc_plonk_preprocess(n, npub, ptr(stl["sel"]), ptr(stl["sigma"]),
+                                    ptr(srs_records), max_degree + 1)
+        assert pk
+        proof = np.zeros(157, dtype=np.uint64)
+        rc = o.orc_plonk_prove(ctypes.c_void_p(pk), ptr(stl["wires"]), ptr(stl["pubs"]),
+                               ctypes.c_uint64(5), ptr(proof))
+        assert rc == 0
+        tau = np.zeros(4, dtype=np.uint64)
+        o.orc_derive_tau(42, ptr(tau))
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(stl["pubs"]), ptr(proof),
+                                  ptr(tau)) == 1
+        # tamper a fee-rate public input -> reject
+        bad = stl["pubs"].copy()
+        bad[14 * 4] ^= np.uint64(1)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(bad), ptr(proof), ptr(tau)) != 1
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk))
+
+    def test_seeds_vary(self, plib):
+        lib = plib.lib
+        for seed in [1, 2, 3]:
+            h = lib.rng_circ_build_settlement(seed)
+            assert h, f"seed {seed} unsatisfied"
+            lib.rng_circ_free(h)

@@ -135,3 +135,74 @@ class TestGpuProver:
         p3 = gpu_prove(s, seed=22)
         assert np.array_equal(p1, p2)
         assert not np.array_equal(p1, p3)
+
+
+@pytest.mark.parametrize("builder,seed", [("rng_circ_build_settlement", 42),
+                                          ("rng_circ_build_vbc", 42)])
+def test_real_circuit_gpu_parity(orc, builder, seed):
+    """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
+    BASELINE config #4, VBC = config #1)."""
+    from renegade_amd import load_prover
+    plib = load_prover()
+    if not plib.gpu_available:
+        pytest.skip("no GPU")
+    lib = plib.lib
+    fn = getattr(lib, builder)
+    fn.restype = ctypes.c_void_p
+    fn.argtypes = [ctypes.c_uint64]
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    lib.rng_preprocess.restype = ctypes.c_void_p
+    lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                              ctypes.c_uint64, U64P, U64P]
+    h = fn(seed)
+    assert h
+    n = lib.rng_circ_n(h)
+    npub = lib.rng_circ_npub(h)
+    sel = np.zeros(13 * n * 4, dtype=np.uint64)
+    sigma = np.zeros(5 * n, dtype=np.uint64)
+    wires = np.zeros(5 * n * 4, dtype=np.uint64)
+    pubs = np.zeros(npub * 4, dtype=np.uint64)
+    lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+    lib.rng_circ_free(h)
+    power = max(4, int(n).bit_length())
+    ptau = orc.srs_generate_ptau(power, seed=42)
+    max_degree = (1 << power) + 2
+    g1, _, _ = orc.srs_parse(ptau, max_degree)
+    srs_records = np.ascontiguousarray(g1).reshape(-1)
+    ctx = plib.init(ptau, max_degree)
+
+    class Desc(ctypes.Structure):
+        _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
+                    ("selectors", U64P), ("sigma", U64P),
+                    ("num_link_groups", ctypes.c_uint64), ("link_offsets", U64P)]
+
+    pk = lib.rng_preprocess(ctx.h, ctypes.byref(Desc(n, npub, ptr(sel), ptr(sigma), 0,
+                                                     None)))
+    assert pk
+    proof = np.zeros(157, dtype=np.uint64)
+    assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(wires), ptr(pubs), 7,
+                         ptr(proof), None) == 0
+    o = orc.lib
+    o.orc_plonk_preprocess.restype = ctypes.c_void_p
+    o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P, U64P,
+                                       U64P, ctypes.c_uint64]
+    o.orc_plonk_prove.argtypes = [ctypes.c_void_p, U64P, U64P, ctypes.c_uint64, U64P]
+    o.orc_plonk_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P]
+    o.orc_derive_tau.argtypes = [ctypes.c_uint64, U64P]
+    opk = o.orc_plonk_preprocess(n, npub, ptr(sel), ptr(sigma), ptr(srs_records),
+                                 max_degree + 1)
+    orc_proof = np.zeros(157, dtype=np.uint64)
+    assert o.orc_plonk_prove(ctypes.c_void_p(opk), ptr(wires), ptr(pubs),
+                             ctypes.c_uint64(7), ptr(orc_proof)) == 0
+    assert np.array_equal(proof, orc_proof)
+    tau = np.zeros(4, dtype=np.uint64)
+    o.orc_derive_tau(42, ptr(tau))
+    assert o.orc_plonk_verify(ctypes.c_void_p(opk), ptr(pubs), ptr(proof), ptr(tau)) == 1
+    lib.rng_pk_free(ctypes.c_void_p(pk))
+    ctx.close()
