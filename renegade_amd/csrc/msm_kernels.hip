@@ -25,7 +25,18 @@
 namespace rng {
 
 constexpr uint32_t MSM_SENTINEL = 0x1FFFFFu;  // > any (w<<16|mag); 21 bits
-constexpr uint32_t MSM_CHUNK = 32;            // buckets per window-sum thread
+constexpr uint32_t MSM_CHUNK = 16;            // buckets per window-sum thread
+
+// window size by problem size: bucket-phase work ~ n*W(c) while the
+// aggregation phase ~ W(c)*2^(c-1); balance at c ~ log2(n) - 3
+__host__ __device__ inline uint32_t msm_auto_c(uint64_t n) {
+    uint32_t log2n = 0;
+    while ((1ull << log2n) < n) log2n++;
+    uint32_t c = log2n > 3 ? log2n - 3 : 8;
+    if (c < 8) c = 8;
+    if (c > 16) c = 16;
+    return c;
+}
 
 // ---- 1. digit decomposition ----
 // scalars: canonical LE 4xu64. keys/vals: n*W entries, window-major

@@ -477,7 +477,7 @@ static int commit_dev(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, G1Aff* out
                        s->canon, (uint32_t)mdeg);
     HIP_CHECK(hipGetLastError());
     G1Jac res;
-    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, mdeg, 16, &res);
+    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, mdeg, msm_auto_c(mdeg), &res);
     if (rc != RNG_OK) return rc;
     uint64_t rec[9];
     jac_to_affine_record(res, rec);
@@ -998,7 +998,7 @@ int rng_msm_g1_dev(RngCtx* ctx, const void* dev_bases, const void* dev_scalars,
                    uint64_t n, uint64_t* out9, int window_c) {
     if (!gpu_ok()) return RNG_ERR_NO_GPU;
     if (!dev_bases || !dev_scalars || !out9 || n == 0) return RNG_ERR_BAD_ARG;
-    uint32_t c = window_c > 0 ? (uint32_t)window_c : 16;
+    uint32_t c = window_c > 0 ? (uint32_t)window_c : msm_auto_c(n);
     if (c < 8 || c > 16) return RNG_ERR_BAD_ARG;
     G1Jac res;
     int rc = msm_dev_run((const G1Aff*)dev_bases, (const uint64_t*)dev_scalars, n, c, &res);
