@@ -229,7 +229,8 @@ struct MsmScratch {
     G1Jac* window_sums = nullptr;
     G1Jac* result = nullptr;
     uint64_t cap_entries = 0;
-    uint32_t cap_c = 0;
+    uint64_t cap_nb = 0;
+    uint64_t cap_nchunks = 0;
 
     ~MsmScratch() {
         for (void* b : {(void*)keys_in, (void*)keys_out, (void*)vals_in, (void*)vals_out,
@@ -252,18 +253,23 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
 
     if (!tls_msm_scratch) tls_msm_scratch = std::make_unique<MsmScratch>();
     MsmScratch* s = tls_msm_scratch.get();
-    if (s->cap_entries < total || s->cap_c != c) {
+    if (s->cap_entries < total || s->cap_nb < nb || s->cap_nchunks < nchunks) {
+        // size with 1.5x slack so nearby problem sizes (n, n+2, n+3 commits,
+        // different auto window sizes) never thrash device allocations
+        uint64_t cap_total = total + total / 2 + 64;
+        uint64_t cap_nb2 = nb + nb / 2 + 64;
+        uint64_t cap_nch = nchunks + nchunks / 2 + 64;
         tls_msm_scratch = std::make_unique<MsmScratch>();
         s = tls_msm_scratch.get();
-        HIP_CHECK(hipMalloc(&s->keys_in, total * 4));
-        HIP_CHECK(hipMalloc(&s->keys_out, total * 4));
-        HIP_CHECK(hipMalloc(&s->vals_in, total * 4));
-        HIP_CHECK(hipMalloc(&s->vals_out, total * 4));
+        HIP_CHECK(hipMalloc(&s->keys_in, cap_total * 4));
+        HIP_CHECK(hipMalloc(&s->keys_out, cap_total * 4));
+        HIP_CHECK(hipMalloc(&s->vals_in, cap_total * 4));
+        HIP_CHECK(hipMalloc(&s->vals_out, cap_total * 4));
         rocprim::radix_sort_pairs(nullptr, s->sort_temp_bytes, s->keys_in, s->keys_out,
-                                  s->vals_in, s->vals_out, total, 0, 21, stream);
+                                  s->vals_in, s->vals_out, cap_total, 0, 21, stream);
         HIP_CHECK(hipMalloc(&s->sort_temp, s->sort_temp_bytes));
-        HIP_CHECK(hipMalloc(&s->head_flags, total));
-        uint64_t max_heads_cap = (nb < total ? nb : total) + 1;
+        HIP_CHECK(hipMalloc(&s->head_flags, cap_total));
+        uint64_t max_heads_cap = (cap_nb2 < cap_total ? cap_nb2 : cap_total) + 1;
         HIP_CHECK(hipMalloc(&s->heads, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->heads_sorted, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->lens, max_heads_cap * 4));
@@ -272,15 +278,16 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         {
             rocprim::counting_iterator<uint32_t> cit(0);
             (void)rocprim::select(nullptr, s->select_temp_bytes, cit, s->head_flags,
-                                  s->heads, s->head_count, total, stream);
+                                  s->heads, s->head_count, cap_total, stream);
             HIP_CHECK(hipMalloc(&s->select_temp, s->select_temp_bytes));
         }
-        HIP_CHECK(hipMalloc(&s->buckets, nb * sizeof(G1Jac)));
-        HIP_CHECK(hipMalloc(&s->partials, 2 * nchunks * sizeof(G1Jac)));
+        HIP_CHECK(hipMalloc(&s->buckets, cap_nb2 * sizeof(G1Jac)));
+        HIP_CHECK(hipMalloc(&s->partials, 2 * cap_nch * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->window_sums, 32 * MSM_SUBB * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->result, sizeof(G1Jac)));
-        s->cap_entries = total;
-        s->cap_c = c;
+        s->cap_entries = cap_total;
+        s->cap_nb = cap_nb2;
+        s->cap_nchunks = cap_nch;
     }
 
     uint32_t tb = 256;
