@@ -19,6 +19,7 @@
 #include <hip/hip_runtime.h>
 #include <rocprim/device/device_radix_sort.hpp>
 #include <rocprim/device/device_select.hpp>
+#include <rocprim/device/device_scan.hpp>
 #include <rocprim/iterator/counting_iterator.hpp>
 #include "gpu_curve.hpp"
 
@@ -90,12 +91,19 @@ __global__ void k_msm_head_flags(const uint32_t* keys, uint32_t total, uint8_t* 
     flags[t] = (key != MSM_SENTINEL && (t == 0 || keys[t - 1] != key)) ? 1 : 0;
 }
 
-// ---- 3b. segment lengths (for length-uniform wave scheduling) ----
+// Segment walks are capped at MSM_MAX_SEG entries: skewed digit
+// distributions (the top window of <2^253 scalars, duplicate scalars) create
+// segments thousands of entries long, and a single lane walking one is the
+// whole kernel's critical path.  Long segments are split into sub-segments
+// reduced in parallel, then merged per segment.
+constexpr uint32_t MSM_MAX_SEG = 128;
+
+// ---- 3b. segment lengths + sub-segment counts ----
 // heads are in increasing order (rocprim::select is stable); seg i spans
 // [heads[i], heads[i+1] or first sentinel/total).
 __global__ void k_msm_seg_lengths(const uint32_t* keys, const uint32_t* heads,
                                   const uint32_t* head_count, uint32_t total,
-                                  uint32_t* lens) {
+                                  uint32_t* lens, uint32_t* nsub) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     uint32_t hc = *head_count;
     if (t >= hc) return;
@@ -109,30 +117,58 @@ __global__ void k_msm_seg_lengths(const uint32_t* keys, const uint32_t* heads,
         while (e < total && keys[e] == key) ++e;
         end = e;
     }
-    lens[t] = end - start;
+    uint32_t len = end - start;
+    lens[t] = len;
+    nsub[t] = (len + MSM_MAX_SEG - 1) / MSM_MAX_SEG;
 }
 
-// ---- 3c. bucket segmented reduction, one thread per segment ----
-// heads_sorted: head indices sorted by segment length so all 64 lanes of a
-// wave walk near-equal-length segments (kills divergence).
-// buckets: W * 2^(c-1) Jacobian points (fully overwritten for live buckets;
-// zero-initialised so untouched buckets read as identity).
-__global__ __launch_bounds__(256) void k_msm_bucket_reduce(const uint32_t* keys, const uint32_t* vals,
-                                    const uint32_t* heads_sorted,
-                                    const uint32_t* lens_sorted,
-                                    const uint32_t* head_count, const G1Aff* bases,
-                                    G1Jac* buckets, uint32_t c) {
+// ---- 3c. emit sub-segment records (after exclusive scan of nsub) ----
+// subs: per sub-segment (start, len, head index)
+__global__ void k_msm_make_subs(const uint32_t* heads, const uint32_t* lens,
+                                const uint32_t* sub_off, const uint32_t* head_count,
+                                uint32_t* sub_start, uint32_t* sub_len) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= *head_count) return;
-    uint32_t start = heads_sorted[t];
-    uint32_t len = lens_sorted[t];
-    uint32_t key = keys[start];
+    uint32_t start = heads[t], len = lens[t], off = sub_off[t];
+    uint32_t k = 0;
+    while (len > 0) {
+        uint32_t l = len < MSM_MAX_SEG ? len : MSM_MAX_SEG;
+        sub_start[off + k] = start;
+        sub_len[off + k] = l;
+        start += l;
+        len -= l;
+        ++k;
+    }
+}
+
+// ---- 3d. sub-segment reduction: one thread per sub-segment (<= MAX_SEG) ----
+__global__ __launch_bounds__(256) void k_msm_bucket_reduce(
+    const uint32_t* vals, const uint32_t* sub_start, const uint32_t* sub_len,
+    uint32_t sub_count, const G1Aff* bases, G1Jac* partials2) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= sub_count) return;
+    uint32_t start = sub_start[t], len = sub_len[t];
     G1Jac acc = G1Jac::identity();
     for (uint32_t j = start; j < start + len; ++j) {
         uint32_t v = vals[j];
         G1Aff p = bases[v & 0x7FFFFFFFu];
         acc = acc.madd(p, (v >> 31) != 0);
     }
+    partials2[t] = acc;
+}
+
+// ---- 3e. merge sub-partials per segment -> bucket ----
+// buckets: W * 2^(c-1) Jacobian points (zero-init = identity for untouched).
+__global__ __launch_bounds__(256) void k_msm_seg_merge(
+    const uint32_t* keys, const uint32_t* heads, const uint32_t* sub_off,
+    const uint32_t* nsub, uint32_t head_count, const G1Jac* partials2,
+    G1Jac* buckets, uint32_t c) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= head_count) return;
+    uint32_t off = sub_off[t], ns = nsub[t];
+    G1Jac acc = partials2[off];
+    for (uint32_t k = 1; k < ns; ++k) acc = acc.add(partials2[off + k]);
+    uint32_t key = keys[heads[t]];
     uint32_t w = key >> 16;
     uint32_t mag = key & 0xFFFFu;  // 1 .. 2^(c-1)
     buckets[(uint64_t)w * (1u << (c - 1)) + (mag - 1)] = acc;

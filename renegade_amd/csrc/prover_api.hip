@@ -218,12 +218,17 @@ struct MsmScratch {
     size_t sort_temp_bytes = 0;
     uint8_t* head_flags = nullptr;
     uint32_t* heads = nullptr;
-    uint32_t* heads_sorted = nullptr;
     uint32_t* lens = nullptr;
-    uint32_t* lens_sorted = nullptr;
+    uint32_t* nsub = nullptr;
+    uint32_t* sub_off = nullptr;
+    uint32_t* sub_start = nullptr;
+    uint32_t* sub_len = nullptr;
+    G1Jac* partials2 = nullptr;
     uint32_t* head_count = nullptr;  // device u32
     void* select_temp = nullptr;
     size_t select_temp_bytes = 0;
+    void* scan_temp = nullptr;
+    size_t scan_temp_bytes = 0;
     G1Jac* buckets = nullptr;
     G1Jac* partials = nullptr;
     G1Jac* window_sums = nullptr;
@@ -234,9 +239,10 @@ struct MsmScratch {
 
     ~MsmScratch() {
         for (void* b : {(void*)keys_in, (void*)keys_out, (void*)vals_in, (void*)vals_out,
-                        sort_temp, (void*)head_flags, (void*)heads, (void*)heads_sorted,
-                        (void*)lens, (void*)lens_sorted, (void*)head_count,
-                        select_temp, (void*)buckets, (void*)partials, (void*)window_sums,
+                        sort_temp, (void*)head_flags, (void*)heads, (void*)lens,
+                        (void*)nsub, (void*)sub_off, (void*)sub_start, (void*)sub_len,
+                        (void*)partials2, (void*)head_count, select_temp, scan_temp,
+                        (void*)buckets, (void*)partials, (void*)window_sums,
                         (void*)result})
             if (b) hipFree(b);
     }
@@ -270,16 +276,24 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->sort_temp, s->sort_temp_bytes));
         HIP_CHECK(hipMalloc(&s->head_flags, cap_total));
         uint64_t max_heads_cap = (cap_nb2 < cap_total ? cap_nb2 : cap_total) + 1;
+        uint64_t max_subs_cap = max_heads_cap + cap_total / MSM_MAX_SEG + 1;
         HIP_CHECK(hipMalloc(&s->heads, max_heads_cap * 4));
-        HIP_CHECK(hipMalloc(&s->heads_sorted, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->lens, max_heads_cap * 4));
-        HIP_CHECK(hipMalloc(&s->lens_sorted, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->nsub, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->sub_off, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->sub_start, max_subs_cap * 4));
+        HIP_CHECK(hipMalloc(&s->sub_len, max_subs_cap * 4));
+        HIP_CHECK(hipMalloc(&s->partials2, max_subs_cap * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->head_count, 4));
         {
             rocprim::counting_iterator<uint32_t> cit(0);
             (void)rocprim::select(nullptr, s->select_temp_bytes, cit, s->head_flags,
                                   s->heads, s->head_count, cap_total, stream);
             HIP_CHECK(hipMalloc(&s->select_temp, s->select_temp_bytes));
+            (void)rocprim::exclusive_scan(nullptr, s->scan_temp_bytes, s->nsub,
+                                          s->sub_off, 0u, max_heads_cap,
+                                          rocprim::plus<uint32_t>(), stream);
+            HIP_CHECK(hipMalloc(&s->scan_temp, s->scan_temp_bytes));
         }
         HIP_CHECK(hipMalloc(&s->buckets, cap_nb2 * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->partials, 2 * cap_nch * sizeof(G1Jac)));
@@ -311,22 +325,35 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint64_t max_heads = nb < total ? nb : total;
     hipLaunchKernelGGL(k_msm_seg_lengths, dim3((uint32_t)((max_heads + tb - 1) / tb)),
                        dim3(tb), 0, stream, s->keys_out, s->heads, s->head_count,
-                       (uint32_t)total, s->lens);
+                       (uint32_t)total, s->lens, s->nsub);
     HIP_CHECK(hipGetLastError());
-    // sort heads by segment length -> every wave walks near-equal segments
-    // (rocprim needs a host-side item count: 4-byte DtoH + sync, ~10us)
     uint32_t hc = 0;
     HIP_CHECK(hipMemcpyAsync(&hc, s->head_count, 4, hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     if (hc > 0) {
-        // descending: longest segments first so the tail waves pack short work
-        rocprim::radix_sort_pairs_desc(s->sort_temp, s->sort_temp_bytes, s->lens,
-                                       s->lens_sorted, s->heads, s->heads_sorted, hc, 0,
-                                       25, stream);
+        // sub-segment split (caps any lane's walk at MSM_MAX_SEG entries)
+        (void)rocprim::exclusive_scan(s->scan_temp, s->scan_temp_bytes, s->nsub,
+                                      s->sub_off, 0u, hc, rocprim::plus<uint32_t>(),
+                                      stream);
+        uint32_t last_off = 0, last_n = 0;
+        HIP_CHECK(hipMemcpyAsync(&last_off, s->sub_off + hc - 1, 4,
+                                 hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipMemcpyAsync(&last_n, s->nsub + hc - 1, 4, hipMemcpyDeviceToHost,
+                                 stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        uint32_t sub_total = last_off + last_n;
+        hipLaunchKernelGGL(k_msm_make_subs, dim3((hc + tb - 1) / tb), dim3(tb), 0,
+                           stream, s->heads, s->lens, s->sub_off, s->head_count,
+                           s->sub_start, s->sub_len);
+        HIP_CHECK(hipGetLastError());
         et.mark(stream);
-        hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((uint32_t)((hc + tb - 1) / tb)),
-                           dim3(tb), 0, stream, s->keys_out, s->vals_out, s->heads_sorted,
-                           s->lens_sorted, s->head_count, d_bases, s->buckets, c);
+        hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((sub_total + tb - 1) / tb), dim3(tb),
+                           0, stream, s->vals_out, s->sub_start, s->sub_len, sub_total,
+                           d_bases, s->partials2);
+        HIP_CHECK(hipGetLastError());
+        hipLaunchKernelGGL(k_msm_seg_merge, dim3((hc + tb - 1) / tb), dim3(tb), 0,
+                           stream, s->keys_out, s->heads, s->sub_off, s->nsub, hc,
+                           s->partials2, s->buckets, c);
         HIP_CHECK(hipGetLastError());
     } else {
         et.mark(stream);
