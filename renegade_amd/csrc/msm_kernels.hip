@@ -98,12 +98,22 @@ __global__ void k_msm_head_flags(const uint32_t* keys, uint32_t total, uint8_t* 
 // reduced in parallel, then merged per segment.
 constexpr uint32_t MSM_MAX_SEG = 128;
 
+// choose the sub-segment cap so the reduce kernel has >= ~64K lanes in
+// flight (256 CUs want >> 256 workgroups); smaller caps cost extra merge
+// work, so keep within [8, MSM_MAX_SEG]
+__host__ inline uint32_t msm_seg_cap(uint64_t total_entries) {
+    uint64_t cap = total_entries / (1u << 16);
+    if (cap < 8) cap = 8;
+    if (cap > MSM_MAX_SEG) cap = MSM_MAX_SEG;
+    return (uint32_t)cap;
+}
+
 // ---- 3b. segment lengths + sub-segment counts ----
 // heads are in increasing order (rocprim::select is stable); seg i spans
 // [heads[i], heads[i+1] or first sentinel/total).
 __global__ void k_msm_seg_lengths(const uint32_t* keys, const uint32_t* heads,
                                   const uint32_t* head_count, uint32_t total,
-                                  uint32_t* lens, uint32_t* nsub) {
+                                  uint32_t* lens, uint32_t* nsub, uint32_t seg_cap) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     uint32_t hc = *head_count;
     if (t >= hc) return;
@@ -119,20 +129,21 @@ __global__ void k_msm_seg_lengths(const uint32_t* keys, const uint32_t* heads,
     }
     uint32_t len = end - start;
     lens[t] = len;
-    nsub[t] = (len + MSM_MAX_SEG - 1) / MSM_MAX_SEG;
+    nsub[t] = (len + seg_cap - 1) / seg_cap;
 }
 
 // ---- 3c. emit sub-segment records (after exclusive scan of nsub) ----
 // subs: per sub-segment (start, len, head index)
 __global__ void k_msm_make_subs(const uint32_t* heads, const uint32_t* lens,
                                 const uint32_t* sub_off, const uint32_t* head_count,
-                                uint32_t* sub_start, uint32_t* sub_len) {
+                                uint32_t* sub_start, uint32_t* sub_len,
+                                uint32_t seg_cap) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= *head_count) return;
     uint32_t start = heads[t], len = lens[t], off = sub_off[t];
     uint32_t k = 0;
     while (len > 0) {
-        uint32_t l = len < MSM_MAX_SEG ? len : MSM_MAX_SEG;
+        uint32_t l = len < seg_cap ? len : seg_cap;
         sub_start[off + k] = start;
         sub_len[off + k] = l;
         start += l;
@@ -144,9 +155,11 @@ __global__ void k_msm_make_subs(const uint32_t* heads, const uint32_t* lens,
 // ---- 3d. sub-segment reduction: one thread per sub-segment (<= MAX_SEG) ----
 __global__ __launch_bounds__(256) void k_msm_bucket_reduce(
     const uint32_t* vals, const uint32_t* sub_start, const uint32_t* sub_len,
+    const uint32_t* sub_order /* sub ids sorted by length desc */,
     uint32_t sub_count, const G1Aff* bases, G1Jac* partials2) {
-    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
-    if (t >= sub_count) return;
+    uint32_t tt = blockIdx.x * blockDim.x + threadIdx.x;
+    if (tt >= sub_count) return;
+    uint32_t t = sub_order[tt];
     uint32_t start = sub_start[t], len = sub_len[t];
     G1Jac acc = G1Jac::identity();
     for (uint32_t j = start; j < start + len; ++j) {

@@ -223,6 +223,8 @@ struct MsmScratch {
     uint32_t* sub_off = nullptr;
     uint32_t* sub_start = nullptr;
     uint32_t* sub_len = nullptr;
+    uint32_t* sub_len_sorted = nullptr;
+    uint32_t* sub_order = nullptr;
     G1Jac* partials2 = nullptr;
     uint32_t* head_count = nullptr;  // device u32
     void* select_temp = nullptr;
@@ -241,6 +243,7 @@ struct MsmScratch {
         for (void* b : {(void*)keys_in, (void*)keys_out, (void*)vals_in, (void*)vals_out,
                         sort_temp, (void*)head_flags, (void*)heads, (void*)lens,
                         (void*)nsub, (void*)sub_off, (void*)sub_start, (void*)sub_len,
+                        (void*)sub_len_sorted, (void*)sub_order,
                         (void*)partials2, (void*)head_count, select_temp, scan_temp,
                         (void*)buckets, (void*)partials, (void*)window_sums,
                         (void*)result})
@@ -276,13 +279,15 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->sort_temp, s->sort_temp_bytes));
         HIP_CHECK(hipMalloc(&s->head_flags, cap_total));
         uint64_t max_heads_cap = (cap_nb2 < cap_total ? cap_nb2 : cap_total) + 1;
-        uint64_t max_subs_cap = max_heads_cap + cap_total / MSM_MAX_SEG + 1;
+        uint64_t max_subs_cap = max_heads_cap + cap_total / 8 + 1;
         HIP_CHECK(hipMalloc(&s->heads, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->lens, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->nsub, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->sub_off, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->sub_start, max_subs_cap * 4));
         HIP_CHECK(hipMalloc(&s->sub_len, max_subs_cap * 4));
+        HIP_CHECK(hipMalloc(&s->sub_len_sorted, max_subs_cap * 4));
+        HIP_CHECK(hipMalloc(&s->sub_order, max_subs_cap * 4));
         HIP_CHECK(hipMalloc(&s->partials2, max_subs_cap * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->head_count, 4));
         {
@@ -323,9 +328,10 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
                               s->heads, s->head_count, total, stream);
     }
     uint64_t max_heads = nb < total ? nb : total;
+    uint32_t seg_cap = msm_seg_cap(total);
     hipLaunchKernelGGL(k_msm_seg_lengths, dim3((uint32_t)((max_heads + tb - 1) / tb)),
                        dim3(tb), 0, stream, s->keys_out, s->heads, s->head_count,
-                       (uint32_t)total, s->lens, s->nsub);
+                       (uint32_t)total, s->lens, s->nsub, seg_cap);
     HIP_CHECK(hipGetLastError());
     uint32_t hc = 0;
     HIP_CHECK(hipMemcpyAsync(&hc, s->head_count, 4, hipMemcpyDeviceToHost, stream));
@@ -344,12 +350,19 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         uint32_t sub_total = last_off + last_n;
         hipLaunchKernelGGL(k_msm_make_subs, dim3((hc + tb - 1) / tb), dim3(tb), 0,
                            stream, s->heads, s->lens, s->sub_off, s->head_count,
-                           s->sub_start, s->sub_len);
+                           s->sub_start, s->sub_len, seg_cap);
         HIP_CHECK(hipGetLastError());
+        {
+            // order subs by length desc so each wave's lanes walk equal work
+            rocprim::counting_iterator<uint32_t> cit(0);
+            rocprim::radix_sort_pairs_desc(s->sort_temp, s->sort_temp_bytes, s->sub_len,
+                                           s->sub_len_sorted, cit, s->sub_order,
+                                           sub_total, 0, 8, stream);
+        }
         et.mark(stream);
         hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((sub_total + tb - 1) / tb), dim3(tb),
-                           0, stream, s->vals_out, s->sub_start, s->sub_len, sub_total,
-                           d_bases, s->partials2);
+                           0, stream, s->vals_out, s->sub_start, s->sub_len, s->sub_order,
+                           sub_total, d_bases, s->partials2);
         HIP_CHECK(hipGetLastError());
         hipLaunchKernelGGL(k_msm_seg_merge, dim3((hc + tb - 1) / tb), dim3(tb), 0,
                            stream, s->keys_out, s->heads, s->sub_off, s->nsub, hc,
