@@ -25,7 +25,7 @@
 
 namespace rng {
 
-constexpr uint32_t MSM_SENTINEL = 0x1FFFFFu;  // > any (w<<16|mag); 21 bits
+constexpr uint32_t MSM_SENTINEL = 0x3FFFFFFu;  // > any (group<<16|mag); 26 bits
 constexpr uint32_t MSM_CHUNK = 16;            // buckets per window-sum thread
 
 // window size by problem size: bucket-phase work ~ n*W(c) while the
@@ -42,15 +42,18 @@ __host__ __device__ inline uint32_t msm_auto_c(uint64_t n) {
 // ---- 1. digit decomposition ----
 // scalars: canonical LE 4xu64. keys/vals: n*W entries, window-major
 // (out[w*n + i]) so writes coalesce per window.
+// B polys of n scalars each share one base array; digits of poly b window w
+// go to key group g = b*W + w so one sort/reduce handles the whole batch.
 __global__ __launch_bounds__(256) void k_msm_digits(const uint64_t* scalars, uint32_t n, uint32_t c,
-                             uint32_t W, uint32_t* keys, uint32_t* vals) {
-    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
+                             uint32_t W, uint32_t B, uint32_t* keys, uint32_t* vals) {
+    uint32_t idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n * B) return;
+    uint32_t b = idx / n, i = idx % n;
     uint64_t s[4];
-    s[0] = scalars[4 * i];
-    s[1] = scalars[4 * i + 1];
-    s[2] = scalars[4 * i + 2];
-    s[3] = scalars[4 * i + 3];
+    s[0] = scalars[4 * idx];
+    s[1] = scalars[4 * idx + 1];
+    s[2] = scalars[4 * idx + 2];
+    s[3] = scalars[4 * idx + 3];
     uint32_t carry = 0;
     uint32_t half = 1u << (c - 1);
     uint64_t cmask = (c == 64) ? ~0ull : ((1ull << c) - 1);
@@ -76,8 +79,8 @@ __global__ __launch_bounds__(256) void k_msm_digits(const uint64_t* scalars, uin
             sign = 0;
             carry = 0;
         }
-        uint64_t o = (uint64_t)w * n + i;
-        keys[o] = mag == 0 ? MSM_SENTINEL : ((w << 16) | mag);
+        uint64_t o = ((uint64_t)b * W + w) * n + i;
+        keys[o] = mag == 0 ? MSM_SENTINEL : (((b * W + w) << 16) | mag);
         vals[o] = (sign << 31) | i;
     }
     // carry out of the top window must be zero for scalars < 2^(W*c-1)

@@ -253,12 +253,15 @@ struct MsmScratch {
 
 static thread_local std::unique_ptr<MsmScratch> tls_msm_scratch;
 
+// B polynomials sharing one base array -> one fused pipeline; results[B].
 static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t n,
-                       uint32_t c, G1Jac* h_result, hipStream_t stream = 0) {
+                       uint32_t c, G1Jac* h_result, uint32_t B = 1,
+                       hipStream_t stream = 0) {
     uint32_t W = (256 + c - 1) / c;
-    uint64_t total = n * W;
-    uint64_t nb = (1ull << (c - 1)) * W;  // total buckets
-    uint64_t nchunks = ((1ull << (c - 1)) / MSM_CHUNK) * W;
+    uint32_t G = B * W;  // key groups
+    uint64_t total = n * G;
+    uint64_t nb = (1ull << (c - 1)) * G;  // total buckets
+    uint64_t nchunks = ((1ull << (c - 1)) / MSM_CHUNK) * G;
 
     if (!tls_msm_scratch) tls_msm_scratch = std::make_unique<MsmScratch>();
     MsmScratch* s = tls_msm_scratch.get();
@@ -275,7 +278,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->vals_in, cap_total * 4));
         HIP_CHECK(hipMalloc(&s->vals_out, cap_total * 4));
         rocprim::radix_sort_pairs(nullptr, s->sort_temp_bytes, s->keys_in, s->keys_out,
-                                  s->vals_in, s->vals_out, cap_total, 0, 21, stream);
+                                  s->vals_in, s->vals_out, cap_total, 0, 27, stream);
         HIP_CHECK(hipMalloc(&s->sort_temp, s->sort_temp_bytes));
         HIP_CHECK(hipMalloc(&s->head_flags, cap_total));
         uint64_t max_heads_cap = (cap_nb2 < cap_total ? cap_nb2 : cap_total) + 1;
@@ -302,7 +305,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         }
         HIP_CHECK(hipMalloc(&s->buckets, cap_nb2 * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->partials, 2 * cap_nch * sizeof(G1Jac)));
-        HIP_CHECK(hipMalloc(&s->window_sums, 32 * MSM_SUBB * sizeof(G1Jac)));
+        HIP_CHECK(hipMalloc(&s->window_sums, 512 * MSM_SUBB * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->result, sizeof(G1Jac)));
         s->cap_entries = cap_total;
         s->cap_nb = cap_nb2;
@@ -312,12 +315,13 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint32_t tb = 256;
     EvtTimer et;
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_digits, dim3((uint32_t)((n + tb - 1) / tb)), dim3(tb), 0,
-                       stream, d_scalars, (uint32_t)n, c, W, s->keys_in, s->vals_in);
+    hipLaunchKernelGGL(k_msm_digits, dim3((uint32_t)((n * B + tb - 1) / tb)), dim3(tb),
+                       0, stream, d_scalars, (uint32_t)n, c, W, B, s->keys_in,
+                       s->vals_in);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
     rocprim::radix_sort_pairs(s->sort_temp, s->sort_temp_bytes, s->keys_in, s->keys_out,
-                              s->vals_in, s->vals_out, total, 0, 21, stream);
+                              s->vals_in, s->vals_out, total, 0, 27, stream);
     HIP_CHECK(hipMemsetAsync(s->buckets, 0, nb * sizeof(G1Jac), stream));
     hipLaunchKernelGGL(k_msm_head_flags, dim3((uint32_t)((total + tb - 1) / tb)), dim3(tb),
                        0, stream, s->keys_out, (uint32_t)total, s->head_flags);
@@ -373,32 +377,36 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     }
     et.mark(stream);
     hipLaunchKernelGGL(k_msm_window_chunks, dim3((uint32_t)((nchunks + tb - 1) / tb)),
-                       dim3(tb), 0, stream, s->buckets, c, W, s->partials);
+                       dim3(tb), 0, stream, s->buckets, c, G, s->partials);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_window_combine, dim3(W * MSM_SUBB), dim3(64), 0, stream,
+    hipLaunchKernelGGL(k_msm_window_combine, dim3(G * MSM_SUBB), dim3(64), 0, stream,
                        s->partials, c, s->window_sums);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    // host-side fold: W*SUBB Jacobians (~1.5 KB); a single-lane dependent EC
-    // chain is far faster on a host core than on one GPU lane
-    G1Jac wsums[32 * MSM_SUBB];
-    HIP_CHECK(hipMemcpyAsync(wsums, s->window_sums, W * MSM_SUBB * sizeof(G1Jac),
+    // host-side fold: G*SUBB Jacobians; a single-lane dependent EC chain is
+    // far faster on a host core than on one GPU lane
+    std::vector<G1Jac> wsums(G * MSM_SUBB);
+    HIP_CHECK(hipMemcpyAsync(wsums.data(), s->window_sums, G * MSM_SUBB * sizeof(G1Jac),
                              hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     et.collect(tls_msm_times, 5);
-    G1Jac ws[32];
-    for (uint32_t w = 0; w < W; ++w) {
-        G1Jac sum = wsums[w * MSM_SUBB];
-        for (uint32_t i = 1; i < MSM_SUBB; ++i) sum = sum.add(wsums[w * MSM_SUBB + i]);
-        ws[w] = sum;
+    for (uint32_t b = 0; b < B; ++b) {
+        G1Jac ws[32];
+        for (uint32_t w = 0; w < W; ++w) {
+            uint32_t g = b * W + w;
+            G1Jac sum = wsums[(size_t)g * MSM_SUBB];
+            for (uint32_t i = 1; i < MSM_SUBB; ++i)
+                sum = sum.add(wsums[(size_t)g * MSM_SUBB + i]);
+            ws[w] = sum;
+        }
+        G1Jac acc = ws[W - 1];
+        for (int w = (int)W - 2; w >= 0; --w) {
+            for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
+            acc = acc.add(ws[w]);
+        }
+        h_result[b] = acc;
     }
-    G1Jac acc = ws[W - 1];
-    for (int w = (int)W - 2; w >= 0; --w) {
-        for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
-        acc = acc.add(ws[w]);
-    }
-    *h_result = acc;
     return RNG_OK;
 }
 
@@ -505,33 +513,47 @@ static int prove_scratch_ensure(uint64_t n) {
     HIP_CHECK(hipMalloc(&s->pi_coset, m * sizeof(Fr)));
     HIP_CHECK(hipMalloc(&s->q_buf, m * sizeof(Fr)));
     HIP_CHECK(hipMalloc(&s->tmp, m * sizeof(Fr)));
-    HIP_CHECK(hipMalloc(&s->stage, (n + 3) * sizeof(Fr)));
-    HIP_CHECK(hipMalloc(&s->canon, 4 * (n + 3) * 8));
+    HIP_CHECK(hipMalloc(&s->stage, 13 * (n + 3) * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->canon, 13 * 4 * (n + 3) * 8));
     s->n = n;
     s->m = m;
     return RNG_OK;
 }
 
-// commit to host coefficients via GPU MSM over the SRS device bases
+// commit to B host coefficient vectors via ONE fused GPU MSM over the SRS
+// bases (shorter polys zero-padded; zero scalars cost nothing)
+static int commit_dev_batch(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
+                            uint32_t B, G1Aff* out, bool* out_inf) {
+    ProveScratch* s = tls_prove_scratch.get();
+    uint64_t m = 0;
+    for (uint32_t b = 0; b < B; ++b)
+        if (polys[b]->size() > m) m = polys[b]->size();
+    if (m > ctx->srs_count || B > 13) return RNG_ERR_BAD_ARG;
+    HIP_CHECK(hipMemsetAsync(s->stage, 0, B * m * sizeof(Fr), 0));
+    for (uint32_t b = 0; b < B; ++b)
+        HIP_CHECK(hipMemcpyAsync(s->stage + b * m, polys[b]->data(),
+                                 polys[b]->size() * sizeof(Fr), hipMemcpyHostToDevice, 0));
+    uint32_t blocks = (uint32_t)((B * m + 255) / 256);
+    hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, 0, s->stage,
+                       s->canon, (uint32_t)(B * m));
+    HIP_CHECK(hipGetLastError());
+    G1Jac res[13];
+    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, msm_auto_c(m), res, B);
+    if (rc != RNG_OK) return rc;
+    for (uint32_t b = 0; b < B; ++b) {
+        uint64_t rec[9];
+        jac_to_affine_record(res[b], rec);
+        memcpy(out[b].x.l, rec, 32);
+        memcpy(out[b].y.l, rec + 4, 32);
+        out_inf[b] = rec[8] != 0;
+    }
+    return RNG_OK;
+}
+
 static int commit_dev(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, G1Aff* out,
                       bool* out_inf) {
-    ProveScratch* s = tls_prove_scratch.get();
-    uint64_t mdeg = coeffs.size();
-    if (mdeg > ctx->srs_count) return RNG_ERR_BAD_ARG;
-    HIP_CHECK(hipMemcpy(s->stage, coeffs.data(), mdeg * sizeof(Fr), hipMemcpyHostToDevice));
-    uint32_t blocks = (uint32_t)((mdeg + 255) / 256);
-    hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, 0, s->stage,
-                       s->canon, (uint32_t)mdeg);
-    HIP_CHECK(hipGetLastError());
-    G1Jac res;
-    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, mdeg, msm_auto_c(mdeg), &res);
-    if (rc != RNG_OK) return rc;
-    uint64_t rec[9];
-    jac_to_affine_record(res, rec);
-    memcpy(out->x.l, rec, 32);
-    memcpy(out->y.l, rec + 4, 32);
-    *out_inf = rec[8] != 0;
-    return RNG_OK;
+    const std::vector<Fr>* p = &coeffs;
+    return commit_dev_batch(ctx, &p, 1, out, out_inf);
 }
 
 // upload host coeffs, pad to m, coset-forward into dst (device, m elems)
@@ -605,14 +627,17 @@ static int plonk_preprocess_impl(RngCtxImpl* ctx, const RngCircuitDesc* d,
     if (ifft_columns(ctx, sig_evals_flat.data(), n, 5, pk->sigp) != RNG_OK)
         return RNG_ERR_HIP;
 
-    // commitments
-    bool inf;
-    for (int s = 0; s < 13; ++s)
-        if (commit_dev(ctx, pk->selq[s], &pk->sel_comms[s], &inf) != RNG_OK)
+    // commitments (two fused batch MSMs)
+    {
+        bool infs[13];
+        const std::vector<Fr>* ps[13];
+        for (int s = 0; s < 13; ++s) ps[s] = &pk->selq[s];
+        if (commit_dev_batch(ctx, ps, 13, pk->sel_comms, infs) != RNG_OK)
             return RNG_ERR_HIP;
-    for (int j = 0; j < 5; ++j)
-        if (commit_dev(ctx, pk->sigp[j], &pk->sig_comms[j], &inf) != RNG_OK)
+        for (int j = 0; j < 5; ++j) ps[j] = &pk->sigp[j];
+        if (commit_dev_batch(ctx, ps, 5, pk->sig_comms, infs) != RNG_OK)
             return RNG_ERR_HIP;
+    }
 
     // coset caches
     HIP_CHECK(hipMalloc(&pk->sel_coset, 13 * m * sizeof(Fr)));
@@ -673,10 +698,13 @@ static int plonk_prove_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, const Fr* wi
         wpoly[j][1] = wpoly[j][1].sub(b1);
         wpoly[j][n] = wpoly[j][n].add(b0);
         wpoly[j][n + 1] = wpoly[j][n + 1].add(b1);
-        if (commit_dev(ctx, wpoly[j], &comms[j], &comm_inf[j]) != RNG_OK)
-            return RNG_ERR_HIP;
-        tr.append_g1(comms[j], comm_inf[j]);
     }
+    {
+        const std::vector<Fr>* ps[5] = {&wpoly[0], &wpoly[1], &wpoly[2], &wpoly[3],
+                                        &wpoly[4]};
+        if (commit_dev_batch(ctx, ps, 5, comms, comm_inf) != RNG_OK) return RNG_ERR_HIP;
+    }
+    for (int j = 0; j < 5; ++j) tr.append_g1(comms[j], comm_inf[j]);
     if (out_link_hint) {
         // hint = wire-0 polynomial (n+2 coeffs, Montgomery) + its commitment
         memcpy(out_link_hint, wpoly[0].data(), (n + 2) * sizeof(Fr));
@@ -782,11 +810,13 @@ static int plonk_prove_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, const Fr* wi
                 quot_chunks[i].resize(n + 3, Fr::zero());
                 quot_chunks[i][n + 2] = quot_chunks[i][n + 2].add(bnext);
             }
-            if (commit_dev(ctx, quot_chunks[i], &comms[6 + i], &comm_inf[6 + i]) != RNG_OK)
-                return RNG_ERR_HIP;
-            tr.append_g1(comms[6 + i], comm_inf[6 + i]);
             prev = bnext;
         }
+        const std::vector<Fr>* ps[5] = {&quot_chunks[0], &quot_chunks[1], &quot_chunks[2],
+                                        &quot_chunks[3], &quot_chunks[4]};
+        if (commit_dev_batch(ctx, ps, 5, comms + 6, comm_inf + 6) != RNG_OK)
+            return RNG_ERR_HIP;
+        for (int i = 0; i < 5; ++i) tr.append_g1(comms[6 + i], comm_inf[6 + i]);
     }
     Fr zeta = tr.challenge();
 
@@ -848,9 +878,12 @@ static int plonk_prove_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, const Fr* wi
         hpoly_add_scaled(C, pk.sigp[j], vp);
     }
     std::vector<Fr> Wz = hpoly_div_linear(C, zeta);
-    if (commit_dev(ctx, Wz, &comms[11], &comm_inf[11]) != RNG_OK) return RNG_ERR_HIP;
     std::vector<Fr> Wzw = hpoly_div_linear(zpoly[0], zeta.mul(w));
-    if (commit_dev(ctx, Wzw, &comms[12], &comm_inf[12]) != RNG_OK) return RNG_ERR_HIP;
+    {
+        const std::vector<Fr>* ps[2] = {&Wz, &Wzw};
+        if (commit_dev_batch(ctx, ps, 2, comms + 11, comm_inf + 11) != RNG_OK)
+            return RNG_ERR_HIP;
+    }
     tr.append_g1(comms[11], comm_inf[11]);
     tr.append_g1(comms[12], comm_inf[12]);
 
@@ -1048,7 +1081,7 @@ int rng_msm_g1_dev(RngCtx* ctx, const void* dev_bases, const void* dev_scalars,
     uint32_t c = window_c > 0 ? (uint32_t)window_c : msm_auto_c(n);
     if (c < 8 || c > 16) return RNG_ERR_BAD_ARG;
     G1Jac res;
-    int rc = msm_dev_run((const G1Aff*)dev_bases, (const uint64_t*)dev_scalars, n, c, &res);
+    int rc = msm_dev_run((const G1Aff*)dev_bases, (const uint64_t*)dev_scalars, n, c, &res, 1);
     if (rc != RNG_OK) return rc;
     jac_to_affine_record(res, out9);
     return RNG_OK;
