@@ -123,6 +123,9 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--jobs", type=int, default=4,
+                    help="concurrent prover threads per GPU (the reference proves "
+                         "from a rayon pool; ctypes releases the GIL)")
     ap.add_argument("--no-kernel-legs", action="store_true",
                     help="skip the MSM/NTT kernel side-measurements")
     args = ap.parse_args()
@@ -166,9 +169,9 @@ def main():
     assert pk, "rng_preprocess failed"
     proof = np.zeros(157, dtype=np.uint64)
 
-    def step(seed):
+    def step(seed, buf=proof):
         rc = lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
-                           ctypes.c_uint64(seed), ptr(proof), None)
+                           ctypes.c_uint64(seed), ptr(buf), None)
         assert rc == 0, f"rng_prove rc={rc}"
 
     for i in range(args.warmup):
@@ -176,18 +179,27 @@ def main():
     ctx.sync()
 
     # determinism spot check
-    p1 = None
     step(12345)
     p1 = proof.copy()
     step(12345)
     assert np.array_equal(p1, proof), "nondeterministic proof"
 
+    # thread pool: independent proofs in flight per GPU (weak scaling within
+    # the device, mirroring native_proof_manager.rs:143-148)
+    from concurrent.futures import ThreadPoolExecutor
+    pool = ThreadPoolExecutor(max_workers=args.jobs)
+    bufs = [np.zeros(157, dtype=np.uint64) for _ in range(args.jobs)]
+    # warm each worker thread's scratch
+    list(pool.map(lambda j: step(500 + j, bufs[j]), range(args.jobs)))
+
     if dist:
         dist.barrier()
     ctx.sync()
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        step(10_000 + rank * 100_000 + i)
+    futs = [pool.submit(step, 10_000 + rank * 100_000 + i, bufs[i % args.jobs])
+            for i in range(args.steps)]
+    for f in futs:
+        f.result()
     ctx.sync()
     if dist:
         import torch
@@ -296,7 +308,8 @@ def main():
                 "domain_n": int(n),
                 "num_public": int(npub),
                 "srs_power": power,
-                "parallelism": f"independent proofs x{n_gpus}",
+                "prover_threads": args.jobs,
+                "parallelism": f"independent proofs x{n_gpus} gpus x{args.jobs} threads",
             },
             "roofline": roofline,
             "cpu_baseline": cb,

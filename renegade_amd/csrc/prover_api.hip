@@ -33,6 +33,10 @@ namespace rng {
         }                                                              \
     } while (0)
 
+// every prover-path op runs on the calling thread's implicit stream so
+// concurrent rng_prove calls from a thread pool overlap on the GPU
+#define RNG_STREAM hipStreamPerThread
+
 static bool gpu_ok() {
     static int cached = -1;
     if (cached < 0) {
@@ -256,7 +260,7 @@ static thread_local std::unique_ptr<MsmScratch> tls_msm_scratch;
 // B polynomials sharing one base array -> one fused pipeline; results[B].
 static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t n,
                        uint32_t c, G1Jac* h_result, uint32_t B = 1,
-                       hipStream_t stream = 0) {
+                       hipStream_t stream = RNG_STREAM) {
     uint32_t W = (256 + c - 1) / c;
     uint32_t G = B * W;  // key groups
     uint64_t total = n * G;
@@ -429,6 +433,7 @@ static void jac_to_affine_record(const G1Jac& j, uint64_t* out9) {
 // ---------------- TurboPlonk prover (GPU) ----------------
 
 static int ensure_coset_tables(RngCtxImpl* ctx, NttPlan* p) {
+    std::lock_guard<std::mutex> lk(ctx->mu);  // thread-safe lazy build
     if (p->gpow) return RNG_OK;
     uint32_t m = p->n;
     Fr g = Fr::from_u64(FR_GENERATOR);
@@ -451,9 +456,10 @@ static int coset_fwd_dev(RngCtxImpl* ctx, Fr* in, Fr* out, uint32_t m) {
     if (!p) return RNG_ERR_HIP;
     if (ensure_coset_tables(ctx, p) != RNG_OK) return RNG_ERR_HIP;
     uint32_t blocks = (m + 255) / 256;
-    hipLaunchKernelGGL(k_mul_pointwise, dim3(blocks), dim3(256), 0, 0, in, p->gpow, m);
+    hipLaunchKernelGGL(k_mul_pointwise, dim3(blocks), dim3(256), 0, RNG_STREAM, in,
+                       p->gpow, m);
     HIP_CHECK(hipGetLastError());
-    return ntt_dev_run(ctx, in, out, m, 1, false);
+    return ntt_dev_run(ctx, in, out, m, 1, false, RNG_STREAM);
 }
 
 // inverse coset NTT of `in` (m evals), result into `out`
@@ -461,10 +467,11 @@ static int coset_inv_dev(RngCtxImpl* ctx, Fr* in, Fr* out, uint32_t m) {
     NttPlan* p = get_plan(ctx, m, 1);
     if (!p) return RNG_ERR_HIP;
     if (ensure_coset_tables(ctx, p) != RNG_OK) return RNG_ERR_HIP;
-    int rc = ntt_dev_run(ctx, in, out, m, 1, true);
+    int rc = ntt_dev_run(ctx, in, out, m, 1, true, RNG_STREAM);
     if (rc != RNG_OK) return rc;
     uint32_t blocks = (m + 255) / 256;
-    hipLaunchKernelGGL(k_mul_pointwise, dim3(blocks), dim3(256), 0, 0, out, p->gpow_inv, m);
+    hipLaunchKernelGGL(k_mul_pointwise, dim3(blocks), dim3(256), 0, RNG_STREAM, out,
+                       p->gpow_inv, m);
     HIP_CHECK(hipGetLastError());
     return RNG_OK;
 }
@@ -529,13 +536,14 @@ static int commit_dev_batch(RngCtxImpl* ctx, const std::vector<Fr>* const* polys
     for (uint32_t b = 0; b < B; ++b)
         if (polys[b]->size() > m) m = polys[b]->size();
     if (m > ctx->srs_count || B > 13) return RNG_ERR_BAD_ARG;
-    HIP_CHECK(hipMemsetAsync(s->stage, 0, B * m * sizeof(Fr), 0));
+    HIP_CHECK(hipMemsetAsync(s->stage, 0, B * m * sizeof(Fr), RNG_STREAM));
     for (uint32_t b = 0; b < B; ++b)
         HIP_CHECK(hipMemcpyAsync(s->stage + b * m, polys[b]->data(),
-                                 polys[b]->size() * sizeof(Fr), hipMemcpyHostToDevice, 0));
+                                 polys[b]->size() * sizeof(Fr), hipMemcpyHostToDevice,
+                                 RNG_STREAM));
     uint32_t blocks = (uint32_t)((B * m + 255) / 256);
-    hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, 0, s->stage,
-                       s->canon, (uint32_t)(B * m));
+    hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, RNG_STREAM,
+                       s->stage, s->canon, (uint32_t)(B * m));
     HIP_CHECK(hipGetLastError());
     G1Jac res[13];
     int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, msm_auto_c(m), res, B);
@@ -560,10 +568,10 @@ static int commit_dev(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, G1Aff* out
 static int coset_of_coeffs(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, Fr* dst,
                            uint32_t m) {
     ProveScratch* s = tls_prove_scratch.get();
-    HIP_CHECK(hipMemcpy(s->stage, coeffs.data(), coeffs.size() * sizeof(Fr),
-                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpyAsync(s->stage, coeffs.data(), coeffs.size() * sizeof(Fr),
+                             hipMemcpyHostToDevice, RNG_STREAM));
     uint32_t blocks = (m + 255) / 256;
-    hipLaunchKernelGGL(k_copy_pad, dim3(blocks), dim3(256), 0, 0, s->stage,
+    hipLaunchKernelGGL(k_copy_pad, dim3(blocks), dim3(256), 0, RNG_STREAM, s->stage,
                        (uint32_t)coeffs.size(), s->tmp, m);
     HIP_CHECK(hipGetLastError());
     return coset_fwd_dev(ctx, s->tmp, dst, m);
@@ -573,26 +581,22 @@ static int coset_of_coeffs(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, Fr* d
 static int ifft_columns(RngCtxImpl* ctx, const Fr* host_evals, uint64_t n,
                         uint64_t ncols, std::vector<Fr>* out_cols) {
     Fr* d = nullptr;
-    HIP_CHECK(hipMalloc(&d, n * ncols * sizeof(Fr)));
-    if (hipMemcpy(d, host_evals, n * ncols * sizeof(Fr), hipMemcpyHostToDevice) !=
-        hipSuccess) {
+    HIP_CHECK(hipMalloc(&d, 2 * n * ncols * sizeof(Fr)));
+    if (hipMemcpyAsync(d, host_evals, n * ncols * sizeof(Fr), hipMemcpyHostToDevice,
+                       RNG_STREAM) != hipSuccess) {
         hipFree(d);
         return RNG_ERR_HIP;
     }
-    NttPlan* p = get_plan(ctx, (uint32_t)n, ncols);
-    if (!p) {
-        hipFree(d);
-        return RNG_ERR_HIP;
-    }
-    Fr* out = (n <= 4096) ? d : p->scratch;
-    int rc = ntt_dev_run(ctx, d, out, (uint32_t)n, ncols, true);
+    Fr* out = (n <= 4096) ? d : d + n * ncols;
+    int rc = ntt_dev_run(ctx, d, out, (uint32_t)n, ncols, true, RNG_STREAM);
     if (rc == RNG_OK) {
         for (uint64_t c = 0; c < ncols; ++c) {
             out_cols[c].resize(n);
-            if (hipMemcpy(out_cols[c].data(), out + c * n, n * sizeof(Fr),
-                          hipMemcpyDeviceToHost) != hipSuccess)
+            if (hipMemcpyAsync(out_cols[c].data(), out + c * n, n * sizeof(Fr),
+                               hipMemcpyDeviceToHost, RNG_STREAM) != hipSuccess)
                 rc = RNG_ERR_HIP;
         }
+        if (hipStreamSynchronize(RNG_STREAM) != hipSuccess) rc = RNG_ERR_HIP;
     }
     hipFree(d);
     return rc;
@@ -789,15 +793,16 @@ static int plonk_prove_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, const Fr* wi
     if (!mp || ensure_coset_tables(ctx, mp) != RNG_OK) return RNG_ERR_HIP;
     {
         uint32_t blocks = (m + 255) / 256;
-        hipLaunchKernelGGL(k_quotient, dim3(blocks), dim3(256), 0, 0, pk.sel_coset,
-                           pk.sig_coset, sc->w_coset, sc->z_coset, sc->pi_coset,
-                           pk.l1_coset, mp->xpow, sc->q_buf, m, ch);
+        hipLaunchKernelGGL(k_quotient, dim3(blocks), dim3(256), 0, RNG_STREAM,
+                           pk.sel_coset, pk.sig_coset, sc->w_coset, sc->z_coset,
+                           sc->pi_coset, pk.l1_coset, mp->xpow, sc->q_buf, m, ch);
         HIP_CHECK(hipGetLastError());
     }
     if (coset_inv_dev(ctx, sc->q_buf, sc->tmp, m) != RNG_OK) return RNG_ERR_HIP;
     std::vector<Fr> quot(5 * (n + 2));
-    HIP_CHECK(hipMemcpy(quot.data(), sc->tmp, quot.size() * sizeof(Fr),
-                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpyAsync(quot.data(), sc->tmp, quot.size() * sizeof(Fr),
+                             hipMemcpyDeviceToHost, RNG_STREAM));
+    HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
     std::vector<Fr> quot_chunks[5];
     {
         Fr prev = Fr::zero();
