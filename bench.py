@@ -123,7 +123,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--jobs", type=int, default=4,
+    ap.add_argument("--jobs", type=int, default=8,
                     help="concurrent prover threads per GPU (the reference proves "
                          "from a rayon pool; ctypes releases the GIL)")
     ap.add_argument("--no-kernel-legs", action="store_true",
@@ -201,6 +201,13 @@ def main():
     for f in futs:
         f.result()
     ctx.sync()
+    # concurrency correctness: a proof produced under the thread pool must be
+    # bit-identical to the same seed proved alone.  bufs[0] holds the proof of
+    # the LAST step that used it; recompute that seed serially.
+    last0 = args.steps - 1 - ((args.steps - 1) % args.jobs)
+    check = np.zeros(157, dtype=np.uint64)
+    step(10_000 + rank * 100_000 + last0, check)
+    assert np.array_equal(check, bufs[0]), "threaded proof differs from serial"
     if dist:
         import torch
         t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64)
