@@ -234,6 +234,57 @@ int orc_plonk_verify(void* pk_, const u64* pubs, const u64* proof157,
     }
 }
 
+// prove + link hint (wire-0 poly (n+2)*4 u64 + 9 u64 commitment record)
+int orc_plonk_prove_with_hint(void* pk_, const u64* wires, const u64* pubs, u64 seed,
+                              u64* out157, u64* out_hint) {
+    auto* pk = static_cast<OrcProvingKey*>(pk_);
+    try {
+        std::vector<Fr> hint;
+        OrcProof pf = orc_prove(*pk, reinterpret_cast<const Fr*>(wires),
+                                reinterpret_cast<const Fr*>(pubs), seed, &hint);
+        proof_store(pf, out157);
+        memcpy(out_hint, hint.data(), hint.size() * 32);
+        store_affine(pf.wire_comms[0], out_hint + 4 * hint.size());
+        return 0;
+    } catch (...) {
+        return -1;
+    }
+}
+
+// link proof between two wire-0 hints; out = 18 u64 (2 affine records)
+int orc_plonk_link(void* pk_, const u64* hint_a, const u64* hint_b, u64 offset,
+                   u64 count, u64* out18) {
+    auto* pk = static_cast<OrcProvingKey*>(pk_);
+    try {
+        u64 hn = pk->n + 2;
+        std::vector<Fr> pa(hn), pb(hn);
+        memcpy(pa.data(), hint_a, hn * 32);
+        memcpy(pb.data(), hint_b, hn * 32);
+        G1Affine ca = load_affine(hint_a + 4 * hn);
+        G1Affine cb = load_affine(hint_b + 4 * hn);
+        OrcLinkProof lp = orc_link_proofs(*pk, pa, ca, pb, cb, offset, count);
+        store_affine(lp.q_comm, out18);
+        store_affine(lp.opening, out18 + 9);
+        return 0;
+    } catch (...) {
+        return -1;
+    }
+}
+
+int orc_plonk_link_verify(void* pk_, const u64* comm_a9, const u64* comm_b9,
+                          const u64* proof18, u64 offset, u64 count,
+                          const u64* tau_canonical) {
+    auto* pk = static_cast<OrcProvingKey*>(pk_);
+    OrcLinkProof lp;
+    lp.q_comm = load_affine(proof18);
+    lp.opening = load_affine(proof18 + 9);
+    Fr tau = Fr::from_canonical(tau_canonical);
+    return orc_link_verify(*pk, load_affine(comm_a9), load_affine(comm_b9), lp, offset,
+                           count, tau)
+               ? 1
+               : 0;
+}
+
 // Poseidon2 hash (Montgomery limbs in/out)
 void orc_poseidon2_hash(const u64* in_mont, u64 n, u64* out_mont) {
     std::vector<Fr> in(n);

@@ -541,4 +541,109 @@ inline bool orc_verify(const OrcProvingKey& pk, const Fr* pubs, const OrcProof& 
     return lhs.to_affine() == rhs.to_affine();
 }
 
+// ---- proof linking (PlonkKzgSnark::link_proofs, called at
+// circuits-core/src/zk_circuits/proof_linking/intent_and_balance.rs:66-73;
+// hint struct pinned at plonk_proof_def.rs:143-150) ----
+//
+// Spec for this build: for link-group positions
+// S = { w^i : i in [offset, offset+count) } shared by two proofs' wire-0
+// polynomials a(X), b(X):
+//   q(X) = (a(X) - b(X)) / Z_S(X)            (exact iff they agree on S)
+//   transcript: append [a], [b], [q] -> challenge eta
+//   W(X) = F(X) / (X - eta),  F(X) = a(X) - b(X) - Z_S(eta) q(X)  (F(eta)=0)
+// link proof = ([q], [W]).  Verify: F-commitment opens to 0 at eta.
+
+struct OrcLinkProof {
+    G1Affine q_comm, opening;
+};
+
+inline std::vector<Fr> vanishing_of_positions(u64 n, u64 offset, u64 count) {
+    Fr w = fr_root_of_unity(n);
+    Fr wi = w.pow_u64(offset);
+    std::vector<Fr> z{Fr::one()};
+    for (u64 i = 0; i < count; ++i) {
+        // multiply by (X - w^(offset+i))
+        std::vector<Fr> nz(z.size() + 1, Fr::zero());
+        for (size_t j = 0; j < z.size(); ++j) {
+            nz[j + 1] = nz[j + 1] + z[j];
+            nz[j] = nz[j] - z[j] * wi;
+        }
+        z = std::move(nz);
+        wi = wi * w;
+    }
+    return z;
+}
+
+// long division p / d (d monic); returns quotient, ignores remainder
+inline std::vector<Fr> poly_div(const std::vector<Fr>& p, const std::vector<Fr>& d) {
+    if (p.size() < d.size()) return {Fr::zero()};
+    std::vector<Fr> r = p;
+    size_t dd = d.size() - 1;
+    std::vector<Fr> q(p.size() - dd, Fr::zero());
+    for (size_t i = p.size(); i-- > dd;) {
+        Fr c = r[i];  // d is monic
+        q[i - dd] = c;
+        if (c.is_zero()) continue;
+        for (size_t j = 0; j <= dd; ++j) r[i - dd + j] = r[i - dd + j] - d[j] * c;
+    }
+    return q;
+}
+
+inline OrcLinkProof orc_link_proofs(const OrcProvingKey& pk, const std::vector<Fr>& poly_a,
+                                    const G1Affine& comm_a, const std::vector<Fr>& poly_b,
+                                    const G1Affine& comm_b, u64 offset, u64 count) {
+    std::vector<Fr> diff = poly_a;
+    {
+        if (diff.size() < poly_b.size()) diff.resize(poly_b.size(), Fr::zero());
+        for (size_t i = 0; i < poly_b.size(); ++i) diff[i] = diff[i] - poly_b[i];
+    }
+    std::vector<Fr> zs = vanishing_of_positions(pk.n, offset, count);
+    std::vector<Fr> q = poly_div(diff, zs);
+    OrcLinkProof lp;
+    lp.q_comm = commit(pk.srs, q);
+    Transcript tr;
+    tr.append_u64(pk.n);
+    tr.append_u64(offset);
+    tr.append_u64(count);
+    tr.append_g1(comm_a);
+    tr.append_g1(comm_b);
+    tr.append_g1(lp.q_comm);
+    Fr eta = tr.challenge();
+    Fr zs_eta = poly_eval(zs, eta);
+    std::vector<Fr> F = diff;
+    poly_add_scaled(F, q, zs_eta.neg());
+    std::vector<Fr> W = poly_div_linear(F, eta);
+    lp.opening = commit(pk.srs, W);
+    return lp;
+}
+
+inline bool orc_link_verify(const OrcProvingKey& pk, const G1Affine& comm_a,
+                            const G1Affine& comm_b, const OrcLinkProof& lp, u64 offset,
+                            u64 count, const Fr& tau) {
+    Transcript tr;
+    tr.append_u64(pk.n);
+    tr.append_u64(offset);
+    tr.append_u64(count);
+    tr.append_g1(comm_a);
+    tr.append_g1(comm_b);
+    tr.append_g1(lp.q_comm);
+    Fr eta = tr.challenge();
+    std::vector<Fr> zs = vanishing_of_positions(pk.n, offset, count);
+    Fr zs_eta = poly_eval(zs, eta);
+    // F = A - B - zs_eta * Q ; check tau*W == eta*W + F
+    G1Proj F = G1Proj::from_affine(comm_a)
+                   .add(G1Proj::from_affine(comm_b).neg());
+    u64 sc[4];
+    zs_eta.neg().to_canonical(sc);
+    F = F.add(G1Proj::from_affine(lp.q_comm).mul(sc));
+    G1Proj W = G1Proj::from_affine(lp.opening);
+    u64 tc[4];
+    tau.to_canonical(tc);
+    G1Proj lhs = W.mul(tc);
+    u64 ec[4];
+    eta.to_canonical(ec);
+    G1Proj rhs = W.mul(ec).add(F);
+    return lhs.to_affine() == rhs.to_affine();
+}
+
 }  // namespace oracle

@@ -1276,9 +1276,85 @@ void rng_pk_comms(RngProvingKey* pk, uint64_t* out) {
 int rng_verify(RngCtx*, const RngProvingKey*, const uint64_t*, const uint64_t*) {
     return RNG_ERR_VERIFY;
 }
-int rng_link_proofs(RngCtx*, const RngProvingKey*, const uint64_t*, const uint64_t*,
-                    uint64_t, uint64_t, uint64_t*) {
-    return RNG_ERR_BAD_ARG;
+
+// Proof linking (replaces PlonkKzgSnark::link_proofs, called at
+// proof_linking/intent_and_balance.rs:66-73).  Spec: oracle/plonk.hpp §link;
+// hints are (n+2) wire-0 coefficients + a 9-u64 commitment record (the
+// layout rng_prove emits); out = ([q], [W]) as 2 affine records (18 u64).
+int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a,
+                    const uint64_t* hint_b, uint64_t group_offset, uint64_t group_size,
+                    uint64_t* out_link_proof) {
+    if (!gpu_ok()) return RNG_ERR_NO_GPU;
+    if (!ctx || !pk || !hint_a || !hint_b || !out_link_proof || group_size == 0)
+        return RNG_ERR_BAD_ARG;
+    const uint64_t n = pk->impl.n;
+    const uint64_t hn = n + 2;
+    if (prove_scratch_ensure(n) != RNG_OK) return RNG_ERR_HIP;
+    std::vector<Fr> diff(hn);
+    memcpy(diff.data(), hint_a, hn * sizeof(Fr));
+    for (uint64_t i = 0; i < hn; ++i) {
+        Fr b;
+        memcpy(b.l, hint_b + 4 * i, 32);
+        diff[i] = diff[i].sub(b);
+    }
+    // vanishing polynomial of S = {w^i : i in [offset, offset+count)}
+    Fr w = h_fr_root_of_unity((uint32_t)n);
+    Fr wi = w.pow_u64(group_offset);
+    std::vector<Fr> zs{Fr::one()};
+    for (uint64_t i = 0; i < group_size; ++i) {
+        std::vector<Fr> nz(zs.size() + 1, Fr::zero());
+        for (size_t j = 0; j < zs.size(); ++j) {
+            nz[j + 1] = nz[j + 1].add(zs[j]);
+            nz[j] = nz[j].sub(zs[j].mul(wi));
+        }
+        zs = std::move(nz);
+        wi = wi.mul(w);
+    }
+    // q = diff / zs (monic long division)
+    std::vector<Fr> q;
+    {
+        std::vector<Fr> r = diff;
+        size_t dd = zs.size() - 1;
+        if (r.size() < zs.size()) r.resize(zs.size(), Fr::zero());
+        q.assign(r.size() - dd, Fr::zero());
+        for (size_t i = r.size(); i-- > dd;) {
+            Fr c = r[i];
+            q[i - dd] = c;
+            if (c.is_zero()) continue;
+            for (size_t j = 0; j <= dd; ++j) r[i - dd + j] = r[i - dd + j].sub(zs[j].mul(c));
+        }
+    }
+    G1Aff qc;
+    bool qinf;
+    if (commit_dev(&ctx->impl, q, &qc, &qinf) != RNG_OK) return RNG_ERR_HIP;
+    HostTranscript tr;
+    tr.append_u64(n);
+    tr.append_u64(group_offset);
+    tr.append_u64(group_size);
+    auto append_rec = [&](const uint64_t* rec) {
+        G1Aff a;
+        memcpy(a.x.l, rec, 32);
+        memcpy(a.y.l, rec + 4, 32);
+        tr.append_g1(a, rec[8] != 0);
+    };
+    append_rec(hint_a + 4 * hn);
+    append_rec(hint_b + 4 * hn);
+    tr.append_g1(qc, qinf);
+    Fr eta = tr.challenge();
+    Fr zs_eta = hpoly_eval(zs, eta);
+    std::vector<Fr> F = diff;
+    hpoly_add_scaled(F, q, zs_eta.neg());
+    std::vector<Fr> W = hpoly_div_linear(F, eta);
+    G1Aff wc;
+    bool winf;
+    if (commit_dev(&ctx->impl, W, &wc, &winf) != RNG_OK) return RNG_ERR_HIP;
+    memcpy(out_link_proof, qc.x.l, 32);
+    memcpy(out_link_proof + 4, qc.y.l, 32);
+    out_link_proof[8] = qinf ? 1 : 0;
+    memcpy(out_link_proof + 9, wc.x.l, 32);
+    memcpy(out_link_proof + 13, wc.y.l, 32);
+    out_link_proof[17] = winf ? 1 : 0;
+    return RNG_OK;
 }
 
 }  // extern "C"
