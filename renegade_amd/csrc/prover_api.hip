@@ -1212,6 +1212,21 @@ void rng_circ_get(void* t_, uint64_t* selectors, uint64_t* sigma, uint64_t* wire
 
 void rng_circ_free(void* t) { delete static_cast<CircuitTables*>(t); }
 
+// host G1 addition on 9-u64 affine records (for combining per-rank MSM shard
+// results after the RCCL/gloo exchange — SURVEY.md §8e; host-side, tiny)
+void rng_g1_add_affine(const uint64_t* a9, const uint64_t* b9, uint64_t* out9) {
+    auto load = [](const uint64_t* r) {
+        G1Jac j;
+        if (r[8]) return G1Jac::identity();
+        G1Aff a;
+        memcpy(a.x.l, r, 32);
+        memcpy(a.y.l, r + 4, 32);
+        return G1Jac::from_affine(a);
+    };
+    G1Jac s = load(a9).add(load(b9));
+    jac_to_affine_record(s, out9);
+}
+
 // host keccak-256 (exported for known-answer tests of the transcript hash)
 void rng_keccak256(const uint8_t* data, size_t len, uint8_t* out32) {
     keccak256_h(data, len, out32);
