@@ -285,6 +285,16 @@ int orc_plonk_link_verify(void* pk_, const u64* comm_a9, const u64* comm_b9,
                : 0;
 }
 
+// G2 generator multiple (for pairing bilinearity tests): out = 16 u64
+// (x.c0, x.c1, y.c0, y.c1 Montgomery)
+void orc_g2_mul_gen(const u64* scalar_canonical, u64* out16) {
+    G2Affine g = G2Proj::from_affine(G2Affine::generator()).mul(scalar_canonical).to_affine();
+    memcpy(out16, g.x.c0.l, 32);
+    memcpy(out16 + 4, g.x.c1.l, 32);
+    memcpy(out16 + 8, g.y.c0.l, 32);
+    memcpy(out16 + 12, g.y.c1.l, 32);
+}
+
 // Poseidon2 hash (Montgomery limbs in/out)
 void orc_poseidon2_hash(const u64* in_mont, u64 n, u64* out_mont) {
     std::vector<Fr> in(n);

@@ -20,6 +20,7 @@
 #include "plonk_host.hpp"
 #include "test_circuits.hpp"
 #include "circuits_core.hpp"
+#include "pairing_impl.hpp"
 
 namespace rng {
 
@@ -1212,6 +1213,29 @@ void rng_circ_get(void* t_, uint64_t* selectors, uint64_t* sigma, uint64_t* wire
 
 void rng_circ_free(void* t) { delete static_cast<CircuitTables*>(t); }
 
+// pairing equality check on raw records (host; for bilinearity tests and
+// external use): p1/p2 = 9-u64 G1 affine, q1/q2 = 16-u64 G2 affine.
+// Returns 1 if e(p1, q1) == e(p2, q2).
+int rng_pairing_check(const uint64_t* p1, const uint64_t* q1, const uint64_t* p2,
+                      const uint64_t* q2) {
+    if (p1[8] || p2[8]) return -1;  // infinities unsupported here
+    Fq p1x, p1y, p2x, p2y;
+    memcpy(p1x.l, p1, 32);
+    memcpy(p1y.l, p1 + 4, 32);
+    memcpy(p2x.l, p2, 32);
+    memcpy(p2y.l, p2 + 4, 32);
+    PG2 Q1, Q2;
+    memcpy(Q1.x.a.l, q1, 32);
+    memcpy(Q1.x.b.l, q1 + 4, 32);
+    memcpy(Q1.y.a.l, q1 + 8, 32);
+    memcpy(Q1.y.b.l, q1 + 12, 32);
+    memcpy(Q2.x.a.l, q2, 32);
+    memcpy(Q2.x.b.l, q2 + 4, 32);
+    memcpy(Q2.y.a.l, q2 + 8, 32);
+    memcpy(Q2.y.b.l, q2 + 12, 32);
+    return pairing_check_eq(p1x, p1y, Q1, p2x, p2y, Q2) ? 1 : 0;
+}
+
 // host G1 addition on 9-u64 affine records (for combining per-rank MSM shard
 // results after the RCCL/gloo exchange — SURVEY.md §8e; host-side, tiny)
 void rng_g1_add_affine(const uint64_t* a9, const uint64_t* b9, uint64_t* out9) {
@@ -1286,10 +1310,180 @@ void rng_pk_comms(RngProvingKey* pk, uint64_t* out) {
     }
 }
 
-// rng_verify needs the CPU pairing (DESIGN.md roadmap); fails loudly until
-// it lands — tests verify product proofs through the oracle verifier.
-int rng_verify(RngCtx*, const RngProvingKey*, const uint64_t*, const uint64_t*) {
-    return RNG_ERR_VERIFY;
+// Full PlonK verifier with the real BN254 pairing (replaces
+// PlonkKzgSnark::verify::<SolidityTranscript>, called at traits.rs:1012).
+// Host-only CPU path by design: verification is the relayer-side self-check,
+// not the GPU proving hot loop.
+static G1Jac jac_mul_canon(const G1Jac& p, const Fr& s) {
+    u64 e[4];
+    s.to_canonical(e);
+    G1Jac acc = G1Jac::identity();
+    for (int i = 255; i >= 0; --i) {
+        acc = acc.dbl();
+        if ((e[i >> 6] >> (i & 63)) & 1) acc = acc.add(p);
+    }
+    return acc;
+}
+
+int rng_verify(RngCtx* ctx, const RngProvingKey* pkw, const uint64_t* public_inputs,
+               const uint64_t* proof) {
+    if (!ctx || !pkw || !proof) return RNG_ERR_BAD_ARG;
+    const PlonkPkImpl& pk = pkw->impl;
+    const uint64_t n = pk.n;
+    if (pk.npub > 0 && !public_inputs) return RNG_ERR_BAD_ARG;
+    const Fr* pubs = (const Fr*)public_inputs;
+    // --- deserialize proof (layout: include/rng_prover.h) ---
+    G1Jac comms[13];
+    bool infs[13];
+    for (int i = 0; i < 13; ++i) {
+        infs[i] = proof[9 * i + 8] != 0;
+        if (infs[i]) {
+            comms[i] = G1Jac::identity();
+        } else {
+            G1Aff a;
+            memcpy(a.x.l, proof + 9 * i, 32);
+            memcpy(a.y.l, proof + 9 * i + 4, 32);
+            comms[i] = G1Jac::from_affine(a);
+        }
+    }
+    Fr wire_evals[5], sigma_evals[4], z_shift_eval;
+    {
+        const uint64_t* e = proof + 117;
+        for (int i = 0; i < 5; ++i) memcpy(wire_evals[i].l, e + 4 * i, 32);
+        for (int i = 0; i < 4; ++i) memcpy(sigma_evals[i].l, e + 20 + 4 * i, 32);
+        memcpy(z_shift_eval.l, e + 36, 32);
+    }
+    // --- transcript replay ---
+    HostTranscript tr;
+    h_transcript_init(tr, pk, pubs);
+    auto append_comm = [&](int i) {
+        G1Aff a;
+        memcpy(a.x.l, proof + 9 * i, 32);
+        memcpy(a.y.l, proof + 9 * i + 4, 32);
+        tr.append_g1(a, infs[i]);
+    };
+    for (int j = 0; j < 5; ++j) append_comm(j);
+    Fr beta = tr.challenge(), gamma = tr.challenge();
+    append_comm(5);
+    Fr alpha = tr.challenge();
+    for (int i = 6; i < 11; ++i) append_comm(i);
+    Fr zeta = tr.challenge();
+    for (int j = 0; j < 5; ++j) tr.append_fr(wire_evals[j]);
+    for (int j = 0; j < 4; ++j) tr.append_fr(sigma_evals[j]);
+    tr.append_fr(z_shift_eval);
+    Fr v = tr.challenge();
+    append_comm(11);
+    append_comm(12);
+    Fr u = tr.challenge();
+
+    Fr w = h_fr_root_of_unity((uint32_t)n);
+    Fr zeta_n = zeta.pow_u64(n);
+    Fr zh_zeta = zeta_n.sub(Fr::one());
+    Fr l1_zeta = zh_zeta.mul(Fr::from_u64(n).mul(zeta.sub(Fr::one())).inverse());
+    Fr pi_zeta = Fr::zero();
+    {
+        std::vector<Fr> dens(pk.npub);
+        std::vector<Fr> wis(pk.npub);
+        Fr wi = Fr::one();
+        for (uint64_t i = 0; i < pk.npub; ++i) {
+            wis[i] = wi;
+            dens[i] = Fr::from_u64(n).mul(zeta.sub(wi));
+            wi = wi.mul(w);
+        }
+        if (pk.npub) {
+            dens = hbatch_inverse(dens);
+            for (uint64_t i = 0; i < pk.npub; ++i)
+                pi_zeta = pi_zeta.add(pubs[i].mul(wis[i]).mul(zh_zeta).mul(dens[i]));
+        }
+    }
+    auto p5 = [](const Fr& x) {
+        Fr x2 = x.sqr();
+        return x2.sqr().mul(x);
+    };
+    const Fr* wb = wire_evals;
+    Fr fbar = Fr::one(), Bbar = Fr::one();
+    for (int j = 0; j < 5; ++j)
+        fbar = fbar.mul(wb[j].add(beta.mul(pk.k[j]).mul(zeta)).add(gamma));
+    for (int j = 0; j < 4; ++j)
+        Bbar = Bbar.mul(wb[j].add(beta.mul(sigma_evals[j])).add(gamma));
+    Fr ED = pi_zeta.neg()
+                .add(alpha.mul(z_shift_eval).mul(Bbar).mul(wb[4].add(gamma)))
+                .add(alpha.sqr().mul(l1_zeta));
+
+    // [D]
+    G1Jac D = G1Jac::identity();
+    auto addc = [&](const G1Aff& c, const Fr& s) {
+        D = D.add(jac_mul_canon(G1Jac::from_affine(c), s));
+    };
+    addc(pk.sel_comms[11], Fr::one());
+    for (int j = 0; j < 4; ++j) addc(pk.sel_comms[j], wb[j]);
+    addc(pk.sel_comms[4], wb[0].mul(wb[1]));
+    addc(pk.sel_comms[5], wb[2].mul(wb[3]));
+    for (int j = 0; j < 4; ++j) addc(pk.sel_comms[6 + j], p5(wb[j]));
+    addc(pk.sel_comms[12], wb[0].mul(wb[1]).mul(wb[2]).mul(wb[3]).mul(wb[4]));
+    addc(pk.sel_comms[10], wb[4].neg());
+    D = D.add(jac_mul_canon(comms[5], alpha.mul(fbar).add(alpha.sqr().mul(l1_zeta))));
+    addc(pk.sig_comms[4], alpha.mul(beta).mul(z_shift_eval).mul(Bbar).neg());
+    {
+        Fr zpow = zh_zeta.neg();
+        Fr step = zeta.pow_u64(n + 2);
+        for (int i = 0; i < 5; ++i) {
+            D = D.add(jac_mul_canon(comms[6 + i], zpow));
+            zpow = zpow.mul(step);
+        }
+    }
+    // F, E
+    G1Jac F = D;
+    Fr E = ED;
+    Fr vp = Fr::one();
+    for (int j = 0; j < 5; ++j) {
+        vp = vp.mul(v);
+        F = F.add(jac_mul_canon(comms[j], vp));
+        E = E.add(vp.mul(wire_evals[j]));
+    }
+    for (int j = 0; j < 4; ++j) {
+        vp = vp.mul(v);
+        F = F.add(jac_mul_canon(G1Jac::from_affine(pk.sig_comms[j]), vp));
+        E = E.add(vp.mul(sigma_evals[j]));
+    }
+    F = F.add(jac_mul_canon(comms[5], u));
+    E = E.add(u.mul(z_shift_eval));
+
+    // P1 = W + u W'; P2 = zeta W + u zeta w W' + F - E*G
+    G1Jac Wz = comms[11], Wzw = comms[12];
+    G1Jac P1 = Wz.add(jac_mul_canon(Wzw, u));
+    G1Jac P2 = jac_mul_canon(Wz, zeta)
+                   .add(jac_mul_canon(Wzw, u.mul(zeta).mul(w)))
+                   .add(F);
+    {
+        G1Aff gen;
+        static const u64 gx[4] = G1_GEN_X_MONT, gy[4] = G1_GEN_Y_MONT;
+        memcpy(gen.x.l, gx, 32);
+        memcpy(gen.y.l, gy, 32);
+        P2 = P2.add(jac_mul_canon(G1Jac::from_affine(gen), E.neg()));
+    }
+    // to affine
+    auto to_aff = [](const G1Jac& j, Fq& x, Fq& y) {
+        Fq zi = j.Z.inverse();
+        Fq zi2 = zi.sqr();
+        x = j.X.mul(zi2);
+        y = j.Y.mul(zi2.mul(zi));
+    };
+    if (P1.is_identity() || P2.is_identity()) return RNG_ERR_VERIFY;
+    Fq p1x, p1y, p2x, p2y;
+    to_aff(P1, p1x, p1y);
+    to_aff(P2, p2x, p2y);
+    // e(P1, [tau]_2) == e(P2, [1]_2)
+    PG2 qh, qbh;
+    memcpy(qh.x.a.l, ctx->impl.h_g2, 32);
+    memcpy(qh.x.b.l, ctx->impl.h_g2 + 4, 32);
+    memcpy(qh.y.a.l, ctx->impl.h_g2 + 8, 32);
+    memcpy(qh.y.b.l, ctx->impl.h_g2 + 12, 32);
+    memcpy(qbh.x.a.l, ctx->impl.beta_h_g2, 32);
+    memcpy(qbh.x.b.l, ctx->impl.beta_h_g2 + 4, 32);
+    memcpy(qbh.y.a.l, ctx->impl.beta_h_g2 + 8, 32);
+    memcpy(qbh.y.b.l, ctx->impl.beta_h_g2 + 12, 32);
+    return pairing_check_eq(p1x, p1y, qbh, p2x, p2y, qh) ? RNG_OK : RNG_ERR_VERIFY;
 }
 
 // Proof linking (replaces PlonkKzgSnark::link_proofs, called at

@@ -1,0 +1,56 @@
+"""Golden-fixture regression pins (tests/golden/): the oracle prover must
+reproduce the committed proof bytes exactly — catches any protocol drift
+(circuit build, SRS, transcript, DRBG, prover algebra) across rounds."""
+import ctypes
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+U64P = ctypes.POINTER(ctypes.c_uint64)
+GOLD = Path(__file__).resolve().parent / "golden"
+ptr = lambda a: a.ctypes.data_as(U64P)
+
+
+@pytest.mark.parametrize("name,builder", [("settlement", "rng_circ_build_settlement"),
+                                          ("vbc", "rng_circ_build_vbc")])
+def test_oracle_matches_golden(orc, name, builder):
+    from renegade_amd import load_prover
+    lib = load_prover().lib
+    fn = getattr(lib, builder)
+    fn.restype = ctypes.c_void_p
+    fn.argtypes = [ctypes.c_uint64]
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    h = fn(42)
+    n = lib.rng_circ_n(h)
+    npub = lib.rng_circ_npub(h)
+    sel = np.zeros(13 * n * 4, dtype=np.uint64)
+    sigma = np.zeros(5 * n, dtype=np.uint64)
+    wires = np.zeros(5 * n * 4, dtype=np.uint64)
+    pubs = np.zeros(npub * 4, dtype=np.uint64)
+    lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+    lib.rng_circ_free(h)
+    gold_pubs = np.load(GOLD / f"{name}_pubs_cseed42.npy")
+    assert np.array_equal(pubs, gold_pubs), "statement drifted from fixture"
+    power = max(4, int(n).bit_length())
+    ptau = orc.srs_generate_ptau(power, seed=42)
+    max_degree = (1 << power) + 2
+    g1, _, _ = orc.srs_parse(ptau, max_degree)
+    srs_records = np.ascontiguousarray(g1).reshape(-1)
+    o = orc.lib
+    o.orc_plonk_preprocess.restype = ctypes.c_void_p
+    o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P, U64P,
+                                       U64P, ctypes.c_uint64]
+    o.orc_plonk_prove.argtypes = [ctypes.c_void_p, U64P, U64P, ctypes.c_uint64, U64P]
+    pk = o.orc_plonk_preprocess(n, npub, ptr(sel), ptr(sigma), ptr(srs_records),
+                                max_degree + 1)
+    proof = np.zeros(157, dtype=np.uint64)
+    assert o.orc_plonk_prove(ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
+                             ctypes.c_uint64(7), ptr(proof)) == 0
+    gold = np.load(GOLD / f"{name}_proof_cseed42_bseed7.npy")
+    assert np.array_equal(proof, gold), "proof bytes drifted from fixture"

@@ -204,5 +204,15 @@ def test_real_circuit_gpu_parity(orc, builder, seed):
     tau = np.zeros(4, dtype=np.uint64)
     o.orc_derive_tau(42, ptr(tau))
     assert o.orc_plonk_verify(ctypes.c_void_p(opk), ptr(pubs), ptr(proof), ptr(tau)) == 1
+    # product verifier (real pairing, traits.rs:1012 replacement)
+    lib.rng_verify.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P]
+    lib.rng_verify.restype = ctypes.c_int
+    assert lib.rng_verify(ctx.h, ctypes.c_void_p(pk), ptr(pubs), ptr(proof)) == 0
+    bad = proof.copy()
+    bad[118] ^= np.uint64(1)  # tamper a wire eval
+    assert lib.rng_verify(ctx.h, ctypes.c_void_p(pk), ptr(pubs), ptr(bad)) != 0
+    badp = pubs.copy()
+    badp[0] ^= np.uint64(1)
+    assert lib.rng_verify(ctx.h, ctypes.c_void_p(pk), ptr(badp), ptr(proof)) != 0
     lib.rng_pk_free(ctypes.c_void_p(pk))
     ctx.close()
