@@ -499,6 +499,7 @@ struct PlonkPkImpl {
     std::vector<Fr> selq[13], sigp[5];       // host coeffs
     std::vector<Fr> sig_evals[5];            // host evals (z build)
     G1Aff sel_comms[13], sig_comms[5];
+    bool sel_inf[13] = {false}, sig_inf[5] = {false};
     Fr* sel_coset = nullptr;   // device 13*m
     Fr* sig_coset = nullptr;   // device 5*m
     Fr* l1_coset = nullptr;    // device m
@@ -634,13 +635,12 @@ static int plonk_preprocess_impl(RngCtxImpl* ctx, const RngCircuitDesc* d,
 
     // commitments (two fused batch MSMs)
     {
-        bool infs[13];
         const std::vector<Fr>* ps[13];
         for (int s = 0; s < 13; ++s) ps[s] = &pk->selq[s];
-        if (commit_dev_batch(ctx, ps, 13, pk->sel_comms, infs) != RNG_OK)
+        if (commit_dev_batch(ctx, ps, 13, pk->sel_comms, pk->sel_inf) != RNG_OK)
             return RNG_ERR_HIP;
         for (int j = 0; j < 5; ++j) ps[j] = &pk->sigp[j];
-        if (commit_dev_batch(ctx, ps, 5, pk->sig_comms, infs) != RNG_OK)
+        if (commit_dev_batch(ctx, ps, 5, pk->sig_comms, pk->sig_inf) != RNG_OK)
             return RNG_ERR_HIP;
     }
 
@@ -673,8 +673,8 @@ static int plonk_preprocess_impl(RngCtxImpl* ctx, const RngCircuitDesc* d,
 static void h_transcript_init(HostTranscript& tr, const PlonkPkImpl& pk, const Fr* pubs) {
     tr.append_u64(pk.n);
     tr.append_u64(pk.npub);
-    for (int s = 0; s < 13; ++s) tr.append_g1(pk.sel_comms[s]);
-    for (int j = 0; j < 5; ++j) tr.append_g1(pk.sig_comms[j]);
+    for (int s = 0; s < 13; ++s) tr.append_g1(pk.sel_comms[s], pk.sel_inf[s]);
+    for (int j = 0; j < 5; ++j) tr.append_g1(pk.sig_comms[j], pk.sig_inf[j]);
     for (uint64_t i = 0; i < pk.npub; ++i) tr.append_fr(pubs[i]);
 }
 
@@ -1412,18 +1412,21 @@ int rng_verify(RngCtx* ctx, const RngProvingKey* pkw, const uint64_t* public_inp
 
     // [D]
     G1Jac D = G1Jac::identity();
-    auto addc = [&](const G1Aff& c, const Fr& s) {
-        D = D.add(jac_mul_canon(G1Jac::from_affine(c), s));
+    auto addsel = [&](int i, const Fr& s) {
+        if (pk.sel_inf[i]) return;
+        D = D.add(jac_mul_canon(G1Jac::from_affine(pk.sel_comms[i]), s));
     };
-    addc(pk.sel_comms[11], Fr::one());
-    for (int j = 0; j < 4; ++j) addc(pk.sel_comms[j], wb[j]);
-    addc(pk.sel_comms[4], wb[0].mul(wb[1]));
-    addc(pk.sel_comms[5], wb[2].mul(wb[3]));
-    for (int j = 0; j < 4; ++j) addc(pk.sel_comms[6 + j], p5(wb[j]));
-    addc(pk.sel_comms[12], wb[0].mul(wb[1]).mul(wb[2]).mul(wb[3]).mul(wb[4]));
-    addc(pk.sel_comms[10], wb[4].neg());
+    addsel(11, Fr::one());
+    for (int j = 0; j < 4; ++j) addsel(j, wb[j]);
+    addsel(4, wb[0].mul(wb[1]));
+    addsel(5, wb[2].mul(wb[3]));
+    for (int j = 0; j < 4; ++j) addsel(6 + j, p5(wb[j]));
+    addsel(12, wb[0].mul(wb[1]).mul(wb[2]).mul(wb[3]).mul(wb[4]));
+    addsel(10, wb[4].neg());
     D = D.add(jac_mul_canon(comms[5], alpha.mul(fbar).add(alpha.sqr().mul(l1_zeta))));
-    addc(pk.sig_comms[4], alpha.mul(beta).mul(z_shift_eval).mul(Bbar).neg());
+    if (!pk.sig_inf[4])
+        D = D.add(jac_mul_canon(G1Jac::from_affine(pk.sig_comms[4]),
+                                alpha.mul(beta).mul(z_shift_eval).mul(Bbar).neg()));
     {
         Fr zpow = zh_zeta.neg();
         Fr step = zeta.pow_u64(n + 2);
@@ -1443,7 +1446,8 @@ int rng_verify(RngCtx* ctx, const RngProvingKey* pkw, const uint64_t* public_inp
     }
     for (int j = 0; j < 4; ++j) {
         vp = vp.mul(v);
-        F = F.add(jac_mul_canon(G1Jac::from_affine(pk.sig_comms[j]), vp));
+        if (!pk.sig_inf[j])
+            F = F.add(jac_mul_canon(G1Jac::from_affine(pk.sig_comms[j]), vp));
         E = E.add(vp.mul(sigma_evals[j]));
     }
     F = F.add(jac_mul_canon(comms[5], u));
