@@ -333,13 +333,21 @@ def main():
         avg_bucket_ms = float(np.mean([k["bucket_reduce"] for k in kt]))
         bucket_bytes = MSM_WINDOW_C * (64 + 8) * nb
         achieved = bucket_bytes / (avg_bucket_ms / 1e3) / 1e9
+        traffic = None
+        pmc_path = REPO / "profiles" / "pmc_traffic.json"
+        if pmc_path.exists():
+            pmc = json.load(open(pmc_path))
+            traffic = pmc.get("fetch_size_reported_bytes_per_launch")
         roofline = {
             "bound": "hbm",
             "achieved": round(achieved, 1),
             "peak": HBM_PEAK_GBS,
             "unit": "GB/s",
             "frac": round(achieved / HBM_PEAK_GBS, 4),
-            "traffic": None,
+            "traffic": traffic,
+            "traffic_note": "FETCH_SIZE reported bytes/launch from committed PMC "
+                            "profile (profiles/pmc_traffic.json; gfx950 counter "
+                            "caveats in DESIGN.md)" if traffic else None,
             "kernel": "k_msm_bucket_reduce (msm 2^20 leg)",
             "kernel_ms": round(avg_bucket_ms, 3),
         }
