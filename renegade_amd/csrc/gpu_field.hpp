@@ -10,6 +10,7 @@
 #pragma once
 #include <hip/hip_runtime.h>
 #include <stdint.h>
+#include <cstring>
 #include "../../include/bn254_params.h"
 
 namespace rng {
@@ -132,6 +133,53 @@ struct Fp4 {
         }
         Fp4 r{{t[0], t[1], t[2], t[3]}};
         if (t[4] || geq_mod(r.l)) {
+            u128 bw = 0;
+            for (int i = 0; i < 4; ++i) {
+                u128 d = (u128)r.l[i] - P::mod[i] - bw;
+                r.l[i] = (u64)d;
+                bw = (d >> 64) & 1;
+            }
+        }
+        return r;
+    }
+
+    // CIOS with 32-bit words: every inner step is a single
+    // 32x32+32+32 -> 64 multiply-add (v_mad_u64_u32 shape) with the carry in
+    // the same u64 — shorter carry chains than the 64-bit formulation.
+    RNG_HD Fp4 mul32(const Fp4& o) const {
+        uint32_t a[8], b[8], p[8], t[10];
+        memcpy(a, l, 32);
+        memcpy(b, o.l, 32);
+        memcpy(p, P::mod, 32);
+        for (int i = 0; i < 10; ++i) t[i] = 0;
+        const uint32_t inv32 = (uint32_t)P::inv;
+        for (int i = 0; i < 8; ++i) {
+            uint64_t cur = 0;
+            uint32_t carry = 0;
+            for (int j = 0; j < 8; ++j) {
+                cur = (uint64_t)a[j] * b[i] + t[j] + carry;
+                t[j] = (uint32_t)cur;
+                carry = (uint32_t)(cur >> 32);
+            }
+            cur = (uint64_t)t[8] + carry;
+            t[8] = (uint32_t)cur;
+            t[9] = t[9] + (uint32_t)(cur >> 32);
+            uint32_t m = t[0] * inv32;
+            cur = (uint64_t)m * p[0] + t[0];
+            carry = (uint32_t)(cur >> 32);
+            for (int j = 1; j < 8; ++j) {
+                cur = (uint64_t)m * p[j] + t[j] + carry;
+                t[j - 1] = (uint32_t)cur;
+                carry = (uint32_t)(cur >> 32);
+            }
+            cur = (uint64_t)t[8] + carry;
+            t[7] = (uint32_t)cur;
+            t[8] = t[9] + (uint32_t)(cur >> 32);
+            t[9] = 0;
+        }
+        Fp4 r;
+        memcpy(r.l, t, 32);
+        if (t[8] || geq_mod(r.l)) {
             u128 bw = 0;
             for (int i = 0; i < 4; ++i) {
                 u128 d = (u128)r.l[i] - P::mod[i] - bw;

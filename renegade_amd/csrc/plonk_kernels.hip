@@ -73,4 +73,36 @@ __global__ __launch_bounds__(256) void k_copy_pad(const Fr* src, uint32_t len, F
     dst[i] = (i < len) ? src[i] : Fr::zero();
 }
 
+// ---- field-mul microbenchmark kernels (A/B the Montgomery formulations) ----
+// Each thread runs `iters` muls: dep = a dependent chain (latency),
+// otherwise 4 independent chains (throughput).
+template <int VARIANT>
+__global__ __launch_bounds__(256) void k_bench_frmul(Fr* io, uint32_t iters, int dep) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    Fr a = io[t & 1023];
+    Fr b = io[(t + 1) & 1023];
+    if (dep) {
+        for (uint32_t i = 0; i < iters; ++i)
+            a = (VARIANT == 0) ? a.mul(b) : a.mul32(b);
+    } else {
+        Fr c = io[(t + 2) & 1023], d = io[(t + 3) & 1023];
+        Fr e = io[(t + 4) & 1023], f = io[(t + 5) & 1023];
+        for (uint32_t i = 0; i < iters / 4; ++i) {
+            if (VARIANT == 0) {
+                a = a.mul(b);
+                c = c.mul(b);
+                d = d.mul(b);
+                e = e.mul(f);
+            } else {
+                a = a.mul32(b);
+                c = c.mul32(b);
+                d = d.mul32(b);
+                e = e.mul32(f);
+            }
+        }
+        a = a.add(c).add(d).add(e);
+    }
+    io[t & 1023] = a;  // keep alive
+}
+
 }  // namespace rng

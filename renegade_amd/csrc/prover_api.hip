@@ -1213,6 +1213,41 @@ void rng_circ_get(void* t_, uint64_t* selectors, uint64_t* sigma, uint64_t* wire
 
 void rng_circ_free(void* t) { delete static_cast<CircuitTables*>(t); }
 
+// field-mul microbenchmark: returns ms per launch; variant 0 = 64-bit CIOS,
+// 1 = 32-bit CIOS; dep = dependent chain
+double rng_bench_frmul(int variant, uint32_t blocks, uint32_t iters, int dep,
+                       int reps) {
+    if (!gpu_ok()) return -1;
+    Fr* io = nullptr;
+    if (hipMalloc(&io, 1024 * sizeof(Fr)) != hipSuccess) return -1;
+    std::vector<Fr> init(1024);
+    for (int i = 0; i < 1024; ++i) init[i] = Fr::from_u64(i * 2654435761u + 1);
+    hipMemcpy(io, init.data(), 1024 * sizeof(Fr), hipMemcpyHostToDevice);
+    auto launch = [&]() {
+        if (variant == 0)
+            hipLaunchKernelGGL(k_bench_frmul<0>, dim3(blocks), dim3(256), 0, 0, io,
+                               iters, dep);
+        else
+            hipLaunchKernelGGL(k_bench_frmul<1>, dim3(blocks), dim3(256), 0, 0, io,
+                               iters, dep);
+    };
+    launch();
+    hipDeviceSynchronize();
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0);
+    hipEventCreate(&e1);
+    hipEventRecord(e0, 0);
+    for (int r = 0; r < reps; ++r) launch();
+    hipEventRecord(e1, 0);
+    hipEventSynchronize(e1);
+    float ms = 0;
+    hipEventElapsedTime(&ms, e0, e1);
+    hipEventDestroy(e0);
+    hipEventDestroy(e1);
+    hipFree(io);
+    return ms / reps;
+}
+
 // pairing equality check on raw records (host; for bilinearity tests and
 // external use): p1/p2 = 9-u64 G1 affine, q1/q2 = 16-u64 G2 affine.
 // Returns 1 if e(p1, q1) == e(p2, q2).
