@@ -190,6 +190,71 @@ struct Fp4 {
         return r;
     }
 
+    // SOS: full 4x4 product (independent partial products -> ILP), then a
+    // separate 4-round Montgomery reduction.
+    RNG_HD Fp4 mul_sos(const Fp4& b) const {
+        u64 t[8];
+        u64 carry = 0;
+        // i = 0 row initializes
+        {
+            u128 cur;
+            cur = (u128)l[0] * b.l[0];
+            t[0] = (u64)cur;
+            carry = (u64)(cur >> 64);
+            cur = (u128)l[1] * b.l[0] + carry;
+            t[1] = (u64)cur;
+            carry = (u64)(cur >> 64);
+            cur = (u128)l[2] * b.l[0] + carry;
+            t[2] = (u64)cur;
+            carry = (u64)(cur >> 64);
+            cur = (u128)l[3] * b.l[0] + carry;
+            t[3] = (u64)cur;
+            t[4] = (u64)(cur >> 64);
+        }
+        for (int i = 1; i < 4; ++i) {
+            u64 c2 = 0;
+            for (int j = 0; j < 4; ++j) {
+                u128 cur = (u128)l[j] * b.l[i] + t[i + j] + c2;
+                t[i + j] = (u64)cur;
+                c2 = (u64)(cur >> 64);
+            }
+            t[i + 4] = c2;
+        }
+        // Montgomery reduce
+        u64 extra = 0;
+        for (int i = 0; i < 4; ++i) {
+            u64 m = t[i] * P::inv;
+            u64 c2 = 0;
+            for (int j = 0; j < 4; ++j) {
+                u128 cur = (u128)m * P::mod[j] + t[i + j] + c2;
+                t[i + j] = (u64)cur;
+                c2 = (u64)(cur >> 64);
+            }
+            // propagate into t[i+4]
+            u128 cur = (u128)t[i + 4] + c2 + ((i > 0) ? 0 : 0);
+            // also absorb previous extra at the top limb
+            cur += (i + 4 == 7) ? 0 : 0;
+            t[i + 4] = (u64)cur;
+            u64 c3 = (u64)(cur >> 64);
+            for (int k = i + 5; k < 8 && c3; ++k) {
+                u128 c4 = (u128)t[k] + c3;
+                t[k] = (u64)c4;
+                c3 = (u64)(c4 >> 64);
+            }
+            extra += (i + 5 > 7) ? c3 : 0;
+        }
+        Fp4 r{{t[4], t[5], t[6], t[7]}};
+        if (extra || geq_mod(r.l)) {
+            u128 bw = 0;
+            for (int i = 0; i < 4; ++i) {
+                u128 d = (u128)r.l[i] - P::mod[i] - bw;
+                r.l[i] = (u64)d;
+                bw = (d >> 64) & 1;
+            }
+        }
+        return r;
+    }
+
     RNG_HD Fp4 sqr() const { return mul(*this); }
 
     RNG_HD Fp4 pow(const u64 e[4]) const {

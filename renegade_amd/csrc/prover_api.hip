@@ -1227,8 +1227,11 @@ double rng_bench_frmul(int variant, uint32_t blocks, uint32_t iters, int dep,
         if (variant == 0)
             hipLaunchKernelGGL(k_bench_frmul<0>, dim3(blocks), dim3(256), 0, 0, io,
                                iters, dep);
-        else
+        else if (variant == 1)
             hipLaunchKernelGGL(k_bench_frmul<1>, dim3(blocks), dim3(256), 0, 0, io,
+                               iters, dep);
+        else
+            hipLaunchKernelGGL(k_bench_frmul<2>, dim3(blocks), dim3(256), 0, 0, io,
                                iters, dep);
     };
     launch();
