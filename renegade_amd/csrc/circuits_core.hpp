@@ -236,6 +236,16 @@ struct Balance {  // DarkpoolBalance (balance.rs): 8 scalars in field order
         return {mint, owner, relayer_fee_recipient, authority_x, authority_y,
                 relayer_fee_balance, protocol_fee_balance, amount};
     }
+    void from_scalars(const Fr* s) {
+        mint = s[0];
+        owner = s[1];
+        relayer_fee_recipient = s[2];
+        authority_x = s[3];
+        authority_y = s[4];
+        relayer_fee_balance = s[5];
+        protocol_fee_balance = s[6];
+        amount = s[7];
+    }
     static constexpr size_t NUM_SCALARS = 8;
 };
 
@@ -329,6 +339,24 @@ inline void vbc_build_witness_statement(uint64_t seed, VbcWitness& w, VbcStateme
     st.balance_commitment = native_commitment(pads, recovery, share, pub);
 }
 
+// scalar-vector (de)serialization, inverse of to_scalars (BaseType
+// from_scalars semantics, traits.rs:103-118)
+inline void vbc_witness_from_scalars(const Fr* s, VbcWitness& w) {
+    u64 c[4];
+    w.initial_share_stream.seed = s[0];
+    s[1].to_canonical(c);
+    w.initial_share_stream.index = c[0];
+    w.initial_recovery_stream.seed = s[2];
+    s[3].to_canonical(c);
+    w.initial_recovery_stream.index = c[0];
+    w.balance.from_scalars(s + 4);
+}
+inline void vbc_statement_from_scalars(const Fr* s, VbcStatement& st) {
+    st.deposit = {s[0], s[1], s[2]};
+    st.balance_commitment = s[3];
+    st.recovery_id = s[4];
+    st.new_balance_share.assign(s + 5, s + 13);
+}
 // apply_constraints (valid_balance_create.rs:44-133).  Returns nothing; all
 // witness/statement vars are allocated in order (create_witness then
 // create_public_var, traits.rs:984-991).
@@ -469,6 +497,30 @@ struct SettlementStatement {  // 17 scalars
         return v;
     }
 };
+
+inline void settlement_witness_from_scalars(const Fr* s, SettlementWitness& w) {
+    for (int i = 0; i < 2; ++i) {
+        const Fr* p = s + 32 * i;
+        w.p[i].obligation = {p[0], p[1], p[2], p[3]};
+        w.p[i].intent = {p[4], p[5], p[6], p[7], p[8]};
+        w.p[i].pre_amount_share = p[9];
+        w.p[i].input_balance.from_scalars(p + 10);
+        w.p[i].pre_in_shares = {p[18], p[19], p[20]};
+        w.p[i].output_balance.from_scalars(p + 21);
+        w.p[i].pre_out_shares = {p[29], p[30], p[31]};
+    }
+}
+inline void settlement_statement_from_scalars(const Fr* s, SettlementStatement& st) {
+    for (int i = 0; i < 2; ++i) {
+        const Fr* p = s + 7 * i;
+        st.new_amount_share[i] = p[0];
+        st.new_in_shares[i] = {p[1], p[2], p[3]};
+        st.new_out_shares[i] = {p[4], p[5], p[6]};
+    }
+    st.relayer_fee_repr[0] = s[14];
+    st.relayer_fee_repr[1] = s[15];
+    st.protocol_fee_repr = s[16];
+}
 
 // fixed-seed witness/statement (test_helpers create_witness_statement,
 // intent_and_balance_private_settlement.rs:334-420; f64 price sampling

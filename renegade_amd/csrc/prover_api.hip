@@ -1176,6 +1176,86 @@ void* rng_circ_build_settlement(uint64_t seed) {
     }
 }
 
+// circuit builders from caller-supplied witness/statement scalars (the shape
+// the external prover service receives — api_types.rs requests; Montgomery
+// limbs, field order per the reference structs)
+void* rng_circ_vbc_from_scalars(const uint64_t* witness12, const uint64_t* statement13) {
+    try {
+        VbcWitness w;
+        VbcStatement st;
+        vbc_witness_from_scalars((const Fr*)witness12, w);
+        vbc_statement_from_scalars((const Fr*)statement13, st);
+        PlonkCircuit cs;
+        vbc_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_vbc_from_scalars: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_vbc_from_scalars: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+void* rng_circ_settlement_from_scalars(const uint64_t* witness64,
+                                       const uint64_t* statement17) {
+    try {
+        SettlementWitness w;
+        SettlementStatement st;
+        settlement_witness_from_scalars((const Fr*)witness64, w);
+        settlement_statement_from_scalars((const Fr*)statement17, st);
+        PlonkCircuit cs;
+        settlement_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_settlement_from_scalars: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_settlement_from_scalars: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// expose the fixed-seed witness/statement generators (bench/test data)
+void rng_vbc_witness_statement(uint64_t seed, uint64_t* witness12, uint64_t* statement13) {
+    VbcWitness w;
+    VbcStatement st;
+    vbc_build_witness_statement(seed, w, st);
+    auto ws = w.to_scalars();
+    auto ss = st.to_scalars();
+    memcpy(witness12, ws.data(), ws.size() * sizeof(Fr));
+    memcpy(statement13, ss.data(), ss.size() * sizeof(Fr));
+}
+void rng_settlement_witness_statement(uint64_t seed, uint64_t* witness64,
+                                      uint64_t* statement17) {
+    SettlementWitness w;
+    SettlementStatement st;
+    settlement_build_witness_statement(seed, w, st);
+    std::vector<Fr> ws;
+    for (int i = 0; i < 2; ++i) {
+        auto a = w.p[i].obligation.to_scalars();
+        auto b = w.p[i].intent.to_scalars();
+        auto c = w.p[i].input_balance.to_scalars();
+        auto d = w.p[i].pre_in_shares.to_scalars();
+        auto e = w.p[i].output_balance.to_scalars();
+        auto f = w.p[i].pre_out_shares.to_scalars();
+        ws.insert(ws.end(), a.begin(), a.end());
+        ws.insert(ws.end(), b.begin(), b.end());
+        ws.push_back(w.p[i].pre_amount_share);
+        ws.insert(ws.end(), c.begin(), c.end());
+        ws.insert(ws.end(), d.begin(), d.end());
+        ws.insert(ws.end(), e.begin(), e.end());
+        ws.insert(ws.end(), f.begin(), f.end());
+    }
+    auto ss = st.to_scalars();
+    memcpy(witness64, ws.data(), ws.size() * sizeof(Fr));
+    memcpy(statement17, ss.data(), ss.size() * sizeof(Fr));
+}
+
 uint64_t rng_circ_num_link_groups(void* t) {
     return static_cast<CircuitTables*>(t)->link_groups.size();
 }

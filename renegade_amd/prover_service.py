@@ -1,0 +1,237 @@
+"""External prover service daemon (the reference's process-level seam).
+
+An unmodified relayer selects an external prover with --prover-service-url /
+--prover-service-password (config/src/cli.rs:55-62; client routes in
+proof-manager/src/implementations/external_proof_manager/
+prover_service_client.rs:100-147, request/response types api_types.rs:80-135).
+
+This daemon serves the settlement-path routes on top of the MI355X C ABI:
+
+  POST /prove-valid-balance-create                    -> ProofResponse
+  POST /prove-intent-and-balance-private-settlement   -> PrivateSettlementProofResponse
+  GET  /health
+
+Wire encoding note: the reference's request bodies are serde
+serializations of the Rust witness/statement structs.  This round encodes
+witness/statement as flat decimal-string scalar arrays in the reference
+struct field order, and proofs/link-proofs/hints in the rkyv field order
+(plonk_proof_def.rs) with decimal-string limbs; matching serde's nested JSON
+byte-for-byte is scheduled work.  Authentication mirrors the client's bearer
+password header.
+
+Run:  python -m renegade_amd.prover_service --port 8000 [--srs-power 13]
+"""
+import argparse
+import ctypes
+import os
+import threading
+from pathlib import Path
+
+import numpy as np
+
+U64P = ctypes.POINTER(ctypes.c_uint64)
+REPO = Path(__file__).resolve().parent.parent
+
+
+def ptr(a):
+    return a.ctypes.data_as(U64P)
+
+
+def scalars_to_json(arr: np.ndarray):
+    """4-limb LE records -> decimal strings."""
+    a = arr.reshape(-1, 4)
+    return [str(int(r[0]) | int(r[1]) << 64 | int(r[2]) << 128 | int(r[3]) << 192)
+            for r in a]
+
+
+def json_to_scalars(vals, count):
+    out = np.zeros(4 * count, dtype=np.uint64)
+    assert len(vals) == count, f"expected {count} scalars, got {len(vals)}"
+    for i, v in enumerate(vals):
+        x = int(v)
+        for k in range(4):
+            out[4 * i + k] = (x >> (64 * k)) & 0xFFFFFFFFFFFFFFFF
+    return out
+
+
+class ProverService:
+    """Holds the GPU context + per-circuit PKs (preprocessed once, shared
+    read-only across request threads like the reference's key cache,
+    traits.rs:80-92)."""
+
+    def __init__(self, srs_power=14, srs_seed=42):
+        import sys
+        sys.path.insert(0, str(REPO))
+        from renegade_amd import load_prover
+        from tests.orc_bindings import OracleLib
+        self.plib = load_prover()
+        self.plib.require_gpu()
+        self.lib = self.plib.lib
+        self._sig()
+        orc = OracleLib(str(REPO / "oracle" / "liborc.so"))
+        ptau = orc.srs_generate_ptau(srs_power, seed=srs_seed)
+        self.max_degree = (1 << srs_power) + 2
+        self.ctx = self.plib.init(ptau, self.max_degree)
+        self.pks = {}
+        self.pk_meta = {}
+        self.lock = threading.Lock()
+        self.seed_ctr = int.from_bytes(os.urandom(4), "little")
+
+    def _sig(self):
+        lib = self.lib
+        for name in ["rng_circ_vbc_from_scalars", "rng_circ_settlement_from_scalars"]:
+            fn = getattr(lib, name)
+            fn.restype = ctypes.c_void_p
+            fn.argtypes = [U64P, U64P]
+        lib.rng_circ_n.restype = ctypes.c_uint64
+        lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+        lib.rng_circ_npub.restype = ctypes.c_uint64
+        lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+        lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+        lib.rng_circ_num_link_groups.restype = ctypes.c_uint64
+        lib.rng_circ_num_link_groups.argtypes = [ctypes.c_void_p]
+        lib.rng_circ_link_groups.argtypes = [ctypes.c_void_p, U64P]
+        lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+        lib.rng_preprocess.restype = ctypes.c_void_p
+        lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                                  ctypes.c_uint64, U64P, U64P]
+        lib.rng_link_proofs.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                                        ctypes.c_uint64, ctypes.c_uint64, U64P]
+
+    class _Desc(ctypes.Structure):
+        _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
+                    ("selectors", U64P), ("sigma", U64P),
+                    ("num_link_groups", ctypes.c_uint64), ("link_offsets", U64P)]
+
+    def _prove_tables(self, circuit_name, handle, want_hint):
+        """Preprocess-or-cache, then prove; returns (proof, hint, meta)."""
+        lib = self.lib
+        n = lib.rng_circ_n(handle)
+        npub = lib.rng_circ_npub(handle)
+        nlg = lib.rng_circ_num_link_groups(handle)
+        lg = np.zeros(3 * max(1, nlg), dtype=np.uint64)
+        lib.rng_circ_link_groups(handle, ptr(lg))
+        sel = np.zeros(13 * n * 4, dtype=np.uint64)
+        sigma = np.zeros(5 * n, dtype=np.uint64)
+        wires = np.zeros(5 * n * 4, dtype=np.uint64)
+        pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+        lib.rng_circ_get(handle, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+        lib.rng_circ_free(handle)
+        with self.lock:
+            if circuit_name not in self.pks:
+                desc = self._Desc(n, npub, ptr(sel), ptr(sigma), 0, None)
+                pk = lib.rng_preprocess(self.ctx.h, ctypes.byref(desc))
+                if not pk:
+                    raise RuntimeError("preprocess failed")
+                self.pks[circuit_name] = pk
+                self.pk_meta[circuit_name] = dict(n=n, lg=lg.reshape(-1, 3))
+            self.seed_ctr += 1
+            seed = self.seed_ctr
+        pk = self.pks[circuit_name]
+        proof = np.zeros(157, dtype=np.uint64)
+        hint = np.zeros(4 * (n + 2) + 9, dtype=np.uint64) if want_hint else None
+        rc = lib.rng_prove(self.ctx.h, ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
+                           ctypes.c_uint64(seed), ptr(proof),
+                           ptr(hint) if want_hint else None)
+        if rc != 0:
+            raise RuntimeError(f"rng_prove rc={rc}")
+        return proof, hint, self.pk_meta[circuit_name]
+
+    # ---- route handlers ----
+    def prove_valid_balance_create(self, body):
+        w = json_to_scalars(body["witness"], 12)
+        s = json_to_scalars(body["statement"], 13)
+        h = self.lib.rng_circ_vbc_from_scalars(ptr(w), ptr(s))
+        if not h:
+            raise ValueError("unsatisfied witness/statement")
+        proof, _, _ = self._prove_tables("valid_balance_create", h, False)
+        # proof = flat 157-u64 buffer in the rkyv field order (plonk_proof_def.rs)
+        return {"proof": [str(int(x)) for x in proof]}
+
+    def prove_private_settlement(self, body):
+        w = json_to_scalars(body["witness"], 64)
+        s = json_to_scalars(body["statement"], 17)
+        h = self.lib.rng_circ_settlement_from_scalars(ptr(w), ptr(s))
+        if not h:
+            raise ValueError("unsatisfied witness/statement")
+        proof, hint, meta = self._prove_tables("intent_and_balance_private_settlement",
+                                               h, True)
+        n = meta["n"]
+        # 4 link proofs: settlement hint vs the 4 supplied hints, one per
+        # link group (party0, party1 validity; party0, party1 output balance;
+        # group order = creation order in the circuit)
+        names = ["validity_link_proof_0", "validity_link_proof_1",
+                 "output_balance_link_proof_0", "output_balance_link_proof_1"]
+        group_order = [0, 2, 1, 3]  # groups created party0(v,o), party1(v,o)
+        out = {"proof": [str(int(x)) for x in proof]}
+        for name, gidx in zip(names, group_order):
+            off, _, count = meta["lg"][gidx]
+            # external hints arrive as flat limb arrays of the hint layout
+            ext_hint = np.array([int(x) for x in body[name.replace("_proof", "_hint")]],
+                                dtype=np.uint64)
+            assert ext_hint.size == 4 * (n + 2) + 9, "bad hint size"
+            lp = np.zeros(18, dtype=np.uint64)
+            rc = self.lib.rng_link_proofs(self.ctx.h,
+                                          ctypes.c_void_p(
+                                              self.pks["intent_and_balance_private_settlement"]),
+                                          ptr(hint), ptr(ext_hint), int(off), int(count),
+                                          ptr(lp))
+            if rc != 0:
+                raise RuntimeError(f"rng_link_proofs rc={rc}")
+            out[name] = [str(int(x)) for x in lp]
+        return out
+
+
+def create_app(service=None, password=None):
+    from fastapi import FastAPI, HTTPException, Request
+
+    app = FastAPI(title="renegade_amd prover service")
+    svc = service
+
+    def auth(request: Request):
+        if password is None:
+            return
+        hdr = request.headers.get("authorization", "")
+        if hdr != f"Bearer {password}":
+            raise HTTPException(status_code=401, detail="bad password")
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok", "backend": svc.plib.version}
+
+    @app.post("/prove-valid-balance-create")
+    async def vbc(request: Request):
+        auth(request)
+        body = await request.json()
+        try:
+            return svc.prove_valid_balance_create(body)
+        except ValueError as e:
+            raise HTTPException(status_code=400, detail=str(e))
+
+    @app.post("/prove-intent-and-balance-private-settlement")
+    async def settle(request: Request):
+        auth(request)
+        body = await request.json()
+        try:
+            return svc.prove_private_settlement(body)
+        except ValueError as e:
+            raise HTTPException(status_code=400, detail=str(e))
+
+    return app
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--srs-power", type=int, default=14)
+    ap.add_argument("--password", default=os.environ.get("PROVER_SERVICE_PASSWORD"))
+    args = ap.parse_args()
+    import uvicorn
+    svc = ProverService(srs_power=args.srs_power)
+    app = create_app(svc, args.password)
+    uvicorn.run(app, host="0.0.0.0", port=args.port)
+
+
+if __name__ == "__main__":
+    main()
