@@ -399,6 +399,86 @@ inline void vbc_apply_constraints(PlonkCircuit& cs, const VbcWitness& w,
     cs.enforce_equal(comm, st_commitment);
 }
 
+// ---- equality / zero gadgets (comparators.rs EqGadget/NotEqualGadget) ----
+// z = (a == 0): allocate z (bool) and iw (= a^-1 or 0); constrain
+//   a*iw + z - 1 = 0   and   a*z = 0
+inline Var is_zero_gadget(PlonkCircuit& cs, Var a) {
+    Fr av = cs.witness(a);
+    bool zero = av.is_zero();
+    Var z = cs.create_boolean_variable(zero ? Fr::one() : Fr::zero());
+    Var iw = cs.create_variable(zero ? Fr::zero() : av.inverse());
+    {
+        std::array<Fr, NUM_SELECTORS> q{};
+        for (auto& x : q) x = Fr::zero();
+        q[SEL_MUL0] = Fr::one();
+        q[SEL_LC2] = Fr::one();
+        q[SEL_C] = Fr::one().neg();
+        cs.insert_gate({a, iw, z, 0, 0}, q);
+    }
+    {
+        std::array<Fr, NUM_SELECTORS> q{};
+        for (auto& x : q) x = Fr::zero();
+        q[SEL_MUL0] = Fr::one();
+        cs.insert_gate({a, z, 0, 0, 0}, q);
+    }
+    return z;
+}
+inline void constrain_not_equal(PlonkCircuit& cs, Var a, Var b) {
+    Var d = cs.sub(a, b);
+    Var z = is_zero_gadget(cs, d);
+    cs.enforce_false(z);
+}
+
+// ---- state wrapper vars + nullifier/merkle gadgets ----
+struct StateWrapperVars {  // StateWrapperVar (state_wrapper.rs)
+    CsprngVar recovery, share;
+    std::vector<Var> inner, public_share;  // 8 each for a balance
+};
+
+// NullifierGadget::compute_nullifier (state_primitives/nullifier.rs:21-40)
+inline Var nullifier_gadget(PlonkCircuit& cs, const StateWrapperVars& el) {
+    constrain_not_equal(cs, el.recovery.index, cs.zero());
+    Var last_idx = cs.sub(el.recovery.index, cs.one());
+    PoseidonHashGadget h(cs);
+    Var rid = h.hash(cs, {el.recovery.seed, last_idx});  // CSPRNGGadget::get_ith
+    PoseidonHashGadget h2(cs);
+    return h2.hash(cs, {rid, el.recovery.seed});
+}
+
+// PoseidonMerkleHashGadget::compute_root_prehashed (merkle.rs:29-95)
+inline Var merkle_root_gadget(PlonkCircuit& cs, Var leaf, const std::vector<Var>& elems,
+                              const std::vector<Var>& indices /*bools; true=right child*/) {
+    Var cur = leaf;
+    for (size_t i = 0; i < elems.size(); ++i) {
+        Var left = cs.mux(indices[i], elems[i], cur);
+        std::array<Fr, 4> cf{Fr::one(), Fr::one(), Fr::one().neg(), Fr::zero()};
+        Var right = cs.lc({cur, elems[i], left, cs.zero()}, cf);
+        PoseidonHashGadget h(cs);
+        cur = h.hash(cs, {left, right});
+    }
+    return cur;
+}
+
+// natives
+inline Fr native_nullifier(const Csprng& recovery) {
+    Csprng c = recovery;
+    c.index -= 1;
+    Fr rid = c.next();
+    Fr in[2] = {rid, recovery.seed};
+    return poseidon_hash(in, 2);
+}
+inline Fr native_merkle_root(const Fr& leaf, const std::vector<Fr>& elems,
+                             const std::vector<bool>& indices) {
+    Fr cur = leaf;
+    for (size_t i = 0; i < elems.size(); ++i) {
+        Fr l = indices[i] ? elems[i] : cur;
+        Fr r = indices[i] ? cur : elems[i];
+        Fr in[2] = {l, r};
+        cur = poseidon_hash(in, 2);
+    }
+    return cur;
+}
+
 // ================== Intent And Balance Private Settlement ==================
 // (zk_circuits/settlement/intent_and_balance_private_settlement.rs — the
 //  VALID MATCH MPC successor; BASELINE config #4)
@@ -715,5 +795,297 @@ inline void settlement_apply_constraints(PlonkCircuit& cs, const SettlementWitne
         cs.enforce_equal(exp_out_pfb, new_out[i][2]);
     }
 }
+
+
+// ========================= Valid Deposit =========================
+// (zk_circuits/valid_deposit.rs — deposit into an EXISTING balance with a
+//  full state rotation: Merkle opening of the old version, nullifier,
+//  re-encrypted amount share, commitment to the new version.
+//  MERKLE_HEIGHT = 10, constants/src/lib.rs:50.)
+
+constexpr int MERKLE_HEIGHT = 10;
+
+struct StateBalance {  // DarkpoolStateBalance native (20 scalars)
+    Csprng recovery, share;
+    Balance inner;
+    Fr public_share[8];
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = {recovery.seed, Fr::from_u64(recovery.index), share.seed,
+                             Fr::from_u64(share.index)};
+        auto b = inner.to_scalars();
+        v.insert(v.end(), b.begin(), b.end());
+        v.insert(v.end(), public_share, public_share + 8);
+        return v;
+    }
+};
+
+struct VdWitness {  // ValidDepositWitness<10> (20 + 20 scalars)
+    StateBalance old_balance;
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_indices[MERKLE_HEIGHT];
+};
+struct VdStatement {  // ValidDepositStatement (8 scalars)
+    Deposit deposit;
+    Fr merkle_root, old_nullifier, new_commitment, recovery_id, new_amount_share;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = deposit.to_scalars();
+        v.push_back(merkle_root);
+        v.push_back(old_nullifier);
+        v.push_back(new_commitment);
+        v.push_back(recovery_id);
+        v.push_back(new_amount_share);
+        return v;
+    }
+};
+
+// fixed-seed builder mirroring test_helpers (valid_deposit.rs:199+)
+inline void vd_build_witness_statement(uint64_t seed, VdWitness& w, VdStatement& st) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    Fr amount = Fr::from_u64(rng.next() & ((1ull << 50) - 1));
+    w.old_balance.inner = {addr(), addr(), addr(), rng.fr(), rng.fr(),
+                           Fr::from_u64(rng.next() & 0xFFFF),
+                           Fr::from_u64(rng.next() & 0xFFFF),
+                           Fr::from_u64(rng.next() & ((1ull << 50) - 1))};
+    w.old_balance.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};  // index >= 1
+    w.old_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int i = 0; i < 8; ++i) w.old_balance.public_share[i] = rng.fr();
+    for (int i = 0; i < MERKLE_HEIGHT; ++i) {
+        w.opening_elems[i] = rng.fr();
+        w.opening_indices[i] = rng.next() & 1;
+    }
+    st.deposit = {w.old_balance.inner.owner, w.old_balance.inner.mint, amount};
+
+    // old commitment + merkle root + nullifier
+    auto priv_of = [](const Balance& inner, const Fr pub_[8]) {
+        std::vector<Fr> p;
+        auto iv = inner.to_scalars();
+        for (int i = 0; i < 8; ++i) p.push_back(iv[i].sub(pub_[i]));
+        return p;
+    };
+    std::vector<Fr> old_priv = priv_of(w.old_balance.inner, w.old_balance.public_share);
+    std::vector<Fr> old_pub(w.old_balance.public_share, w.old_balance.public_share + 8);
+    Fr old_comm = native_commitment(old_priv, w.old_balance.recovery, w.old_balance.share,
+                                    old_pub);
+    st.merkle_root = native_merkle_root(
+        old_comm, std::vector<Fr>(w.opening_elems, w.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(w.opening_indices, w.opening_indices + MERKLE_HEIGHT));
+    st.old_nullifier = native_nullifier(w.old_balance.recovery);
+
+    // new balance: amount += deposit, re-encrypt amount, rotate streams
+    StateBalance nb = w.old_balance;
+    nb.inner.amount = nb.inner.amount.add(amount);
+    Csprng share = nb.share;
+    Fr pad = share.next();
+    st.new_amount_share = nb.inner.amount.sub(pad);
+    nb.public_share[7] = st.new_amount_share;
+    nb.share = share;
+    Csprng recovery = nb.recovery;
+    st.recovery_id = recovery.next();
+    nb.recovery = recovery;
+    std::vector<Fr> new_priv = priv_of(nb.inner, nb.public_share);
+    std::vector<Fr> new_pub(nb.public_share, nb.public_share + 8);
+    st.new_commitment = native_commitment(new_priv, nb.recovery, nb.share, new_pub);
+}
+
+inline void vd_apply_constraints(PlonkCircuit& cs, const VdWitness& w,
+                                 const VdStatement& st) {
+    // --- witness allocation (field order) ---
+    StateWrapperVars old_v;
+    old_v.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                      cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    old_v.share = {cs.create_variable(w.old_balance.share.seed),
+                   cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    auto iv = w.old_balance.inner.to_scalars();
+    for (auto& s : iv) old_v.inner.push_back(cs.create_variable(s));
+    for (int i = 0; i < 8; ++i)
+        old_v.public_share.push_back(cs.create_variable(w.old_balance.public_share[i]));
+    std::vector<Var> op_elems, op_idx;
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_elems.push_back(cs.create_variable(w.opening_elems[i]));
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_indices[i] ? Fr::one() : Fr::zero()));
+    // --- statement (public) ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var dep_from = pub[0], dep_token = pub[1], dep_amount = pub[2];
+    Var p_root = pub[3], p_null = pub[4], p_comm = pub[5], p_rid = pub[6],
+        p_amt_share = pub[7];
+
+    // validate_deposit (valid_deposit.rs:79-96)
+    cs.enforce_in_range(dep_amount, 100);
+    cs.enforce_equal(dep_token, old_v.inner[0]);
+    cs.enforce_equal(dep_from, old_v.inner[1]);
+
+    // complementary private shares (shares.rs:45-59)
+    std::vector<Var> old_priv;
+    for (int i = 0; i < 8; ++i)
+        old_priv.push_back(cs.sub(old_v.inner[i], old_v.public_share[i]));
+
+    // create_new_balance (:99-127)
+    StateWrapperVars new_v = old_v;
+    std::vector<Var> new_priv = old_priv;
+    new_v.inner[7] = cs.add(old_v.inner[7], dep_amount);
+    cs.enforce_in_range(new_v.inner[7], 100);
+    std::vector<Var> pads, pubs_enc;
+    {
+        std::vector<Var> vals{new_v.inner[7]};
+        stream_cipher_encrypt(cs, vals, new_v.share, pads, pubs_enc);
+        new_priv[7] = pads[0];
+        new_v.public_share[7] = pubs_enc[0];
+        cs.enforce_equal(pubs_enc[0], p_amt_share);
+    }
+
+    // rotation (state_rotation.rs:83-122)
+    Var rid = csprng_next(cs, new_v.recovery);  // RecoveryIdGadget
+    cs.enforce_equal(rid, p_rid);
+    Var old_comm = commitment_gadget(cs, old_priv, old_v.recovery, old_v.share,
+                                     old_v.public_share);
+    Var new_comm = commitment_gadget(cs, new_priv, new_v.recovery, new_v.share,
+                                     new_v.public_share);
+    cs.enforce_equal(new_comm, p_comm);
+    Var root = merkle_root_gadget(cs, old_comm, op_elems, op_idx);
+    cs.enforce_equal(root, p_root);
+    Var nul = nullifier_gadget(cs, old_v);
+    cs.enforce_equal(nul, p_null);
+}
+
+
+
+// ========================= Valid Withdrawal =========================
+// (zk_circuits/valid_withdrawal.rs — withdraw from an existing balance:
+//  nonzero amount <= balance, zero outstanding fees, full state rotation.)
+
+struct VwStatement {  // ValidWithdrawalStatement (8 scalars)
+    Fr to, token, amount;  // Withdrawal (withdrawal.rs:26-33)
+    Fr merkle_root, old_nullifier, new_commitment, recovery_id, new_amount_share;
+    std::vector<Fr> to_scalars() const {
+        return {to, token, amount, merkle_root, old_nullifier, new_commitment,
+                recovery_id, new_amount_share};
+    }
+};
+
+inline void vw_build_witness_statement(uint64_t seed, VdWitness& w, VwStatement& st) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    uint64_t bal_amount = (rng.next() & ((1ull << 50) - 1)) + 2;
+    uint64_t wd_amount = bal_amount / 2 + 1;  // nonzero, <= balance
+    w.old_balance.inner = {addr(), addr(), addr(), rng.fr(), rng.fr(),
+                           Fr::zero(), Fr::zero(),  // no outstanding fees
+                           Fr::from_u64(bal_amount)};
+    w.old_balance.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+    w.old_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int i = 0; i < 8; ++i) w.old_balance.public_share[i] = rng.fr();
+    for (int i = 0; i < MERKLE_HEIGHT; ++i) {
+        w.opening_elems[i] = rng.fr();
+        w.opening_indices[i] = rng.next() & 1;
+    }
+    st.to = w.old_balance.inner.owner;
+    st.token = w.old_balance.inner.mint;
+    st.amount = Fr::from_u64(wd_amount);
+
+    auto priv_of = [](const Balance& inner, const Fr pub_[8]) {
+        std::vector<Fr> p;
+        auto iv = inner.to_scalars();
+        for (int i = 0; i < 8; ++i) p.push_back(iv[i].sub(pub_[i]));
+        return p;
+    };
+    std::vector<Fr> old_priv = priv_of(w.old_balance.inner, w.old_balance.public_share);
+    std::vector<Fr> old_pub(w.old_balance.public_share, w.old_balance.public_share + 8);
+    Fr old_comm = native_commitment(old_priv, w.old_balance.recovery, w.old_balance.share,
+                                    old_pub);
+    st.merkle_root = native_merkle_root(
+        old_comm, std::vector<Fr>(w.opening_elems, w.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(w.opening_indices, w.opening_indices + MERKLE_HEIGHT));
+    st.old_nullifier = native_nullifier(w.old_balance.recovery);
+
+    StateBalance nb = w.old_balance;
+    nb.inner.amount = nb.inner.amount.sub(st.amount);
+    Csprng share = nb.share;
+    Fr pad = share.next();
+    st.new_amount_share = nb.inner.amount.sub(pad);
+    nb.public_share[7] = st.new_amount_share;
+    nb.share = share;
+    Csprng recovery = nb.recovery;
+    st.recovery_id = recovery.next();
+    nb.recovery = recovery;
+    std::vector<Fr> new_priv = priv_of(nb.inner, nb.public_share);
+    std::vector<Fr> new_pub(nb.public_share, nb.public_share + 8);
+    st.new_commitment = native_commitment(new_priv, nb.recovery, nb.share, new_pub);
+}
+
+inline void vw_apply_constraints(PlonkCircuit& cs, const VdWitness& w,
+                                 const VwStatement& st) {
+    StateWrapperVars old_v;
+    old_v.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                      cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    old_v.share = {cs.create_variable(w.old_balance.share.seed),
+                   cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    auto iv = w.old_balance.inner.to_scalars();
+    for (auto& s : iv) old_v.inner.push_back(cs.create_variable(s));
+    for (int i = 0; i < 8; ++i)
+        old_v.public_share.push_back(cs.create_variable(w.old_balance.public_share[i]));
+    std::vector<Var> op_elems, op_idx;
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_elems.push_back(cs.create_variable(w.opening_elems[i]));
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_indices[i] ? Fr::one() : Fr::zero()));
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var wd_to = pub[0], wd_token = pub[1], wd_amount = pub[2];
+    Var p_root = pub[3], p_null = pub[4], p_comm = pub[5], p_rid = pub[6],
+        p_amt_share = pub[7];
+
+    // validate_withdrawal (valid_withdrawal.rs:76-101)
+    cs.enforce_in_range(wd_amount, 100);
+    Var z = is_zero_gadget(cs, wd_amount);
+    cs.enforce_false(z);
+    cs.enforce_equal(wd_token, old_v.inner[0]);
+    cs.enforce_equal(wd_to, old_v.inner[1]);
+    gte_gadget(cs, old_v.inner[7], wd_amount, AMOUNT_BITS);
+    // no outstanding fees (:103-114)
+    cs.enforce_equal(old_v.inner[5], cs.zero());
+    cs.enforce_equal(old_v.inner[6], cs.zero());
+
+    std::vector<Var> old_priv;
+    for (int i = 0; i < 8; ++i)
+        old_priv.push_back(cs.sub(old_v.inner[i], old_v.public_share[i]));
+
+    StateWrapperVars new_v = old_v;
+    std::vector<Var> new_priv = old_priv;
+    new_v.inner[7] = cs.sub(old_v.inner[7], wd_amount);
+    std::vector<Var> pads, pubs_enc;
+    {
+        std::vector<Var> vals{new_v.inner[7]};
+        stream_cipher_encrypt(cs, vals, new_v.share, pads, pubs_enc);
+        new_priv[7] = pads[0];
+        new_v.public_share[7] = pubs_enc[0];
+        cs.enforce_equal(pubs_enc[0], p_amt_share);
+    }
+    Var rid = csprng_next(cs, new_v.recovery);
+    cs.enforce_equal(rid, p_rid);
+    Var old_comm = commitment_gadget(cs, old_priv, old_v.recovery, old_v.share,
+                                     old_v.public_share);
+    Var new_comm = commitment_gadget(cs, new_priv, new_v.recovery, new_v.share,
+                                     new_v.public_share);
+    cs.enforce_equal(new_comm, p_comm);
+    Var root = merkle_root_gadget(cs, old_comm, op_elems, op_idx);
+    cs.enforce_equal(root, p_root);
+    Var nul = nullifier_gadget(cs, old_v);
+    cs.enforce_equal(nul, p_null);
+}
+
 
 }  // namespace rng

@@ -1256,6 +1256,46 @@ void rng_settlement_witness_statement(uint64_t seed, uint64_t* witness64,
     memcpy(statement17, ss.data(), ss.size() * sizeof(Fr));
 }
 
+// `Valid Deposit` circuit builder (zk_circuits/valid_deposit.rs)
+void* rng_circ_build_valid_deposit(uint64_t seed) {
+    try {
+        VdWitness w;
+        VdStatement st;
+        vd_build_witness_statement(seed, w, st);
+        PlonkCircuit cs;
+        vd_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_valid_deposit: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_valid_deposit: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// `Valid Withdrawal` circuit builder (zk_circuits/valid_withdrawal.rs)
+void* rng_circ_build_valid_withdrawal(uint64_t seed) {
+    try {
+        VdWitness w;
+        VwStatement st;
+        vw_build_witness_statement(seed, w, st);
+        PlonkCircuit cs;
+        vw_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_valid_withdrawal: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_valid_withdrawal: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 uint64_t rng_circ_num_link_groups(void* t) {
     return static_cast<CircuitTables*>(t)->link_groups.size();
 }

@@ -138,7 +138,9 @@ class TestGpuProver:
 
 
 @pytest.mark.parametrize("builder,seed", [("rng_circ_build_settlement", 42),
-                                          ("rng_circ_build_vbc", 42)])
+                                          ("rng_circ_build_vbc", 42),
+                                          ("rng_circ_build_valid_deposit", 42),
+                                          ("rng_circ_build_valid_withdrawal", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
