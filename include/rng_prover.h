@@ -149,8 +149,12 @@ int rng_prove(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* wires,
 int rng_verify(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* public_inputs,
                const uint64_t* proof);
 
+/* group placement per mpc-relation GroupLayout: values live on the shared
+ * subgroup H_{2^alignment} at grid offsets [offset, offset+size) — the
+ * placement rng_circ_link_groups reports. */
 int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a,
-                    const uint64_t* hint_b, uint64_t group_offset, uint64_t group_size,
+                    const uint64_t* hint_b, uint64_t group_alignment,
+                    uint64_t group_offset, uint64_t group_size,
                     uint64_t* out_link_proof);
 
 #ifdef __cplusplus

@@ -252,8 +252,8 @@ int orc_plonk_prove_with_hint(void* pk_, const u64* wires, const u64* pubs, u64 
 }
 
 // link proof between two wire-0 hints; out = 18 u64 (2 affine records)
-int orc_plonk_link(void* pk_, const u64* hint_a, const u64* hint_b, u64 offset,
-                   u64 count, u64* out18) {
+int orc_plonk_link(void* pk_, const u64* hint_a, const u64* hint_b, u64 alignment,
+                   u64 offset, u64 count, u64* out18) {
     auto* pk = static_cast<OrcProvingKey*>(pk_);
     try {
         u64 hn = pk->n + 2;
@@ -262,7 +262,7 @@ int orc_plonk_link(void* pk_, const u64* hint_a, const u64* hint_b, u64 offset,
         memcpy(pb.data(), hint_b, hn * 32);
         G1Affine ca = load_affine(hint_a + 4 * hn);
         G1Affine cb = load_affine(hint_b + 4 * hn);
-        OrcLinkProof lp = orc_link_proofs(*pk, pa, ca, pb, cb, offset, count);
+        OrcLinkProof lp = orc_link_proofs(*pk, pa, ca, pb, cb, alignment, offset, count);
         store_affine(lp.q_comm, out18);
         store_affine(lp.opening, out18 + 9);
         return 0;
@@ -272,15 +272,15 @@ int orc_plonk_link(void* pk_, const u64* hint_a, const u64* hint_b, u64 offset,
 }
 
 int orc_plonk_link_verify(void* pk_, const u64* comm_a9, const u64* comm_b9,
-                          const u64* proof18, u64 offset, u64 count,
+                          const u64* proof18, u64 alignment, u64 offset, u64 count,
                           const u64* tau_canonical) {
     auto* pk = static_cast<OrcProvingKey*>(pk_);
     OrcLinkProof lp;
     lp.q_comm = load_affine(proof18);
     lp.opening = load_affine(proof18 + 9);
     Fr tau = Fr::from_canonical(tau_canonical);
-    return orc_link_verify(*pk, load_affine(comm_a9), load_affine(comm_b9), lp, offset,
-                           count, tau)
+    return orc_link_verify(*pk, load_affine(comm_a9), load_affine(comm_b9), lp,
+                           alignment, offset, count, tau)
                ? 1
                : 0;
 }

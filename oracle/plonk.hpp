@@ -557,8 +557,9 @@ struct OrcLinkProof {
     G1Affine q_comm, opening;
 };
 
-inline std::vector<Fr> vanishing_of_positions(u64 n, u64 offset, u64 count) {
-    Fr w = fr_root_of_unity(n);
+// positions live on the shared alignment grid H_{2^a}: x_k = w_{2^a}^(off+k)
+inline std::vector<Fr> vanishing_of_positions(u64 alignment, u64 offset, u64 count) {
+    Fr w = fr_root_of_unity(u64(1) << alignment);
     Fr wi = w.pow_u64(offset);
     std::vector<Fr> z{Fr::one()};
     for (u64 i = 0; i < count; ++i) {
@@ -591,18 +592,19 @@ inline std::vector<Fr> poly_div(const std::vector<Fr>& p, const std::vector<Fr>&
 
 inline OrcLinkProof orc_link_proofs(const OrcProvingKey& pk, const std::vector<Fr>& poly_a,
                                     const G1Affine& comm_a, const std::vector<Fr>& poly_b,
-                                    const G1Affine& comm_b, u64 offset, u64 count) {
+                                    const G1Affine& comm_b, u64 alignment, u64 offset,
+                                    u64 count) {
     std::vector<Fr> diff = poly_a;
     {
         if (diff.size() < poly_b.size()) diff.resize(poly_b.size(), Fr::zero());
         for (size_t i = 0; i < poly_b.size(); ++i) diff[i] = diff[i] - poly_b[i];
     }
-    std::vector<Fr> zs = vanishing_of_positions(pk.n, offset, count);
+    std::vector<Fr> zs = vanishing_of_positions(alignment, offset, count);
     std::vector<Fr> q = poly_div(diff, zs);
     OrcLinkProof lp;
     lp.q_comm = commit(pk.srs, q);
     Transcript tr;
-    tr.append_u64(pk.n);
+    tr.append_u64(alignment);
     tr.append_u64(offset);
     tr.append_u64(count);
     tr.append_g1(comm_a);
@@ -618,17 +620,17 @@ inline OrcLinkProof orc_link_proofs(const OrcProvingKey& pk, const std::vector<F
 }
 
 inline bool orc_link_verify(const OrcProvingKey& pk, const G1Affine& comm_a,
-                            const G1Affine& comm_b, const OrcLinkProof& lp, u64 offset,
-                            u64 count, const Fr& tau) {
+                            const G1Affine& comm_b, const OrcLinkProof& lp,
+                            u64 alignment, u64 offset, u64 count, const Fr& tau) {
     Transcript tr;
-    tr.append_u64(pk.n);
+    tr.append_u64(alignment);
     tr.append_u64(offset);
     tr.append_u64(count);
     tr.append_g1(comm_a);
     tr.append_g1(comm_b);
     tr.append_g1(lp.q_comm);
     Fr eta = tr.challenge();
-    std::vector<Fr> zs = vanishing_of_positions(pk.n, offset, count);
+    std::vector<Fr> zs = vanishing_of_positions(alignment, offset, count);
     Fr zs_eta = poly_eval(zs, eta);
     // F = A - B - zs_eta * Q ; check tau*W == eta*W + F
     G1Proj F = G1Proj::from_affine(comm_a)

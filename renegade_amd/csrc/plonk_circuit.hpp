@@ -13,6 +13,7 @@
 //   q_c + PI(X) + sum_i q_lc[i] w_i + q_mul[0] w0 w1 + q_mul[1] w2 w3
 //       + sum_i q_hash[i] w_i^5 + q_ecc w0 w1 w2 w3 w4 - q_o w4  =  0
 #pragma once
+#include <algorithm>
 #include <array>
 #include <cassert>
 #include <cstdint>
@@ -41,14 +42,18 @@ struct GateRow {
     std::array<Fr, NUM_SELECTORS> q{};        // selector values (Montgomery)
 };
 
+// Proof-linking group (mpc-relation GroupLayout semantics): the group's
+// wire-0 values are placed on the shared subgroup H_{2^alignment} at grid
+// positions [offset, offset+size), i.e. gate indices (offset+k) * (n/2^a).
+// Because w_n^(n/2^a) = w_{2^a} for every domain n, two circuits of
+// DIFFERENT sizes that place a group at the same (alignment, offset) put its
+// values at the same field points — which is what cross-circuit linking
+// opens against.
 struct LinkGroup {
     std::string id;
-    // placement (assigned at finalize or specified): offset/stride into the
-    // gate index space; stride = n/2^alignment
-    int alignment = -1;   // subdomain size = 2^alignment; -1 = auto
-    uint64_t offset = 0;  // first gate index used (assigned at finalize)
+    int alignment = -1;   // grid = 2^alignment points; -1 = auto-assign
+    int64_t offset = -1;  // grid offset; -1 = auto-assign
     std::vector<Var> vars;
-    bool placed = false;
 };
 
 // Finalized tables (feeds RngCircuitDesc / the provers)
@@ -59,10 +64,11 @@ struct CircuitTables {
     std::vector<uint64_t> sigma;  // 5*n slot permutation: sigma[j*n+i] = j'*n+i'
     std::vector<Fr> wires;      // 5 columns * n wire VALUES [j*n + i]
     std::vector<Fr> public_inputs;
-    // link groups: id -> (offset, stride, count) in gate indices
+    // link groups: id -> (alignment, grid offset, count); gate index of
+    // element k = (offset+k) * (n >> alignment)
     struct PlacedGroup {
         std::string id;
-        uint64_t offset, stride, count;
+        uint64_t alignment, offset, count;
     };
     std::vector<PlacedGroup> link_groups;
 };
@@ -300,8 +306,9 @@ class PlonkCircuit {
     void enforce_in_range(Var x, int bits) { (void)to_bits(x, bits); }
 
     // ---- proof linking ----
-    void create_link_group(const std::string& id, int alignment = -1) {
-        link_groups_.push_back(LinkGroup{id, alignment, 0, {}, false});
+    void create_link_group(const std::string& id, int alignment = -1,
+                           int64_t offset = -1) {
+        link_groups_.push_back(LinkGroup{id, alignment, offset, {}});
     }
     void add_to_link_group(Var v, const std::string& id) {
         for (auto& g : link_groups_) {
@@ -367,37 +374,86 @@ inline CircuitTables PlonkCircuit::finalize() {
     CircuitTables T;
     size_t n_pub = public_vars_.size();
 
-    // assemble final gate order: pub gates first
-    std::vector<GateRow> all;
-    all.reserve(n_pub + gates_.size());
+    // main gate order: pub gates first, then the explicit gates
+    std::vector<GateRow> main;
+    main.reserve(n_pub + gates_.size());
     for (Var pv : public_vars_) {
         auto q = zq();
         q[SEL_LC0] = Fr::one().neg();  // -pub + PI(w^i) = 0
-        all.push_back(GateRow{{pv, 0, 0, 0, 0}, q});
+        main.push_back(GateRow{{pv, 0, 0, 0, 0}, q});
     }
-    for (auto& g : gates_) all.push_back(g);
+    for (auto& g : gates_) main.push_back(g);
 
-    // link-group gates: one gate per linked var, wire0 carries the var.
-    // Placed contiguously after the main gates; stride 1.
-    std::vector<CircuitTables::PlacedGroup> placed;
-    for (auto& lg : link_groups_) {
-        uint64_t off = all.size();
-        for (Var v : lg.vars) {
-            auto q = zq();  // all-zero selectors: gate is trivially satisfied
-            all.push_back(GateRow{{v, 0, 0, 0, 0}, q});
+    // ---- link-group placement on the alignment grid ----
+    uint64_t total_link = 0;
+    for (auto& lg : link_groups_) total_link += lg.vars.size();
+    // auto alignment: smallest grid holding offset-1 start + all groups
+    int auto_align = 0;
+    while ((1ull << auto_align) < total_link + 2) auto_align++;
+    int align = auto_align;
+    for (auto& lg : link_groups_)
+        if (lg.alignment > align) align = lg.alignment;
+    // assign grid offsets (consecutive from 1) where unspecified
+    {
+        int64_t next_off = 1;
+        for (auto& lg : link_groups_) {
+            if (lg.offset >= 0) {
+                next_off = std::max(next_off, lg.offset + (int64_t)lg.vars.size());
+            }
         }
-        placed.push_back({lg.id, off, 1, lg.vars.size()});
+        int64_t cursor = 1;
+        for (auto& lg : link_groups_) {
+            if (lg.offset < 0) {
+                lg.offset = cursor;
+                cursor += (int64_t)lg.vars.size();
+            } else {
+                cursor = std::max(cursor, lg.offset + (int64_t)lg.vars.size());
+            }
+        }
+        if ((1ll << align) < cursor) {
+            while ((1ll << align) < cursor) align++;
+        }
     }
 
-    uint64_t n = 1;
-    while (n < all.size()) n <<= 1;
-    if (n < 8) n = 8;
+    // domain size: must fit main gates + link gates and give stride > n_pub
+    uint64_t n = 8;
+    auto fits = [&](uint64_t nn) {
+        if (nn < main.size() + total_link) return false;
+        if (link_groups_.empty()) return true;
+        if (nn < (1ull << align)) return false;
+        uint64_t stride = nn >> align;
+        if (stride <= n_pub) return false;  // grid slots must avoid PI gates
+        return true;
+    };
+    while (!fits(n)) n <<= 1;
 
-    // pad with empty gates (all wires = var 0, selectors = 0)
+    // lay out: reserved slots for link gates, others fill in order
+    std::vector<GateRow> all(n);
     GateRow pad{};
     pad.wires = {0, 0, 0, 0, 0};
     pad.q = zq();
-    while (all.size() < n) all.push_back(pad);
+    for (auto& g : all) g = pad;
+    std::vector<bool> reserved(n, false);
+    std::vector<CircuitTables::PlacedGroup> placed;
+    uint64_t stride = link_groups_.empty() ? 0 : (n >> align);
+    for (auto& lg : link_groups_) {
+        for (size_t k = 0; k < lg.vars.size(); ++k) {
+            uint64_t idx = (uint64_t)(lg.offset + (int64_t)k) * stride;
+            auto q = zq();  // all-zero selectors: trivially satisfied
+            all[idx] = GateRow{{lg.vars[k], 0, 0, 0, 0}, q};
+            reserved[idx] = true;
+        }
+        placed.push_back({lg.id, (uint64_t)align, (uint64_t)lg.offset, lg.vars.size()});
+    }
+    {
+        uint64_t slot = 0;
+        for (auto& g : main) {
+            while (reserved[slot]) ++slot;
+            all[slot++] = g;
+        }
+        // note: PI gates land at 0..n_pub-1 because stride > n_pub keeps the
+        // reserved grid slots out of that prefix
+    }
 
     T.n = n;
     T.num_public = n_pub;

@@ -1256,6 +1256,34 @@ void rng_settlement_witness_statement(uint64_t seed, uint64_t* witness64,
     memcpy(statement17, ss.data(), ss.size() * sizeof(Fr));
 }
 
+// linked test circuit: `count` link-group values derived from value_seed,
+// placed at (alignment, offset), plus filler gates scaled by `scale` —
+// two different scales give two circuits of different domain size whose
+// group values coincide (the cross-circuit linking shape: a validity proof
+// linking into the settlement proof across domains).
+void* rng_testcirc_build_linked(uint64_t value_seed, uint64_t scale, uint64_t alignment,
+                                uint64_t offset, uint64_t count) {
+    try {
+        PlonkCircuit cs;
+        cs.create_link_group("xlink", (int)alignment, (int64_t)offset);
+        Lcg vr(value_seed);
+        for (uint64_t k = 0; k < count; ++k) {
+            Var v = cs.create_variable(vr.fr());
+            cs.add_to_link_group(v, "xlink");
+        }
+        build_mixed_circuit(cs, value_seed + 1000 * scale, scale);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_testcirc_build_linked: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_testcirc_build_linked: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // `Valid Deposit` circuit builder (zk_circuits/valid_deposit.rs)
 void* rng_circ_build_valid_deposit(uint64_t seed) {
     try {
@@ -1299,12 +1327,12 @@ void* rng_circ_build_valid_withdrawal(uint64_t seed) {
 uint64_t rng_circ_num_link_groups(void* t) {
     return static_cast<CircuitTables*>(t)->link_groups.size();
 }
-// out: per group 3 u64 (offset, stride, count), in creation order
+// out: per group 3 u64 (alignment, grid offset, count), in creation order
 void rng_circ_link_groups(void* t_, uint64_t* out) {
     auto* t = static_cast<CircuitTables*>(t_);
     for (size_t i = 0; i < t->link_groups.size(); ++i) {
-        out[3 * i] = t->link_groups[i].offset;
-        out[3 * i + 1] = t->link_groups[i].stride;
+        out[3 * i] = t->link_groups[i].alignment;
+        out[3 * i + 1] = t->link_groups[i].offset;
         out[3 * i + 2] = t->link_groups[i].count;
     }
 }
@@ -1653,7 +1681,8 @@ int rng_verify(RngCtx* ctx, const RngProvingKey* pkw, const uint64_t* public_inp
 // hints are (n+2) wire-0 coefficients + a 9-u64 commitment record (the
 // layout rng_prove emits); out = ([q], [W]) as 2 affine records (18 u64).
 int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a,
-                    const uint64_t* hint_b, uint64_t group_offset, uint64_t group_size,
+                    const uint64_t* hint_b, uint64_t group_alignment,
+                    uint64_t group_offset, uint64_t group_size,
                     uint64_t* out_link_proof) {
     if (!gpu_ok()) return RNG_ERR_NO_GPU;
     if (!ctx || !pk || !hint_a || !hint_b || !out_link_proof || group_size == 0)
@@ -1668,8 +1697,8 @@ int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a
         memcpy(b.l, hint_b + 4 * i, 32);
         diff[i] = diff[i].sub(b);
     }
-    // vanishing polynomial of S = {w^i : i in [offset, offset+count)}
-    Fr w = h_fr_root_of_unity((uint32_t)n);
+    // vanishing polynomial of S = {w_{2^a}^(off+k)} on the alignment grid
+    Fr w = h_fr_root_of_unity(1u << group_alignment);
     Fr wi = w.pow_u64(group_offset);
     std::vector<Fr> zs{Fr::one()};
     for (uint64_t i = 0; i < group_size; ++i) {
@@ -1699,7 +1728,7 @@ int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a
     bool qinf;
     if (commit_dev(&ctx->impl, q, &qc, &qinf) != RNG_OK) return RNG_ERR_HIP;
     HostTranscript tr;
-    tr.append_u64(n);
+    tr.append_u64(group_alignment);
     tr.append_u64(group_offset);
     tr.append_u64(group_size);
     auto append_rec = [&](const uint64_t* rec) {

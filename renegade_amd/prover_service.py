@@ -97,7 +97,8 @@ class ProverService:
         lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
                                   ctypes.c_uint64, U64P, U64P]
         lib.rng_link_proofs.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
-                                        ctypes.c_uint64, ctypes.c_uint64, U64P]
+                                        ctypes.c_uint64, ctypes.c_uint64,
+                                        ctypes.c_uint64, U64P]
 
     class _Desc(ctypes.Structure):
         _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
@@ -166,7 +167,7 @@ class ProverService:
         group_order = [0, 2, 1, 3]  # groups created party0(v,o), party1(v,o)
         out = {"proof": [str(int(x)) for x in proof]}
         for name, gidx in zip(names, group_order):
-            off, _, count = meta["lg"][gidx]
+            align, off, count = meta["lg"][gidx]
             # external hints arrive as flat limb arrays of the hint layout
             ext_hint = np.array([int(x) for x in body[name.replace("_proof", "_hint")]],
                                 dtype=np.uint64)
@@ -175,8 +176,8 @@ class ProverService:
             rc = self.lib.rng_link_proofs(self.ctx.h,
                                           ctypes.c_void_p(
                                               self.pks["intent_and_balance_private_settlement"]),
-                                          ptr(hint), ptr(ext_hint), int(off), int(count),
-                                          ptr(lp))
+                                          ptr(hint), ptr(ext_hint), int(align), int(off),
+                                          int(count), ptr(lp))
             if rc != 0:
                 raise RuntimeError(f"rng_link_proofs rc={rc}")
             out[name] = [str(int(x)) for x in lp]

@@ -58,6 +58,9 @@ class TestSettlementCircuit:
         assert stl["nlg"] == 4
         counts = sorted(int(c) for c in stl["lg"][:, 2])
         assert counts == [11, 11, 17, 17]
+        # grid-aligned placement: all on one shared subgroup, disjoint offsets
+        aligns = set(int(a) for a in stl["lg"][:, 0])
+        assert len(aligns) == 1
         print("settlement n =", stl["n"])
 
     def test_oracle_prove_verify(self, stl, orc):

@@ -62,9 +62,10 @@ def linkset(orc):
     o.orc_plonk_prove_with_hint.argtypes = [ctypes.c_void_p, U64P, U64P,
                                             ctypes.c_uint64, U64P, U64P]
     o.orc_plonk_link.argtypes = [ctypes.c_void_p, U64P, U64P, ctypes.c_uint64,
-                                 ctypes.c_uint64, U64P]
+                                 ctypes.c_uint64, ctypes.c_uint64, U64P]
     o.orc_plonk_link_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P,
-                                        ctypes.c_uint64, ctypes.c_uint64, U64P]
+                                        ctypes.c_uint64, ctypes.c_uint64,
+                                        ctypes.c_uint64, U64P]
     o.orc_derive_tau.argtypes = [ctypes.c_uint64, U64P]
     pk = o.orc_plonk_preprocess(n, npub, ptr(t["sel"]), ptr(t["sigma"]),
                                 ptr(srs_records), max_degree + 1)
@@ -91,15 +92,15 @@ class TestLinkOracle:
         o = s["orc"].lib
         _, ha = s["oprove"](s["t"], 7)
         _, hb = s["oprove"](s["t"], 8)  # different blinders, same wire values
-        for off, stride, count in s["t"]["lg"]:
-            assert stride == 1
+        for align, off, count in s["t"]["lg"]:
             lp = np.zeros(18, dtype=np.uint64)
             rc = o.orc_plonk_link(ctypes.c_void_p(s["pk"]), ptr(ha), ptr(hb),
-                                  ctypes.c_uint64(int(off)), ctypes.c_uint64(int(count)),
-                                  ptr(lp))
+                                  ctypes.c_uint64(int(align)), ctypes.c_uint64(int(off)),
+                                  ctypes.c_uint64(int(count)), ptr(lp))
             assert rc == 0
             ok = o.orc_plonk_link_verify(ctypes.c_void_p(s["pk"]), ptr(ha[-9:].copy()),
                                          ptr(hb[-9:].copy()), ptr(lp),
+                                         ctypes.c_uint64(int(align)),
                                          ctypes.c_uint64(int(off)),
                                          ctypes.c_uint64(int(count)), ptr(s["tau"]))
             assert ok == 1, f"link group at {off} failed to verify"
@@ -109,12 +110,14 @@ class TestLinkOracle:
         o = s["orc"].lib
         _, ha = s["oprove"](s["t"], 7)
         _, hb = s["oprove"](s["t2"], 7)  # different witness values
-        off, _, count = s["t"]["lg"][0]
+        align, off, count = s["t"]["lg"][0]
         lp = np.zeros(18, dtype=np.uint64)
         o.orc_plonk_link(ctypes.c_void_p(s["pk"]), ptr(ha), ptr(hb),
-                         ctypes.c_uint64(int(off)), ctypes.c_uint64(int(count)), ptr(lp))
+                         ctypes.c_uint64(int(align)), ctypes.c_uint64(int(off)),
+                         ctypes.c_uint64(int(count)), ptr(lp))
         ok = o.orc_plonk_link_verify(ctypes.c_void_p(s["pk"]), ptr(ha[-9:].copy()),
                                      ptr(hb[-9:].copy()), ptr(lp),
+                                     ctypes.c_uint64(int(align)),
                                      ctypes.c_uint64(int(off)),
                                      ctypes.c_uint64(int(count)), ptr(s["tau"]))
         assert ok != 1
@@ -133,7 +136,8 @@ class TestLinkGpu:
         lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
                                   ctypes.c_uint64, U64P, U64P]
         lib.rng_link_proofs.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
-                                        ctypes.c_uint64, ctypes.c_uint64, U64P]
+                                        ctypes.c_uint64, ctypes.c_uint64,
+                                        ctypes.c_uint64, U64P]
         n, npub = s["n"], s["npub"]
         ctx = plib.init(s["ptau"], s["max_degree"])
 
@@ -160,17 +164,19 @@ class TestLinkGpu:
         _, hb_o = s["oprove"](t, 8)
         assert np.array_equal(ha_g, ha_o)  # hints bit-exact
         o = s["orc"].lib
-        for off, _, count in t["lg"]:
+        for align, off, count in t["lg"]:
             lp_g = np.zeros(18, dtype=np.uint64)
             assert lib.rng_link_proofs(ctx.h, ctypes.c_void_p(pk), ptr(ha_g), ptr(hb_g),
-                                       int(off), int(count), ptr(lp_g)) == 0
+                                       int(align), int(off), int(count), ptr(lp_g)) == 0
             lp_o = np.zeros(18, dtype=np.uint64)
             assert o.orc_plonk_link(ctypes.c_void_p(s["pk"]), ptr(ha_o), ptr(hb_o),
+                                    ctypes.c_uint64(int(align)),
                                     ctypes.c_uint64(int(off)),
                                     ctypes.c_uint64(int(count)), ptr(lp_o)) == 0
             assert np.array_equal(lp_g, lp_o), f"link proof mismatch at group {off}"
             ok = o.orc_plonk_link_verify(ctypes.c_void_p(s["pk"]), ptr(ha_g[-9:].copy()),
                                          ptr(hb_g[-9:].copy()), ptr(lp_g),
+                                         ctypes.c_uint64(int(align)),
                                          ctypes.c_uint64(int(off)),
                                          ctypes.c_uint64(int(count)), ptr(s["tau"]))
             assert ok == 1
