@@ -165,7 +165,7 @@ def msm_shard_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
     ctx.sync()
     if dist:
         import torch
-        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64)
+        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device="cuda")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
     else:
@@ -298,7 +298,8 @@ def main():
     assert np.array_equal(check, bufs[0]), "threaded proof differs from serial"
     if dist:
         import torch
-        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64)
+        # NCCL collectives operate on device tensors
+        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device="cuda")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
     else:

@@ -33,8 +33,13 @@ def combine_shard_results(dist_mod, record9: np.ndarray, add_fn):
     """
     import torch
     t = torch.from_numpy(record9.view(np.int64).copy())
+    on_gpu = dist_mod.get_backend() == "nccl"
+    if on_gpu:
+        t = t.cuda()  # NCCL collectives need device tensors
     out = [torch.zeros_like(t) for _ in range(dist_mod.get_world_size())]
     dist_mod.all_gather(out, t)
+    if on_gpu:
+        out = [r.cpu() for r in out]
     acc = out[0].numpy().view(np.uint64).copy()
     for r in out[1:]:
         acc = add_fn(acc, r.numpy().view(np.uint64).copy())
