@@ -165,7 +165,8 @@ def msm_shard_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
     ctx.sync()
     if dist:
         import torch
-        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device="cuda")
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
     else:
@@ -227,7 +228,10 @@ def main():
     orc = OracleLib(str(REPO / "oracle" / "liborc.so"))
     plib = load_prover()
     plib.require_gpu()
-    plib.set_device(local_rank)
+    try:
+        plib.set_device(local_rank)
+    except Exception:
+        plib.set_device(0)  # over-subscribed smoke runs on fewer GPUs
     lib = plib.lib
     lib.rng_preprocess.restype = ctypes.c_void_p
     lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
@@ -239,8 +243,10 @@ def main():
         import torch
         import torch.distributed as td
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        td.init_process_group(backend="nccl", rank=rank, world_size=world)
-        torch.cuda.set_device(local_rank)
+        backend = os.environ.get("RNG_DIST_BACKEND", "nccl")
+        td.init_process_group(backend=backend, rank=rank, world_size=world)
+        if backend == "nccl":
+            torch.cuda.set_device(local_rank)
         dist = td
 
     if args.mode == "msm-shard":
@@ -299,7 +305,8 @@ def main():
     if dist:
         import torch
         # NCCL collectives operate on device tensors
-        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device="cuda")
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
     else:
