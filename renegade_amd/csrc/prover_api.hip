@@ -924,8 +924,9 @@ int rng_gpu_available(void) { return gpu_ok() ? 1 : 0; }
 
 int rng_set_device(int device) {
     if (!gpu_ok()) return RNG_ERR_NO_GPU;
-    HIP_CHECK(hipSetDevice(device));
-    return RNG_OK;
+    hipError_t e = hipSetDevice(device);
+    (void)hipGetLastError();  // clear the sticky last-error slot on failure
+    return e == hipSuccess ? RNG_OK : RNG_ERR_HIP;
 }
 
 int rng_msm_last_times(double out_ms[5]) {
