@@ -536,10 +536,11 @@ struct Intent {  // intent.rs:49-70 (5 scalars; min_price = FixedPoint repr)
         return {in_token, out_token, owner, min_price_repr, amount_in};
     }
 };
-struct PostMatchShare {  // balance.rs PostMatchBalanceShare (3 scalars)
-    Fr amount, relayer_fee_balance, protocol_fee_balance;
+struct PostMatchShare {  // balance.rs:145-152 PostMatchBalance(Share) (3 scalars)
+    // Reference field order: relayer_fee_balance, protocol_fee_balance, amount.
+    Fr relayer_fee_balance, protocol_fee_balance, amount;
     std::vector<Fr> to_scalars() const {
-        return {amount, relayer_fee_balance, protocol_fee_balance};
+        return {relayer_fee_balance, protocol_fee_balance, amount};
     }
 };
 
@@ -677,12 +678,12 @@ inline void settlement_build_witness_statement(uint64_t seed, SettlementWitness&
         Fr pf = floor_mul(st.protocol_fee_repr, ob.amount_out);
         Fr net = ob.amount_out.sub(rf).sub(pf);
         st.new_amount_share[i] = w.p[i].pre_amount_share.sub(ob.amount_in);
-        st.new_in_shares[i] = {w.p[i].pre_in_shares.amount.sub(ob.amount_in),
-                               w.p[i].pre_in_shares.relayer_fee_balance,
-                               w.p[i].pre_in_shares.protocol_fee_balance};
-        st.new_out_shares[i] = {w.p[i].pre_out_shares.amount.add(net),
-                                w.p[i].pre_out_shares.relayer_fee_balance.add(rf),
-                                w.p[i].pre_out_shares.protocol_fee_balance.add(pf)};
+        st.new_in_shares[i] = {w.p[i].pre_in_shares.relayer_fee_balance,
+                               w.p[i].pre_in_shares.protocol_fee_balance,
+                               w.p[i].pre_in_shares.amount.sub(ob.amount_in)};
+        st.new_out_shares[i] = {w.p[i].pre_out_shares.relayer_fee_balance.add(rf),
+                                w.p[i].pre_out_shares.protocol_fee_balance.add(pf),
+                                w.p[i].pre_out_shares.amount.add(net)};
     }
 }
 
@@ -781,18 +782,20 @@ inline void settlement_apply_constraints(PlonkCircuit& cs, const SettlementWitne
         cs.enforce_in_range(new_rfb, AMOUNT_BITS);
         cs.enforce_in_range(new_pfb, AMOUNT_BITS);
         // --- 4. state updates (settlement_lib.rs:141-199) ---
+        // share tuples are in PostMatchBalanceShare field order:
+        // [0]=relayer_fee_balance, [1]=protocol_fee_balance, [2]=amount
         Var exp_amount = cs.sub(pv[i].pre_amount, pv[i].ob[2]);
         cs.enforce_equal(exp_amount, new_amount[i]);
-        Var exp_in_amt = cs.sub(pv[i].pre_in[0], pv[i].ob[2]);
-        cs.enforce_equal(exp_in_amt, new_in[i][0]);
+        Var exp_in_amt = cs.sub(pv[i].pre_in[2], pv[i].ob[2]);
+        cs.enforce_equal(exp_in_amt, new_in[i][2]);
+        cs.enforce_equal(pv[i].pre_in[0], new_in[i][0]);
         cs.enforce_equal(pv[i].pre_in[1], new_in[i][1]);
-        cs.enforce_equal(pv[i].pre_in[2], new_in[i][2]);
-        Var exp_out_amt = cs.add(pv[i].pre_out[0], net_receive);
-        cs.enforce_equal(exp_out_amt, new_out[i][0]);
-        Var exp_out_rfb = cs.add(pv[i].pre_out[1], ft.relayer_fee);
-        cs.enforce_equal(exp_out_rfb, new_out[i][1]);
-        Var exp_out_pfb = cs.add(pv[i].pre_out[2], ft.protocol_fee);
-        cs.enforce_equal(exp_out_pfb, new_out[i][2]);
+        Var exp_out_amt = cs.add(pv[i].pre_out[2], net_receive);
+        cs.enforce_equal(exp_out_amt, new_out[i][2]);
+        Var exp_out_rfb = cs.add(pv[i].pre_out[0], ft.relayer_fee);
+        cs.enforce_equal(exp_out_rfb, new_out[i][0]);
+        Var exp_out_pfb = cs.add(pv[i].pre_out[1], ft.protocol_fee);
+        cs.enforce_equal(exp_out_pfb, new_out[i][1]);
     }
 }
 
@@ -1087,5 +1090,359 @@ inline void vw_apply_constraints(PlonkCircuit& cs, const VdWitness& w,
     cs.enforce_equal(nul, p_null);
 }
 
+
+// ================== Intent And Balance Validity ==================
+// (zk_circuits/validity_proofs/intent_and_balance.rs — the per-party
+//  validity proof whose witness proof-links into the settlement proof's
+//  intent_and_balance_settlement_party{0,1} groups.)
+
+// intent_and_balance.rs:56 (IntentShare::NUM_SCALARS - 1: omit amount_in)
+constexpr size_t INTENT_PARTIAL_COMMITMENT_SIZE = 4;
+// intent_and_balance_first_fill.rs:53-54
+// (DarkpoolBalance::NUM_SCALARS - PostMatchBalance::NUM_SCALARS = 8 - 3)
+constexpr size_t BALANCE_PARTIAL_COMMITMENT_SIZE = 5;
+
+struct StateIntent {  // DarkpoolStateIntent = StateWrapper<Intent> (14 scalars)
+    Csprng recovery, share;
+    Intent inner;
+    Fr public_share[5];
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = {recovery.seed, Fr::from_u64(recovery.index), share.seed,
+                             Fr::from_u64(share.index)};
+        auto iv = inner.to_scalars();
+        v.insert(v.end(), iv.begin(), iv.end());
+        v.insert(v.end(), public_share, public_share + 5);
+        return v;
+    }
+};
+
+struct ValidityWitness {  // intent_and_balance.rs:236-272 (field order)
+    StateIntent old_intent;
+    Fr intent_opening_elems[MERKLE_HEIGHT];
+    bool intent_opening_idx[MERKLE_HEIGHT];
+    Intent intent;               // linked (denormalized from old_intent.inner)
+    Fr new_amount_public_share;  // linked (re-encrypted amount_in share)
+    StateBalance old_balance;
+    Fr balance_opening_elems[MERKLE_HEIGHT];
+    bool balance_opening_idx[MERKLE_HEIGHT];
+    Balance balance;                      // linked
+    PostMatchShare post_match_balance_shares;  // linked (rfb, pfb, amount order)
+};
+
+struct ValidityStatement {  // intent_and_balance.rs:277-309 (10 scalars)
+    Fr intent_merkle_root, old_intent_nullifier;
+    Fr intent_partial_private, intent_partial_public;  // PartialCommitment
+    Fr intent_recovery_id;
+    Fr balance_merkle_root, old_balance_nullifier;
+    Fr balance_partial_private, balance_partial_public;
+    Fr balance_recovery_id;
+    std::vector<Fr> to_scalars() const {
+        return {intent_merkle_root, old_intent_nullifier, intent_partial_private,
+                intent_partial_public, intent_recovery_id, balance_merkle_root,
+                old_balance_nullifier, balance_partial_private, balance_partial_public,
+                balance_recovery_id};
+    }
+};
+
+// native partial commitment (commitment.rs:149-181): commitment to the
+// private shares + streams, and resumable commitment over the first K
+// public shares.
+inline void native_partial_commitment(const std::vector<Fr>& private_shares,
+                                      const Csprng& recovery, const Csprng& share,
+                                      const std::vector<Fr>& public_shares, size_t K,
+                                      Fr& out_priv, Fr& out_pub) {
+    std::vector<Fr> in = private_shares;
+    in.push_back(recovery.seed);
+    in.push_back(Fr::from_u64(recovery.index));
+    in.push_back(share.seed);
+    in.push_back(Fr::from_u64(share.index));
+    out_priv = poseidon_hash(in.data(), in.size());
+    Fr comm = public_shares[0];
+    for (size_t i = 1; i < K; ++i) {
+        Fr two[2] = {comm, public_shares[i]};
+        comm = poseidon_hash(two, 2);
+    }
+    out_pub = comm;
+}
+
+// in-circuit partial commitment (commitment.rs:149-181)
+inline std::pair<Var, Var> partial_commitment_gadget(
+    PlonkCircuit& cs, const std::vector<Var>& private_share, const CsprngVar& recovery,
+    const CsprngVar& share, const std::vector<Var>& public_share, size_t K) {
+    PoseidonHashGadget h(cs);
+    std::vector<Var> in = private_share;
+    in.push_back(recovery.seed);
+    in.push_back(recovery.index);
+    in.push_back(share.seed);
+    in.push_back(share.index);
+    Var priv = h.hash(cs, in);
+    std::vector<Var> firstK(public_share.begin(), public_share.begin() + K);
+    Var pp = resumable_commitment(cs, firstK);
+    return {priv, pp};
+}
+
+// The full bundle: one settlement witness/statement plus the two parties'
+// validity witnesses/statements, mutually consistent so the validity proofs
+// link into the settlement proof (the production proof bundle the reference
+// relayer submits).
+struct ValidityBundle {
+    SettlementWitness sw;
+    SettlementStatement sst;
+    ValidityWitness vw[2];
+    ValidityStatement vst[2];
+};
+
+inline void validity_bundle_build(uint64_t seed, ValidityBundle& b) {
+    settlement_build_witness_statement(seed, b.sw, b.sst);
+    Lcg rng(seed ^ 0x9E3779B97F4A7C15ull);
+    auto priv_of = [](const std::vector<Fr>& inner, const Fr* pub_, size_t n) {
+        std::vector<Fr> p;
+        for (size_t i = 0; i < n; ++i) p.push_back(inner[i].sub(pub_[i]));
+        return p;
+    };
+    for (int i = 0; i < 2; ++i) {
+        ValidityWitness& v = b.vw[i];
+        ValidityStatement& st = b.vst[i];
+        // old intent state element wrapping the party's intent
+        v.old_intent.inner = b.sw.p[i].intent;
+        v.old_intent.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+        v.old_intent.share = {rng.fr(), rng.next() & 0xFFFFFF};
+        for (int k = 0; k < 5; ++k) v.old_intent.public_share[k] = rng.fr();
+        for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+            v.intent_opening_elems[k] = rng.fr();
+            v.intent_opening_idx[k] = rng.next() & 1;
+        }
+        v.intent = b.sw.p[i].intent;
+        // old balance state element wrapping the party's input balance
+        v.old_balance.inner = b.sw.p[i].input_balance;
+        v.old_balance.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+        v.old_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+        for (int k = 0; k < 8; ++k) v.old_balance.public_share[k] = rng.fr();
+        for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+            v.balance_opening_elems[k] = rng.fr();
+            v.balance_opening_idx[k] = rng.next() & 1;
+        }
+        v.balance = b.sw.p[i].input_balance;
+
+        // --- intent rotation natives (build_new_intent :135-158) ---
+        auto old_iv = v.old_intent.inner.to_scalars();
+        std::vector<Fr> old_priv_i = priv_of(old_iv, v.old_intent.public_share, 5);
+        Fr old_comm_i = native_commitment(
+            old_priv_i, v.old_intent.recovery, v.old_intent.share,
+            std::vector<Fr>(v.old_intent.public_share, v.old_intent.public_share + 5));
+        st.intent_merkle_root = native_merkle_root(
+            old_comm_i,
+            std::vector<Fr>(v.intent_opening_elems, v.intent_opening_elems + MERKLE_HEIGHT),
+            std::vector<bool>(v.intent_opening_idx, v.intent_opening_idx + MERKLE_HEIGHT));
+        st.old_intent_nullifier = native_nullifier(v.old_intent.recovery);
+        StateIntent ni = v.old_intent;
+        Fr pad = ni.share.next();
+        v.new_amount_public_share = ni.inner.amount_in.sub(pad);
+        ni.public_share[4] = v.new_amount_public_share;
+        std::vector<Fr> new_priv_i = old_priv_i;
+        new_priv_i[4] = pad;
+        st.intent_recovery_id = ni.recovery.next();
+        native_partial_commitment(
+            new_priv_i, ni.recovery, ni.share,
+            std::vector<Fr>(ni.public_share, ni.public_share + 5),
+            INTENT_PARTIAL_COMMITMENT_SIZE, st.intent_partial_private,
+            st.intent_partial_public);
+
+        // --- balance rotation natives (build_new_balance :199-238) ---
+        auto old_bv = v.old_balance.inner.to_scalars();
+        std::vector<Fr> old_priv_b = priv_of(old_bv, v.old_balance.public_share, 8);
+        Fr old_comm_b = native_commitment(
+            old_priv_b, v.old_balance.recovery, v.old_balance.share,
+            std::vector<Fr>(v.old_balance.public_share, v.old_balance.public_share + 8));
+        st.balance_merkle_root = native_merkle_root(
+            old_comm_b,
+            std::vector<Fr>(v.balance_opening_elems, v.balance_opening_elems + MERKLE_HEIGHT),
+            std::vector<bool>(v.balance_opening_idx, v.balance_opening_idx + MERKLE_HEIGHT));
+        st.old_balance_nullifier = native_nullifier(v.old_balance.recovery);
+        StateBalance nb = v.old_balance;
+        // re-encrypt the post-match trio in PostMatchBalanceShare field order:
+        // relayer_fee_balance (idx 5), protocol_fee_balance (6), amount (7)
+        Fr p0 = nb.share.next(), p1 = nb.share.next(), p2 = nb.share.next();
+        v.post_match_balance_shares = {nb.inner.relayer_fee_balance.sub(p0),
+                                       nb.inner.protocol_fee_balance.sub(p1),
+                                       nb.inner.amount.sub(p2)};
+        nb.public_share[5] = v.post_match_balance_shares.relayer_fee_balance;
+        nb.public_share[6] = v.post_match_balance_shares.protocol_fee_balance;
+        nb.public_share[7] = v.post_match_balance_shares.amount;
+        std::vector<Fr> new_priv_b = old_priv_b;
+        new_priv_b[5] = p0;
+        new_priv_b[6] = p1;
+        new_priv_b[7] = p2;
+        st.balance_recovery_id = nb.recovery.next();
+        native_partial_commitment(
+            new_priv_b, nb.recovery, nb.share,
+            std::vector<Fr>(nb.public_share, nb.public_share + 8),
+            BALANCE_PARTIAL_COMMITMENT_SIZE, st.balance_partial_private,
+            st.balance_partial_public);
+
+        // feed the re-encrypted shares back into the settlement witness: the
+        // validity circuit's outputs ARE the settlement's pre-update shares
+        b.sw.p[i].pre_amount_share = v.new_amount_public_share;
+        b.sw.p[i].pre_in_shares = v.post_match_balance_shares;
+    }
+    // recompute the statement fields that depend on the pre-update shares
+    for (int i = 0; i < 2; ++i) {
+        const auto& ob = b.sw.p[i].obligation;
+        b.sst.new_amount_share[i] = b.sw.p[i].pre_amount_share.sub(ob.amount_in);
+        b.sst.new_in_shares[i] = {b.sw.p[i].pre_in_shares.relayer_fee_balance,
+                                  b.sw.p[i].pre_in_shares.protocol_fee_balance,
+                                  b.sw.p[i].pre_in_shares.amount.sub(ob.amount_in)};
+    }
+}
+
+// apply_constraints (intent_and_balance.rs:71-232).  The two link groups are
+// placed at the SETTLEMENT circuit's layout (proof_linking_groups inherits
+// the settlement placement, :316-341): pass that placement in.
+inline void validity_apply_constraints(PlonkCircuit& cs, const ValidityWitness& w,
+                                       const ValidityStatement& st, int alignment,
+                                       int64_t party_offset0, int64_t party_offset1) {
+    const char* g0 = "intent_and_balance_settlement_party0";
+    const char* g1 = "intent_and_balance_settlement_party1";
+    cs.create_link_group(g0, alignment, party_offset0);
+    cs.create_link_group(g1, alignment, party_offset1);
+    auto link_both = [&](Var x) {
+        cs.add_to_link_group(x, g0);
+        cs.add_to_link_group(x, g1);
+    };
+
+    // --- witness allocation (struct field order) ---
+    StateWrapperVars oi;
+    oi.recovery = {cs.create_variable(w.old_intent.recovery.seed),
+                   cs.create_variable(Fr::from_u64(w.old_intent.recovery.index))};
+    oi.share = {cs.create_variable(w.old_intent.share.seed),
+                cs.create_variable(Fr::from_u64(w.old_intent.share.index))};
+    for (auto& s : w.old_intent.inner.to_scalars())
+        oi.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 5; ++k)
+        oi.public_share.push_back(cs.create_variable(w.old_intent.public_share[k]));
+    std::vector<Var> iop_elems, iop_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        iop_elems.push_back(cs.create_variable(w.intent_opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        iop_idx.push_back(cs.create_boolean_variable(
+            w.intent_opening_idx[k] ? Fr::one() : Fr::zero()));
+    std::array<Var, 5> intent_v;
+    {
+        auto iv = w.intent.to_scalars();
+        for (int k = 0; k < 5; ++k) {
+            intent_v[k] = cs.create_variable(iv[k]);
+            link_both(intent_v[k]);
+        }
+    }
+    Var new_amt_share = cs.create_variable(w.new_amount_public_share);
+    link_both(new_amt_share);
+    StateWrapperVars ob_;
+    ob_.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                    cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    ob_.share = {cs.create_variable(w.old_balance.share.seed),
+                 cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    for (auto& s : w.old_balance.inner.to_scalars())
+        ob_.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 8; ++k)
+        ob_.public_share.push_back(cs.create_variable(w.old_balance.public_share[k]));
+    std::vector<Var> bop_elems, bop_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        bop_elems.push_back(cs.create_variable(w.balance_opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        bop_idx.push_back(cs.create_boolean_variable(
+            w.balance_opening_idx[k] ? Fr::one() : Fr::zero()));
+    std::array<Var, 8> bal_v;
+    {
+        auto bv = w.balance.to_scalars();
+        for (int k = 0; k < 8; ++k) {
+            bal_v[k] = cs.create_variable(bv[k]);
+            link_both(bal_v[k]);
+        }
+    }
+    std::array<Var, 3> pms_v;
+    {
+        auto pv = w.post_match_balance_shares.to_scalars();
+        for (int k = 0; k < 3; ++k) {
+            pms_v[k] = cs.create_variable(pv[k]);
+            link_both(pms_v[k]);
+        }
+    }
+
+    // --- statement (public inputs, field order) ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var p_iroot = pub[0], p_inull = pub[1], p_ipriv = pub[2], p_ipub = pub[3],
+        p_irid = pub[4];
+    Var p_broot = pub[5], p_bnull = pub[6], p_bpriv = pub[7], p_bpub = pub[8],
+        p_brid = pub[9];
+
+    // --- validate_intent (:95-131) ---
+    for (int k = 0; k < 5; ++k) cs.enforce_equal(intent_v[k], oi.inner[k]);
+    std::vector<Var> old_priv_i;
+    for (int k = 0; k < 5; ++k)
+        old_priv_i.push_back(cs.sub(oi.inner[k], oi.public_share[k]));
+    // build_new_intent (:135-158): re-encrypt amount_in (last share)
+    CsprngVar ni_share = oi.share;
+    std::vector<Var> ipads, icipher;
+    stream_cipher_encrypt(cs, {oi.inner[4]}, ni_share, ipads, icipher);
+    std::vector<Var> new_priv_i = old_priv_i;
+    new_priv_i[4] = ipads[0];
+    std::vector<Var> new_pub_i = oi.public_share;
+    new_pub_i[4] = icipher[0];
+    cs.enforce_equal(icipher[0], new_amt_share);
+    // rotate_version_with_partial_commitment (state_rotation.rs:130-168)
+    CsprngVar ni_rec = oi.recovery;
+    Var irid = csprng_next(cs, ni_rec);
+    cs.enforce_equal(irid, p_irid);
+    auto ipc = partial_commitment_gadget(cs, new_priv_i, ni_rec, ni_share, new_pub_i,
+                                         INTENT_PARTIAL_COMMITMENT_SIZE);
+    cs.enforce_equal(ipc.first, p_ipriv);
+    cs.enforce_equal(ipc.second, p_ipub);
+    Var old_comm_i = commitment_gadget(cs, old_priv_i, oi.recovery, oi.share,
+                                       oi.public_share);
+    Var iroot = merkle_root_gadget(cs, old_comm_i, iop_elems, iop_idx);
+    cs.enforce_equal(iroot, p_iroot);
+    Var inull = nullifier_gadget(cs, oi);
+    cs.enforce_equal(inull, p_inull);
+
+    // --- validate_balance (:160-196) ---
+    for (int k = 0; k < 8; ++k) cs.enforce_equal(bal_v[k], ob_.inner[k]);
+    std::vector<Var> old_priv_b;
+    for (int k = 0; k < 8; ++k)
+        old_priv_b.push_back(cs.sub(ob_.inner[k], ob_.public_share[k]));
+    // build_new_balance (:199-238): re-encrypt the post-match trio in
+    // PostMatchBalanceShare field order (rfb@5, pfb@6, amount@7)
+    CsprngVar nb_share = ob_.share;
+    std::vector<Var> bpads, bcipher;
+    stream_cipher_encrypt(cs, {ob_.inner[5], ob_.inner[6], ob_.inner[7]}, nb_share,
+                          bpads, bcipher);
+    std::vector<Var> new_priv_b = old_priv_b;
+    new_priv_b[5] = bpads[0];
+    new_priv_b[6] = bpads[1];
+    new_priv_b[7] = bpads[2];
+    std::vector<Var> new_pub_b = ob_.public_share;
+    new_pub_b[5] = bcipher[0];
+    new_pub_b[6] = bcipher[1];
+    new_pub_b[7] = bcipher[2];
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(bcipher[k], pms_v[k]);
+    CsprngVar nb_rec = ob_.recovery;
+    Var brid = csprng_next(cs, nb_rec);
+    cs.enforce_equal(brid, p_brid);
+    auto bpc = partial_commitment_gadget(cs, new_priv_b, nb_rec, nb_share, new_pub_b,
+                                         BALANCE_PARTIAL_COMMITMENT_SIZE);
+    cs.enforce_equal(bpc.first, p_bpriv);
+    cs.enforce_equal(bpc.second, p_bpub);
+    Var old_comm_b = commitment_gadget(cs, old_priv_b, ob_.recovery, ob_.share,
+                                       ob_.public_share);
+    Var broot = merkle_root_gadget(cs, old_comm_b, bop_elems, bop_idx);
+    cs.enforce_equal(broot, p_broot);
+    Var bnull = nullifier_gadget(cs, ob_);
+    cs.enforce_equal(bnull, p_bnull);
+
+    // --- intent <-> balance cross constraints (:84-88) ---
+    cs.enforce_equal(intent_v[0], bal_v[0]);  // in_token == mint
+    cs.enforce_equal(intent_v[2], bal_v[1]);  // owner == owner
+}
 
 }  // namespace rng

@@ -1177,6 +1177,66 @@ void* rng_circ_build_settlement(uint64_t seed) {
     }
 }
 
+// --- validity <-> settlement proof bundle builders ---
+// (zk_circuits/validity_proofs/intent_and_balance.rs; the validity circuit's
+//  two link groups are placed at the SETTLEMENT circuit's layout, :316-341.)
+
+// settlement circuit whose pre-update shares come from the two validity
+// witnesses of the same seed (the consistent production bundle)
+void* rng_circ_build_settlement_bundle(uint64_t seed) {
+    try {
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        PlonkCircuit cs;
+        settlement_apply_constraints(cs, b.sw, b.sst);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_settlement_bundle: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_settlement_bundle: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// INTENT AND BALANCE VALIDITY circuit for party 0 or 1 of the seed's bundle
+void* rng_circ_build_validity(uint64_t seed, uint64_t party) {
+    try {
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        // read the settlement circuit's link-group placement (inherited layout)
+        uint64_t align = 0;
+        int64_t off[2] = {0, 0};
+        {
+            PlonkCircuit scs;
+            settlement_apply_constraints(scs, b.sw, b.sst);
+            CircuitTables stt = scs.finalize();
+            const char* names[2] = {"intent_and_balance_settlement_party0",
+                                    "intent_and_balance_settlement_party1"};
+            for (auto& g : stt.link_groups)
+                for (int i = 0; i < 2; ++i)
+                    if (g.id == names[i]) {
+                        align = g.alignment;
+                        off[i] = (int64_t)g.offset;
+                    }
+        }
+        int p = (int)(party & 1);
+        PlonkCircuit cs;
+        validity_apply_constraints(cs, b.vw[p], b.vst[p], (int)align, off[0], off[1]);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_validity: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_validity: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // circuit builders from caller-supplied witness/statement scalars (the shape
 // the external prover service receives — api_types.rs requests; Montgomery
 // limbs, field order per the reference structs)

@@ -1,0 +1,267 @@
+"""`INTENT AND BALANCE VALIDITY` circuit
+(zk_circuits/validity_proofs/intent_and_balance.rs) — build, shape, oracle
+prove/verify, and the PRODUCTION cross-domain proof link: the validity
+proof's party group links into the settlement proof of the same bundle
+(proof_linking_groups inherits the settlement layout, :316-341)."""
+import ctypes
+
+import numpy as np
+import pytest
+
+U64P = ctypes.POINTER(ctypes.c_uint64)
+
+
+def ptr(a):
+    return a.ctypes.data_as(U64P)
+
+
+@pytest.fixture(scope="module")
+def vb(orc):
+    from renegade_amd import load_prover
+    plib = load_prover()
+    lib = plib.lib
+    lib.rng_circ_build_validity.restype = ctypes.c_void_p
+    lib.rng_circ_build_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+    lib.rng_circ_build_settlement_bundle.restype = ctypes.c_void_p
+    lib.rng_circ_build_settlement_bundle.argtypes = [ctypes.c_uint64]
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_num_link_groups.restype = ctypes.c_uint64
+    lib.rng_circ_num_link_groups.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_link_groups.argtypes = [ctypes.c_void_p, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+
+    def fetch(h):
+        assert h
+        n = lib.rng_circ_n(h)
+        npub = lib.rng_circ_npub(h)
+        nlg = lib.rng_circ_num_link_groups(h)
+        lg = np.zeros(3 * max(1, nlg), dtype=np.uint64)
+        lib.rng_circ_link_groups(h, ptr(lg))
+        sel = np.zeros(13 * n * 4, dtype=np.uint64)
+        sigma = np.zeros(5 * n, dtype=np.uint64)
+        wires = np.zeros(5 * n * 4, dtype=np.uint64)
+        pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+        lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+        lib.rng_circ_free(h)
+        return dict(n=n, npub=npub, nlg=nlg, lg=lg.reshape(-1, 3), sel=sel,
+                    sigma=sigma, wires=wires, pubs=pubs)
+
+    o = orc.lib
+    o.orc_plonk_preprocess.restype = ctypes.c_void_p
+    o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P, U64P,
+                                       U64P, ctypes.c_uint64]
+    o.orc_plonk_prove_with_hint.argtypes = [ctypes.c_void_p, U64P, U64P,
+                                            ctypes.c_uint64, U64P, U64P]
+    o.orc_plonk_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P]
+    o.orc_plonk_link.argtypes = [ctypes.c_void_p, U64P, U64P] + \
+        [ctypes.c_uint64] * 3 + [U64P]
+    o.orc_plonk_link_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P] + \
+        [ctypes.c_uint64] * 3 + [U64P]
+    o.orc_derive_tau.argtypes = [ctypes.c_uint64, U64P]
+
+    def setup(t):
+        power = max(4, int(t["n"]).bit_length())
+        ptau = orc.srs_generate_ptau(power, seed=42)
+        md = (1 << power) + 2
+        g1, _, _ = orc.srs_parse(ptau, md)
+        srs = np.ascontiguousarray(g1).reshape(-1)
+        pk = o.orc_plonk_preprocess(t["n"], t["npub"], ptr(t["sel"]), ptr(t["sigma"]),
+                                    ptr(srs), md + 1)
+        assert pk
+        return pk
+
+    def prove(pk, t, seed):
+        proof = np.zeros(157, dtype=np.uint64)
+        hint = np.zeros(4 * (t["n"] + 2) + 9, dtype=np.uint64)
+        assert o.orc_plonk_prove_with_hint(ctypes.c_void_p(pk), ptr(t["wires"]),
+                                           ptr(t["pubs"]), ctypes.c_uint64(seed),
+                                           ptr(proof), ptr(hint)) == 0
+        return proof, hint
+
+    tau = np.zeros(4, dtype=np.uint64)
+    o.orc_derive_tau(42, ptr(tau))
+    return dict(lib=lib, o=o, fetch=fetch, setup=setup, prove=prove, tau=tau)
+
+
+class TestValidityCircuit:
+    def test_shape(self, vb):
+        lib = vb["lib"]
+        t = vb["fetch"](lib.rng_circ_build_validity(42, 0))
+        # statement = 10 scalars (intent_and_balance.rs:277-309)
+        assert t["npub"] == 10
+        # two link groups, both 17 vars, inheriting the settlement placement
+        assert t["nlg"] == 2
+        ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(42))
+        stl_groups = {tuple(int(x) for x in row) for row in ts["lg"]}
+        for row in t["lg"]:
+            assert int(row[2]) == 17
+            assert tuple(int(x) for x in row) in stl_groups, \
+                "validity group not at the settlement's layout"
+        assert t["n"] != ts["n"], "expect distinct domain sizes (cross-domain link)"
+
+    def test_seeds_vary(self, vb):
+        lib = vb["lib"]
+        for seed in [1, 2, 3]:
+            for party in (0, 1):
+                h = lib.rng_circ_build_validity(seed, party)
+                assert h, f"seed {seed} party {party} unsatisfied"
+                lib.rng_circ_free(h)
+
+    def test_oracle_prove_verify(self, vb):
+        lib, o = vb["lib"], vb["o"]
+        t = vb["fetch"](lib.rng_circ_build_validity(42, 0))
+        pk = vb["setup"](t)
+        proof, _ = vb["prove"](pk, t, 11)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(t["pubs"]), ptr(proof),
+                                  ptr(vb["tau"])) == 1
+        # tamper the intent nullifier -> reject
+        bad = t["pubs"].copy()
+        bad[1 * 4] ^= np.uint64(1)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(bad), ptr(proof),
+                                  ptr(vb["tau"])) != 1
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk))
+
+
+class TestBundleLink:
+    """The production proof bundle: settlement proof + two validity proofs,
+    linked party-by-party across DIFFERENT domain sizes."""
+
+    def test_validity_settlement_link(self, vb):
+        lib, o = vb["lib"], vb["o"]
+        ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(7))
+        t0 = vb["fetch"](lib.rng_circ_build_validity(7, 0))
+        t1 = vb["fetch"](lib.rng_circ_build_validity(7, 1))
+        pk_s = vb["setup"](ts)
+        pk_v = vb["setup"](t0)  # same shape for both parties
+        _, hs = vb["prove"](pk_s, ts, 3)
+        _, h0 = vb["prove"](pk_v, t0, 4)
+        _, h1 = vb["prove"](pk_v, t1, 5)
+        n_big = max(int(ts["n"]), int(t0["n"]))
+        pk_big = pk_v if int(t0["n"]) == n_big else pk_s
+
+        def ext(h, n_small):
+            out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+            out[:4 * (n_small + 2)] = h[:4 * (n_small + 2)]
+            out[-9:] = h[-9:]
+            return out
+
+        hs_e = ext(hs, int(ts["n"]))
+        groups = {int(r[1]): (int(r[0]), int(r[1]), int(r[2])) for r in ts["lg"]
+                  if int(r[2]) == 17}
+        offs = sorted(groups)  # party0 offset < party1 offset (creation order)
+        party_group = {0: groups[offs[0]], 1: groups[offs[1]]}
+        for party, hv in ((0, h0), (1, h1)):
+            a, off, cnt = party_group[party]
+            hv_e = ext(hv, int(t0["n"]))
+            lp = np.zeros(18, dtype=np.uint64)
+            rc = o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hs_e),
+                                  ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                  ctypes.c_uint64(cnt), ptr(lp))
+            assert rc == 0
+            ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big),
+                                         ptr(hv_e[-9:].copy()), ptr(hs_e[-9:].copy()),
+                                         ptr(lp), ctypes.c_uint64(a),
+                                         ctypes.c_uint64(off), ctypes.c_uint64(cnt),
+                                         ptr(vb["tau"]))
+            assert ok == 1, f"validity party{party} <-> settlement link failed"
+
+        # negative: party0's validity proof must NOT link at party1's offset
+        a, off, cnt = party_group[1]
+        h0_e = ext(h0, int(t0["n"]))
+        lp = np.zeros(18, dtype=np.uint64)
+        o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(h0_e), ptr(hs_e),
+                         ctypes.c_uint64(a), ctypes.c_uint64(off),
+                         ctypes.c_uint64(cnt), ptr(lp))
+        ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big), ptr(h0_e[-9:].copy()),
+                                     ptr(hs_e[-9:].copy()), ptr(lp),
+                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                     ctypes.c_uint64(cnt), ptr(vb["tau"]))
+        assert ok != 1, "cross-party link should not verify"
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_s))
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_v))
+
+
+@pytest.mark.gpu
+class TestBundleLinkGpu:
+    """GPU end-to-end production bundle: settlement + both validity proofs on
+    the HIP prover, cross-domain links computed on the GPU, verified by the
+    oracle's trapdoor verifier."""
+
+    def test_gpu_bundle(self, vb, orc):
+        from renegade_amd import load_prover
+        plib = load_prover()
+        if not plib.gpu_available:
+            pytest.skip("no GPU")
+        lib = vb["lib"]
+        o = vb["o"]
+        lib.rng_preprocess.restype = ctypes.c_void_p
+        lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                                  ctypes.c_uint64, U64P, U64P]
+        lib.rng_link_proofs.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                                        ctypes.c_uint64, ctypes.c_uint64,
+                                        ctypes.c_uint64, U64P]
+        ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(7))
+        t0 = vb["fetch"](lib.rng_circ_build_validity(7, 0))
+        t1 = vb["fetch"](lib.rng_circ_build_validity(7, 1))
+        n_big = max(int(ts["n"]), int(t0["n"]))
+        power = max(4, int(n_big).bit_length())  # covers n_big + 2
+        ptau = orc.srs_generate_ptau(power, seed=42)
+        max_degree = (1 << power) + 2
+        ctx = plib.init(ptau, max_degree)
+
+        class Desc(ctypes.Structure):
+            _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
+                        ("selectors", U64P), ("sigma", U64P),
+                        ("num_link_groups", ctypes.c_uint64), ("link_offsets", U64P)]
+
+        def gprove(t, seed):
+            pk = lib.rng_preprocess(ctx.h, ctypes.byref(
+                Desc(t["n"], t["npub"], ptr(t["sel"]), ptr(t["sigma"]), 0, None)))
+            assert pk
+            proof = np.zeros(157, dtype=np.uint64)
+            hint = np.zeros(4 * (int(t["n"]) + 2) + 9, dtype=np.uint64)
+            assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(t["wires"]),
+                                 ptr(t["pubs"]), seed, ptr(proof), ptr(hint)) == 0
+            return pk, proof, hint
+
+        pk_s, _, hs = gprove(ts, 3)
+        pk_v, _, h0 = gprove(t0, 4)
+        _, _, h1 = gprove(t1, 5)
+
+        def ext(h, n_small):
+            out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+            out[:4 * (n_small + 2)] = h[:4 * (n_small + 2)]
+            out[-9:] = h[-9:]
+            return out
+
+        hs_e = ext(hs, int(ts["n"]))
+        pk_big = pk_v if int(t0["n"]) == n_big else pk_s
+        groups = {int(r[1]): (int(r[0]), int(r[1]), int(r[2])) for r in ts["lg"]
+                  if int(r[2]) == 17}
+        offs = sorted(groups)
+        for party, hv in ((0, h0), (1, h1)):
+            a, off, cnt = groups[offs[party]]
+            hv_e = ext(hv, int(t0["n"]))
+            lp_g = np.zeros(18, dtype=np.uint64)
+            assert lib.rng_link_proofs(ctx.h, ctypes.c_void_p(pk_big), ptr(hv_e),
+                                       ptr(hs_e), a, off, cnt, ptr(lp_g)) == 0
+            # oracle computes the same link proof bit-exact
+            opk_big = vb["setup"](t0 if int(t0["n"]) == n_big else ts)
+            lp_o = np.zeros(18, dtype=np.uint64)
+            assert o.orc_plonk_link(ctypes.c_void_p(opk_big), ptr(hv_e), ptr(hs_e),
+                                    ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                    ctypes.c_uint64(cnt), ptr(lp_o)) == 0
+            assert np.array_equal(lp_g, lp_o), f"party{party} GPU link != oracle"
+            ok = o.orc_plonk_link_verify(ctypes.c_void_p(opk_big),
+                                         ptr(hv_e[-9:].copy()), ptr(hs_e[-9:].copy()),
+                                         ptr(lp_g), ctypes.c_uint64(a),
+                                         ctypes.c_uint64(off), ctypes.c_uint64(cnt),
+                                         ptr(vb["tau"]))
+            assert ok == 1
+            o.orc_plonk_pk_free(ctypes.c_void_p(opk_big))
+        ctx.close()

@@ -140,7 +140,8 @@ class TestGpuProver:
 @pytest.mark.parametrize("builder,seed", [("rng_circ_build_settlement", 42),
                                           ("rng_circ_build_vbc", 42),
                                           ("rng_circ_build_valid_deposit", 42),
-                                          ("rng_circ_build_valid_withdrawal", 42)])
+                                          ("rng_circ_build_valid_withdrawal", 42),
+                                          ("rng_circ_build_validity", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
@@ -151,7 +152,8 @@ def test_real_circuit_gpu_parity(orc, builder, seed):
     lib = plib.lib
     fn = getattr(lib, builder)
     fn.restype = ctypes.c_void_p
-    fn.argtypes = [ctypes.c_uint64]
+    two_arg = builder == "rng_circ_build_validity"  # (seed, party)
+    fn.argtypes = [ctypes.c_uint64, ctypes.c_uint64] if two_arg else [ctypes.c_uint64]
     lib.rng_circ_n.restype = ctypes.c_uint64
     lib.rng_circ_n.argtypes = [ctypes.c_void_p]
     lib.rng_circ_npub.restype = ctypes.c_uint64
@@ -162,7 +164,7 @@ def test_real_circuit_gpu_parity(orc, builder, seed):
     lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
     lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
                               ctypes.c_uint64, U64P, U64P]
-    h = fn(seed)
+    h = fn(seed, 0) if two_arg else fn(seed)
     assert h
     n = lib.rng_circ_n(h)
     npub = lib.rng_circ_npub(h)
