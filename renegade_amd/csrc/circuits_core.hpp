@@ -603,6 +603,37 @@ inline void settlement_statement_from_scalars(const Fr* s, SettlementStatement& 
     st.protocol_fee_repr = s[16];
 }
 
+// native statement share updates from the witness pre-update shares
+// (test_helpers :356-390): recomputable after the pre shares are replaced by
+// validity-circuit outputs in a bundle.
+inline void settlement_update_statement(const SettlementWitness& w,
+                                        SettlementStatement& st) {
+    auto floor_mul = [](const Fr& rate_repr, const Fr& amt) {
+        // floor(rate * amt / 2^63) over canonical integers (fits 128 bits here)
+        u64 rl[4], al[4];
+        rate_repr.to_canonical(rl);
+        amt.to_canonical(al);
+        unsigned __int128 rate = ((unsigned __int128)rl[1] << 64) | rl[0];
+        unsigned __int128 prod = rate * al[0];  // amounts < 2^64 here
+        unsigned __int128 fl = prod >> 63;
+        u64 l[4] = {(u64)fl, (u64)(fl >> 64), 0, 0};
+        return Fr::from_canonical(l);
+    };
+    for (int i = 0; i < 2; ++i) {
+        const auto& ob = w.p[i].obligation;
+        Fr rf = floor_mul(st.relayer_fee_repr[i], ob.amount_out);
+        Fr pf = floor_mul(st.protocol_fee_repr, ob.amount_out);
+        Fr net = ob.amount_out.sub(rf).sub(pf);
+        st.new_amount_share[i] = w.p[i].pre_amount_share.sub(ob.amount_in);
+        st.new_in_shares[i] = {w.p[i].pre_in_shares.relayer_fee_balance,
+                               w.p[i].pre_in_shares.protocol_fee_balance,
+                               w.p[i].pre_in_shares.amount.sub(ob.amount_in)};
+        st.new_out_shares[i] = {w.p[i].pre_out_shares.relayer_fee_balance.add(rf),
+                                w.p[i].pre_out_shares.protocol_fee_balance.add(pf),
+                                w.p[i].pre_out_shares.amount.add(net)};
+    }
+}
+
 // fixed-seed witness/statement (test_helpers create_witness_statement,
 // intent_and_balance_private_settlement.rs:334-420; f64 price sampling
 // replaced by exact integer fixed-point construction)
@@ -660,31 +691,7 @@ inline void settlement_build_witness_statement(uint64_t seed, SettlementWitness&
     st.relayer_fee_repr[1] = fee_repr();
     st.protocol_fee_repr = fee_repr();
 
-    // native share updates (test_helpers :356-390)
-    auto floor_mul = [&](const Fr& rate_repr, const Fr& amt) {
-        // floor(rate * amt / 2^63) over canonical integers (fits 128 bits here)
-        u64 rl[4], al[4];
-        rate_repr.to_canonical(rl);
-        amt.to_canonical(al);
-        unsigned __int128 rate = ((unsigned __int128)rl[1] << 64) | rl[0];
-        unsigned __int128 prod = rate * al[0];  // amounts < 2^64 here
-        unsigned __int128 fl = prod >> 63;
-        u64 l[4] = {(u64)fl, (u64)(fl >> 64), 0, 0};
-        return Fr::from_canonical(l);
-    };
-    for (int i = 0; i < 2; ++i) {
-        const auto& ob = w.p[i].obligation;
-        Fr rf = floor_mul(st.relayer_fee_repr[i], ob.amount_out);
-        Fr pf = floor_mul(st.protocol_fee_repr, ob.amount_out);
-        Fr net = ob.amount_out.sub(rf).sub(pf);
-        st.new_amount_share[i] = w.p[i].pre_amount_share.sub(ob.amount_in);
-        st.new_in_shares[i] = {w.p[i].pre_in_shares.relayer_fee_balance,
-                               w.p[i].pre_in_shares.protocol_fee_balance,
-                               w.p[i].pre_in_shares.amount.sub(ob.amount_in)};
-        st.new_out_shares[i] = {w.p[i].pre_out_shares.relayer_fee_balance.add(rf),
-                                w.p[i].pre_out_shares.protocol_fee_balance.add(pf),
-                                w.p[i].pre_out_shares.amount.add(net)};
-    }
+    settlement_update_statement(w, st);
 }
 
 struct SettlementVars {
@@ -1181,15 +1188,42 @@ inline std::pair<Var, Var> partial_commitment_gadget(
     return {priv, pp};
 }
 
-// The full bundle: one settlement witness/statement plus the two parties'
-// validity witnesses/statements, mutually consistent so the validity proofs
+// ---- OUTPUT BALANCE VALIDITY (validity_proofs/output_balance.rs) ----
+// The balance-only counterpart: rotate the party's OUTPUT balance with a
+// partial commitment; links into the settlement's
+// output_balance_settlement_party{0,1} groups (11 vars).
+
+struct ObValidityWitness {  // output_balance.rs:137-155 (field order)
+    StateBalance old_balance;
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_idx[MERKLE_HEIGHT];
+    Balance balance;                           // linked (denormalized)
+    PostMatchShare post_match_balance_shares;  // linked (rfb, pfb, amount)
+};
+
+struct ObValidityStatement {  // output_balance.rs:163-176 (5 scalars)
+    Fr merkle_root, old_balance_nullifier;
+    Fr partial_private, partial_public;  // PartialCommitment
+    Fr recovery_id;
+    std::vector<Fr> to_scalars() const {
+        return {merkle_root, old_balance_nullifier, partial_private, partial_public,
+                recovery_id};
+    }
+};
+
+// The full bundle: one settlement witness/statement plus, per party, the
+// INTENT AND BALANCE VALIDITY and OUTPUT BALANCE VALIDITY
+// witnesses/statements — mutually consistent so all four validity proofs
 // link into the settlement proof (the production proof bundle the reference
-// relayer submits).
+// relayer submits: 1 settlement proof + 4 link proofs,
+// native_proof_manager.rs:554-590).
 struct ValidityBundle {
     SettlementWitness sw;
     SettlementStatement sst;
     ValidityWitness vw[2];
     ValidityStatement vst[2];
+    ObValidityWitness ow[2];
+    ObValidityStatement ost[2];
 };
 
 inline void validity_bundle_build(uint64_t seed, ValidityBundle& b) {
@@ -1284,15 +1318,139 @@ inline void validity_bundle_build(uint64_t seed, ValidityBundle& b) {
         // validity circuit's outputs ARE the settlement's pre-update shares
         b.sw.p[i].pre_amount_share = v.new_amount_public_share;
         b.sw.p[i].pre_in_shares = v.post_match_balance_shares;
+
+        // --- OUTPUT BALANCE VALIDITY for this party (output_balance.rs) ---
+        ObValidityWitness& o = b.ow[i];
+        ObValidityStatement& os = b.ost[i];
+        o.old_balance.inner = b.sw.p[i].output_balance;
+        o.old_balance.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+        o.old_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+        for (int k = 0; k < 8; ++k) o.old_balance.public_share[k] = rng.fr();
+        for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+            o.opening_elems[k] = rng.fr();
+            o.opening_idx[k] = rng.next() & 1;
+        }
+        o.balance = b.sw.p[i].output_balance;
+        auto old_ov = o.old_balance.inner.to_scalars();
+        std::vector<Fr> old_priv_o = priv_of(old_ov, o.old_balance.public_share, 8);
+        Fr old_comm_o = native_commitment(
+            old_priv_o, o.old_balance.recovery, o.old_balance.share,
+            std::vector<Fr>(o.old_balance.public_share, o.old_balance.public_share + 8));
+        os.merkle_root = native_merkle_root(
+            old_comm_o,
+            std::vector<Fr>(o.opening_elems, o.opening_elems + MERKLE_HEIGHT),
+            std::vector<bool>(o.opening_idx, o.opening_idx + MERKLE_HEIGHT));
+        os.old_balance_nullifier = native_nullifier(o.old_balance.recovery);
+        StateBalance no = o.old_balance;
+        Fr q0 = no.share.next(), q1 = no.share.next(), q2 = no.share.next();
+        o.post_match_balance_shares = {no.inner.relayer_fee_balance.sub(q0),
+                                       no.inner.protocol_fee_balance.sub(q1),
+                                       no.inner.amount.sub(q2)};
+        no.public_share[5] = o.post_match_balance_shares.relayer_fee_balance;
+        no.public_share[6] = o.post_match_balance_shares.protocol_fee_balance;
+        no.public_share[7] = o.post_match_balance_shares.amount;
+        std::vector<Fr> new_priv_o = old_priv_o;
+        new_priv_o[5] = q0;
+        new_priv_o[6] = q1;
+        new_priv_o[7] = q2;
+        os.recovery_id = no.recovery.next();
+        native_partial_commitment(
+            new_priv_o, no.recovery, no.share,
+            std::vector<Fr>(no.public_share, no.public_share + 8),
+            BALANCE_PARTIAL_COMMITMENT_SIZE, os.partial_private, os.partial_public);
+        b.sw.p[i].pre_out_shares = o.post_match_balance_shares;
     }
     // recompute the statement fields that depend on the pre-update shares
-    for (int i = 0; i < 2; ++i) {
-        const auto& ob = b.sw.p[i].obligation;
-        b.sst.new_amount_share[i] = b.sw.p[i].pre_amount_share.sub(ob.amount_in);
-        b.sst.new_in_shares[i] = {b.sw.p[i].pre_in_shares.relayer_fee_balance,
-                                  b.sw.p[i].pre_in_shares.protocol_fee_balance,
-                                  b.sw.p[i].pre_in_shares.amount.sub(ob.amount_in)};
+    settlement_update_statement(b.sw, b.sst);
+}
+
+// apply_constraints (output_balance.rs:62-130); link groups placed at the
+// settlement's output_balance layout.
+inline void ob_validity_apply_constraints(PlonkCircuit& cs, const ObValidityWitness& w,
+                                          const ObValidityStatement& st, int alignment,
+                                          int64_t party_offset0, int64_t party_offset1) {
+    const char* g0 = "output_balance_settlement_party0";
+    const char* g1 = "output_balance_settlement_party1";
+    cs.create_link_group(g0, alignment, party_offset0);
+    cs.create_link_group(g1, alignment, party_offset1);
+    auto link_both = [&](Var x) {
+        cs.add_to_link_group(x, g0);
+        cs.add_to_link_group(x, g1);
+    };
+
+    // --- witness allocation (struct field order) ---
+    StateWrapperVars ob_;
+    ob_.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                    cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    ob_.share = {cs.create_variable(w.old_balance.share.seed),
+                 cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    for (auto& s : w.old_balance.inner.to_scalars())
+        ob_.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 8; ++k)
+        ob_.public_share.push_back(cs.create_variable(w.old_balance.public_share[k]));
+    std::vector<Var> op_elems, op_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_elems.push_back(cs.create_variable(w.opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_idx[k] ? Fr::one() : Fr::zero()));
+    std::array<Var, 8> bal_v;
+    {
+        auto bv = w.balance.to_scalars();
+        for (int k = 0; k < 8; ++k) {
+            bal_v[k] = cs.create_variable(bv[k]);
+            link_both(bal_v[k]);
+        }
     }
+    std::array<Var, 3> pms_v;
+    {
+        auto pv = w.post_match_balance_shares.to_scalars();
+        for (int k = 0; k < 3; ++k) {
+            pms_v[k] = cs.create_variable(pv[k]);
+            link_both(pms_v[k]);
+        }
+    }
+
+    // --- statement (public inputs, field order) ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var p_root = pub[0], p_null = pub[1], p_priv = pub[2], p_pub = pub[3],
+        p_rid = pub[4];
+
+    // complementary private shares (:70-75)
+    std::vector<Var> old_priv;
+    for (int k = 0; k < 8; ++k)
+        old_priv.push_back(cs.sub(ob_.inner[k], ob_.public_share[k]));
+    // build_new_balance (:104-130): re-encrypt the post-match trio
+    CsprngVar nb_share = ob_.share;
+    std::vector<Var> pads, cipher;
+    stream_cipher_encrypt(cs, {ob_.inner[5], ob_.inner[6], ob_.inner[7]}, nb_share,
+                          pads, cipher);
+    std::vector<Var> new_priv = old_priv;
+    new_priv[5] = pads[0];
+    new_priv[6] = pads[1];
+    new_priv[7] = pads[2];
+    std::vector<Var> new_pub = ob_.public_share;
+    new_pub[5] = cipher[0];
+    new_pub[6] = cipher[1];
+    new_pub[7] = cipher[2];
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(cipher[k], pms_v[k]);
+    for (int k = 0; k < 8; ++k) cs.enforce_equal(ob_.inner[k], bal_v[k]);
+    // rotation with partial commitment (state_rotation.rs:130-168)
+    CsprngVar nb_rec = ob_.recovery;
+    Var rid = csprng_next(cs, nb_rec);
+    cs.enforce_equal(rid, p_rid);
+    auto pc = partial_commitment_gadget(cs, new_priv, nb_rec, nb_share, new_pub,
+                                        BALANCE_PARTIAL_COMMITMENT_SIZE);
+    cs.enforce_equal(pc.first, p_priv);
+    cs.enforce_equal(pc.second, p_pub);
+    Var old_comm = commitment_gadget(cs, old_priv, ob_.recovery, ob_.share,
+                                     ob_.public_share);
+    Var root = merkle_root_gadget(cs, old_comm, op_elems, op_idx);
+    cs.enforce_equal(root, p_root);
+    Var nul = nullifier_gadget(cs, ob_);
+    cs.enforce_equal(nul, p_null);
 }
 
 // apply_constraints (intent_and_balance.rs:71-232).  The two link groups are

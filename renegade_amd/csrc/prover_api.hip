@@ -1237,6 +1237,43 @@ void* rng_circ_build_validity(uint64_t seed, uint64_t party) {
     }
 }
 
+// OUTPUT BALANCE VALIDITY circuit for party 0 or 1 of the seed's bundle
+// (validity_proofs/output_balance.rs; links at the settlement's
+//  output_balance_settlement_party{0,1} layout)
+void* rng_circ_build_ob_validity(uint64_t seed, uint64_t party) {
+    try {
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        uint64_t align = 0;
+        int64_t off[2] = {0, 0};
+        {
+            PlonkCircuit scs;
+            settlement_apply_constraints(scs, b.sw, b.sst);
+            CircuitTables stt = scs.finalize();
+            const char* names[2] = {"output_balance_settlement_party0",
+                                    "output_balance_settlement_party1"};
+            for (auto& g : stt.link_groups)
+                for (int i = 0; i < 2; ++i)
+                    if (g.id == names[i]) {
+                        align = g.alignment;
+                        off[i] = (int64_t)g.offset;
+                    }
+        }
+        int p = (int)(party & 1);
+        PlonkCircuit cs;
+        ob_validity_apply_constraints(cs, b.ow[p], b.ost[p], (int)align, off[0], off[1]);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_ob_validity: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_ob_validity: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // circuit builders from caller-supplied witness/statement scalars (the shape
 // the external prover service receives — api_types.rs requests; Montgomery
 // limbs, field order per the reference structs)

@@ -22,6 +22,8 @@ def vb(orc):
     lib = plib.lib
     lib.rng_circ_build_validity.restype = ctypes.c_void_p
     lib.rng_circ_build_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+    lib.rng_circ_build_ob_validity.restype = ctypes.c_void_p
+    lib.rng_circ_build_ob_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
     lib.rng_circ_build_settlement_bundle.restype = ctypes.c_void_p
     lib.rng_circ_build_settlement_bundle.argtypes = [ctypes.c_uint64]
     lib.rng_circ_n.restype = ctypes.c_uint64
@@ -126,22 +128,64 @@ class TestValidityCircuit:
         o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
 
-class TestBundleLink:
-    """The production proof bundle: settlement proof + two validity proofs,
-    linked party-by-party across DIFFERENT domain sizes."""
+class TestObValidityCircuit:
+    def test_shape_and_prove(self, vb):
+        lib, o = vb["lib"], vb["o"]
+        t = vb["fetch"](lib.rng_circ_build_ob_validity(42, 0))
+        # statement = 5 scalars (output_balance.rs:163-176)
+        assert t["npub"] == 5
+        # both out-groups, 11 vars, at the settlement's layout
+        assert t["nlg"] == 2
+        ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(42))
+        stl_groups = {tuple(int(x) for x in row) for row in ts["lg"]}
+        for row in t["lg"]:
+            assert int(row[2]) == 11
+            assert tuple(int(x) for x in row) in stl_groups
+        pk = vb["setup"](t)
+        proof, _ = vb["prove"](pk, t, 13)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(t["pubs"]), ptr(proof),
+                                  ptr(vb["tau"])) == 1
+        bad = t["pubs"].copy()
+        bad[4 * 4] ^= np.uint64(1)  # recovery id
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(bad), ptr(proof),
+                                  ptr(vb["tau"])) != 1
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
-    def test_validity_settlement_link(self, vb):
+
+def bundle_groups(ts_lg):
+    """Settlement link groups by (kind, party): count-17 rows are the
+    intent_and_balance party groups, count-11 the output_balance groups;
+    within a kind, party0 has the smaller grid offset (creation order)."""
+    party = sorted((int(r[1]), (int(r[0]), int(r[1]), int(r[2])))
+                   for r in ts_lg if int(r[2]) == 17)
+    out = sorted((int(r[1]), (int(r[0]), int(r[1]), int(r[2])))
+                 for r in ts_lg if int(r[2]) == 11)
+    return {("party", 0): party[0][1], ("party", 1): party[1][1],
+            ("out", 0): out[0][1], ("out", 1): out[1][1]}
+
+
+class TestBundleLink:
+    """The production proof bundle (native_proof_manager.rs:554-590):
+    settlement proof + 4 validity proofs, linked across DIFFERENT domain
+    sizes."""
+
+    def test_full_bundle_links(self, vb):
         lib, o = vb["lib"], vb["o"]
         ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(7))
         t0 = vb["fetch"](lib.rng_circ_build_validity(7, 0))
         t1 = vb["fetch"](lib.rng_circ_build_validity(7, 1))
+        u0 = vb["fetch"](lib.rng_circ_build_ob_validity(7, 0))
+        u1 = vb["fetch"](lib.rng_circ_build_ob_validity(7, 1))
         pk_s = vb["setup"](ts)
         pk_v = vb["setup"](t0)  # same shape for both parties
+        pk_u = vb["setup"](u0)
         _, hs = vb["prove"](pk_s, ts, 3)
         _, h0 = vb["prove"](pk_v, t0, 4)
         _, h1 = vb["prove"](pk_v, t1, 5)
-        n_big = max(int(ts["n"]), int(t0["n"]))
-        pk_big = pk_v if int(t0["n"]) == n_big else pk_s
+        _, g0 = vb["prove"](pk_u, u0, 6)
+        _, g1 = vb["prove"](pk_u, u1, 7)
+        n_big = max(int(ts["n"]), int(t0["n"]), int(u0["n"]))
+        pk_big = {int(ts["n"]): pk_s, int(t0["n"]): pk_v, int(u0["n"]): pk_u}[n_big]
 
         def ext(h, n_small):
             out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
@@ -150,13 +194,13 @@ class TestBundleLink:
             return out
 
         hs_e = ext(hs, int(ts["n"]))
-        groups = {int(r[1]): (int(r[0]), int(r[1]), int(r[2])) for r in ts["lg"]
-                  if int(r[2]) == 17}
-        offs = sorted(groups)  # party0 offset < party1 offset (creation order)
-        party_group = {0: groups[offs[0]], 1: groups[offs[1]]}
-        for party, hv in ((0, h0), (1, h1)):
-            a, off, cnt = party_group[party]
-            hv_e = ext(hv, int(t0["n"]))
+        groups = bundle_groups(ts["lg"])
+        legs = [(("party", 0), ext(h0, int(t0["n"]))),
+                (("party", 1), ext(h1, int(t1["n"]))),
+                (("out", 0), ext(g0, int(u0["n"]))),
+                (("out", 1), ext(g1, int(u1["n"])))]
+        for key, hv_e in legs:
+            a, off, cnt = groups[key]
             lp = np.zeros(18, dtype=np.uint64)
             rc = o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hs_e),
                                   ctypes.c_uint64(a), ctypes.c_uint64(off),
@@ -167,10 +211,10 @@ class TestBundleLink:
                                          ptr(lp), ctypes.c_uint64(a),
                                          ctypes.c_uint64(off), ctypes.c_uint64(cnt),
                                          ptr(vb["tau"]))
-            assert ok == 1, f"validity party{party} <-> settlement link failed"
+            assert ok == 1, f"{key} <-> settlement link failed"
 
         # negative: party0's validity proof must NOT link at party1's offset
-        a, off, cnt = party_group[1]
+        a, off, cnt = groups[("party", 1)]
         h0_e = ext(h0, int(t0["n"]))
         lp = np.zeros(18, dtype=np.uint64)
         o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(h0_e), ptr(hs_e),
@@ -181,8 +225,8 @@ class TestBundleLink:
                                      ctypes.c_uint64(a), ctypes.c_uint64(off),
                                      ctypes.c_uint64(cnt), ptr(vb["tau"]))
         assert ok != 1, "cross-party link should not verify"
-        o.orc_plonk_pk_free(ctypes.c_void_p(pk_s))
-        o.orc_plonk_pk_free(ctypes.c_void_p(pk_v))
+        for pk in (pk_s, pk_v, pk_u):
+            o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
 
 @pytest.mark.gpu
@@ -208,7 +252,9 @@ class TestBundleLinkGpu:
         ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(7))
         t0 = vb["fetch"](lib.rng_circ_build_validity(7, 0))
         t1 = vb["fetch"](lib.rng_circ_build_validity(7, 1))
-        n_big = max(int(ts["n"]), int(t0["n"]))
+        u0 = vb["fetch"](lib.rng_circ_build_ob_validity(7, 0))
+        u1 = vb["fetch"](lib.rng_circ_build_ob_validity(7, 1))
+        n_big = max(int(ts["n"]), int(t0["n"]), int(u0["n"]))
         power = max(4, int(n_big).bit_length())  # covers n_big + 2
         ptau = orc.srs_generate_ptau(power, seed=42)
         max_degree = (1 << power) + 2
@@ -232,6 +278,8 @@ class TestBundleLinkGpu:
         pk_s, _, hs = gprove(ts, 3)
         pk_v, _, h0 = gprove(t0, 4)
         _, _, h1 = gprove(t1, 5)
+        pk_u, _, g0 = gprove(u0, 6)
+        _, _, g1 = gprove(u1, 7)
 
         def ext(h, n_small):
             out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
@@ -240,28 +288,30 @@ class TestBundleLinkGpu:
             return out
 
         hs_e = ext(hs, int(ts["n"]))
-        pk_big = pk_v if int(t0["n"]) == n_big else pk_s
-        groups = {int(r[1]): (int(r[0]), int(r[1]), int(r[2])) for r in ts["lg"]
-                  if int(r[2]) == 17}
-        offs = sorted(groups)
-        for party, hv in ((0, h0), (1, h1)):
-            a, off, cnt = groups[offs[party]]
-            hv_e = ext(hv, int(t0["n"]))
+        pk_big = {int(ts["n"]): pk_s, int(t0["n"]): pk_v, int(u0["n"]): pk_u}[n_big]
+        t_big = {int(ts["n"]): ts, int(t0["n"]): t0, int(u0["n"]): u0}[n_big]
+        opk_big = vb["setup"](t_big)
+        groups = bundle_groups(ts["lg"])
+        legs = [(("party", 0), ext(h0, int(t0["n"]))),
+                (("party", 1), ext(h1, int(t1["n"]))),
+                (("out", 0), ext(g0, int(u0["n"]))),
+                (("out", 1), ext(g1, int(u1["n"])))]
+        for key, hv_e in legs:
+            a, off, cnt = groups[key]
             lp_g = np.zeros(18, dtype=np.uint64)
             assert lib.rng_link_proofs(ctx.h, ctypes.c_void_p(pk_big), ptr(hv_e),
                                        ptr(hs_e), a, off, cnt, ptr(lp_g)) == 0
             # oracle computes the same link proof bit-exact
-            opk_big = vb["setup"](t0 if int(t0["n"]) == n_big else ts)
             lp_o = np.zeros(18, dtype=np.uint64)
             assert o.orc_plonk_link(ctypes.c_void_p(opk_big), ptr(hv_e), ptr(hs_e),
                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
                                     ctypes.c_uint64(cnt), ptr(lp_o)) == 0
-            assert np.array_equal(lp_g, lp_o), f"party{party} GPU link != oracle"
+            assert np.array_equal(lp_g, lp_o), f"{key} GPU link != oracle"
             ok = o.orc_plonk_link_verify(ctypes.c_void_p(opk_big),
                                          ptr(hv_e[-9:].copy()), ptr(hs_e[-9:].copy()),
                                          ptr(lp_g), ctypes.c_uint64(a),
                                          ctypes.c_uint64(off), ctypes.c_uint64(cnt),
                                          ptr(vb["tau"]))
             assert ok == 1
-            o.orc_plonk_pk_free(ctypes.c_void_p(opk_big))
+        o.orc_plonk_pk_free(ctypes.c_void_p(opk_big))
         ctx.close()
