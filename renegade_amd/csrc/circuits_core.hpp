@@ -1211,6 +1211,86 @@ struct ObValidityStatement {  // output_balance.rs:163-176 (5 scalars)
     }
 };
 
+// ================== Valid Order Cancellation ==================
+// (zk_circuits/valid_order_cancellation.rs — prove the intent exists and
+//  spend its nullifier; the owner is leaked for contract authorization.)
+
+struct VocWitness {  // valid_order_cancellation.rs:77-84
+    StateIntent old_intent;
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_idx[MERKLE_HEIGHT];
+};
+struct VocStatement {  // :95-105 (3 scalars)
+    Fr merkle_root, old_intent_nullifier, owner;
+    std::vector<Fr> to_scalars() const {
+        return {merkle_root, old_intent_nullifier, owner};
+    }
+};
+
+inline void voc_build_witness_statement(uint64_t seed, VocWitness& w, VocStatement& st) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    Fr owner = addr();
+    w.old_intent.inner = {addr(), addr(), owner, rng.fr(), Fr::from_u64(rng.next())};
+    w.old_intent.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+    w.old_intent.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int k = 0; k < 5; ++k) w.old_intent.public_share[k] = rng.fr();
+    for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+        w.opening_elems[k] = rng.fr();
+        w.opening_idx[k] = rng.next() & 1;
+    }
+    auto iv = w.old_intent.inner.to_scalars();
+    std::vector<Fr> priv;
+    for (int k = 0; k < 5; ++k) priv.push_back(iv[k].sub(w.old_intent.public_share[k]));
+    Fr comm = native_commitment(
+        priv, w.old_intent.recovery, w.old_intent.share,
+        std::vector<Fr>(w.old_intent.public_share, w.old_intent.public_share + 5));
+    st.merkle_root = native_merkle_root(
+        comm, std::vector<Fr>(w.opening_elems, w.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(w.opening_idx, w.opening_idx + MERKLE_HEIGHT));
+    st.old_intent_nullifier = native_nullifier(w.old_intent.recovery);
+    st.owner = owner;
+}
+
+inline void voc_apply_constraints(PlonkCircuit& cs, const VocWitness& w,
+                                  const VocStatement& st) {
+    StateWrapperVars oi;
+    oi.recovery = {cs.create_variable(w.old_intent.recovery.seed),
+                   cs.create_variable(Fr::from_u64(w.old_intent.recovery.index))};
+    oi.share = {cs.create_variable(w.old_intent.share.seed),
+                cs.create_variable(Fr::from_u64(w.old_intent.share.index))};
+    for (auto& s : w.old_intent.inner.to_scalars())
+        oi.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 5; ++k)
+        oi.public_share.push_back(cs.create_variable(w.old_intent.public_share[k]));
+    std::vector<Var> op_elems, op_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_elems.push_back(cs.create_variable(w.opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_idx[k] ? Fr::one() : Fr::zero()));
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+
+    // 1. intent exists in the tree (:48-60)
+    std::vector<Var> priv;
+    for (int k = 0; k < 5; ++k)
+        priv.push_back(cs.sub(oi.inner[k], oi.public_share[k]));
+    Var comm = commitment_gadget(cs, priv, oi.recovery, oi.share, oi.public_share);
+    Var root = merkle_root_gadget(cs, comm, op_elems, op_idx);
+    cs.enforce_equal(root, pub[0]);
+    // 2. nullifier (:62-64)
+    Var nul = nullifier_gadget(cs, oi);
+    cs.enforce_equal(nul, pub[1]);
+    // 3. owner leak (:66-67)
+    cs.enforce_equal(oi.inner[2], pub[2]);
+}
+
 // The full bundle: one settlement witness/statement plus, per party, the
 // INTENT AND BALANCE VALIDITY and OUTPUT BALANCE VALIDITY
 // witnesses/statements — mutually consistent so all four validity proofs

@@ -1274,6 +1274,26 @@ void* rng_circ_build_ob_validity(uint64_t seed, uint64_t party) {
     }
 }
 
+// VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
+void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
+    try {
+        VocWitness w;
+        VocStatement st;
+        voc_build_witness_statement(seed, w, st);
+        PlonkCircuit cs;
+        voc_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_valid_order_cancellation: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_valid_order_cancellation: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // circuit builders from caller-supplied witness/statement scalars (the shape
 // the external prover service receives — api_types.rs requests; Montgomery
 // limbs, field order per the reference structs)

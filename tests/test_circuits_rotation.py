@@ -1,5 +1,6 @@
-"""State-rotation circuits (`Valid Deposit`, `Valid Withdrawal` —
-zk_circuits/valid_deposit.rs, valid_withdrawal.rs): build, satisfiability,
+"""State-rotation circuits (`Valid Deposit`, `Valid Withdrawal`,
+`Valid Order Cancellation` — zk_circuits/valid_deposit.rs,
+valid_withdrawal.rs, valid_order_cancellation.rs): build, satisfiability,
 oracle prove/verify, tamper rejection (CPU)."""
 import ctypes
 
@@ -9,7 +10,13 @@ import pytest
 U64P = ctypes.POINTER(ctypes.c_uint64)
 ptr = lambda a: a.ctypes.data_as(U64P)
 
-BUILDERS = ["rng_circ_build_valid_deposit", "rng_circ_build_valid_withdrawal"]
+# builder -> (statement scalars, tampered public-input word offsets)
+BUILDER_INFO = {
+    "rng_circ_build_valid_deposit": (8, [3 * 4, 4 * 4, 5 * 4, 6 * 4]),
+    "rng_circ_build_valid_withdrawal": (8, [3 * 4, 4 * 4, 5 * 4, 6 * 4]),
+    "rng_circ_build_valid_order_cancellation": (3, [0, 1 * 4, 2 * 4]),
+}
+BUILDERS = list(BUILDER_INFO)
 
 
 @pytest.fixture(scope="module")
@@ -44,7 +51,7 @@ def build_tables(lib, builder, seed):
 @pytest.mark.parametrize("builder", BUILDERS)
 def test_rotation_circuit_prove_verify(plib, orc, builder):
     n, npub, sel, sigma, wires, pubs = build_tables(plib.lib, builder, 42)
-    assert npub == 8
+    assert npub == BUILDER_INFO[builder][0]
     power = max(4, int(n).bit_length())
     ptau = orc.srs_generate_ptau(power, seed=42)
     md = (1 << power) + 2
@@ -65,7 +72,7 @@ def test_rotation_circuit_prove_verify(plib, orc, builder):
     tau = np.zeros(4, dtype=np.uint64)
     o.orc_derive_tau(42, ptr(tau))
     assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(pubs), ptr(proof), ptr(tau)) == 1
-    for idx in [3 * 4, 4 * 4, 5 * 4, 6 * 4]:  # root, nullifier, commitment, rid
+    for idx in BUILDER_INFO[builder][1]:
         bad = pubs.copy()
         bad[idx] ^= np.uint64(1)
         assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(bad), ptr(proof),

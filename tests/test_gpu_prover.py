@@ -141,7 +141,9 @@ class TestGpuProver:
                                           ("rng_circ_build_vbc", 42),
                                           ("rng_circ_build_valid_deposit", 42),
                                           ("rng_circ_build_valid_withdrawal", 42),
-                                          ("rng_circ_build_validity", 42)])
+                                          ("rng_circ_build_validity", 42),
+                                          ("rng_circ_build_ob_validity", 42),
+                                          ("rng_circ_build_valid_order_cancellation", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
@@ -152,7 +154,8 @@ def test_real_circuit_gpu_parity(orc, builder, seed):
     lib = plib.lib
     fn = getattr(lib, builder)
     fn.restype = ctypes.c_void_p
-    two_arg = builder == "rng_circ_build_validity"  # (seed, party)
+    two_arg = builder in ("rng_circ_build_validity",
+                          "rng_circ_build_ob_validity")  # (seed, party)
     fn.argtypes = [ctypes.c_uint64, ctypes.c_uint64] if two_arg else [ctypes.c_uint64]
     lib.rng_circ_n.restype = ctypes.c_uint64
     lib.rng_circ_n.argtypes = [ctypes.c_void_p]
