@@ -1211,6 +1211,100 @@ struct ObValidityStatement {  // output_balance.rs:163-176 (5 scalars)
     }
 };
 
+// ================== Intent And Balance Public Settlement ==================
+// (settlement/intent_and_balance_public_settlement.rs — single-party public
+//  settlement by a relayer cluster: the obligation and pre-update shares are
+//  LEAKED in the statement and applied on-chain; links against the party-0
+//  groups of the validity proofs at the private settlement's layout.)
+
+struct PubSettlementStatement {  // :148-180 (14 scalars)
+    Obligation obligation;
+    Fr amount_public_share;
+    PostMatchShare in_shares, out_shares;
+    Fr relayer_fee_repr, protocol_fee_repr;  // FeeRates (fee.rs:43-48)
+    Fr relayer_fee_recipient;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = obligation.to_scalars();
+        v.push_back(amount_public_share);
+        auto a = in_shares.to_scalars();
+        v.insert(v.end(), a.begin(), a.end());
+        auto b = out_shares.to_scalars();
+        v.insert(v.end(), b.begin(), b.end());
+        v.push_back(relayer_fee_repr);
+        v.push_back(protocol_fee_repr);
+        v.push_back(relayer_fee_recipient);
+        return v;
+    }
+};
+
+// witness = the party's linked vars (SettlementParty minus the obligation,
+// :100-143); built from bundle party 0 so its links match the validity
+// proofs.  Groups are placed at the private settlement's layout
+// (proof_linking_groups, :194-213).
+inline void pub_settlement_apply_constraints(PlonkCircuit& cs,
+                                             const SettlementParty& w,
+                                             const PubSettlementStatement& st,
+                                             int alignment, int64_t pg_offset,
+                                             int64_t og_offset) {
+    const char* pg = "intent_and_balance_settlement_party0";
+    const char* og = "output_balance_settlement_party0";
+    cs.create_link_group(pg, alignment, pg_offset);
+    cs.create_link_group(og, alignment, og_offset);
+    auto alloc_list = [&](const std::vector<Fr>& vals, const char* group) {
+        std::vector<Var> out;
+        for (auto& v : vals) {
+            Var x = cs.create_variable(v);
+            if (group) cs.add_to_link_group(x, group);
+            out.push_back(x);
+        }
+        return out;
+    };
+    // --- witness (field order :100-143; linked groups per annotations) ---
+    auto intent_v = alloc_list(w.intent.to_scalars(), pg);
+    auto pre_amt = alloc_list({w.pre_amount_share}, pg);
+    auto in_bal = alloc_list(w.input_balance.to_scalars(), pg);
+    auto pre_in = alloc_list(w.pre_in_shares.to_scalars(), pg);
+    auto out_bal = alloc_list(w.output_balance.to_scalars(), og);
+    auto pre_out = alloc_list(w.pre_out_shares.to_scalars(), og);
+
+    // --- statement (public inputs, field order) ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    std::array<Var, 4> ob{pub[0], pub[1], pub[2], pub[3]};
+    Var p_amt_share = pub[4];
+    std::array<Var, 3> p_in{pub[5], pub[6], pub[7]};
+    std::array<Var, 3> p_out{pub[8], pub[9], pub[10]};
+    Var relayer_rate = pub[11], protocol_rate = pub[12], p_fee_recipient = pub[13];
+
+    // 1. fee take from the statement rates (:52-56)
+    FeeTakeVars ft = fee_take_gadget(cs, ob[3], relayer_rate, protocol_rate);
+    // 2. intent/balance obligation constraints (settlement_lib.rs:29-126)
+    cs.enforce_equal(ob[0], intent_v[0]);
+    cs.enforce_equal(ob[1], intent_v[1]);
+    gte_gadget(cs, intent_v[4], ob[2], AMOUNT_BITS);
+    Var min_out_fp = cs.mul(intent_v[3], ob[2]);
+    Var min_out = fp_floor_gadget(cs, min_out_fp);
+    gte_gadget(cs, ob[3], min_out, AMOUNT_BITS);
+    gte_gadget(cs, in_bal[7], ob[2], AMOUNT_BITS);
+    cs.enforce_equal(out_bal[0], ob[1]);
+    cs.enforce_equal(out_bal[1], intent_v[2]);
+    Var total_fee = cs.add(ft.relayer_fee, ft.protocol_fee);
+    Var net_receive = cs.sub(ob[3], total_fee);
+    Var new_bal_amount = cs.add(out_bal[7], net_receive);
+    cs.enforce_in_range(new_bal_amount, AMOUNT_BITS);
+    Var new_rfb = cs.add(out_bal[5], ft.relayer_fee);
+    Var new_pfb = cs.add(out_bal[6], ft.protocol_fee);
+    cs.enforce_in_range(new_rfb, AMOUNT_BITS);
+    cs.enforce_in_range(new_pfb, AMOUNT_BITS);
+    // 3. leaked pre-update shares match the proof-linked witness (:68-86)
+    cs.enforce_equal(pre_amt[0], p_amt_share);
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(pre_in[k], p_in[k]);
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(pre_out[k], p_out[k]);
+    // 4. relayer fee recipient leak (:88-93)
+    cs.enforce_equal(out_bal[2], p_fee_recipient);
+}
+
 // ================== Valid Order Cancellation ==================
 // (zk_circuits/valid_order_cancellation.rs — prove the intent exists and
 //  spend its nullifier; the owner is leaked for contract authorization.)
@@ -1442,6 +1536,18 @@ inline void validity_bundle_build(uint64_t seed, ValidityBundle& b) {
     }
     // recompute the statement fields that depend on the pre-update shares
     settlement_update_statement(b.sw, b.sst);
+}
+
+// statement for bundle party 0 (test_helpers :263-303 semantics)
+inline void pub_settlement_statement_from_bundle(const ValidityBundle& b,
+                                                 PubSettlementStatement& st) {
+    st.obligation = b.sw.p[0].obligation;
+    st.amount_public_share = b.sw.p[0].pre_amount_share;
+    st.in_shares = b.sw.p[0].pre_in_shares;
+    st.out_shares = b.sw.p[0].pre_out_shares;
+    st.relayer_fee_repr = b.sst.relayer_fee_repr[0];
+    st.protocol_fee_repr = b.sst.protocol_fee_repr;
+    st.relayer_fee_recipient = b.sw.p[0].output_balance.relayer_fee_recipient;
 }
 
 // apply_constraints (output_balance.rs:62-130); link groups placed at the

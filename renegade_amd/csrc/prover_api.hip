@@ -1274,6 +1274,43 @@ void* rng_circ_build_ob_validity(uint64_t seed, uint64_t party) {
     }
 }
 
+// INTENT AND BALANCE PUBLIC SETTLEMENT circuit for party 0 of the seed's
+// bundle (settlement/intent_and_balance_public_settlement.rs); its two link
+// groups are placed at the private settlement's party-0 layout.
+void* rng_circ_build_public_settlement(uint64_t seed) {
+    try {
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        uint64_t align = 0;
+        int64_t pg_off = 0, og_off = 0;
+        {
+            PlonkCircuit scs;
+            settlement_apply_constraints(scs, b.sw, b.sst);
+            CircuitTables stt = scs.finalize();
+            for (auto& g : stt.link_groups) {
+                if (g.id == "intent_and_balance_settlement_party0") {
+                    align = g.alignment;
+                    pg_off = (int64_t)g.offset;
+                }
+                if (g.id == "output_balance_settlement_party0") og_off = (int64_t)g.offset;
+            }
+        }
+        PubSettlementStatement st;
+        pub_settlement_statement_from_bundle(b, st);
+        PlonkCircuit cs;
+        pub_settlement_apply_constraints(cs, b.sw.p[0], st, (int)align, pg_off, og_off);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_public_settlement: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_public_settlement: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
 void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
     try {

@@ -225,6 +225,29 @@ class TestBundleLink:
                                      ctypes.c_uint64(a), ctypes.c_uint64(off),
                                      ctypes.c_uint64(cnt), ptr(vb["tau"]))
         assert ok != 1, "cross-party link should not verify"
+
+        # --- public settlement: same validity proofs link into the
+        #     single-party public settlement at the party-0 layouts ---
+        lib.rng_circ_build_public_settlement.restype = ctypes.c_void_p
+        lib.rng_circ_build_public_settlement.argtypes = [ctypes.c_uint64]
+        tp = vb["fetch"](lib.rng_circ_build_public_settlement(7))
+        assert tp["npub"] == 14
+        pk_p = vb["setup"](tp)
+        _, hp = vb["prove"](pk_p, tp, 8)
+        hp_e = ext(hp, int(tp["n"]))
+        for key, hv_e in legs[:1] + legs[2:3]:  # (party,0) and (out,0)
+            a, off, cnt = groups[key]
+            lp = np.zeros(18, dtype=np.uint64)
+            assert o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hp_e),
+                                    ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                    ctypes.c_uint64(cnt), ptr(lp)) == 0
+            ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big),
+                                         ptr(hv_e[-9:].copy()), ptr(hp_e[-9:].copy()),
+                                         ptr(lp), ctypes.c_uint64(a),
+                                         ctypes.c_uint64(off), ctypes.c_uint64(cnt),
+                                         ptr(vb["tau"]))
+            assert ok == 1, f"{key} <-> public settlement link failed"
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_p))
         for pk in (pk_s, pk_v, pk_u):
             o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
