@@ -1305,6 +1305,195 @@ inline void pub_settlement_apply_constraints(PlonkCircuit& cs,
     cs.enforce_equal(out_bal[2], p_fee_recipient);
 }
 
+// ================== Intent Only Settlement + Validity ==================
+// (settlement/intent_only_public_settlement.rs + validity_proofs/intent_only.rs
+//  — a private intent capitalized by a PUBLIC balance: the settlement leaks
+//  the obligation; one 5-var link group "intent_only_settlement" placed by
+//  the settlement circuit and inherited by the validity circuit.)
+
+struct IoSettlementStatement {  // intent_only_public_settlement.rs:85-112 (6)
+    Obligation obligation;
+    Fr relayer_fee_repr;      // transcript-only (non-malleable), unconstrained
+    Fr relayer_fee_recipient;  // likewise
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = obligation.to_scalars();
+        v.push_back(relayer_fee_repr);
+        v.push_back(relayer_fee_recipient);
+        return v;
+    }
+};
+
+struct IoValidityWitness {  // intent_only.rs:148-166
+    StateIntent old_intent;
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_idx[MERKLE_HEIGHT];
+    Intent intent;  // linked (denormalized new-intent inner)
+};
+struct IoValidityStatement {  // intent_only.rs:174-204 (7 scalars)
+    Fr owner, merkle_root, old_intent_nullifier, new_amount_public_share;
+    Fr partial_private, partial_public;
+    Fr recovery_id;
+    std::vector<Fr> to_scalars() const {
+        return {owner, merkle_root, old_intent_nullifier, new_amount_public_share,
+                partial_private, partial_public, recovery_id};
+    }
+};
+
+// consistent pair: the validity witness wraps the same intent the settlement
+// witness links, and the obligation respects the intent's constraints
+inline void io_bundle_build(uint64_t seed, IoValidityWitness& vw, IoValidityStatement& vs,
+                            IoSettlementStatement& ss) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    Fr token0 = addr(), token1 = addr(), owner = addr();
+    uint64_t amount = (rng.next() & ((1ull << 52) - 1)) + 2;
+    uint64_t traded_in = amount / 2 + 1;
+    uint64_t traded_out = (rng.next() & ((1ull << 52) - 1)) + 2;
+    auto half_price_repr = [&](uint64_t out_amt, uint64_t in_amt) {
+        unsigned __int128 r = ((unsigned __int128)out_amt << 63) / in_amt / 2;
+        u64 l[4] = {(u64)r, (u64)(r >> 64), 0, 0};
+        return Fr::from_canonical(l);
+    };
+    vw.intent = {token0, token1, owner, half_price_repr(traded_out, traded_in),
+                 Fr::from_u64(amount)};
+    ss.obligation = {token0, token1, Fr::from_u64(traded_in), Fr::from_u64(traded_out)};
+    auto fee_repr = [&]() {
+        unsigned __int128 r = ((unsigned __int128)1 << 63) / (1000 + (rng.next() & 1023));
+        u64 l[4] = {(u64)r, (u64)(r >> 64), 0, 0};
+        return Fr::from_canonical(l);
+    };
+    ss.relayer_fee_repr = fee_repr();
+    ss.relayer_fee_recipient = addr();
+
+    vw.old_intent.inner = vw.intent;
+    vw.old_intent.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+    vw.old_intent.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int k = 0; k < 5; ++k) vw.old_intent.public_share[k] = rng.fr();
+    for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+        vw.opening_elems[k] = rng.fr();
+        vw.opening_idx[k] = rng.next() & 1;
+    }
+    auto iv = vw.old_intent.inner.to_scalars();
+    std::vector<Fr> old_priv;
+    for (int k = 0; k < 5; ++k)
+        old_priv.push_back(iv[k].sub(vw.old_intent.public_share[k]));
+    Fr old_comm = native_commitment(
+        old_priv, vw.old_intent.recovery, vw.old_intent.share,
+        std::vector<Fr>(vw.old_intent.public_share, vw.old_intent.public_share + 5));
+    vs.owner = owner;
+    vs.merkle_root = native_merkle_root(
+        old_comm, std::vector<Fr>(vw.opening_elems, vw.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(vw.opening_idx, vw.opening_idx + MERKLE_HEIGHT));
+    vs.old_intent_nullifier = native_nullifier(vw.old_intent.recovery);
+    StateIntent ni = vw.old_intent;
+    Fr pad = ni.share.next();
+    vs.new_amount_public_share = ni.inner.amount_in.sub(pad);
+    ni.public_share[4] = vs.new_amount_public_share;
+    std::vector<Fr> new_priv = old_priv;
+    new_priv[4] = pad;
+    vs.recovery_id = ni.recovery.next();
+    native_partial_commitment(
+        new_priv, ni.recovery, ni.share,
+        std::vector<Fr>(ni.public_share, ni.public_share + 5),
+        INTENT_PARTIAL_COMMITMENT_SIZE, vs.partial_private, vs.partial_public);
+}
+
+// intent_only_public_settlement.rs:48-62: intent constraints only; the group
+// is PLACED by this circuit (auto layout)
+inline void io_settlement_apply_constraints(PlonkCircuit& cs, const Intent& w,
+                                            const IoSettlementStatement& st) {
+    cs.create_link_group("intent_only_settlement", -1, -1);
+    std::array<Var, 5> intent_v;
+    auto iv = w.to_scalars();
+    for (int k = 0; k < 5; ++k) {
+        intent_v[k] = cs.create_variable(iv[k]);
+        cs.add_to_link_group(intent_v[k], "intent_only_settlement");
+    }
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    // verify_intent_constraints (settlement_lib.rs:45-74)
+    cs.enforce_equal(pub[0], intent_v[0]);
+    cs.enforce_equal(pub[1], intent_v[1]);
+    gte_gadget(cs, intent_v[4], pub[2], AMOUNT_BITS);
+    Var min_out_fp = cs.mul(intent_v[3], pub[2]);
+    Var min_out = fp_floor_gadget(cs, min_out_fp);
+    gte_gadget(cs, pub[3], min_out, AMOUNT_BITS);
+}
+
+// intent_only.rs:63-116; the link group inherits the settlement's placement
+inline void io_validity_apply_constraints(PlonkCircuit& cs, const IoValidityWitness& w,
+                                          const IoValidityStatement& st, int alignment,
+                                          int64_t offset) {
+    const char* g = "intent_only_settlement";
+    cs.create_link_group(g, alignment, offset);
+
+    // --- witness (field order) ---
+    StateWrapperVars oi;
+    oi.recovery = {cs.create_variable(w.old_intent.recovery.seed),
+                   cs.create_variable(Fr::from_u64(w.old_intent.recovery.index))};
+    oi.share = {cs.create_variable(w.old_intent.share.seed),
+                cs.create_variable(Fr::from_u64(w.old_intent.share.index))};
+    for (auto& s : w.old_intent.inner.to_scalars())
+        oi.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 5; ++k)
+        oi.public_share.push_back(cs.create_variable(w.old_intent.public_share[k]));
+    std::vector<Var> op_elems, op_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_elems.push_back(cs.create_variable(w.opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_idx[k] ? Fr::one() : Fr::zero()));
+    std::array<Var, 5> intent_v;
+    {
+        auto iv = w.intent.to_scalars();
+        for (int k = 0; k < 5; ++k) {
+            intent_v[k] = cs.create_variable(iv[k]);
+            cs.add_to_link_group(intent_v[k], g);
+        }
+    }
+    // --- statement ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var p_owner = pub[0], p_root = pub[1], p_null = pub[2], p_amt = pub[3],
+        p_priv = pub[4], p_pub = pub[5], p_rid = pub[6];
+
+    // 1-2. complementary shares + new intent (:69-79, :119-143)
+    std::vector<Var> old_priv;
+    for (int k = 0; k < 5; ++k)
+        old_priv.push_back(cs.sub(oi.inner[k], oi.public_share[k]));
+    CsprngVar ni_share = oi.share;
+    std::vector<Var> pads, cipher;
+    stream_cipher_encrypt(cs, {oi.inner[4]}, ni_share, pads, cipher);
+    std::vector<Var> new_priv = old_priv;
+    new_priv[4] = pads[0];
+    std::vector<Var> new_pub = oi.public_share;
+    new_pub[4] = cipher[0];
+    // 3. denormalized intent + leaks (:81-92)
+    for (int k = 0; k < 5; ++k) cs.enforce_equal(oi.inner[k], intent_v[k]);
+    cs.enforce_equal(cipher[0], p_amt);
+    cs.enforce_equal(oi.inner[2], p_owner);
+    // 4. rotation with partial commitment (:94-113)
+    CsprngVar ni_rec = oi.recovery;
+    Var rid = csprng_next(cs, ni_rec);
+    cs.enforce_equal(rid, p_rid);
+    auto pc = partial_commitment_gadget(cs, new_priv, ni_rec, ni_share, new_pub,
+                                        INTENT_PARTIAL_COMMITMENT_SIZE);
+    cs.enforce_equal(pc.first, p_priv);
+    cs.enforce_equal(pc.second, p_pub);
+    Var old_comm = commitment_gadget(cs, old_priv, oi.recovery, oi.share,
+                                     oi.public_share);
+    Var root = merkle_root_gadget(cs, old_comm, op_elems, op_idx);
+    cs.enforce_equal(root, p_root);
+    Var nul = nullifier_gadget(cs, oi);
+    cs.enforce_equal(nul, p_null);
+}
+
 // ================== Valid Order Cancellation ==================
 // (zk_circuits/valid_order_cancellation.rs — prove the intent exists and
 //  spend its nullifier; the owner is leaked for contract authorization.)

@@ -1311,6 +1311,61 @@ void* rng_circ_build_public_settlement(uint64_t seed) {
     }
 }
 
+// INTENT ONLY PUBLIC SETTLEMENT circuit (intent_only_public_settlement.rs)
+void* rng_circ_build_io_settlement(uint64_t seed) {
+    try {
+        IoValidityWitness vw;
+        IoValidityStatement vs;
+        IoSettlementStatement ss;
+        io_bundle_build(seed, vw, vs, ss);
+        PlonkCircuit cs;
+        io_settlement_apply_constraints(cs, vw.intent, ss);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_io_settlement: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_io_settlement: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// INTENT ONLY VALIDITY circuit (validity_proofs/intent_only.rs); its link
+// group inherits the INTENT ONLY PUBLIC SETTLEMENT placement
+void* rng_circ_build_io_validity(uint64_t seed) {
+    try {
+        IoValidityWitness vw;
+        IoValidityStatement vs;
+        IoSettlementStatement ss;
+        io_bundle_build(seed, vw, vs, ss);
+        uint64_t align = 0;
+        int64_t off = 0;
+        {
+            PlonkCircuit scs;
+            io_settlement_apply_constraints(scs, vw.intent, ss);
+            CircuitTables stt = scs.finalize();
+            for (auto& g : stt.link_groups)
+                if (g.id == "intent_only_settlement") {
+                    align = g.alignment;
+                    off = (int64_t)g.offset;
+                }
+        }
+        PlonkCircuit cs;
+        io_validity_apply_constraints(cs, vw, vs, (int)align, off);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_io_validity: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_io_validity: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
 void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
     try {

@@ -252,6 +252,68 @@ class TestBundleLink:
             o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
 
+class TestIntentOnly:
+    """INTENT ONLY VALIDITY <-> INTENT ONLY PUBLIC SETTLEMENT
+    (validity_proofs/intent_only.rs, intent_only_public_settlement.rs):
+    one 5-var link group placed by the settlement, inherited cross-domain."""
+
+    def test_pair_and_link(self, vb):
+        lib, o = vb["lib"], vb["o"]
+        lib.rng_circ_build_io_settlement.restype = ctypes.c_void_p
+        lib.rng_circ_build_io_settlement.argtypes = [ctypes.c_uint64]
+        lib.rng_circ_build_io_validity.restype = ctypes.c_void_p
+        lib.rng_circ_build_io_validity.argtypes = [ctypes.c_uint64]
+        ts = vb["fetch"](lib.rng_circ_build_io_settlement(5))
+        tv = vb["fetch"](lib.rng_circ_build_io_validity(5))
+        assert ts["npub"] == 6 and tv["npub"] == 7
+        assert [tuple(map(int, r)) for r in ts["lg"]] == \
+            [tuple(map(int, r)) for r in tv["lg"]]
+        pk_s = vb["setup"](ts)
+        pk_v = vb["setup"](tv)
+        ps, hs = vb["prove"](pk_s, ts, 3)
+        pv_, hv = vb["prove"](pk_v, tv, 4)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk_s), ptr(ts["pubs"]), ptr(ps),
+                                  ptr(vb["tau"])) == 1
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk_v), ptr(tv["pubs"]), ptr(pv_),
+                                  ptr(vb["tau"])) == 1
+        n_big = max(int(ts["n"]), int(tv["n"]))
+        pk_big = pk_v if int(tv["n"]) == n_big else pk_s
+
+        def ext(h, n_small):
+            out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+            out[:4 * (n_small + 2)] = h[:4 * (n_small + 2)]
+            out[-9:] = h[-9:]
+            return out
+
+        a, off, cnt = (int(x) for x in tv["lg"][0])
+        hv_e, hs_e = ext(hv, int(tv["n"])), ext(hs, int(ts["n"]))
+        lp = np.zeros(18, dtype=np.uint64)
+        assert o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hs_e),
+                                ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                ctypes.c_uint64(cnt), ptr(lp)) == 0
+        ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big), ptr(hv_e[-9:].copy()),
+                                     ptr(hs_e[-9:].copy()), ptr(lp),
+                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                     ctypes.c_uint64(cnt), ptr(vb["tau"]))
+        assert ok == 1, "intent-only validity <-> settlement link failed"
+
+        # a settlement over a DIFFERENT intent must not link
+        ts2 = vb["fetch"](lib.rng_circ_build_io_settlement(6))
+        pk_s2 = vb["setup"](ts2)
+        _, hs2 = vb["prove"](pk_s2, ts2, 3)
+        hs2_e = ext(hs2, int(ts2["n"]))
+        o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hs2_e),
+                         ctypes.c_uint64(a), ctypes.c_uint64(off),
+                         ctypes.c_uint64(cnt), ptr(lp))
+        ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big), ptr(hv_e[-9:].copy()),
+                                     ptr(hs2_e[-9:].copy()), ptr(lp),
+                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                     ctypes.c_uint64(cnt), ptr(vb["tau"]))
+        assert ok != 1
+        for pk in (pk_s, pk_v, pk_s2):
+            o.orc_plonk_pk_free(ctypes.c_void_p(pk))
+
+
 @pytest.mark.gpu
 class TestBundleLinkGpu:
     """GPU end-to-end production bundle: settlement + both validity proofs on
