@@ -1494,6 +1494,158 @@ inline void io_validity_apply_constraints(PlonkCircuit& cs, const IoValidityWitn
     cs.enforce_equal(nul, p_null);
 }
 
+// ================== Bounded Settlements ==================
+// (settlement/intent_only_bounded_settlement.rs +
+//  intent_and_balance_bounded_settlement.rs — the trade size is chosen at
+//  runtime by an external party within [min, max]; constraints bound the
+//  WORST CASE: max amount in, price lower bound, max output overflow.)
+
+constexpr int PRICE_BITS = FP_PRECISION + 64;  // circuit-types/src/lib.rs:64
+
+struct BoundedMatchResult {  // bounded_match_result.rs:41-58 (6 scalars)
+    Fr internal_party_input_token, internal_party_output_token;
+    Fr min_internal_party_amount_in, max_internal_party_amount_in;
+    Fr price_repr;
+    Fr block_deadline;
+    std::vector<Fr> to_scalars() const {
+        return {internal_party_input_token, internal_party_output_token,
+                min_internal_party_amount_in, max_internal_party_amount_in,
+                price_repr, block_deadline};
+    }
+};
+
+struct IoBoundedStatement {  // intent_only_bounded_settlement.rs:88-114 (9)
+    BoundedMatchResult bmr;
+    Fr internal_relayer_fee_repr, external_relayer_fee_repr, relayer_fee_recipient;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = bmr.to_scalars();
+        v.push_back(internal_relayer_fee_repr);
+        v.push_back(external_relayer_fee_repr);
+        v.push_back(relayer_fee_recipient);
+        return v;
+    }
+};
+
+struct IbBoundedStatement {  // intent_and_balance_bounded_settlement.rs:141-182
+    BoundedMatchResult bmr;   // (16 scalars)
+    Fr amount_public_share;
+    PostMatchShare in_shares, out_shares;
+    Fr internal_relayer_fee_repr, external_relayer_fee_repr, relayer_fee_recipient;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = bmr.to_scalars();
+        v.push_back(amount_public_share);
+        auto a = in_shares.to_scalars();
+        v.insert(v.end(), a.begin(), a.end());
+        auto b = out_shares.to_scalars();
+        v.insert(v.end(), b.begin(), b.end());
+        v.push_back(internal_relayer_fee_repr);
+        v.push_back(external_relayer_fee_repr);
+        v.push_back(relayer_fee_recipient);
+        return v;
+    }
+};
+
+// BoundedSettlementGadget::verify_intent_constraints (settlement_lib.rs:224-261)
+inline void bounded_intent_constraints(PlonkCircuit& cs,
+                                       const std::array<Var, 5>& intent_v,
+                                       const std::array<Var, 6>& bmr) {
+    cs.enforce_equal(bmr[0], intent_v[0]);
+    cs.enforce_equal(bmr[1], intent_v[1]);
+    gte_gadget(cs, intent_v[4], bmr[3], AMOUNT_BITS);  // amount_in >= max_in
+    gte_gadget(cs, bmr[4], intent_v[3], PRICE_BITS);   // price >= min_price
+}
+
+// intent_only_bounded_settlement.rs:52-67 (group inherits the intent-only
+// public settlement placement)
+inline void io_bounded_apply_constraints(PlonkCircuit& cs, const Intent& w,
+                                         const IoBoundedStatement& st, int alignment,
+                                         int64_t offset) {
+    cs.create_link_group("intent_only_settlement", alignment, offset);
+    std::array<Var, 5> intent_v;
+    auto iv = w.to_scalars();
+    for (int k = 0; k < 5; ++k) {
+        intent_v[k] = cs.create_variable(iv[k]);
+        cs.add_to_link_group(intent_v[k], "intent_only_settlement");
+    }
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    std::array<Var, 6> bmr{pub[0], pub[1], pub[2], pub[3], pub[4], pub[5]};
+    bounded_intent_constraints(cs, intent_v, bmr);
+}
+
+// intent_and_balance_bounded_settlement.rs:45-86 (groups inherit the private
+// settlement party-0 placement)
+inline void ib_bounded_apply_constraints(PlonkCircuit& cs, const SettlementParty& w,
+                                         const IbBoundedStatement& st, int alignment,
+                                         int64_t pg_offset, int64_t og_offset) {
+    const char* pg = "intent_and_balance_settlement_party0";
+    const char* og = "output_balance_settlement_party0";
+    cs.create_link_group(pg, alignment, pg_offset);
+    cs.create_link_group(og, alignment, og_offset);
+    auto alloc_list = [&](const std::vector<Fr>& vals, const char* group) {
+        std::vector<Var> out;
+        for (auto& v : vals) {
+            Var x = cs.create_variable(v);
+            if (group) cs.add_to_link_group(x, group);
+            out.push_back(x);
+        }
+        return out;
+    };
+    auto intent_l = alloc_list(w.intent.to_scalars(), pg);
+    auto pre_amt = alloc_list({w.pre_amount_share}, pg);
+    auto in_bal = alloc_list(w.input_balance.to_scalars(), pg);
+    auto pre_in = alloc_list(w.pre_in_shares.to_scalars(), pg);
+    auto out_bal = alloc_list(w.output_balance.to_scalars(), og);
+    auto pre_out = alloc_list(w.pre_out_shares.to_scalars(), og);
+    std::array<Var, 5> intent_v;
+    std::copy(intent_l.begin(), intent_l.end(), intent_v.begin());
+
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    std::array<Var, 6> bmr{pub[0], pub[1], pub[2], pub[3], pub[4], pub[5]};
+    Var p_amt_share = pub[6];
+    std::array<Var, 3> p_in{pub[7], pub[8], pub[9]};
+    std::array<Var, 3> p_out{pub[10], pub[11], pub[12]};
+    Var p_fee_recipient = pub[15];
+
+    // 1. bounded match-result constraints (settlement_lib.rs:208-320)
+    bounded_intent_constraints(cs, intent_v, bmr);
+    gte_gadget(cs, in_bal[7], bmr[3], AMOUNT_BITS);  // balance >= max_in
+    cs.enforce_equal(out_bal[0], bmr[1]);            // mint == output token
+    cs.enforce_equal(out_bal[1], intent_v[2]);       // owner
+    Var max_out_fp = cs.mul(bmr[4], bmr[3]);         // price * max_in
+    Var max_out = fp_floor_gadget(cs, max_out_fp);
+    Var max_bal_amount = cs.add(out_bal[7], max_out);
+    cs.enforce_in_range(max_bal_amount, AMOUNT_BITS);
+    // 2. leaked pre-update shares (:61-78)
+    cs.enforce_equal(pre_amt[0], p_amt_share);
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(pre_in[k], p_in[k]);
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(pre_out[k], p_out[k]);
+    // 3. relayer fee recipient leak (:80-85)
+    cs.enforce_equal(out_bal[2], p_fee_recipient);
+}
+
+// consistent bounded statements from the bundles
+inline void io_bounded_statement_build(uint64_t seed, const IoValidityWitness& vw,
+                                       const IoSettlementStatement& ss,
+                                       IoBoundedStatement& st) {
+    Lcg rng(seed ^ 0xB0DEDB0DEDB0DEDull);
+    u64 max_l[4], min_l[4];
+    ss.obligation.amount_in.to_canonical(max_l);
+    st.bmr.internal_party_input_token = vw.intent.in_token;
+    st.bmr.internal_party_output_token = vw.intent.out_token;
+    st.bmr.max_internal_party_amount_in = ss.obligation.amount_in;
+    min_l[0] = max_l[0] / 2 + 1;
+    st.bmr.min_internal_party_amount_in = Fr::from_u64(min_l[0]);
+    st.bmr.price_repr = vw.intent.min_price_repr.dbl();  // >= min_price
+    st.bmr.block_deadline = Fr::from_u64(rng.next() & 0xFFFFFFFF);
+    st.internal_relayer_fee_repr = ss.relayer_fee_repr;
+    st.external_relayer_fee_repr = ss.relayer_fee_repr;
+    st.relayer_fee_recipient = ss.relayer_fee_recipient;
+}
+
 // ================== Valid Order Cancellation ==================
 // (zk_circuits/valid_order_cancellation.rs — prove the intent exists and
 //  spend its nullifier; the owner is leaked for contract authorization.)
@@ -1725,6 +1877,28 @@ inline void validity_bundle_build(uint64_t seed, ValidityBundle& b) {
     }
     // recompute the statement fields that depend on the pre-update shares
     settlement_update_statement(b.sw, b.sst);
+}
+
+// bounded statement for bundle party 0 (worst-case bounds consistent with
+// the bundle's obligation: max_in = obligation.amount_in, price = the trade
+// price = 2 * intent.min_price)
+inline void ib_bounded_statement_from_bundle(const ValidityBundle& b,
+                                             IbBoundedStatement& st) {
+    const auto& p = b.sw.p[0];
+    st.bmr.internal_party_input_token = p.intent.in_token;
+    st.bmr.internal_party_output_token = p.intent.out_token;
+    st.bmr.max_internal_party_amount_in = p.obligation.amount_in;
+    u64 l[4];
+    p.obligation.amount_in.to_canonical(l);
+    st.bmr.min_internal_party_amount_in = Fr::from_u64(l[0] / 2 + 1);
+    st.bmr.price_repr = p.intent.min_price_repr.dbl();
+    st.bmr.block_deadline = Fr::from_u64(123456);
+    st.amount_public_share = p.pre_amount_share;
+    st.in_shares = p.pre_in_shares;
+    st.out_shares = p.pre_out_shares;
+    st.internal_relayer_fee_repr = b.sst.relayer_fee_repr[0];
+    st.external_relayer_fee_repr = b.sst.relayer_fee_repr[1];
+    st.relayer_fee_recipient = p.output_balance.relayer_fee_recipient;
 }
 
 // statement for bundle party 0 (test_helpers :263-303 semantics)

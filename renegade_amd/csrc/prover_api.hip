@@ -1366,6 +1366,78 @@ void* rng_circ_build_io_validity(uint64_t seed) {
     }
 }
 
+// INTENT ONLY BOUNDED SETTLEMENT circuit (intent_only_bounded_settlement.rs)
+void* rng_circ_build_io_bounded_settlement(uint64_t seed) {
+    try {
+        IoValidityWitness vw;
+        IoValidityStatement vs;
+        IoSettlementStatement ss;
+        io_bundle_build(seed, vw, vs, ss);
+        uint64_t align = 0;
+        int64_t off = 0;
+        {
+            PlonkCircuit scs;
+            io_settlement_apply_constraints(scs, vw.intent, ss);
+            CircuitTables stt = scs.finalize();
+            for (auto& g : stt.link_groups)
+                if (g.id == "intent_only_settlement") {
+                    align = g.alignment;
+                    off = (int64_t)g.offset;
+                }
+        }
+        IoBoundedStatement st;
+        io_bounded_statement_build(seed, vw, ss, st);
+        PlonkCircuit cs;
+        io_bounded_apply_constraints(cs, vw.intent, st, (int)align, off);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_io_bounded_settlement: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_io_bounded_settlement: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// INTENT AND BALANCE BOUNDED SETTLEMENT circuit
+// (intent_and_balance_bounded_settlement.rs; party-0 groups at the private
+//  settlement layout)
+void* rng_circ_build_ib_bounded_settlement(uint64_t seed) {
+    try {
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        uint64_t align = 0;
+        int64_t pg_off = 0, og_off = 0;
+        {
+            PlonkCircuit scs;
+            settlement_apply_constraints(scs, b.sw, b.sst);
+            CircuitTables stt = scs.finalize();
+            for (auto& g : stt.link_groups) {
+                if (g.id == "intent_and_balance_settlement_party0") {
+                    align = g.alignment;
+                    pg_off = (int64_t)g.offset;
+                }
+                if (g.id == "output_balance_settlement_party0") og_off = (int64_t)g.offset;
+            }
+        }
+        IbBoundedStatement st;
+        ib_bounded_statement_from_bundle(b, st);
+        PlonkCircuit cs;
+        ib_bounded_apply_constraints(cs, b.sw.p[0], st, (int)align, pg_off, og_off);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_ib_bounded_settlement: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_ib_bounded_settlement: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
 void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
     try {

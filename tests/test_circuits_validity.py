@@ -248,6 +248,29 @@ class TestBundleLink:
                                          ptr(vb["tau"]))
             assert ok == 1, f"{key} <-> public settlement link failed"
         o.orc_plonk_pk_free(ctypes.c_void_p(pk_p))
+
+        # --- bounded settlement: same validity proofs link into the
+        #     bounded variant at the party-0 layouts ---
+        lib.rng_circ_build_ib_bounded_settlement.restype = ctypes.c_void_p
+        lib.rng_circ_build_ib_bounded_settlement.argtypes = [ctypes.c_uint64]
+        tbb = vb["fetch"](lib.rng_circ_build_ib_bounded_settlement(7))
+        assert tbb["npub"] == 16
+        pk_bb = vb["setup"](tbb)
+        _, hbb = vb["prove"](pk_bb, tbb, 9)
+        hbb_e = ext(hbb, int(tbb["n"]))
+        for key, hv_e in legs[:1] + legs[2:3]:  # (party,0) and (out,0)
+            a, off, cnt = groups[key]
+            lp = np.zeros(18, dtype=np.uint64)
+            assert o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hbb_e),
+                                    ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                    ctypes.c_uint64(cnt), ptr(lp)) == 0
+            ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big),
+                                         ptr(hv_e[-9:].copy()), ptr(hbb_e[-9:].copy()),
+                                         ptr(lp), ctypes.c_uint64(a),
+                                         ctypes.c_uint64(off), ctypes.c_uint64(cnt),
+                                         ptr(vb["tau"]))
+            assert ok == 1, f"{key} <-> bounded settlement link failed"
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_bb))
         for pk in (pk_s, pk_v, pk_u):
             o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
@@ -296,6 +319,28 @@ class TestIntentOnly:
                                      ctypes.c_uint64(a), ctypes.c_uint64(off),
                                      ctypes.c_uint64(cnt), ptr(vb["tau"]))
         assert ok == 1, "intent-only validity <-> settlement link failed"
+
+        # bounded settlement (intent_only_bounded_settlement.rs): same intent
+        # links into the bounded variant at the same inherited layout
+        lib.rng_circ_build_io_bounded_settlement.restype = ctypes.c_void_p
+        lib.rng_circ_build_io_bounded_settlement.argtypes = [ctypes.c_uint64]
+        tb = vb["fetch"](lib.rng_circ_build_io_bounded_settlement(5))
+        assert tb["npub"] == 9
+        pk_b = vb["setup"](tb)
+        pb, hb = vb["prove"](pk_b, tb, 9)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk_b), ptr(tb["pubs"]), ptr(pb),
+                                  ptr(vb["tau"])) == 1
+        hb_e = ext(hb, int(tb["n"]))
+        lp2 = np.zeros(18, dtype=np.uint64)
+        assert o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hv_e), ptr(hb_e),
+                                ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                ctypes.c_uint64(cnt), ptr(lp2)) == 0
+        ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big), ptr(hv_e[-9:].copy()),
+                                     ptr(hb_e[-9:].copy()), ptr(lp2),
+                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                     ctypes.c_uint64(cnt), ptr(vb["tau"]))
+        assert ok == 1, "intent-only validity <-> bounded settlement link failed"
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_b))
 
         # a settlement over a DIFFERENT intent must not link
         ts2 = vb["fetch"](lib.rng_circ_build_io_settlement(6))

@@ -146,7 +146,9 @@ class TestGpuProver:
                                           ("rng_circ_build_valid_order_cancellation", 42),
                                           ("rng_circ_build_public_settlement", 42),
                                           ("rng_circ_build_io_settlement", 42),
-                                          ("rng_circ_build_io_validity", 42)])
+                                          ("rng_circ_build_io_validity", 42),
+                                          ("rng_circ_build_io_bounded_settlement", 42),
+                                          ("rng_circ_build_ib_bounded_settlement", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
