@@ -2152,4 +2152,220 @@ inline void validity_apply_constraints(PlonkCircuit& cs, const ValidityWitness& 
     cs.enforce_equal(intent_v[2], bal_v[1]);  // owner == owner
 }
 
+// ================== Fee payment circuits (zk_circuits/fees/) ==================
+// Notes are plaintext 4-tuples committed with Poseidon2; fee payments rotate
+// the payer's balance with the fee field re-encrypted to zero.
+
+struct Note {  // darkpool-types/src/note.rs:47-59 (4 scalars)
+    Fr mint, amount, receiver, blinder;
+    std::vector<Fr> to_scalars() const { return {mint, amount, receiver, blinder}; }
+};
+// note.rs:77-87: commitment = H(fields); nullifier = H(commitment, blinder)
+inline Fr native_note_commitment(const Note& n) {
+    Fr in[4] = {n.mint, n.amount, n.receiver, n.blinder};
+    return poseidon_hash(in, 4);
+}
+inline Fr native_note_nullifier(const Fr& comm, const Fr& blinder) {
+    Fr in[2] = {comm, blinder};
+    return poseidon_hash(in, 2);
+}
+
+// ---- VALID NOTE REDEMPTION (fees/valid_note_redemption.rs) ----
+struct NoteRedemptionStatement {  // :88-97 (6 scalars)
+    Note note;
+    Fr note_root, note_nullifier;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = note.to_scalars();
+        v.push_back(note_root);
+        v.push_back(note_nullifier);
+        return v;
+    }
+};
+struct NoteRedemptionWitness {  // :73-77
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_idx[MERKLE_HEIGHT];
+};
+
+inline void note_redemption_build(uint64_t seed, NoteRedemptionWitness& w,
+                                  NoteRedemptionStatement& st) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    st.note = {addr(), Fr::from_u64(rng.next() & ((1ull << 50) - 1)), addr(), rng.fr()};
+    for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+        w.opening_elems[k] = rng.fr();
+        w.opening_idx[k] = rng.next() & 1;
+    }
+    Fr comm = native_note_commitment(st.note);
+    st.note_root = native_merkle_root(
+        comm, std::vector<Fr>(w.opening_elems, w.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(w.opening_idx, w.opening_idx + MERKLE_HEIGHT));
+    st.note_nullifier = native_note_nullifier(comm, st.note.blinder);
+}
+
+inline void note_redemption_apply_constraints(PlonkCircuit& cs,
+                                              const NoteRedemptionWitness& w,
+                                              const NoteRedemptionStatement& st) {
+    std::vector<Var> op_elems, op_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_elems.push_back(cs.create_variable(w.opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_idx[k] ? Fr::one() : Fr::zero()));
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    // NoteGadget::compute_note_commitment (note hash over the 4 fields)
+    PoseidonHashGadget h(cs);
+    Var comm = h.hash(cs, {pub[0], pub[1], pub[2], pub[3]});
+    Var root = merkle_root_gadget(cs, comm, op_elems, op_idx);
+    cs.enforce_equal(root, pub[4]);
+    PoseidonHashGadget h2(cs);
+    Var nul = h2.hash(cs, {comm, pub[3]});
+    cs.enforce_equal(nul, pub[5]);
+}
+
+// ---- VALID {PUBLIC, PRIVATE} x {RELAYER, PROTOCOL} FEE PAYMENT ----
+// (fees/valid_public_relayer_fee_payment.rs etc.) — shared machinery: the
+// payer balance rotates with fee field `field` (5 = relayer, 6 = protocol)
+// re-encrypted to zero; public variants carry the full plaintext note in the
+// statement, the private relayer variant only the receiver + note commitment
+// (encryption verified out of circuit).
+
+struct FeePaymentStatement {  // balance-rotation half (5 scalars), shared
+    Fr merkle_root, old_balance_nullifier, new_balance_commitment, recovery_id;
+    Fr new_fee_balance_share;
+};
+
+// builder: balance with nonzero fee balance at `field`, full rotation natives
+inline void fee_payment_build(uint64_t seed, int field, VdWitness& w,
+                              FeePaymentStatement& st, Note& note) {
+    Lcg rng(seed);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    w.old_balance.inner = {addr(), addr(), addr(), rng.fr(), rng.fr(),
+                           Fr::from_u64((rng.next() & 0xFFFF) + 1),
+                           Fr::from_u64((rng.next() & 0xFFFF) + 1),
+                           Fr::from_u64(rng.next() & ((1ull << 50) - 1))};
+    w.old_balance.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+    w.old_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int i = 0; i < 8; ++i) w.old_balance.public_share[i] = rng.fr();
+    for (int i = 0; i < MERKLE_HEIGHT; ++i) {
+        w.opening_elems[i] = rng.fr();
+        w.opening_indices[i] = rng.next() & 1;
+    }
+    auto iv = w.old_balance.inner.to_scalars();
+    std::vector<Fr> old_priv;
+    for (int i = 0; i < 8; ++i)
+        old_priv.push_back(iv[i].sub(w.old_balance.public_share[i]));
+    Fr old_comm = native_commitment(
+        old_priv, w.old_balance.recovery, w.old_balance.share,
+        std::vector<Fr>(w.old_balance.public_share, w.old_balance.public_share + 8));
+    st.merkle_root = native_merkle_root(
+        old_comm, std::vector<Fr>(w.opening_elems, w.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(w.opening_indices, w.opening_indices + MERKLE_HEIGHT));
+    st.old_balance_nullifier = native_nullifier(w.old_balance.recovery);
+    // the note pays out the full fee balance to the recipient on the balance
+    note.mint = w.old_balance.inner.mint;
+    note.amount = iv[field];
+    note.receiver = w.old_balance.inner.relayer_fee_recipient;
+    note.blinder = rng.fr();
+    // rotate with fee field := 0, re-encrypted
+    StateBalance nb = w.old_balance;
+    auto nv = nb.inner.to_scalars();
+    Fr pad = nb.share.next();
+    st.new_fee_balance_share = Fr::zero().sub(pad);
+    nb.public_share[field] = st.new_fee_balance_share;
+    std::vector<Fr> new_priv = old_priv;
+    new_priv[field] = pad;
+    (void)nv;
+    if (field == 5) nb.inner.relayer_fee_balance = Fr::zero();
+    else nb.inner.protocol_fee_balance = Fr::zero();
+    st.recovery_id = nb.recovery.next();
+    auto nv2 = nb.inner.to_scalars();
+    std::vector<Fr> new_pub(nb.public_share, nb.public_share + 8);
+    st.new_balance_commitment = native_commitment(new_priv, nb.recovery, nb.share, new_pub);
+}
+
+// shared constraint body; note_mode: 0 = public note (9 statement scalars:
+// rotation 5 + note 4), 1 = private note (7: rotation 5 + receiver +
+// note_commitment); check_receiver: constrain note.receiver (false only for
+// the public protocol variant, whose receiver the contract checks)
+inline void fee_payment_apply_constraints(PlonkCircuit& cs, const VdWitness& w,
+                                          const Fr& blinder, int field, int note_mode,
+                                          bool check_receiver,
+                                          const std::vector<Fr>& st_scalars) {
+    StateWrapperVars old_v;
+    old_v.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                      cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    old_v.share = {cs.create_variable(w.old_balance.share.seed),
+                   cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    auto iv = w.old_balance.inner.to_scalars();
+    for (auto& s : iv) old_v.inner.push_back(cs.create_variable(s));
+    for (int i = 0; i < 8; ++i)
+        old_v.public_share.push_back(cs.create_variable(w.old_balance.public_share[i]));
+    std::vector<Var> op_elems, op_idx;
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_elems.push_back(cs.create_variable(w.opening_elems[i]));
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_indices[i] ? Fr::one() : Fr::zero()));
+    Var blinder_v = (note_mode == 1) ? cs.create_variable(blinder) : cs.zero();
+
+    std::vector<Var> pub;
+    for (auto& s : st_scalars) pub.push_back(cs.create_public_variable(s));
+    Var p_root = pub[0], p_null = pub[1], p_comm = pub[2], p_rid = pub[3],
+        p_fee_share = pub[4];
+
+    // the fee balance must be nonzero (verify_note, :85-88)
+    Var z = is_zero_gadget(cs, old_v.inner[field]);
+    cs.enforce_false(z);
+    // note checks
+    if (note_mode == 0) {
+        Var n_mint = pub[5], n_amount = pub[6], n_receiver = pub[7];
+        cs.enforce_equal(n_mint, old_v.inner[0]);
+        cs.enforce_equal(n_amount, old_v.inner[field]);
+        if (check_receiver) cs.enforce_equal(n_receiver, old_v.inner[2]);
+    } else {
+        Var p_receiver = pub[5], p_note_comm = pub[6];
+        cs.enforce_equal(p_receiver, old_v.inner[2]);
+        PoseidonHashGadget h(cs);
+        Var ncomm = h.hash(cs, {old_v.inner[0], old_v.inner[field], old_v.inner[2],
+                                blinder_v});
+        cs.enforce_equal(ncomm, p_note_comm);
+    }
+
+    // complementary shares + post-payment balance (fee field := 0)
+    std::vector<Var> old_priv;
+    for (int i = 0; i < 8; ++i)
+        old_priv.push_back(cs.sub(old_v.inner[i], old_v.public_share[i]));
+    StateWrapperVars new_v = old_v;
+    std::vector<Var> new_priv = old_priv;
+    new_v.inner[field] = cs.zero();
+    std::vector<Var> pads, pubs_enc;
+    stream_cipher_encrypt(cs, {new_v.inner[field]}, new_v.share, pads, pubs_enc);
+    new_priv[field] = pads[0];
+    new_v.public_share[field] = pubs_enc[0];
+    cs.enforce_equal(pubs_enc[0], p_fee_share);
+
+    // full rotation (state_rotation.rs:83-122)
+    Var rid = csprng_next(cs, new_v.recovery);
+    cs.enforce_equal(rid, p_rid);
+    Var old_comm = commitment_gadget(cs, old_priv, old_v.recovery, old_v.share,
+                                     old_v.public_share);
+    Var new_comm = commitment_gadget(cs, new_priv, new_v.recovery, new_v.share,
+                                     new_v.public_share);
+    cs.enforce_equal(new_comm, p_comm);
+    Var root = merkle_root_gadget(cs, old_comm, op_elems, op_idx);
+    cs.enforce_equal(root, p_root);
+    Var nul = nullifier_gadget(cs, old_v);
+    cs.enforce_equal(nul, p_null);
+}
+
 }  // namespace rng

@@ -1438,6 +1438,75 @@ void* rng_circ_build_ib_bounded_settlement(uint64_t seed) {
     }
 }
 
+// ---- fee circuits (zk_circuits/fees/) ----
+
+// VALID NOTE REDEMPTION (fees/valid_note_redemption.rs)
+void* rng_circ_build_note_redemption(uint64_t seed) {
+    try {
+        NoteRedemptionWitness w;
+        NoteRedemptionStatement st;
+        note_redemption_build(seed, w, st);
+        PlonkCircuit cs;
+        note_redemption_apply_constraints(cs, w, st);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_note_redemption: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_note_redemption: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// shared fee-payment builder; variant: 0 = public relayer, 1 = public
+// protocol, 2 = private relayer (fees/valid_{public,private}_{relayer,
+// protocol}_fee_payment.rs; private protocol needs in-circuit ElGamal and
+// lands with the embedded-curve gadget)
+static void* build_fee_payment(uint64_t seed, int variant, const char* name) {
+    try {
+        VdWitness w;
+        FeePaymentStatement st;
+        Note note;
+        int field = (variant == 1) ? 6 : 5;
+        fee_payment_build(seed, field, w, st, note);
+        std::vector<Fr> ss = {st.merkle_root, st.old_balance_nullifier,
+                              st.new_balance_commitment, st.recovery_id,
+                              st.new_fee_balance_share};
+        int note_mode = (variant == 2) ? 1 : 0;
+        if (note_mode == 0) {
+            auto nv = note.to_scalars();
+            ss.insert(ss.end(), nv.begin(), nv.end());
+        } else {
+            ss.push_back(note.receiver);
+            ss.push_back(native_note_commitment(note));
+        }
+        PlonkCircuit cs;
+        fee_payment_apply_constraints(cs, w, note.blinder, field, note_mode,
+                                      /*check_receiver=*/variant != 1, ss);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "%s: %s\n", name, why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "%s: %s\n", name, e.what());
+        return nullptr;
+    }
+}
+
+void* rng_circ_build_fee_public_relayer(uint64_t seed) {
+    return build_fee_payment(seed, 0, "rng_circ_build_fee_public_relayer");
+}
+void* rng_circ_build_fee_public_protocol(uint64_t seed) {
+    return build_fee_payment(seed, 1, "rng_circ_build_fee_public_protocol");
+}
+void* rng_circ_build_fee_private_relayer(uint64_t seed) {
+    return build_fee_payment(seed, 2, "rng_circ_build_fee_private_relayer");
+}
+
 // VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
 void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
     try {

@@ -15,6 +15,10 @@ BUILDER_INFO = {
     "rng_circ_build_valid_deposit": (8, [3 * 4, 4 * 4, 5 * 4, 6 * 4]),
     "rng_circ_build_valid_withdrawal": (8, [3 * 4, 4 * 4, 5 * 4, 6 * 4]),
     "rng_circ_build_valid_order_cancellation": (3, [0, 1 * 4, 2 * 4]),
+    "rng_circ_build_note_redemption": (6, [4 * 4, 5 * 4]),
+    "rng_circ_build_fee_public_relayer": (9, [0, 1 * 4, 2 * 4, 3 * 4]),
+    "rng_circ_build_fee_public_protocol": (9, [0, 1 * 4, 2 * 4, 3 * 4]),
+    "rng_circ_build_fee_private_relayer": (7, [0, 1 * 4, 2 * 4, 3 * 4, 6 * 4]),
 }
 BUILDERS = list(BUILDER_INFO)
 

@@ -148,7 +148,11 @@ class TestGpuProver:
                                           ("rng_circ_build_io_settlement", 42),
                                           ("rng_circ_build_io_validity", 42),
                                           ("rng_circ_build_io_bounded_settlement", 42),
-                                          ("rng_circ_build_ib_bounded_settlement", 42)])
+                                          ("rng_circ_build_ib_bounded_settlement", 42),
+                                          ("rng_circ_build_note_redemption", 42),
+                                          ("rng_circ_build_fee_public_relayer", 42),
+                                          ("rng_circ_build_fee_public_protocol", 42),
+                                          ("rng_circ_build_fee_private_relayer", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
