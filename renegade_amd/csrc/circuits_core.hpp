@@ -1646,6 +1646,98 @@ inline void io_bounded_statement_build(uint64_t seed, const IoValidityWitness& v
     st.relayer_fee_recipient = ss.relayer_fee_recipient;
 }
 
+// ---- INTENT ONLY FIRST FILL VALIDITY (intent_only_first_fill.rs) ----
+// First fill: the intent is NOT yet in the tree; prove well-formedness and
+// commit to the private shares (the contract absorbs the public shares).
+
+struct IoffWitness {  // :105-123 (field order)
+    Intent intent;  // linked "intent_only_settlement"
+    Csprng share_stream, recovery_stream;
+    Fr private_shares[5];
+};
+struct IoffStatement {  // :132-149 (8 scalars)
+    Fr owner, intent_private_commitment, recovery_id;
+    Fr intent_public_share[5];
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = {owner, intent_private_commitment, recovery_id};
+        v.insert(v.end(), intent_public_share, intent_public_share + 5);
+        return v;
+    }
+};
+
+// uses the SAME intent as io_bundle_build(seed), so the first-fill proof
+// links against rng_circ_build_io_settlement(seed)'s proof
+inline void ioff_build(uint64_t seed, IoffWitness& w, IoffStatement& st) {
+    {
+        IoValidityWitness vw;
+        IoValidityStatement vs;
+        IoSettlementStatement ss;
+        io_bundle_build(seed, vw, vs, ss);
+        w.intent = vw.intent;
+    }
+    Lcg rng(seed ^ 0xF157F111F157F111ull);
+    w.share_stream = {rng.fr(), 0};
+    w.recovery_stream = {rng.fr(), 0};
+    for (int k = 0; k < 5; ++k) w.private_shares[k] = rng.fr();
+    st.owner = w.intent.owner;
+    auto iv = w.intent.to_scalars();
+    for (int k = 0; k < 5; ++k)
+        st.intent_public_share[k] = iv[k].sub(w.private_shares[k]);
+    Csprng rec = w.recovery_stream;
+    st.recovery_id = rec.next();
+    std::vector<Fr> in(w.private_shares, w.private_shares + 5);
+    in.push_back(rec.seed);
+    in.push_back(Fr::from_u64(rec.index));
+    in.push_back(w.share_stream.seed);
+    in.push_back(Fr::from_u64(w.share_stream.index));
+    st.intent_private_commitment = poseidon_hash(in.data(), in.size());
+}
+
+inline void ioff_apply_constraints(PlonkCircuit& cs, const IoffWitness& w,
+                                   const IoffStatement& st, int alignment,
+                                   int64_t offset) {
+    const char* g = "intent_only_settlement";
+    cs.create_link_group(g, alignment, offset);
+    // --- witness (field order) ---
+    std::array<Var, 5> intent_v;
+    auto iv = w.intent.to_scalars();
+    for (int k = 0; k < 5; ++k) {
+        intent_v[k] = cs.create_variable(iv[k]);
+        cs.add_to_link_group(intent_v[k], g);
+    }
+    CsprngVar share{cs.create_variable(w.share_stream.seed),
+                    cs.create_variable(Fr::from_u64(w.share_stream.index))};
+    CsprngVar rec{cs.create_variable(w.recovery_stream.seed),
+                  cs.create_variable(Fr::from_u64(w.recovery_stream.index))};
+    std::vector<Var> priv;
+    for (int k = 0; k < 5; ++k) priv.push_back(cs.create_variable(w.private_shares[k]));
+    // --- statement ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+
+    // build_and_validate_intent (:68-96)
+    cs.enforce_in_range(intent_v[4], AMOUNT_BITS);
+    cs.enforce_in_range(intent_v[3], PRICE_BITS);
+    cs.enforce_equal(intent_v[2], pub[0]);
+    for (int k = 0; k < 5; ++k) {
+        Var share_k = cs.sub(intent_v[k], priv[k]);
+        cs.enforce_equal(share_k, pub[3 + k]);
+    }
+    // recovery id over the wrapper's recovery stream (:56-58)
+    Var rid = csprng_next(cs, rec);
+    cs.enforce_equal(rid, pub[2]);
+    // private commitment with the ADVANCED recovery stream (:60-62)
+    PoseidonHashGadget h(cs);
+    std::vector<Var> in = priv;
+    in.push_back(rec.seed);
+    in.push_back(rec.index);
+    in.push_back(share.seed);
+    in.push_back(share.index);
+    Var pc = h.hash(cs, in);
+    cs.enforce_equal(pc, pub[1]);
+}
+
 // ================== Valid Order Cancellation ==================
 // (zk_circuits/valid_order_cancellation.rs — prove the intent exists and
 //  spend its nullifier; the owner is leaked for contract authorization.)

@@ -1438,6 +1438,43 @@ void* rng_circ_build_ib_bounded_settlement(uint64_t seed) {
     }
 }
 
+// INTENT ONLY FIRST FILL VALIDITY (validity_proofs/intent_only_first_fill.rs);
+// shares the seed's intent with rng_circ_build_io_settlement so the proofs link
+void* rng_circ_build_ioff(uint64_t seed) {
+    try {
+        IoffWitness w;
+        IoffStatement st;
+        ioff_build(seed, w, st);
+        uint64_t align = 0;
+        int64_t off = 0;
+        {
+            IoValidityWitness vw;
+            IoValidityStatement vs;
+            IoSettlementStatement ss;
+            io_bundle_build(seed, vw, vs, ss);
+            PlonkCircuit scs;
+            io_settlement_apply_constraints(scs, vw.intent, ss);
+            CircuitTables stt = scs.finalize();
+            for (auto& g : stt.link_groups)
+                if (g.id == "intent_only_settlement") {
+                    align = g.alignment;
+                    off = (int64_t)g.offset;
+                }
+        }
+        PlonkCircuit cs;
+        ioff_apply_constraints(cs, w, st, (int)align, off);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_ioff: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_ioff: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // ---- fee circuits (zk_circuits/fees/) ----
 
 // VALID NOTE REDEMPTION (fees/valid_note_redemption.rs)

@@ -342,6 +342,28 @@ class TestIntentOnly:
         assert ok == 1, "intent-only validity <-> bounded settlement link failed"
         o.orc_plonk_pk_free(ctypes.c_void_p(pk_b))
 
+        # first-fill validity (intent_only_first_fill.rs): same intent links
+        # into the settlement from a first-fill proof as well
+        lib.rng_circ_build_ioff.restype = ctypes.c_void_p
+        lib.rng_circ_build_ioff.argtypes = [ctypes.c_uint64]
+        tf = vb["fetch"](lib.rng_circ_build_ioff(5))
+        assert tf["npub"] == 8
+        pk_f = vb["setup"](tf)
+        pf, hf = vb["prove"](pk_f, tf, 10)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk_f), ptr(tf["pubs"]), ptr(pf),
+                                  ptr(vb["tau"])) == 1
+        hf_e = ext(hf, int(tf["n"]))
+        lp3 = np.zeros(18, dtype=np.uint64)
+        assert o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hf_e), ptr(hs_e),
+                                ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                ctypes.c_uint64(cnt), ptr(lp3)) == 0
+        ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big), ptr(hf_e[-9:].copy()),
+                                     ptr(hs_e[-9:].copy()), ptr(lp3),
+                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                     ctypes.c_uint64(cnt), ptr(vb["tau"]))
+        assert ok == 1, "first-fill validity <-> settlement link failed"
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_f))
+
         # a settlement over a DIFFERENT intent must not link
         ts2 = vb["fetch"](lib.rng_circ_build_io_settlement(6))
         pk_s2 = vb["setup"](ts2)

@@ -152,7 +152,8 @@ class TestGpuProver:
                                           ("rng_circ_build_note_redemption", 42),
                                           ("rng_circ_build_fee_public_relayer", 42),
                                           ("rng_circ_build_fee_public_protocol", 42),
-                                          ("rng_circ_build_fee_private_relayer", 42)])
+                                          ("rng_circ_build_fee_private_relayer", 42),
+                                          ("rng_circ_build_ioff", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
