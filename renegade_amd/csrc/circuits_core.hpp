@@ -16,6 +16,7 @@
 //    witness/statement builder mirroring test_helpers (:225-283).
 #pragma once
 #include <vector>
+#include "jubjub.hpp"
 #include "plonk_circuit.hpp"
 #include "poseidon2.hpp"
 #include "test_circuits.hpp"  // Lcg
@@ -477,6 +478,134 @@ inline Fr native_merkle_root(const Fr& leaf, const std::vector<Fr>& elems,
         cur = poseidon_hash(in, 2);
     }
     return cur;
+}
+
+// ================ embedded-curve gadgets (Baby Jubjub) ================
+// In-circuit twisted Edwards arithmetic over the embedded curve
+// (mpc-relation gadgets/ecc + jf-primitives circuit/signature/schnorr.rs,
+// circuit/elgamal.rs shapes; see jubjub.hpp header for the hash-choice
+// parity note).  Addition is COMPLETE (a QR, d non-QR — verified by
+// scripts/gen_babyjubjub_params.py), so identity needs no special casing.
+
+struct JjPointVars {
+    Var x, y;
+};
+
+inline JjPointVars jj_const_point_gadget(PlonkCircuit& cs, const JjPoint& p) {
+    Var x = cs.create_variable(p.x);
+    cs.enforce_constant(x, p.x);
+    Var y = cs.create_variable(p.y);
+    cs.enforce_constant(y, p.y);
+    return {x, y};
+}
+
+// constrain a*x^2 + y^2 == 1 + d*x^2*y^2
+inline void jj_on_curve_gadget(PlonkCircuit& cs, const JjPointVars& p) {
+    Var x2 = cs.mul(p.x, p.x);
+    Var y2 = cs.mul(p.y, p.y);
+    Var x2y2 = cs.mul(x2, y2);
+    std::array<Fr, NUM_SELECTORS> q{};
+    q[SEL_LC0] = jj_a();
+    q[SEL_LC1] = Fr::one();
+    q[SEL_LC2] = jj_d().neg();
+    q[SEL_C] = Fr::one().neg();
+    cs.insert_gate({x2, y2, x2y2, 0, 0}, q);
+}
+
+// complete addition: 5 mul gates + 2 rational-constraint gates
+inline JjPointVars jj_add_gadget(PlonkCircuit& cs, const JjPointVars& p,
+                                 const JjPointVars& q_) {
+    Var x1x2 = cs.mul(p.x, q_.x);
+    Var y1y2 = cs.mul(p.y, q_.y);
+    Var x1y2 = cs.mul(p.x, q_.y);
+    Var y1x2 = cs.mul(p.y, q_.x);
+    Var tau = cs.mul(x1x2, y1y2);
+    Fr t = jj_d().mul(cs.witness(tau));
+    Fr x3v = cs.witness(x1y2).add(cs.witness(y1x2)).mul(Fr::one().add(t).inverse());
+    Fr y3v = cs.witness(y1y2)
+                 .sub(jj_a().mul(cs.witness(x1x2)))
+                 .mul(Fr::one().sub(t).inverse());
+    Var x3 = cs.create_variable(x3v);
+    Var y3 = cs.create_variable(y3v);
+    {  // x3 + d*x3*tau - x1y2 - y1x2 = 0
+        std::array<Fr, NUM_SELECTORS> q{};
+        q[SEL_LC0] = Fr::one();
+        q[SEL_MUL0] = jj_d();
+        q[SEL_LC2] = Fr::one().neg();
+        q[SEL_LC3] = Fr::one().neg();
+        cs.insert_gate({x3, tau, x1y2, y1x2, 0}, q);
+    }
+    {  // y3 - d*y3*tau - y1y2 + a*x1x2 = 0
+        std::array<Fr, NUM_SELECTORS> q{};
+        q[SEL_LC0] = Fr::one();
+        q[SEL_MUL0] = jj_d().neg();
+        q[SEL_LC2] = Fr::one().neg();
+        q[SEL_LC3] = jj_a();
+        cs.insert_gate({y3, tau, y1y2, x1x2, 0}, q);
+    }
+    return {x3, y3};
+}
+
+inline JjPointVars jj_select_gadget(PlonkCircuit& cs, Var b, const JjPointVars& t,
+                                    const JjPointVars& f) {
+    return {cs.mux(b, t.x, f.x), cs.mux(b, t.y, f.y)};
+}
+
+// bits little-endian (boolean vars); double-and-add from the LSB
+inline JjPointVars jj_scalar_mul_gadget(PlonkCircuit& cs, const std::vector<Var>& bits,
+                                        JjPointVars base) {
+    JjPointVars acc{cs.zero(), cs.one()};
+    JjPointVars addend = base;
+    for (size_t i = 0; i < bits.size(); ++i) {
+        JjPointVars sum = jj_add_gadget(cs, acc, addend);
+        acc = jj_select_gadget(cs, bits[i], sum, acc);
+        if (i + 1 < bits.size()) addend = jj_add_gadget(cs, addend, addend);
+    }
+    return acc;
+}
+
+// Schnorr verification (schnorr.rs:43-54 shape): s*B == R + c*V with
+// c = low 248 bits of Poseidon2(vk || R || msg).  `s` is the plain scalar
+// witnessed as a field element (< 2^251).
+inline void schnorr_verify_gadget(PlonkCircuit& cs, const JjPointVars& vk,
+                                  const JjPointVars& R, Var s,
+                                  const std::vector<Var>& msg) {
+    jj_on_curve_gadget(cs, vk);
+    jj_on_curve_gadget(cs, R);
+    PoseidonHashGadget h(cs);
+    std::vector<Var> in = {vk.x, vk.y, R.x, R.y};
+    in.insert(in.end(), msg.begin(), msg.end());
+    Var c = h.hash(cs, in);
+    auto cbits = cs.to_bits(c, 254);
+    cbits.resize(248);  // challenge scalar = low 248 bits (< l)
+    auto sbits = cs.to_bits(s, JJ_ORDER_BITS);
+    JjPointVars B = jj_const_point_gadget(cs, jj_base());
+    JjPointVars sB = jj_scalar_mul_gadget(cs, sbits, B);
+    JjPointVars cV = jj_scalar_mul_gadget(cs, cbits, vk);
+    JjPointVars rhs = jj_add_gadget(cs, R, cV);
+    cs.enforce_equal(sB.x, rhs.x);
+    cs.enforce_equal(sB.y, rhs.y);
+}
+
+// ElGamal hybrid encryption (elgamal.rs shape): eph = k*B, shared = k*pk,
+// pad_i = Poseidon2(shared.x, shared.y, i), c_i = m_i + pad_i.  Returns the
+// ephemeral key and ciphertext vars; `k` witnessed as a field element.
+inline void elgamal_encrypt_gadget(PlonkCircuit& cs, const JjPointVars& pk, Var k,
+                                   const std::vector<Var>& msg, JjPointVars& out_eph,
+                                   std::vector<Var>& out_cipher) {
+    jj_on_curve_gadget(cs, pk);
+    auto kbits = cs.to_bits(k, JJ_ORDER_BITS);
+    JjPointVars B = jj_const_point_gadget(cs, jj_base());
+    out_eph = jj_scalar_mul_gadget(cs, kbits, B);
+    JjPointVars shared = jj_scalar_mul_gadget(cs, kbits, pk);
+    out_cipher.clear();
+    for (size_t i = 0; i < msg.size(); ++i) {
+        PoseidonHashGadget h(cs);
+        Var idx = cs.create_variable(Fr::from_u64((u64)i));
+        cs.enforce_constant(idx, Fr::from_u64((u64)i));
+        Var pad = h.hash(cs, {shared.x, shared.y, idx});
+        out_cipher.push_back(cs.add(msg[i], pad));
+    }
 }
 
 // ================== Intent And Balance Private Settlement ==================

@@ -1475,6 +1475,82 @@ void* rng_circ_build_ioff(uint64_t seed) {
     }
 }
 
+// embedded-curve gadget self-tests (Schnorr / ElGamal over Baby Jubjub):
+// native-sign -> in-circuit verify; returns the finalized tables or null if
+// the circuit is unsatisfied (tamper != 0 flips a signature/ciphertext bit
+// and MUST yield null)
+void* rng_testcirc_schnorr(uint64_t seed, int tamper) {
+    try {
+        Lcg rng(seed);
+        auto sc = [&]() {
+            JjScalar s{{rng.next() | (rng.next() << 52),
+                        rng.next() | (rng.next() << 52),
+                        rng.next() | (rng.next() << 52), rng.next() & 0x3FFFFFFFFFFull}};
+            return s;  // < 2^250 < l
+        };
+        JjScalar sk = sc(), k = sc();
+        Fr msg[3] = {rng.fr(), rng.fr(), rng.fr()};
+        JjPoint vk = jj_pubkey(sk);
+        JjSignature sig = jj_sign(sk, k, msg, 3);
+        if (!jj_verify(vk, sig, msg, 3)) {
+            fprintf(stderr, "rng_testcirc_schnorr: native verify failed\n");
+            return nullptr;
+        }
+        if (tamper) sig.s.v[0] ^= 1;
+        PlonkCircuit cs;
+        JjPointVars vkv{cs.create_variable(vk.x), cs.create_variable(vk.y)};
+        JjPointVars Rv{cs.create_variable(sig.R.x), cs.create_variable(sig.R.y)};
+        Fr s_fr = Fr::from_canonical(sig.s.v);
+        Var sv = cs.create_variable(s_fr);
+        std::vector<Var> mv;
+        for (int i = 0; i < 3; ++i) mv.push_back(cs.create_variable(msg[i]));
+        schnorr_verify_gadget(cs, vkv, Rv, sv, mv);
+        std::string why;
+        if (!cs.check_satisfied(&why)) return nullptr;
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_testcirc_schnorr: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+void* rng_testcirc_elgamal(uint64_t seed, int tamper) {
+    try {
+        Lcg rng(seed);
+        JjScalar dk{{rng.next() | (rng.next() << 52), rng.next() | (rng.next() << 52),
+                     rng.next() | (rng.next() << 52), rng.next() & 0x3FFFFFFFFFFull}};
+        JjScalar k{{rng.next() | (rng.next() << 52), rng.next() | (rng.next() << 52),
+                    rng.next() | (rng.next() << 52), rng.next() & 0x3FFFFFFFFFFull}};
+        JjPoint pk = jj_pubkey(dk);
+        Fr msg[3] = {rng.fr(), rng.fr(), rng.fr()};
+        JjCiphertext<3> ct = jj_elgamal_encrypt<3>(pk, k, msg);
+        if (tamper) ct.ciphertext[1] = ct.ciphertext[1].add(Fr::one());
+        PlonkCircuit cs;
+        JjPointVars pkv{cs.create_variable(pk.x), cs.create_variable(pk.y)};
+        Var kv = cs.create_variable(Fr::from_canonical(k.v));
+        std::vector<Var> mv;
+        for (int i = 0; i < 3; ++i) mv.push_back(cs.create_variable(msg[i]));
+        JjPointVars eph;
+        std::vector<Var> cipher;
+        elgamal_encrypt_gadget(cs, pkv, kv, mv, eph, cipher);
+        // constrain against the native ciphertext (the statement shape)
+        Var ex = cs.create_public_variable(ct.ephemeral_key.x);
+        Var ey = cs.create_public_variable(ct.ephemeral_key.y);
+        cs.enforce_equal(eph.x, ex);
+        cs.enforce_equal(eph.y, ey);
+        for (int i = 0; i < 3; ++i) {
+            Var ci = cs.create_public_variable(ct.ciphertext[i]);
+            cs.enforce_equal(cipher[i], ci);
+        }
+        std::string why;
+        if (!cs.check_satisfied(&why)) return nullptr;
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_testcirc_elgamal: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // ---- fee circuits (zk_circuits/fees/) ----
 
 // VALID NOTE REDEMPTION (fees/valid_note_redemption.rs)
