@@ -1960,11 +1960,29 @@ struct ValidityBundle {
     ValidityStatement vst[2];
     ObValidityWitness ow[2];
     ObValidityStatement ost[2];
+    // Schnorr secret keys whose public keys are the input balances'
+    // `authority` fields (needed by the first-fill circuits that bootstrap
+    // intent authorization from the balance)
+    JjScalar auth_sk[2];
 };
+
+inline JjScalar jj_random_scalar(Lcg& rng) {
+    JjScalar s{{rng.next() | (rng.next() << 52), rng.next() | (rng.next() << 52),
+                rng.next() | (rng.next() << 52), rng.next() & 0x3FFFFFFFFFFull}};
+    return s;  // < 2^250 < l
+}
 
 inline void validity_bundle_build(uint64_t seed, ValidityBundle& b) {
     settlement_build_witness_statement(seed, b.sw, b.sst);
     Lcg rng(seed ^ 0x9E3779B97F4A7C15ull);
+    // give the input balances REAL Schnorr authorities (first-fill circuits
+    // verify signatures under these keys)
+    for (int i = 0; i < 2; ++i) {
+        b.auth_sk[i] = jj_random_scalar(rng);
+        JjPoint vk = jj_pubkey(b.auth_sk[i]);
+        b.sw.p[i].input_balance.authority_x = vk.x;
+        b.sw.p[i].input_balance.authority_y = vk.y;
+    }
     auto priv_of = [](const std::vector<Fr>& inner, const Fr* pub_, size_t n) {
         std::vector<Fr> p;
         for (size_t i = 0; i < n; ++i) p.push_back(inner[i].sub(pub_[i]));
@@ -2373,6 +2391,440 @@ inline void validity_apply_constraints(PlonkCircuit& cs, const ValidityWitness& 
     cs.enforce_equal(intent_v[2], bal_v[1]);  // owner == owner
 }
 
+// ========== INTENT AND BALANCE FIRST FILL VALIDITY ==========
+// (validity_proofs/intent_and_balance_first_fill.rs — first fill: the intent
+//  is NOT yet in the tree; its authorization is bootstrapped from the
+//  balance's authority key via a Schnorr signature over the intent's
+//  commitment.)
+
+struct FfWitness {  // :285-329 (field order)
+    Intent intent;  // linked party0+party1
+    Csprng share_stream, recovery_stream;
+    Fr private_intent_shares[5];
+    Fr new_amount_public_share;  // linked
+    JjSignature sig;             // SchnorrSignature {s, R}
+    StateBalance old_balance;
+    Balance balance;                           // linked
+    PostMatchShare post_match_balance_shares;  // linked
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_idx[MERKLE_HEIGHT];
+};
+struct FfStatement {  // :341-368 (11 scalars)
+    Fr merkle_root;
+    Fr intent_public_share[4];  // PreMatchIntentShare (no amount_in)
+    Fr intent_private_share_commitment, intent_recovery_id;
+    Fr balance_partial_private, balance_partial_public;
+    Fr old_balance_nullifier, balance_recovery_id;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = {merkle_root};
+        v.insert(v.end(), intent_public_share, intent_public_share + 4);
+        v.push_back(intent_private_share_commitment);
+        v.push_back(intent_recovery_id);
+        v.push_back(balance_partial_private);
+        v.push_back(balance_partial_public);
+        v.push_back(old_balance_nullifier);
+        v.push_back(balance_recovery_id);
+        return v;
+    }
+};
+
+// First-fill witness for bundle party p: shares the intent, balance state and
+// re-encrypted shares with the bundle's intent_and_balance validity witness,
+// so the first-fill proof links into the same settlement proof.
+inline void ff_build(const ValidityBundle& b, int party, uint64_t seed, FfWitness& w,
+                     FfStatement& st) {
+    const ValidityWitness& v = b.vw[party];
+    const ValidityStatement& vs = b.vst[party];
+    Lcg rng(seed ^ 0xFF00FF00FF00FF0ull);
+    w.intent = b.sw.p[party].intent;
+    w.share_stream = {rng.fr(), rng.next() & 0xFFFF};
+    w.recovery_stream = {rng.fr(), rng.next() & 0xFFFF};
+    auto iv = w.intent.to_scalars();
+    for (int k = 0; k < 4; ++k) w.private_intent_shares[k] = rng.fr();
+    // arrange the fresh amount_in public share to equal the bundle's
+    // (the settlement's pre-update share)
+    w.new_amount_public_share = v.new_amount_public_share;
+    w.private_intent_shares[4] = iv[4].sub(w.new_amount_public_share);
+    w.old_balance = v.old_balance;
+    w.balance = v.balance;
+    w.post_match_balance_shares = v.post_match_balance_shares;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+        w.opening_elems[k] = v.balance_opening_elems[k];
+        w.opening_idx[k] = v.balance_opening_idx[k];
+    }
+    // public shares + commitments
+    Fr pub[5];
+    for (int k = 0; k < 5; ++k) pub[k] = iv[k].sub(w.private_intent_shares[k]);
+    std::vector<Fr> priv(w.private_intent_shares, w.private_intent_shares + 5);
+    auto priv_comm_at = [&](uint64_t rec_index) {
+        std::vector<Fr> in = priv;
+        in.push_back(w.recovery_stream.seed);
+        in.push_back(Fr::from_u64(rec_index));
+        in.push_back(w.share_stream.seed);
+        in.push_back(Fr::from_u64(w.share_stream.index));
+        return poseidon_hash(in.data(), in.size());
+    };
+    Fr old_pc = priv_comm_at(w.recovery_stream.index);
+    Fr new_pc = priv_comm_at(w.recovery_stream.index + 1);
+    Fr pub_comm = pub[0];
+    for (int k = 1; k < 5; ++k) {
+        Fr two[2] = {pub_comm, pub[k]};
+        pub_comm = poseidon_hash(two, 2);
+    }
+    Fr two[2] = {old_pc, pub_comm};
+    Fr full_comm = poseidon_hash(two, 2);
+    Csprng rec = w.recovery_stream;
+    st.intent_recovery_id = rec.next();
+    st.intent_private_share_commitment = new_pc;
+    for (int k = 0; k < 4; ++k) st.intent_public_share[k] = pub[k];
+    st.merkle_root = vs.balance_merkle_root;
+    st.balance_partial_private = vs.balance_partial_private;
+    st.balance_partial_public = vs.balance_partial_public;
+    st.old_balance_nullifier = vs.old_balance_nullifier;
+    st.balance_recovery_id = vs.balance_recovery_id;
+    // sign the ORIGINAL intent commitment with the balance's authority key
+    JjScalar k_nonce = jj_random_scalar(rng);
+    w.sig = jj_sign(b.auth_sk[party], k_nonce, &full_comm, 1);
+}
+
+inline void ff_apply_constraints(PlonkCircuit& cs, const FfWitness& w,
+                                 const FfStatement& st, int alignment,
+                                 int64_t party_offset0, int64_t party_offset1) {
+    const char* g0 = "intent_and_balance_settlement_party0";
+    const char* g1 = "intent_and_balance_settlement_party1";
+    cs.create_link_group(g0, alignment, party_offset0);
+    cs.create_link_group(g1, alignment, party_offset1);
+    auto link_both = [&](Var x) {
+        cs.add_to_link_group(x, g0);
+        cs.add_to_link_group(x, g1);
+    };
+
+    // --- witness allocation (struct field order) ---
+    std::array<Var, 5> intent_v;
+    {
+        auto iv = w.intent.to_scalars();
+        for (int k = 0; k < 5; ++k) {
+            intent_v[k] = cs.create_variable(iv[k]);
+            link_both(intent_v[k]);
+        }
+    }
+    CsprngVar share{cs.create_variable(w.share_stream.seed),
+                    cs.create_variable(Fr::from_u64(w.share_stream.index))};
+    CsprngVar rec{cs.create_variable(w.recovery_stream.seed),
+                  cs.create_variable(Fr::from_u64(w.recovery_stream.index))};
+    std::vector<Var> ipriv;
+    for (int k = 0; k < 5; ++k)
+        ipriv.push_back(cs.create_variable(w.private_intent_shares[k]));
+    Var new_amt = cs.create_variable(w.new_amount_public_share);
+    link_both(new_amt);
+    Var sig_s = cs.create_variable(Fr::from_canonical(w.sig.s.v));
+    JjPointVars sig_R{cs.create_variable(w.sig.R.x), cs.create_variable(w.sig.R.y)};
+    StateWrapperVars ob_;
+    ob_.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                    cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    ob_.share = {cs.create_variable(w.old_balance.share.seed),
+                 cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    for (auto& s : w.old_balance.inner.to_scalars())
+        ob_.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 8; ++k)
+        ob_.public_share.push_back(cs.create_variable(w.old_balance.public_share[k]));
+    std::array<Var, 8> bal_v;
+    {
+        auto bv = w.balance.to_scalars();
+        for (int k = 0; k < 8; ++k) {
+            bal_v[k] = cs.create_variable(bv[k]);
+            link_both(bal_v[k]);
+        }
+    }
+    std::array<Var, 3> pms_v;
+    {
+        auto pv = w.post_match_balance_shares.to_scalars();
+        for (int k = 0; k < 3; ++k) {
+            pms_v[k] = cs.create_variable(pv[k]);
+            link_both(pms_v[k]);
+        }
+    }
+    std::vector<Var> op_elems, op_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_elems.push_back(cs.create_variable(w.opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_idx[k] ? Fr::one() : Fr::zero()));
+
+    // --- statement ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var p_root = pub[0];
+    std::array<Var, 4> p_ipub{pub[1], pub[2], pub[3], pub[4]};
+    Var p_ipc = pub[5], p_irid = pub[6];
+    Var p_bpriv = pub[7], p_bpub = pub[8], p_bnull = pub[9], p_brid = pub[10];
+
+    // --- validate_balance (:218-238 + create_new_balance :240-277) ---
+    for (int k = 0; k < 8; ++k) cs.enforce_equal(bal_v[k], ob_.inner[k]);
+    std::vector<Var> old_priv_b;
+    for (int k = 0; k < 8; ++k)
+        old_priv_b.push_back(cs.sub(ob_.inner[k], ob_.public_share[k]));
+    CsprngVar nb_share = ob_.share;
+    std::vector<Var> bpads, bcipher;
+    stream_cipher_encrypt(cs, {ob_.inner[5], ob_.inner[6], ob_.inner[7]}, nb_share,
+                          bpads, bcipher);
+    std::vector<Var> new_priv_b = old_priv_b;
+    new_priv_b[5] = bpads[0];
+    new_priv_b[6] = bpads[1];
+    new_priv_b[7] = bpads[2];
+    std::vector<Var> new_pub_b = ob_.public_share;
+    new_pub_b[5] = bcipher[0];
+    new_pub_b[6] = bcipher[1];
+    new_pub_b[7] = bcipher[2];
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(bcipher[k], pms_v[k]);
+    CsprngVar nb_rec = ob_.recovery;
+    Var brid = csprng_next(cs, nb_rec);
+    cs.enforce_equal(brid, p_brid);
+    auto bpc = partial_commitment_gadget(cs, new_priv_b, nb_rec, nb_share, new_pub_b,
+                                         BALANCE_PARTIAL_COMMITMENT_SIZE);
+    cs.enforce_equal(bpc.first, p_bpriv);
+    cs.enforce_equal(bpc.second, p_bpub);
+    Var old_comm_b = commitment_gadget(cs, old_priv_b, ob_.recovery, ob_.share,
+                                       ob_.public_share);
+    Var broot = merkle_root_gadget(cs, old_comm_b, op_elems, op_idx);
+    cs.enforce_equal(broot, p_root);
+    Var bnull = nullifier_gadget(cs, ob_);
+    cs.enforce_equal(bnull, p_bnull);
+
+    // --- verify_intent_fields (:120-140) ---
+    cs.enforce_equal(intent_v[2], bal_v[1]);  // owner
+    cs.enforce_in_range(intent_v[3], PRICE_BITS);
+    cs.enforce_in_range(intent_v[4], AMOUNT_BITS);
+    cs.enforce_equal(bal_v[0], intent_v[0]);  // balance.mint == intent.in_token
+    // --- build_intent_state_wrapper (:142-169) ---
+    std::vector<Var> ipub;
+    for (int k = 0; k < 5; ++k) ipub.push_back(cs.sub(intent_v[k], ipriv[k]));
+    for (int k = 0; k < 4; ++k) cs.enforce_equal(ipub[k], p_ipub[k]);
+    cs.enforce_equal(ipub[4], new_amt);
+    // --- recovery id + commitments (:106-116, :171-203) ---
+    CsprngVar ni_rec = rec;
+    Var irid = csprng_next(cs, ni_rec);
+    cs.enforce_equal(irid, p_irid);
+    auto priv_comm_at = [&](const CsprngVar& r) {
+        PoseidonHashGadget h(cs);
+        std::vector<Var> in = ipriv;
+        in.push_back(r.seed);
+        in.push_back(r.index);
+        in.push_back(share.seed);
+        in.push_back(share.index);
+        return h.hash(cs, in);
+    };
+    Var old_pc = priv_comm_at(rec);
+    Var new_pc = priv_comm_at(ni_rec);
+    cs.enforce_equal(new_pc, p_ipc);
+    Var pub_comm = resumable_commitment(cs, ipub);
+    PoseidonHashGadget hf(cs);
+    Var full_comm = hf.hash(cs, {old_pc, pub_comm});
+    // --- Schnorr authorization by the balance's authority (:79-86) ---
+    JjPointVars vk{bal_v[3], bal_v[4]};
+    schnorr_verify_gadget(cs, vk, sig_R, sig_s, {full_comm});
+}
+
+// ========== NEW OUTPUT BALANCE VALIDITY ==========
+// (validity_proofs/new_output_balance.rs — create a fresh output balance
+//  (zero amount/fees), authorized by a Schnorr signature from an EXISTING
+//  balance's authority; links at the settlement's output-balance layout.)
+
+struct NobWitness {  // :246-269 (field order)
+    StateBalance new_balance;
+    Balance balance;                           // linked out0+out1
+    PostMatchShare post_match_balance_shares;  // linked
+    StateBalance existing_balance;
+    Fr opening_elems[MERKLE_HEIGHT];
+    bool opening_idx[MERKLE_HEIGHT];
+    JjSignature sig;
+};
+struct NobStatement {  // :281-299 (10 scalars)
+    Fr existing_merkle_root, existing_nullifier;
+    Fr pre_match_shares[5];  // mint, owner, relayer_fee_recipient, authority x/y
+    Fr partial_private, partial_public, recovery_id;
+    std::vector<Fr> to_scalars() const {
+        std::vector<Fr> v = {existing_merkle_root, existing_nullifier};
+        v.insert(v.end(), pre_match_shares, pre_match_shares + 5);
+        v.push_back(partial_private);
+        v.push_back(partial_public);
+        v.push_back(recovery_id);
+        return v;
+    }
+};
+
+inline void nob_build(uint64_t seed, NobWitness& w, NobStatement& st) {
+    Lcg rng(seed ^ 0x0B0B0B0B0B0B0B0Bull);
+    auto addr = [&]() {
+        u64 l[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                    rng.next() & 0xFFFFFFFF, 0};
+        return Fr::from_canonical(l);
+    };
+    JjScalar sk = jj_random_scalar(rng);
+    JjPoint vk = jj_pubkey(sk);
+    // existing balance (the authorization bootstrap)
+    w.existing_balance.inner = {addr(), addr(), addr(), vk.x, vk.y,
+                                Fr::from_u64(rng.next() & 0xFFFF),
+                                Fr::from_u64(rng.next() & 0xFFFF),
+                                Fr::from_u64(rng.next() & ((1ull << 50) - 1))};
+    w.existing_balance.recovery = {rng.fr(), (rng.next() & 0xFFFF) + 1};
+    w.existing_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int k = 0; k < 8; ++k) w.existing_balance.public_share[k] = rng.fr();
+    for (int k = 0; k < MERKLE_HEIGHT; ++k) {
+        w.opening_elems[k] = rng.fr();
+        w.opening_idx[k] = rng.next() & 1;
+    }
+    // fresh output balance: zero amount and fees, same owner/authority/rfr
+    w.new_balance.inner = {addr(), w.existing_balance.inner.owner,
+                           w.existing_balance.inner.relayer_fee_recipient, vk.x, vk.y,
+                           Fr::zero(), Fr::zero(), Fr::zero()};
+    w.new_balance.recovery = {rng.fr(), rng.next() & 0xFFFF};
+    w.new_balance.share = {rng.fr(), rng.next() & 0xFFFFFF};
+    for (int k = 0; k < 8; ++k) w.new_balance.public_share[k] = rng.fr();
+    w.balance = w.new_balance.inner;
+    w.post_match_balance_shares = {w.new_balance.public_share[5],
+                                   w.new_balance.public_share[6],
+                                   w.new_balance.public_share[7]};
+    // natives
+    auto priv_of = [](const Balance& inner, const Fr* pub_) {
+        std::vector<Fr> p;
+        auto iv = inner.to_scalars();
+        for (int i = 0; i < 8; ++i) p.push_back(iv[i].sub(pub_[i]));
+        return p;
+    };
+    std::vector<Fr> priv_n = priv_of(w.new_balance.inner, w.new_balance.public_share);
+    Fr full_comm, partial_pub_unused;
+    {
+        // full commitment at the PRE-advance recovery index
+        std::vector<Fr> pubs(w.new_balance.public_share, w.new_balance.public_share + 8);
+        full_comm = native_commitment(priv_n, w.new_balance.recovery, w.new_balance.share,
+                                      pubs);
+    }
+    StateBalance nb = w.new_balance;
+    st.recovery_id = nb.recovery.next();
+    native_partial_commitment(
+        priv_n, nb.recovery, nb.share,
+        std::vector<Fr>(nb.public_share, nb.public_share + 8),
+        BALANCE_PARTIAL_COMMITMENT_SIZE, st.partial_private, st.partial_public);
+    (void)partial_pub_unused;
+    for (int k = 0; k < 5; ++k) st.pre_match_shares[k] = w.new_balance.public_share[k];
+    std::vector<Fr> priv_e = priv_of(w.existing_balance.inner,
+                                     w.existing_balance.public_share);
+    Fr ecomm = native_commitment(
+        priv_e, w.existing_balance.recovery, w.existing_balance.share,
+        std::vector<Fr>(w.existing_balance.public_share,
+                        w.existing_balance.public_share + 8));
+    st.existing_merkle_root = native_merkle_root(
+        ecomm, std::vector<Fr>(w.opening_elems, w.opening_elems + MERKLE_HEIGHT),
+        std::vector<bool>(w.opening_idx, w.opening_idx + MERKLE_HEIGHT));
+    st.existing_nullifier = native_nullifier(w.existing_balance.recovery);
+    JjScalar k_nonce = jj_random_scalar(rng);
+    w.sig = jj_sign(sk, k_nonce, &full_comm, 1);
+}
+
+inline void nob_apply_constraints(PlonkCircuit& cs, const NobWitness& w,
+                                  const NobStatement& st, int alignment,
+                                  int64_t out_offset0, int64_t out_offset1) {
+    const char* g0 = "output_balance_settlement_party0";
+    const char* g1 = "output_balance_settlement_party1";
+    cs.create_link_group(g0, alignment, out_offset0);
+    cs.create_link_group(g1, alignment, out_offset1);
+    auto link_both = [&](Var x) {
+        cs.add_to_link_group(x, g0);
+        cs.add_to_link_group(x, g1);
+    };
+
+    // --- witness allocation (struct field order) ---
+    StateWrapperVars nb;
+    nb.recovery = {cs.create_variable(w.new_balance.recovery.seed),
+                   cs.create_variable(Fr::from_u64(w.new_balance.recovery.index))};
+    nb.share = {cs.create_variable(w.new_balance.share.seed),
+                cs.create_variable(Fr::from_u64(w.new_balance.share.index))};
+    for (auto& s : w.new_balance.inner.to_scalars())
+        nb.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 8; ++k)
+        nb.public_share.push_back(cs.create_variable(w.new_balance.public_share[k]));
+    std::array<Var, 8> bal_v;
+    {
+        auto bv = w.balance.to_scalars();
+        for (int k = 0; k < 8; ++k) {
+            bal_v[k] = cs.create_variable(bv[k]);
+            link_both(bal_v[k]);
+        }
+    }
+    std::array<Var, 3> pms_v;
+    {
+        auto pv = w.post_match_balance_shares.to_scalars();
+        for (int k = 0; k < 3; ++k) {
+            pms_v[k] = cs.create_variable(pv[k]);
+            link_both(pms_v[k]);
+        }
+    }
+    StateWrapperVars eb;
+    eb.recovery = {cs.create_variable(w.existing_balance.recovery.seed),
+                   cs.create_variable(Fr::from_u64(w.existing_balance.recovery.index))};
+    eb.share = {cs.create_variable(w.existing_balance.share.seed),
+                cs.create_variable(Fr::from_u64(w.existing_balance.share.index))};
+    for (auto& s : w.existing_balance.inner.to_scalars())
+        eb.inner.push_back(cs.create_variable(s));
+    for (int k = 0; k < 8; ++k)
+        eb.public_share.push_back(cs.create_variable(w.existing_balance.public_share[k]));
+    std::vector<Var> op_elems, op_idx;
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_elems.push_back(cs.create_variable(w.opening_elems[k]));
+    for (int k = 0; k < MERKLE_HEIGHT; ++k)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_idx[k] ? Fr::one() : Fr::zero()));
+    Var sig_s = cs.create_variable(Fr::from_canonical(w.sig.s.v));
+    JjPointVars sig_R{cs.create_variable(w.sig.R.x), cs.create_variable(w.sig.R.y)};
+
+    // --- statement ---
+    auto ss = st.to_scalars();
+    std::vector<Var> pub;
+    for (auto& s : ss) pub.push_back(cs.create_public_variable(s));
+    Var p_eroot = pub[0], p_enull = pub[1];
+    std::array<Var, 5> p_pre{pub[2], pub[3], pub[4], pub[5], pub[6]};
+    Var p_priv = pub[7], p_pub = pub[8], p_rid = pub[9];
+
+    // --- validate_new_balance (:144-166) ---
+    for (int k = 0; k < 8; ++k) cs.enforce_equal(bal_v[k], nb.inner[k]);
+    cs.enforce_equal(bal_v[7], cs.zero());  // amount
+    cs.enforce_equal(bal_v[5], cs.zero());  // relayer fee balance
+    cs.enforce_equal(bal_v[6], cs.zero());  // protocol fee balance
+    // --- validate_balance_shares (:116-141) ---
+    std::vector<Var> priv_n;
+    for (int k = 0; k < 8; ++k)
+        priv_n.push_back(cs.sub(nb.inner[k], nb.public_share[k]));
+    for (int k = 0; k < 5; ++k) cs.enforce_equal(nb.public_share[k], p_pre[k]);
+    for (int k = 0; k < 3; ++k) cs.enforce_equal(nb.public_share[5 + k], pms_v[k]);
+    // --- recovery id + shared-prefix commitments (:83-107) ---
+    CsprngVar nb_rec2 = nb.recovery;
+    Var rid = csprng_next(cs, nb_rec2);
+    cs.enforce_equal(rid, p_rid);
+    Var full_comm = commitment_gadget(cs, priv_n, nb.recovery, nb.share,
+                                      nb.public_share);
+    auto pc = partial_commitment_gadget(cs, priv_n, nb_rec2, nb.share, nb.public_share,
+                                        BALANCE_PARTIAL_COMMITMENT_SIZE);
+    cs.enforce_equal(pc.first, p_priv);
+    cs.enforce_equal(pc.second, p_pub);
+    // --- authorize_new_balance (:168-228) ---
+    std::vector<Var> priv_e;
+    for (int k = 0; k < 8; ++k)
+        priv_e.push_back(cs.sub(eb.inner[k], eb.public_share[k]));
+    Var ecomm = commitment_gadget(cs, priv_e, eb.recovery, eb.share, eb.public_share);
+    Var eroot = merkle_root_gadget(cs, ecomm, op_elems, op_idx);
+    cs.enforce_equal(eroot, p_eroot);
+    Var enull = nullifier_gadget(cs, eb);
+    cs.enforce_equal(enull, p_enull);
+    JjPointVars vk{eb.inner[3], eb.inner[4]};
+    schnorr_verify_gadget(cs, vk, sig_R, sig_s, {full_comm});
+    cs.enforce_equal(bal_v[1], eb.inner[1]);  // owner
+    cs.enforce_equal(bal_v[3], eb.inner[3]);  // authority
+    cs.enforce_equal(bal_v[4], eb.inner[4]);
+    cs.enforce_equal(bal_v[2], eb.inner[2]);  // relayer fee recipient
+}
+
 // ================== Fee payment circuits (zk_circuits/fees/) ==================
 // Notes are plaintext 4-tuples committed with Poseidon2; fee payments rotate
 // the payer's balance with the fee field re-encrypted to zero.
@@ -2576,6 +3028,84 @@ inline void fee_payment_apply_constraints(PlonkCircuit& cs, const VdWitness& w,
     cs.enforce_equal(pubs_enc[0], p_fee_share);
 
     // full rotation (state_rotation.rs:83-122)
+    Var rid = csprng_next(cs, new_v.recovery);
+    cs.enforce_equal(rid, p_rid);
+    Var old_comm = commitment_gadget(cs, old_priv, old_v.recovery, old_v.share,
+                                     old_v.public_share);
+    Var new_comm = commitment_gadget(cs, new_priv, new_v.recovery, new_v.share,
+                                     new_v.public_share);
+    cs.enforce_equal(new_comm, p_comm);
+    Var root = merkle_root_gadget(cs, old_comm, op_elems, op_idx);
+    cs.enforce_equal(root, p_root);
+    Var nul = nullifier_gadget(cs, old_v);
+    cs.enforce_equal(nul, p_null);
+}
+
+// ---- VALID PRIVATE PROTOCOL FEE PAYMENT (ElGamal note encryption) ----
+// (fees/valid_private_protocol_fee_payment.rs — the note is encrypted
+//  IN-circuit under the protocol key; statement = rotation half + receiver +
+//  note commitment + ciphertext (eph 2 + 3) + encryption key (2) = 14.)
+inline void fee_private_protocol_apply_constraints(PlonkCircuit& cs, const VdWitness& w,
+                                                   const Fr& blinder, const Fr& enc_k,
+                                                   const std::vector<Fr>& st_scalars) {
+    const int field = 6;  // protocol_fee_balance
+    StateWrapperVars old_v;
+    old_v.recovery = {cs.create_variable(w.old_balance.recovery.seed),
+                      cs.create_variable(Fr::from_u64(w.old_balance.recovery.index))};
+    old_v.share = {cs.create_variable(w.old_balance.share.seed),
+                   cs.create_variable(Fr::from_u64(w.old_balance.share.index))};
+    for (auto& s : w.old_balance.inner.to_scalars())
+        old_v.inner.push_back(cs.create_variable(s));
+    for (int i = 0; i < 8; ++i)
+        old_v.public_share.push_back(cs.create_variable(w.old_balance.public_share[i]));
+    std::vector<Var> op_elems, op_idx;
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_elems.push_back(cs.create_variable(w.opening_elems[i]));
+    for (int i = 0; i < MERKLE_HEIGHT; ++i)
+        op_idx.push_back(cs.create_boolean_variable(
+            w.opening_indices[i] ? Fr::one() : Fr::zero()));
+    Var blinder_v = cs.create_variable(blinder);
+    Var k_v = cs.create_variable(enc_k);
+
+    std::vector<Var> pub;
+    for (auto& s : st_scalars) pub.push_back(cs.create_public_variable(s));
+    Var p_root = pub[0], p_null = pub[1], p_comm = pub[2], p_rid = pub[3],
+        p_fee_share = pub[4], p_receiver = pub[5], p_note_comm = pub[6];
+    JjPointVars p_eph{pub[7], pub[8]};
+    Var p_c0 = pub[9], p_c1 = pub[10], p_c2 = pub[11];
+    JjPointVars p_pk{pub[12], pub[13]};
+
+    // fee balance nonzero (:84-88)
+    Var z = is_zero_gadget(cs, old_v.inner[field]);
+    cs.enforce_false(z);
+    // note = {mint, protocol_fee_balance, statement.receiver, blinder}
+    // encryption: plaintext = [mint, amount, blinder] (note.rs:91-94)
+    JjPointVars eph;
+    std::vector<Var> cipher;
+    elgamal_encrypt_gadget(cs, p_pk, k_v,
+                           {old_v.inner[0], old_v.inner[field], blinder_v}, eph, cipher);
+    cs.enforce_equal(eph.x, p_eph.x);
+    cs.enforce_equal(eph.y, p_eph.y);
+    cs.enforce_equal(cipher[0], p_c0);
+    cs.enforce_equal(cipher[1], p_c1);
+    cs.enforce_equal(cipher[2], p_c2);
+    // note commitment (H over the 4 fields)
+    PoseidonHashGadget hn(cs);
+    Var ncomm = hn.hash(cs, {old_v.inner[0], old_v.inner[field], p_receiver, blinder_v});
+    cs.enforce_equal(ncomm, p_note_comm);
+
+    // post-payment rotation (protocol fee := 0)
+    std::vector<Var> old_priv;
+    for (int i = 0; i < 8; ++i)
+        old_priv.push_back(cs.sub(old_v.inner[i], old_v.public_share[i]));
+    StateWrapperVars new_v = old_v;
+    std::vector<Var> new_priv = old_priv;
+    new_v.inner[field] = cs.zero();
+    std::vector<Var> pads, pubs_enc;
+    stream_cipher_encrypt(cs, {new_v.inner[field]}, new_v.share, pads, pubs_enc);
+    new_priv[field] = pads[0];
+    new_v.public_share[field] = pubs_enc[0];
+    cs.enforce_equal(pubs_enc[0], p_fee_share);
     Var rid = csprng_next(cs, new_v.recovery);
     cs.enforce_equal(rid, p_rid);
     Var old_comm = commitment_gadget(cs, old_priv, old_v.recovery, old_v.share,

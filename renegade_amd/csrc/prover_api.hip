@@ -1620,6 +1620,122 @@ void* rng_circ_build_fee_private_relayer(uint64_t seed) {
     return build_fee_payment(seed, 2, "rng_circ_build_fee_private_relayer");
 }
 
+// VALID PRIVATE PROTOCOL FEE PAYMENT (in-circuit ElGamal note encryption)
+void* rng_circ_build_fee_private_protocol(uint64_t seed) {
+    try {
+        VdWitness w;
+        FeePaymentStatement st;
+        Note note;
+        fee_payment_build(seed, 6, w, st, note);
+        Lcg rng(seed ^ 0xE161A3A1E161A3A1ull);
+        // protocol receiver comes from the statement, not the balance
+        u64 al[4] = {rng.next() | (rng.next() << 53), rng.next() | (rng.next() << 53),
+                     rng.next() & 0xFFFFFFFF, 0};
+        note.receiver = Fr::from_canonical(al);
+        JjScalar dk = jj_random_scalar(rng);
+        JjScalar k = jj_random_scalar(rng);
+        JjPoint pk = jj_pubkey(dk);
+        Fr plain[3] = {note.mint, note.amount, note.blinder};
+        JjCiphertext<3> ct = jj_elgamal_encrypt<3>(pk, k, plain);
+        std::vector<Fr> ss = {st.merkle_root, st.old_balance_nullifier,
+                              st.new_balance_commitment, st.recovery_id,
+                              st.new_fee_balance_share, note.receiver,
+                              native_note_commitment(note), ct.ephemeral_key.x,
+                              ct.ephemeral_key.y, ct.ciphertext[0], ct.ciphertext[1],
+                              ct.ciphertext[2], pk.x, pk.y};
+        PlonkCircuit cs;
+        fee_private_protocol_apply_constraints(cs, w, note.blinder,
+                                               Fr::from_canonical(k.v), ss);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_fee_private_protocol: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_fee_private_protocol: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// INTENT AND BALANCE FIRST FILL VALIDITY for bundle party 0/1
+// (validity_proofs/intent_and_balance_first_fill.rs; links at the private
+//  settlement's party layouts)
+void* rng_circ_build_ff_validity(uint64_t seed, uint64_t party) {
+    try {
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        uint64_t align = 0;
+        int64_t off[2] = {0, 0};
+        {
+            PlonkCircuit scs;
+            settlement_apply_constraints(scs, b.sw, b.sst);
+            CircuitTables stt = scs.finalize();
+            const char* names[2] = {"intent_and_balance_settlement_party0",
+                                    "intent_and_balance_settlement_party1"};
+            for (auto& g : stt.link_groups)
+                for (int i = 0; i < 2; ++i)
+                    if (g.id == names[i]) {
+                        align = g.alignment;
+                        off[i] = (int64_t)g.offset;
+                    }
+        }
+        int p = (int)(party & 1);
+        FfWitness w;
+        FfStatement st;
+        ff_build(b, p, seed, w, st);
+        PlonkCircuit cs;
+        ff_apply_constraints(cs, w, st, (int)align, off[0], off[1]);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_ff_validity: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_ff_validity: %s\n", e.what());
+        return nullptr;
+    }
+}
+
+// NEW OUTPUT BALANCE VALIDITY (validity_proofs/new_output_balance.rs; links
+// at the settlement's output-balance layouts)
+void* rng_circ_build_nob_validity(uint64_t seed) {
+    try {
+        NobWitness w;
+        NobStatement st;
+        nob_build(seed, w, st);
+        uint64_t align = 0;
+        int64_t off[2] = {0, 0};
+        {
+            ValidityBundle b;
+            validity_bundle_build(seed, b);
+            PlonkCircuit scs;
+            settlement_apply_constraints(scs, b.sw, b.sst);
+            CircuitTables stt = scs.finalize();
+            const char* names[2] = {"output_balance_settlement_party0",
+                                    "output_balance_settlement_party1"};
+            for (auto& g : stt.link_groups)
+                for (int i = 0; i < 2; ++i)
+                    if (g.id == names[i]) {
+                        align = g.alignment;
+                        off[i] = (int64_t)g.offset;
+                    }
+        }
+        PlonkCircuit cs;
+        nob_apply_constraints(cs, w, st, (int)align, off[0], off[1]);
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_build_nob_validity: %s\n", why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_build_nob_validity: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
 void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
     try {

@@ -381,6 +381,78 @@ class TestIntentOnly:
             o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
 
+class TestSchnorrCircuits:
+    """The three embedded-curve circuits (Schnorr / ElGamal): build +
+    satisfiability (the builders return null on an unsatisfied circuit), and
+    the first-fill proof linking into the settlement bundle."""
+
+    def test_builders_satisfied(self, vb):
+        lib = vb["lib"]
+        lib.rng_circ_build_fee_private_protocol.restype = ctypes.c_void_p
+        lib.rng_circ_build_fee_private_protocol.argtypes = [ctypes.c_uint64]
+        lib.rng_circ_build_nob_validity.restype = ctypes.c_void_p
+        lib.rng_circ_build_nob_validity.argtypes = [ctypes.c_uint64]
+        lib.rng_circ_build_ff_validity.restype = ctypes.c_void_p
+        lib.rng_circ_build_ff_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+        for h, what in [(lib.rng_circ_build_fee_private_protocol(3), "fee pp"),
+                        (lib.rng_circ_build_nob_validity(3), "nob"),
+                        (lib.rng_circ_build_ff_validity(3, 1), "ff p1")]:
+            assert h, f"{what} unsatisfied"
+            lib.rng_circ_free(h)
+
+    def test_schnorr_gadget_tamper(self, vb):
+        lib = vb["lib"]
+        lib.rng_testcirc_schnorr.restype = ctypes.c_void_p
+        lib.rng_testcirc_schnorr.argtypes = [ctypes.c_uint64, ctypes.c_int]
+        lib.rng_testcirc_elgamal.restype = ctypes.c_void_p
+        lib.rng_testcirc_elgamal.argtypes = [ctypes.c_uint64, ctypes.c_int]
+        h = lib.rng_testcirc_schnorr(9, 0)
+        assert h
+        lib.rng_circ_free(h)
+        assert not lib.rng_testcirc_schnorr(9, 1), "tampered signature accepted"
+        h = lib.rng_testcirc_elgamal(9, 0)
+        assert h
+        lib.rng_circ_free(h)
+        assert not lib.rng_testcirc_elgamal(9, 1), "tampered ciphertext accepted"
+
+    def test_ff_links_into_settlement(self, vb):
+        lib, o = vb["lib"], vb["o"]
+        lib.rng_circ_build_ff_validity.restype = ctypes.c_void_p
+        lib.rng_circ_build_ff_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+        ts = vb["fetch"](lib.rng_circ_build_settlement_bundle(7))
+        tf = vb["fetch"](lib.rng_circ_build_ff_validity(7, 0))
+        assert tf["npub"] == 11
+        pk_s = vb["setup"](ts)
+        pk_f = vb["setup"](tf)
+        _, hs = vb["prove"](pk_s, ts, 3)
+        pf, hf = vb["prove"](pk_f, tf, 4)
+        assert o.orc_plonk_verify(ctypes.c_void_p(pk_f), ptr(tf["pubs"]), ptr(pf),
+                                  ptr(vb["tau"])) == 1
+        n_big = max(int(ts["n"]), int(tf["n"]))
+        pk_big = pk_f if int(tf["n"]) == n_big else pk_s
+
+        def ext(h, n_small):
+            out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+            out[:4 * (n_small + 2)] = h[:4 * (n_small + 2)]
+            out[-9:] = h[-9:]
+            return out
+
+        groups = bundle_groups(ts["lg"])
+        a, off, cnt = groups[("party", 0)]
+        hf_e, hs_e = ext(hf, int(tf["n"])), ext(hs, int(ts["n"]))
+        lp = np.zeros(18, dtype=np.uint64)
+        assert o.orc_plonk_link(ctypes.c_void_p(pk_big), ptr(hf_e), ptr(hs_e),
+                                ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                ctypes.c_uint64(cnt), ptr(lp)) == 0
+        ok = o.orc_plonk_link_verify(ctypes.c_void_p(pk_big), ptr(hf_e[-9:].copy()),
+                                     ptr(hs_e[-9:].copy()), ptr(lp),
+                                     ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                     ctypes.c_uint64(cnt), ptr(vb["tau"]))
+        assert ok == 1, "first-fill validity <-> settlement link failed"
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_s))
+        o.orc_plonk_pk_free(ctypes.c_void_p(pk_f))
+
+
 @pytest.mark.gpu
 class TestBundleLinkGpu:
     """GPU end-to-end production bundle: settlement + both validity proofs on

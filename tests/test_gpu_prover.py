@@ -153,7 +153,10 @@ class TestGpuProver:
                                           ("rng_circ_build_fee_public_relayer", 42),
                                           ("rng_circ_build_fee_public_protocol", 42),
                                           ("rng_circ_build_fee_private_relayer", 42),
-                                          ("rng_circ_build_ioff", 42)])
+                                          ("rng_circ_build_ioff", 42),
+                                          ("rng_circ_build_ff_validity", 42),
+                                          ("rng_circ_build_nob_validity", 42),
+                                          ("rng_circ_build_fee_private_protocol", 42)])
 def test_real_circuit_gpu_parity(orc, builder, seed):
     """GPU prover bit-exact vs oracle on the REAL circuits (settlement =
     BASELINE config #4, VBC = config #1)."""
@@ -164,8 +167,8 @@ def test_real_circuit_gpu_parity(orc, builder, seed):
     lib = plib.lib
     fn = getattr(lib, builder)
     fn.restype = ctypes.c_void_p
-    two_arg = builder in ("rng_circ_build_validity",
-                          "rng_circ_build_ob_validity")  # (seed, party)
+    two_arg = builder in ("rng_circ_build_validity", "rng_circ_build_ob_validity",
+                          "rng_circ_build_ff_validity")  # (seed, party)
     fn.argtypes = [ctypes.c_uint64, ctypes.c_uint64] if two_arg else [ctypes.c_uint64]
     lib.rng_circ_n.restype = ctypes.c_uint64
     lib.rng_circ_n.argtypes = [ctypes.c_void_p]
