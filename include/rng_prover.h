@@ -157,6 +157,44 @@ int rng_link_proofs(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* hint_a
                     uint64_t group_offset, uint64_t group_size,
                     uint64_t* out_link_proof);
 
+/* ---- circuit builders (arithmetization of the reference's 20 circuits;
+ *      prover_service_client.rs:100-147 route bodies) ----
+ *
+ * rng_circ_from_scalars(kind, witness64, statement64) builds any circuit
+ * from flat scalar arrays in the reference struct field order (Montgomery
+ * limbs; CSPRNG indexes and Merkle-path bits as canonical integers).
+ * Returns an opaque CircuitTables handle, or NULL if the witness does not
+ * satisfy the circuit.  Scalar counts per kind via rng_ws_sizes; fixed-seed
+ * test vectors via rng_witness_statement.
+ *
+ * kind  circuit (reference file under circuits-core/src/zk_circuits/)
+ *   1   VALID DEPOSIT                       valid_deposit.rs
+ *   2   VALID WITHDRAWAL                    valid_withdrawal.rs
+ *   3   VALID ORDER CANCELLATION            valid_order_cancellation.rs
+ *   4   INTENT AND BALANCE VALIDITY         validity_proofs/intent_and_balance.rs
+ *   5   ... FIRST FILL VALIDITY             validity_proofs/intent_and_balance_first_fill.rs
+ *   6   INTENT ONLY VALIDITY                validity_proofs/intent_only.rs
+ *   7   INTENT ONLY FIRST FILL VALIDITY     validity_proofs/intent_only_first_fill.rs
+ *   8   NEW OUTPUT BALANCE VALIDITY         validity_proofs/new_output_balance.rs
+ *   9   OUTPUT BALANCE VALIDITY             validity_proofs/output_balance.rs
+ *  11   INTENT AND BALANCE PUBLIC SETTLEMENT   settlement/intent_and_balance_public_settlement.rs
+ *  12   INTENT AND BALANCE BOUNDED SETTLEMENT  settlement/intent_and_balance_bounded_settlement.rs
+ *  13   INTENT ONLY PUBLIC SETTLEMENT       settlement/intent_only_public_settlement.rs
+ *  14   INTENT ONLY BOUNDED SETTLEMENT      settlement/intent_only_bounded_settlement.rs
+ *  15   VALID NOTE REDEMPTION               fees/valid_note_redemption.rs
+ *  16   VALID PUBLIC RELAYER FEE PAYMENT    fees/valid_public_relayer_fee_payment.rs
+ *  17   VALID PUBLIC PROTOCOL FEE PAYMENT   fees/valid_public_protocol_fee_payment.rs
+ *  18   VALID PRIVATE RELAYER FEE PAYMENT   fees/valid_private_relayer_fee_payment.rs
+ *  19   VALID PRIVATE PROTOCOL FEE PAYMENT  fees/valid_private_protocol_fee_payment.rs
+ * (VALID BALANCE CREATE and the private settlement keep their dedicated
+ *  entry points rng_circ_vbc_from_scalars / rng_circ_settlement_from_scalars.)
+ */
+void* rng_circ_from_scalars(int kind, const uint64_t* witness64,
+                            const uint64_t* statement64);
+int rng_ws_sizes(int kind, uint64_t* out_num_witness, uint64_t* out_num_statement);
+int rng_witness_statement(int kind, uint64_t seed, uint64_t* out_witness,
+                          uint64_t* out_statement);
+
 #ifdef __cplusplus
 }
 #endif

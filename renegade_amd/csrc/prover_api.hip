@@ -1736,6 +1736,715 @@ void* rng_circ_build_nob_validity(uint64_t seed) {
     }
 }
 
+// ---- link-group layouts of the placing circuits (computed once) ----
+// The settlement circuits PLACE the groups; validity circuits inherit them.
+// Layout depends only on circuit structure, so compute from a fixed seed.
+struct LinkLayouts {
+    uint64_t ib_align = 0;
+    int64_t pg0 = 0, og0 = 0, pg1 = 0, og1 = 0;  // private-settlement groups
+    uint64_t io_align = 0;
+    int64_t io_off = 0;  // intent-only settlement group
+};
+static const LinkLayouts& link_layouts() {
+    static LinkLayouts L;
+    static std::once_flag once;
+    std::call_once(once, [] {
+        ValidityBundle b;
+        validity_bundle_build(42, b);
+        PlonkCircuit scs;
+        settlement_apply_constraints(scs, b.sw, b.sst);
+        CircuitTables t = scs.finalize();
+        for (auto& g : t.link_groups) {
+            if (g.id == "intent_and_balance_settlement_party0") {
+                L.ib_align = g.alignment;
+                L.pg0 = (int64_t)g.offset;
+            } else if (g.id == "intent_and_balance_settlement_party1") {
+                L.pg1 = (int64_t)g.offset;
+            } else if (g.id == "output_balance_settlement_party0") {
+                L.og0 = (int64_t)g.offset;
+            } else if (g.id == "output_balance_settlement_party1") {
+                L.og1 = (int64_t)g.offset;
+            }
+        }
+        IoValidityWitness vw;
+        IoValidityStatement vs;
+        IoSettlementStatement ss;
+        io_bundle_build(42, vw, vs, ss);
+        PlonkCircuit c2;
+        io_settlement_apply_constraints(c2, vw.intent, ss);
+        CircuitTables t2 = c2.finalize();
+        for (auto& g : t2.link_groups)
+            if (g.id == "intent_only_settlement") {
+                L.io_align = g.alignment;
+                L.io_off = (int64_t)g.offset;
+            }
+    });
+    return L;
+}
+
+// ---- witness/statement parsers (scalar cursor over the field orders the
+//      route bodies use; see rng_prover.h kind table) ----
+namespace {
+struct Rd {
+    const Fr* p;
+    Fr f() { return *p++; }
+    uint64_t u() {
+        u64 l[4];
+        (*p).to_canonical(l);
+        ++p;
+        return l[0];
+    }
+    void frs(Fr* dst, int n) {
+        for (int i = 0; i < n; ++i) dst[i] = f();
+    }
+    Csprng cs() {
+        Csprng c;
+        c.seed = f();
+        c.index = u();
+        return c;
+    }
+    Balance bal() {
+        Balance b;
+        Fr s[8];
+        frs(s, 8);
+        b.from_scalars(s);
+        return b;
+    }
+    Intent in() {
+        Intent i;
+        i.in_token = f();
+        i.out_token = f();
+        i.owner = f();
+        i.min_price_repr = f();
+        i.amount_in = f();
+        return i;
+    }
+    PostMatchShare pms() {
+        PostMatchShare s;
+        s.relayer_fee_balance = f();
+        s.protocol_fee_balance = f();
+        s.amount = f();
+        return s;
+    }
+    Obligation ob() {
+        Obligation o;
+        o.input_token = f();
+        o.output_token = f();
+        o.amount_in = f();
+        o.amount_out = f();
+        return o;
+    }
+    StateBalance sbal() {
+        StateBalance s;
+        s.recovery = cs();
+        s.share = cs();
+        s.inner = bal();
+        frs(s.public_share, 8);
+        return s;
+    }
+    StateIntent sint() {
+        StateIntent s;
+        s.recovery = cs();
+        s.share = cs();
+        s.inner = in();
+        frs(s.public_share, 5);
+        return s;
+    }
+    void opening(Fr* elems, bool* idx) {
+        frs(elems, MERKLE_HEIGHT);
+        for (int i = 0; i < MERKLE_HEIGHT; ++i) idx[i] = u() != 0;
+    }
+    JjSignature sig() {
+        JjSignature s;
+        Fr sf = f();
+        sf.to_canonical(s.s.v);
+        s.R.x = f();
+        s.R.y = f();
+        return s;
+    }
+    Note note() {
+        Note n;
+        n.mint = f();
+        n.amount = f();
+        n.receiver = f();
+        n.blinder = f();
+        return n;
+    }
+    SettlementParty party_no_ob() {  // public/bounded settlement witness (28)
+        SettlementParty p;
+        p.intent = in();
+        p.pre_amount_share = f();
+        p.input_balance = bal();
+        p.pre_in_shares = pms();
+        p.output_balance = bal();
+        p.pre_out_shares = pms();
+        return p;
+    }
+};
+}  // namespace
+
+namespace {
+struct Wt {  // writer mirroring Rd
+    Fr* p;
+    void f(const Fr& v) { *p++ = v; }
+    void u(uint64_t x) { *p++ = Fr::from_u64(x); }
+    void frs(const Fr* src, int n) {
+        for (int i = 0; i < n; ++i) f(src[i]);
+    }
+    void cs(const Csprng& c) {
+        f(c.seed);
+        u(c.index);
+    }
+    void bal(const Balance& b) {
+        for (auto& s : b.to_scalars()) f(s);
+    }
+    void in(const Intent& i) {
+        for (auto& s : i.to_scalars()) f(s);
+    }
+    void pms(const PostMatchShare& s_) {
+        for (auto& s : s_.to_scalars()) f(s);
+    }
+    void sbal(const StateBalance& s_) {
+        for (auto& s : s_.to_scalars()) f(s);
+    }
+    void sint(const StateIntent& s_) {
+        for (auto& s : s_.to_scalars()) f(s);
+    }
+    void opening(const Fr* elems, const bool* idx) {
+        frs(elems, MERKLE_HEIGHT);
+        for (int i = 0; i < MERKLE_HEIGHT; ++i) u(idx[i] ? 1 : 0);
+    }
+    void sig(const JjSignature& s_) {
+        f(Fr::from_canonical(s_.s.v));
+        f(s_.R.x);
+        f(s_.R.y);
+    }
+    void st(const std::vector<Fr>& v) {
+        for (auto& s : v) f(s);
+    }
+};
+}  // namespace
+
+// per-kind witness/statement scalar counts (the route body sizes)
+int rng_ws_sizes(int kind, uint64_t* out_nw, uint64_t* out_ns) {
+    struct {
+        int k, nw, ns;
+    } T[] = {{1, 40, 8},  {2, 40, 8},  {3, 34, 3},  {4, 91, 10}, {5, 69, 11},
+             {6, 39, 7},  {7, 14, 8},  {8, 74, 10}, {9, 51, 5},  {11, 28, 14},
+             {12, 28, 16}, {13, 5, 6},  {14, 5, 9},  {15, 20, 6}, {16, 40, 9},
+             {17, 40, 9}, {18, 41, 7}, {19, 42, 14}};
+    for (auto& t : T)
+        if (t.k == kind) {
+            *out_nw = t.nw;
+            *out_ns = t.ns;
+            return 0;
+        }
+    return RNG_ERR_BAD_ARG;
+}
+
+// fixed-seed witness/statement generator for any kind (test vectors for the
+// prover-service routes; same serialization rng_circ_from_scalars parses)
+int rng_witness_statement(int kind, uint64_t seed, uint64_t* out_w, uint64_t* out_s) {
+    try {
+        Wt w{(Fr*)out_w};
+        Wt s{(Fr*)out_s};
+        switch (kind) {
+            case 1: {
+                VdWitness vw;
+                VdStatement st;
+                vd_build_witness_statement(seed, vw, st);
+                w.sbal(vw.old_balance);
+                w.opening(vw.opening_elems, vw.opening_indices);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 2: {
+                VdWitness vw;
+                VwStatement st;
+                vw_build_witness_statement(seed, vw, st);
+                w.sbal(vw.old_balance);
+                w.opening(vw.opening_elems, vw.opening_indices);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 3: {
+                VocWitness vw;
+                VocStatement st;
+                voc_build_witness_statement(seed, vw, st);
+                w.sint(vw.old_intent);
+                w.opening(vw.opening_elems, vw.opening_idx);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 4: {
+                ValidityBundle b;
+                validity_bundle_build(seed, b);
+                const ValidityWitness& vw = b.vw[0];
+                w.sint(vw.old_intent);
+                w.opening(vw.intent_opening_elems, vw.intent_opening_idx);
+                w.in(vw.intent);
+                w.f(vw.new_amount_public_share);
+                w.sbal(vw.old_balance);
+                w.opening(vw.balance_opening_elems, vw.balance_opening_idx);
+                w.bal(vw.balance);
+                w.pms(vw.post_match_balance_shares);
+                s.st(b.vst[0].to_scalars());
+                break;
+            }
+            case 5: {
+                ValidityBundle b;
+                validity_bundle_build(seed, b);
+                FfWitness vw;
+                FfStatement st;
+                ff_build(b, 0, seed, vw, st);
+                w.in(vw.intent);
+                w.cs(vw.share_stream);
+                w.cs(vw.recovery_stream);
+                w.frs(vw.private_intent_shares, 5);
+                w.f(vw.new_amount_public_share);
+                w.sig(vw.sig);
+                w.sbal(vw.old_balance);
+                w.bal(vw.balance);
+                w.pms(vw.post_match_balance_shares);
+                w.opening(vw.opening_elems, vw.opening_idx);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 6: {
+                IoValidityWitness vw;
+                IoValidityStatement vs;
+                IoSettlementStatement ss;
+                io_bundle_build(seed, vw, vs, ss);
+                w.sint(vw.old_intent);
+                w.opening(vw.opening_elems, vw.opening_idx);
+                w.in(vw.intent);
+                s.st(vs.to_scalars());
+                break;
+            }
+            case 7: {
+                IoffWitness vw;
+                IoffStatement st;
+                ioff_build(seed, vw, st);
+                w.in(vw.intent);
+                w.cs(vw.share_stream);
+                w.cs(vw.recovery_stream);
+                w.frs(vw.private_shares, 5);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 8: {
+                NobWitness vw;
+                NobStatement st;
+                nob_build(seed, vw, st);
+                w.sbal(vw.new_balance);
+                w.bal(vw.balance);
+                w.pms(vw.post_match_balance_shares);
+                w.sbal(vw.existing_balance);
+                w.opening(vw.opening_elems, vw.opening_idx);
+                w.sig(vw.sig);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 9: {
+                ValidityBundle b;
+                validity_bundle_build(seed, b);
+                const ObValidityWitness& vw = b.ow[0];
+                w.sbal(vw.old_balance);
+                w.opening(vw.opening_elems, vw.opening_idx);
+                w.bal(vw.balance);
+                w.pms(vw.post_match_balance_shares);
+                s.st(b.ost[0].to_scalars());
+                break;
+            }
+            case 11: {
+                ValidityBundle b;
+                validity_bundle_build(seed, b);
+                const SettlementParty& p = b.sw.p[0];
+                w.in(p.intent);
+                w.f(p.pre_amount_share);
+                w.bal(p.input_balance);
+                w.pms(p.pre_in_shares);
+                w.bal(p.output_balance);
+                w.pms(p.pre_out_shares);
+                PubSettlementStatement st;
+                pub_settlement_statement_from_bundle(b, st);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 12: {
+                ValidityBundle b;
+                validity_bundle_build(seed, b);
+                const SettlementParty& p = b.sw.p[0];
+                w.in(p.intent);
+                w.f(p.pre_amount_share);
+                w.bal(p.input_balance);
+                w.pms(p.pre_in_shares);
+                w.bal(p.output_balance);
+                w.pms(p.pre_out_shares);
+                IbBoundedStatement st;
+                ib_bounded_statement_from_bundle(b, st);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 13: {
+                IoValidityWitness vw;
+                IoValidityStatement vs;
+                IoSettlementStatement ss;
+                io_bundle_build(seed, vw, vs, ss);
+                w.in(vw.intent);
+                s.st(ss.to_scalars());
+                break;
+            }
+            case 14: {
+                IoValidityWitness vw;
+                IoValidityStatement vs;
+                IoSettlementStatement ss;
+                io_bundle_build(seed, vw, vs, ss);
+                IoBoundedStatement st;
+                io_bounded_statement_build(seed, vw, ss, st);
+                w.in(vw.intent);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 15: {
+                NoteRedemptionWitness vw;
+                NoteRedemptionStatement st;
+                note_redemption_build(seed, vw, st);
+                w.opening(vw.opening_elems, vw.opening_idx);
+                s.st(st.to_scalars());
+                break;
+            }
+            case 16:
+            case 17:
+            case 18: {
+                VdWitness vw;
+                FeePaymentStatement st;
+                Note note;
+                int field = (kind == 17) ? 6 : 5;
+                fee_payment_build(seed, field, vw, st, note);
+                w.sbal(vw.old_balance);
+                w.opening(vw.opening_elems, vw.opening_indices);
+                std::vector<Fr> ss = {st.merkle_root, st.old_balance_nullifier,
+                                      st.new_balance_commitment, st.recovery_id,
+                                      st.new_fee_balance_share};
+                if (kind == 18) {
+                    w.f(note.blinder);
+                    ss.push_back(note.receiver);
+                    ss.push_back(native_note_commitment(note));
+                } else {
+                    auto nv = note.to_scalars();
+                    ss.insert(ss.end(), nv.begin(), nv.end());
+                }
+                s.st(ss);
+                break;
+            }
+            case 19: {
+                VdWitness vw;
+                FeePaymentStatement st;
+                Note note;
+                fee_payment_build(seed, 6, vw, st, note);
+                Lcg rng(seed ^ 0xE161A3A1E161A3A1ull);
+                u64 al[4] = {rng.next() | (rng.next() << 53),
+                             rng.next() | (rng.next() << 53), rng.next() & 0xFFFFFFFF, 0};
+                note.receiver = Fr::from_canonical(al);
+                JjScalar dk = jj_random_scalar(rng);
+                JjScalar k = jj_random_scalar(rng);
+                JjPoint pk = jj_pubkey(dk);
+                Fr plain[3] = {note.mint, note.amount, note.blinder};
+                JjCiphertext<3> ct = jj_elgamal_encrypt<3>(pk, k, plain);
+                w.sbal(vw.old_balance);
+                w.opening(vw.opening_elems, vw.opening_indices);
+                w.f(note.blinder);
+                w.f(Fr::from_canonical(k.v));
+                s.st({st.merkle_root, st.old_balance_nullifier, st.new_balance_commitment,
+                      st.recovery_id, st.new_fee_balance_share, note.receiver,
+                      native_note_commitment(note), ct.ephemeral_key.x,
+                      ct.ephemeral_key.y, ct.ciphertext[0], ct.ciphertext[1],
+                      ct.ciphertext[2], pk.x, pk.y});
+                break;
+            }
+            default:
+                return RNG_ERR_BAD_ARG;
+        }
+        return RNG_OK;
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_witness_statement: %s\n", e.what());
+        return RNG_ERR_BAD_ARG;
+    }
+}
+
+// Build any circuit from caller-supplied witness/statement scalars (the
+// prover-service request shape).  `kind` table in include/rng_prover.h;
+// returns null if the witness does not satisfy the circuit.
+void* rng_circ_from_scalars(int kind, const uint64_t* witness64,
+                            const uint64_t* statement64) {
+    try {
+        Rd w{(const Fr*)witness64};
+        Rd s{(const Fr*)statement64};
+        const LinkLayouts& L = link_layouts();
+        PlonkCircuit cs;
+        switch (kind) {
+            case 1:
+            case 2: {  // valid deposit / withdrawal (witness 40)
+                VdWitness vw;
+                vw.old_balance = w.sbal();
+                w.opening(vw.opening_elems, vw.opening_indices);
+                if (kind == 1) {
+                    VdStatement st;
+                    st.deposit.from = s.f();
+                    st.deposit.token = s.f();
+                    st.deposit.amount = s.f();
+                    st.merkle_root = s.f();
+                    st.old_nullifier = s.f();
+                    st.new_commitment = s.f();
+                    st.recovery_id = s.f();
+                    st.new_amount_share = s.f();
+                    vd_apply_constraints(cs, vw, st);
+                } else {
+                    VwStatement st;
+                    st.to = s.f();
+                    st.token = s.f();
+                    st.amount = s.f();
+                    st.merkle_root = s.f();
+                    st.old_nullifier = s.f();
+                    st.new_commitment = s.f();
+                    st.recovery_id = s.f();
+                    st.new_amount_share = s.f();
+                    vw_apply_constraints(cs, vw, st);
+                }
+                break;
+            }
+            case 3: {  // valid order cancellation (witness 34)
+                VocWitness vw;
+                vw.old_intent = w.sint();
+                w.opening(vw.opening_elems, vw.opening_idx);
+                VocStatement st;
+                st.merkle_root = s.f();
+                st.old_intent_nullifier = s.f();
+                st.owner = s.f();
+                voc_apply_constraints(cs, vw, st);
+                break;
+            }
+            case 4: {  // intent-and-balance validity (witness 91)
+                ValidityWitness vw;
+                vw.old_intent = w.sint();
+                w.opening(vw.intent_opening_elems, vw.intent_opening_idx);
+                vw.intent = w.in();
+                vw.new_amount_public_share = w.f();
+                vw.old_balance = w.sbal();
+                w.opening(vw.balance_opening_elems, vw.balance_opening_idx);
+                vw.balance = w.bal();
+                vw.post_match_balance_shares = w.pms();
+                ValidityStatement st;
+                st.intent_merkle_root = s.f();
+                st.old_intent_nullifier = s.f();
+                st.intent_partial_private = s.f();
+                st.intent_partial_public = s.f();
+                st.intent_recovery_id = s.f();
+                st.balance_merkle_root = s.f();
+                st.old_balance_nullifier = s.f();
+                st.balance_partial_private = s.f();
+                st.balance_partial_public = s.f();
+                st.balance_recovery_id = s.f();
+                validity_apply_constraints(cs, vw, st, (int)L.ib_align, L.pg0, L.pg1);
+                break;
+            }
+            case 5: {  // first-fill validity (witness 69)
+                FfWitness vw;
+                vw.intent = w.in();
+                vw.share_stream = w.cs();
+                vw.recovery_stream = w.cs();
+                w.frs(vw.private_intent_shares, 5);
+                vw.new_amount_public_share = w.f();
+                vw.sig = w.sig();
+                vw.old_balance = w.sbal();
+                vw.balance = w.bal();
+                vw.post_match_balance_shares = w.pms();
+                w.opening(vw.opening_elems, vw.opening_idx);
+                FfStatement st;
+                st.merkle_root = s.f();
+                s.frs(st.intent_public_share, 4);
+                st.intent_private_share_commitment = s.f();
+                st.intent_recovery_id = s.f();
+                st.balance_partial_private = s.f();
+                st.balance_partial_public = s.f();
+                st.old_balance_nullifier = s.f();
+                st.balance_recovery_id = s.f();
+                ff_apply_constraints(cs, vw, st, (int)L.ib_align, L.pg0, L.pg1);
+                break;
+            }
+            case 6: {  // intent-only validity (witness 39)
+                IoValidityWitness vw;
+                vw.old_intent = w.sint();
+                w.opening(vw.opening_elems, vw.opening_idx);
+                vw.intent = w.in();
+                IoValidityStatement st;
+                st.owner = s.f();
+                st.merkle_root = s.f();
+                st.old_intent_nullifier = s.f();
+                st.new_amount_public_share = s.f();
+                st.partial_private = s.f();
+                st.partial_public = s.f();
+                st.recovery_id = s.f();
+                io_validity_apply_constraints(cs, vw, st, (int)L.io_align, L.io_off);
+                break;
+            }
+            case 7: {  // intent-only first fill (witness 14)
+                IoffWitness vw;
+                vw.intent = w.in();
+                vw.share_stream = w.cs();
+                vw.recovery_stream = w.cs();
+                w.frs(vw.private_shares, 5);
+                IoffStatement st;
+                st.owner = s.f();
+                st.intent_private_commitment = s.f();
+                st.recovery_id = s.f();
+                s.frs(st.intent_public_share, 5);
+                ioff_apply_constraints(cs, vw, st, (int)L.io_align, L.io_off);
+                break;
+            }
+            case 8: {  // new output balance (witness 74)
+                NobWitness vw;
+                vw.new_balance = w.sbal();
+                vw.balance = w.bal();
+                vw.post_match_balance_shares = w.pms();
+                vw.existing_balance = w.sbal();
+                w.opening(vw.opening_elems, vw.opening_idx);
+                vw.sig = w.sig();
+                NobStatement st;
+                st.existing_merkle_root = s.f();
+                st.existing_nullifier = s.f();
+                s.frs(st.pre_match_shares, 5);
+                st.partial_private = s.f();
+                st.partial_public = s.f();
+                st.recovery_id = s.f();
+                nob_apply_constraints(cs, vw, st, (int)L.ib_align, L.og0, L.og1);
+                break;
+            }
+            case 9: {  // output-balance validity (witness 51)
+                ObValidityWitness vw;
+                vw.old_balance = w.sbal();
+                w.opening(vw.opening_elems, vw.opening_idx);
+                vw.balance = w.bal();
+                vw.post_match_balance_shares = w.pms();
+                ObValidityStatement st;
+                st.merkle_root = s.f();
+                st.old_balance_nullifier = s.f();
+                st.partial_private = s.f();
+                st.partial_public = s.f();
+                st.recovery_id = s.f();
+                ob_validity_apply_constraints(cs, vw, st, (int)L.ib_align, L.og0, L.og1);
+                break;
+            }
+            case 11: {  // ib public settlement (witness 28)
+                SettlementParty p = w.party_no_ob();
+                PubSettlementStatement st;
+                st.obligation = s.ob();
+                st.amount_public_share = s.f();
+                st.in_shares = s.pms();
+                st.out_shares = s.pms();
+                st.relayer_fee_repr = s.f();
+                st.protocol_fee_repr = s.f();
+                st.relayer_fee_recipient = s.f();
+                pub_settlement_apply_constraints(cs, p, st, (int)L.ib_align, L.pg0,
+                                                 L.og0);
+                break;
+            }
+            case 12: {  // ib bounded settlement (witness 28)
+                SettlementParty p = w.party_no_ob();
+                IbBoundedStatement st;
+                st.bmr.internal_party_input_token = s.f();
+                st.bmr.internal_party_output_token = s.f();
+                st.bmr.min_internal_party_amount_in = s.f();
+                st.bmr.max_internal_party_amount_in = s.f();
+                st.bmr.price_repr = s.f();
+                st.bmr.block_deadline = s.f();
+                st.amount_public_share = s.f();
+                st.in_shares = s.pms();
+                st.out_shares = s.pms();
+                st.internal_relayer_fee_repr = s.f();
+                st.external_relayer_fee_repr = s.f();
+                st.relayer_fee_recipient = s.f();
+                ib_bounded_apply_constraints(cs, p, st, (int)L.ib_align, L.pg0, L.og0);
+                break;
+            }
+            case 13: {  // io public settlement (witness 5)
+                Intent in = w.in();
+                IoSettlementStatement st;
+                st.obligation = s.ob();
+                st.relayer_fee_repr = s.f();
+                st.relayer_fee_recipient = s.f();
+                io_settlement_apply_constraints(cs, in, st);
+                break;
+            }
+            case 14: {  // io bounded settlement (witness 5)
+                Intent in = w.in();
+                IoBoundedStatement st;
+                st.bmr.internal_party_input_token = s.f();
+                st.bmr.internal_party_output_token = s.f();
+                st.bmr.min_internal_party_amount_in = s.f();
+                st.bmr.max_internal_party_amount_in = s.f();
+                st.bmr.price_repr = s.f();
+                st.bmr.block_deadline = s.f();
+                st.internal_relayer_fee_repr = s.f();
+                st.external_relayer_fee_repr = s.f();
+                st.relayer_fee_recipient = s.f();
+                io_bounded_apply_constraints(cs, in, st, (int)L.io_align, L.io_off);
+                break;
+            }
+            case 15: {  // note redemption (witness 20)
+                NoteRedemptionWitness vw;
+                w.opening(vw.opening_elems, vw.opening_idx);
+                NoteRedemptionStatement st;
+                st.note = s.note();
+                st.note_root = s.f();
+                st.note_nullifier = s.f();
+                note_redemption_apply_constraints(cs, vw, st);
+                break;
+            }
+            case 16:
+            case 17:
+            case 18: {  // fee payments (witness 40 [+1 blinder for 18])
+                VdWitness vw;
+                vw.old_balance = w.sbal();
+                w.opening(vw.opening_elems, vw.opening_indices);
+                Fr blinder = (kind == 18) ? w.f() : Fr::zero();
+                int field = (kind == 17) ? 6 : 5;
+                int note_mode = (kind == 18) ? 1 : 0;
+                std::vector<Fr> ss;
+                int nst = (note_mode == 0) ? 9 : 7;
+                for (int i = 0; i < nst; ++i) ss.push_back(s.f());
+                fee_payment_apply_constraints(cs, vw, blinder, field, note_mode,
+                                              kind != 17, ss);
+                break;
+            }
+            case 19: {  // private protocol fee (witness 42)
+                VdWitness vw;
+                vw.old_balance = w.sbal();
+                w.opening(vw.opening_elems, vw.opening_indices);
+                Fr blinder = w.f();
+                Fr enc_k = w.f();
+                std::vector<Fr> ss;
+                for (int i = 0; i < 14; ++i) ss.push_back(s.f());
+                fee_private_protocol_apply_constraints(cs, vw, blinder, enc_k, ss);
+                break;
+            }
+            default:
+                fprintf(stderr, "rng_circ_from_scalars: unknown kind %d\n", kind);
+                return nullptr;
+        }
+        std::string why;
+        if (!cs.check_satisfied(&why)) {
+            fprintf(stderr, "rng_circ_from_scalars(kind=%d): %s\n", kind, why.c_str());
+            return nullptr;
+        }
+        return new CircuitTables(cs.finalize());
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_circ_from_scalars: %s\n", e.what());
+        return nullptr;
+    }
+}
+
 // VALID ORDER CANCELLATION circuit (valid_order_cancellation.rs)
 void* rng_circ_build_valid_order_cancellation(uint64_t seed) {
     try {

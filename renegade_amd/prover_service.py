@@ -83,6 +83,11 @@ class ProverService:
             fn = getattr(lib, name)
             fn.restype = ctypes.c_void_p
             fn.argtypes = [U64P, U64P]
+        lib.rng_circ_from_scalars.restype = ctypes.c_void_p
+        lib.rng_circ_from_scalars.argtypes = [ctypes.c_int, U64P, U64P]
+        lib.rng_ws_sizes.restype = ctypes.c_int
+        lib.rng_ws_sizes.argtypes = [ctypes.c_int, ctypes.POINTER(ctypes.c_uint64),
+                                     ctypes.POINTER(ctypes.c_uint64)]
         lib.rng_circ_n.restype = ctypes.c_uint64
         lib.rng_circ_n.argtypes = [ctypes.c_void_p]
         lib.rng_circ_npub.restype = ctypes.c_uint64
@@ -149,6 +154,24 @@ class ProverService:
         proof, _, _ = self._prove_tables("valid_balance_create", h, False)
         # proof = flat 157-u64 buffer in the rkyv field order (plonk_proof_def.rs)
         return {"proof": [str(int(x)) for x in proof]}
+
+    def prove_kind(self, route_name, kind, body, want_hint):
+        """Generic route: witness/statement scalar arrays -> proof
+        (+ link hint for circuits that later enter link proofs)."""
+        nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+        if self.lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) != 0:
+            raise ValueError(f"unknown kind {kind}")
+        w = json_to_scalars(body["witness"], nw.value) if nw.value else \
+            np.zeros(0, dtype=np.uint64)
+        s = json_to_scalars(body["statement"], ns.value)
+        h = self.lib.rng_circ_from_scalars(kind, ptr(w), ptr(s))
+        if not h:
+            raise ValueError("unsatisfied witness/statement")
+        proof, hint, _ = self._prove_tables(route_name, h, want_hint)
+        out = {"proof": [str(int(x)) for x in proof]}
+        if want_hint:
+            out["link_hint"] = [str(int(x)) for x in hint]
+        return out
 
     def prove_private_settlement(self, body):
         w = json_to_scalars(body["witness"], 64)
@@ -218,6 +241,43 @@ def create_app(service=None, password=None):
             return svc.prove_private_settlement(body)
         except ValueError as e:
             raise HTTPException(status_code=400, detail=str(e))
+
+    # the remaining 18 routes of prover_service_client.rs:100-147, served by
+    # the generic from-scalars prover; validity/settlement circuits that enter
+    # later link proofs also return their link hint
+    ROUTES = {
+        "/prove-valid-deposit": (1, False),
+        "/prove-valid-withdrawal": (2, False),
+        "/prove-valid-order-cancellation": (3, False),
+        "/prove-intent-and-balance-validity": (4, True),
+        "/prove-intent-and-balance-first-fill-validity": (5, True),
+        "/prove-intent-only-validity": (6, True),
+        "/prove-intent-only-first-fill-validity": (7, True),
+        "/prove-new-output-balance-validity": (8, True),
+        "/prove-output-balance-validity": (9, True),
+        "/prove-intent-and-balance-public-settlement": (11, True),
+        "/prove-intent-and-balance-bounded-settlement": (12, True),
+        "/prove-intent-only-public-settlement": (13, True),
+        "/prove-intent-only-bounded-settlement": (14, True),
+        "/prove-valid-note-redemption": (15, False),
+        "/prove-valid-public-relayer-fee-payment": (16, False),
+        "/prove-valid-public-protocol-fee-payment": (17, False),
+        "/prove-valid-private-relayer-fee-payment": (18, False),
+        "/prove-valid-private-protocol-fee-payment": (19, False),
+    }
+
+    def make_route(path, kind, want_hint):
+        async def handler(request: Request):
+            auth(request)
+            body = await request.json()
+            try:
+                return svc.prove_kind(path.lstrip("/"), kind, body, want_hint)
+            except ValueError as e:
+                raise HTTPException(status_code=400, detail=str(e))
+        app.post(path)(handler)
+
+    for path, (kind, want_hint) in ROUTES.items():
+        make_route(path, kind, want_hint)
 
     return app
 

@@ -20,7 +20,7 @@ def client(orc):
         pytest.skip("no GPU")
     from fastapi.testclient import TestClient
     from renegade_amd.prover_service import ProverService, create_app
-    svc = ProverService(srs_power=14, srs_seed=42)
+    svc = ProverService(srs_power=15, srs_seed=42)  # covers n up to 32768
     app = create_app(svc, password="hunter2")
     return TestClient(app), svc
 
@@ -89,3 +89,52 @@ class TestProverService:
         for k in ["validity_link_proof_0", "validity_link_proof_1",
                   "output_balance_link_proof_0", "output_balance_link_proof_1"]:
             assert len(body[k]) == 18  # ([q], [W]) affine records
+
+    def test_all_generic_routes(self, client):
+        """Every remaining route of prover_service_client.rs:100-147: fixed
+        test vectors from rng_witness_statement, proof returned; a tampered
+        statement is rejected with 400."""
+        c, svc = client
+        lib = svc.lib
+        lib.rng_ws_sizes.restype = ctypes.c_int
+        lib.rng_ws_sizes.argtypes = [ctypes.c_int, ctypes.POINTER(ctypes.c_uint64),
+                                     ctypes.POINTER(ctypes.c_uint64)]
+        lib.rng_witness_statement.restype = ctypes.c_int
+        lib.rng_witness_statement.argtypes = [ctypes.c_int, ctypes.c_uint64, U64P, U64P]
+        from renegade_amd.prover_service import create_app  # noqa: F401
+        routes = {
+            "/prove-valid-deposit": 1,
+            "/prove-valid-withdrawal": 2,
+            "/prove-valid-order-cancellation": 3,
+            "/prove-intent-and-balance-validity": 4,
+            "/prove-intent-and-balance-first-fill-validity": 5,
+            "/prove-intent-only-validity": 6,
+            "/prove-intent-only-first-fill-validity": 7,
+            "/prove-new-output-balance-validity": 8,
+            "/prove-output-balance-validity": 9,
+            "/prove-intent-and-balance-public-settlement": 11,
+            "/prove-intent-and-balance-bounded-settlement": 12,
+            "/prove-intent-only-public-settlement": 13,
+            "/prove-intent-only-bounded-settlement": 14,
+            "/prove-valid-note-redemption": 15,
+            "/prove-valid-public-relayer-fee-payment": 16,
+            "/prove-valid-public-protocol-fee-payment": 17,
+            "/prove-valid-private-relayer-fee-payment": 18,
+            "/prove-valid-private-protocol-fee-payment": 19,
+        }
+        hdr = {"authorization": "Bearer hunter2"}
+        for path, kind in routes.items():
+            nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+            assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
+            w = np.zeros(4 * nw.value, dtype=np.uint64)
+            s = np.zeros(4 * ns.value, dtype=np.uint64)
+            assert lib.rng_witness_statement(kind, 9, ptr(w), ptr(s)) == 0
+            r = c.post(path, headers=hdr,
+                       json={"witness": _scal_json(w), "statement": _scal_json(s)})
+            assert r.status_code == 200, f"{path}: {r.text}"
+            assert len(r.json()["proof"]) == 157
+        # tamper one
+        s[0] ^= np.uint64(1)
+        r = c.post("/prove-valid-private-protocol-fee-payment", headers=hdr,
+                   json={"witness": _scal_json(w), "statement": _scal_json(s)})
+        assert r.status_code == 400
