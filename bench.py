@@ -117,6 +117,146 @@ def cpu_baseline_proofs(orc, n, npub, sel, sigma, wires, pubs, srs_records, max_
     }
 
 
+def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
+    """Full production bundle throughput: one step = the complete response of
+    native_proof_manager.rs:554-590 — 1 settlement proof (n=4096) + 2
+    INTENT AND BALANCE VALIDITY proofs (n=16384) + 2 OUTPUT BALANCE VALIDITY
+    proofs (n=8192) + 4 cross-domain link proofs, all on the GPU."""
+    lib = plib.lib
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_num_link_groups.restype = ctypes.c_uint64
+    lib.rng_circ_num_link_groups.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_link_groups.argtypes = [ctypes.c_void_p, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_build_settlement_bundle.restype = ctypes.c_void_p
+    lib.rng_circ_build_settlement_bundle.argtypes = [ctypes.c_uint64]
+    lib.rng_circ_build_validity.restype = ctypes.c_void_p
+    lib.rng_circ_build_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+    lib.rng_circ_build_ob_validity.restype = ctypes.c_void_p
+    lib.rng_circ_build_ob_validity.argtypes = [ctypes.c_uint64, ctypes.c_uint64]
+    lib.rng_link_proofs.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                                    ctypes.c_uint64, ctypes.c_uint64,
+                                    ctypes.c_uint64, U64P]
+
+    def fetch(h):
+        assert h
+        n = lib.rng_circ_n(h)
+        npub = lib.rng_circ_npub(h)
+        nlg = lib.rng_circ_num_link_groups(h)
+        lg = np.zeros(3 * max(1, nlg), dtype=np.uint64)
+        lib.rng_circ_link_groups(h, ptr(lg))
+        sel = np.zeros(13 * n * 4, dtype=np.uint64)
+        sigma = np.zeros(5 * n, dtype=np.uint64)
+        wires = np.zeros(5 * n * 4, dtype=np.uint64)
+        pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+        lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+        lib.rng_circ_free(h)
+        return dict(n=n, npub=npub, lg=lg.reshape(-1, 3), sel=sel, sigma=sigma,
+                    wires=wires, pubs=pubs)
+
+    ts = fetch(lib.rng_circ_build_settlement_bundle(42))
+    tv = [fetch(lib.rng_circ_build_validity(42, p)) for p in range(2)]
+    to = [fetch(lib.rng_circ_build_ob_validity(42, p)) for p in range(2)]
+    n_big = max(int(ts["n"]), int(tv[0]["n"]), int(to[0]["n"]))
+    power = max(4, int(n_big).bit_length())
+    ptau = orc.srs_generate_ptau(power, seed=42)
+    max_degree = (1 << power) + 2
+    ctx = plib.init(ptau, max_degree)
+
+    def mkpk(t):
+        desc = Desc(t["n"], t["npub"], ptr(t["sel"]), ptr(t["sigma"]), 0, None)
+        pk = lib.rng_preprocess(ctx.h, ctypes.byref(desc))
+        assert pk
+        return pk
+
+    pk_s, pk_v, pk_o = mkpk(ts), mkpk(tv[0]), mkpk(to[0])
+    groups = sorted((int(r[1]), int(r[0]), int(r[2])) for r in ts["lg"])
+    # offsets ascending = party0(17), out0(11), party1(17), out1(11)
+    legs = [(groups[0], pk_v, tv[0]), (groups[1], pk_o, to[0]),
+            (groups[2], pk_v, tv[1]), (groups[3], pk_o, to[1])]
+    pk_big = pk_v if int(tv[0]["n"]) == n_big else pk_s
+
+    def ext(h, n_small, out):
+        out[:4 * (n_small + 2)] = h[:4 * (n_small + 2)]
+        out[4 * (n_small + 2):-9] = 0
+        out[-9:] = h[-9:]
+
+    def bundle(seed):
+        proofs = np.zeros((5, 157), dtype=np.uint64)
+        hs = np.zeros(4 * (int(ts["n"]) + 2) + 9, dtype=np.uint64)
+        assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk_s), ptr(ts["wires"]),
+                             ptr(ts["pubs"]), seed, ptr(proofs[0]), ptr(hs)) == 0
+        hs_e = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+        ext(hs, int(ts["n"]), hs_e)
+        hv_e = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+        lps = np.zeros((4, 18), dtype=np.uint64)
+        for i, ((off, align, cnt), pk, t) in enumerate(legs):
+            hv = np.zeros(4 * (int(t["n"]) + 2) + 9, dtype=np.uint64)
+            assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(t["wires"]),
+                                 ptr(t["pubs"]), seed + 1 + i, ptr(proofs[1 + i]),
+                                 ptr(hv)) == 0
+            ext(hv, int(t["n"]), hv_e)
+            assert lib.rng_link_proofs(ctx.h, ctypes.c_void_p(pk_big), ptr(hv_e),
+                                       ptr(hs_e), align, off, cnt, ptr(lps[i])) == 0
+        return proofs, lps
+
+    for i in range(args.warmup):
+        bundle(100 + i)
+    ctx.sync()
+    from concurrent.futures import ThreadPoolExecutor
+    pool = ThreadPoolExecutor(max_workers=args.jobs)
+    list(pool.map(lambda j: bundle(500 + j), range(args.jobs)))
+    if dist:
+        dist.barrier()
+    ctx.sync()
+    t0 = time.perf_counter()
+    futs = [pool.submit(bundle, 10_000 + rank * 100_000 + 10 * i)
+            for i in range(args.steps)]
+    for f in futs:
+        f.result()
+    ctx.sync()
+    if dist:
+        import torch
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([time.perf_counter() - t0], dtype=torch.float64, device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    else:
+        elapsed = time.perf_counter() - t0
+    if rank == 0:
+        out = {
+            "metric": "settlement_bundles_per_s",
+            "value": round(args.steps * n_gpus / elapsed, 3),
+            "unit": "bundles/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000.0 * elapsed / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u64",
+            "data": "synthetic",
+            "config": {
+                "workload": "production proof bundle: 1 settlement (n=4096) + 2 "
+                            "intent-and-balance validity (n=16384) + 2 output-"
+                            "balance validity (n=8192) + 4 cross-domain link "
+                            "proofs (native_proof_manager.rs:554-590)",
+                "jobs": args.jobs,
+            },
+        }
+        print(json.dumps(out), flush=True)
+    # join worker threads BEFORE interpreter teardown so their TLS HIP
+    # scratch frees while the runtime is still alive
+    pool.shutdown(wait=True)
+    if dist:
+        dist.destroy_process_group()
+
+
 def msm_shard_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
     """BASELINE configs[4]: 2^20-point MSM base-split across ranks; exchange =
     all-gather of per-rank G1 partials (EC add is not an RCCL reduce op) +
@@ -211,9 +351,12 @@ def main():
                          "from a rayon pool; ctypes releases the GIL)")
     ap.add_argument("--no-kernel-legs", action="store_true",
                     help="skip the MSM/NTT kernel side-measurements")
-    ap.add_argument("--mode", choices=["proofs", "msm-shard"], default="proofs",
+    ap.add_argument("--mode", choices=["proofs", "msm-shard", "bundle"],
+                    default="proofs",
                     help="msm-shard = BASELINE configs[4]: one 2^20 MSM base-split "
-                         "across ranks, RCCL all-gather of partials + host EC fold")
+                         "across ranks, RCCL all-gather of partials + host EC fold; "
+                         "bundle = full production bundle (1 settlement + 4 validity "
+                         "proofs + 4 link proofs) per step")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -251,6 +394,8 @@ def main():
 
     if args.mode == "msm-shard":
         return msm_shard_mode(args, plib, orc, dist, rank, local_rank, n_gpus)
+    if args.mode == "bundle":
+        return bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus)
 
     # --- setup: circuit tables + SRS + PK ---
     n, npub, sel, sigma, wires, pubs = build_settlement_tables(lib)
@@ -428,6 +573,7 @@ def main():
         }
         print(json.dumps(result), flush=True)
 
+    pool.shutdown(wait=True)
     if dist:
         dist.destroy_process_group()
 
