@@ -346,7 +346,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--jobs", type=int, default=8,
+    ap.add_argument("--jobs", type=int, default=16,
                     help="concurrent prover threads per GPU (the reference proves "
                          "from a rayon pool; ctypes releases the GIL)")
     ap.add_argument("--no-kernel-legs", action="store_true",

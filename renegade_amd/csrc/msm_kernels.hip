@@ -28,15 +28,18 @@ namespace rng {
 constexpr uint32_t MSM_SENTINEL = 0x3FFFFFFu;  // > any (group<<16|mag); 26 bits
 constexpr uint32_t MSM_CHUNK = 16;            // buckets per window-sum thread
 
-// window size by problem size: bucket-phase work ~ n*W(c) while the
-// aggregation phase ~ W(c)*2^(c-1); balance at c ~ log2(n) - 3
+// window size by problem size.  Besides balancing bucket-phase work
+// (~ n*W(c)) against aggregation (~ W(c)*2^(c-1)), c must keep the TOP
+// window well-populated: with 254-bit scalars the top window holds only
+// 254-(W-1)*c significant bits, and when that is small every scalar's top
+// digit lands in a handful of buckets — giant segments whose serial merge
+// chains dominate (measured 3x slowdown at c=9/11 vs c=8 for n=4096).
+// Top-window distinct digits D(c) = 2^(254-(W-1)*c): c=8 -> 64, c=13 -> 128,
+// c=16 -> 2^14; those three tiers are the sweet spots.
 __host__ __device__ inline uint32_t msm_auto_c(uint64_t n) {
-    uint32_t log2n = 0;
-    while ((1ull << log2n) < n) log2n++;
-    uint32_t c = log2n > 3 ? log2n - 3 : 8;
-    if (c < 8) c = 8;
-    if (c > 16) c = 16;
-    return c;
+    if (n <= (1ull << 16)) return 8;
+    if (n <= (1ull << 18)) return 13;
+    return 16;
 }
 
 // ---- 1. digit decomposition ----
@@ -194,23 +197,24 @@ __global__ __launch_bounds__(256) void k_msm_seg_merge(
 // grid: W * (2^(c-1) / MSM_CHUNK) threads total; partials: per thread
 // (T = plain sum, S = locally-weighted sum) -> 2 Jacobians.
 __global__ __launch_bounds__(256) void k_msm_window_chunks(const G1Jac* buckets, uint32_t c, uint32_t W,
+                                    uint32_t chunk_sz,
                                     G1Jac* partials /* 2 per thread: T, S */) {
     uint32_t nb = 1u << (c - 1);
-    uint32_t chunks_per_w = nb / MSM_CHUNK;
+    uint32_t chunks_per_w = nb / chunk_sz;
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= W * chunks_per_w) return;
     uint32_t w = t / chunks_per_w;
     uint32_t chunk = t % chunks_per_w;
-    const G1Jac* b = buckets + (uint64_t)w * nb + (uint64_t)chunk * MSM_CHUNK;
-    // local digits are base + j, j = 1..CHUNK, base = chunk*CHUNK
+    const G1Jac* b = buckets + (uint64_t)w * nb + (uint64_t)chunk * chunk_sz;
+    // local digits are base + j, j = 1..chunk_sz, base = chunk*chunk_sz
     // suffix running sums over j descending: run = sum_{k>=j} S_k ; S += run
     G1Jac run = G1Jac::identity(), S = G1Jac::identity();
-    for (int j = MSM_CHUNK - 1; j >= 0; --j) {
+    for (int j = (int)chunk_sz - 1; j >= 0; --j) {
         run = run.add(b[j]);
         S = S.add(run);
     }
     partials[2 * t] = run;  // T = plain sum of chunk buckets
-    partials[2 * t + 1] = S;  // sum_j (j_local) * bucket, j_local = 1..CHUNK
+    partials[2 * t + 1] = S;  // sum_j (j_local) * bucket, j_local = 1..chunk_sz
 }
 
 // ---- 5a. window combine: grid = W * MSM_SUBB blocks; block (w, sb) folds a
@@ -220,12 +224,13 @@ constexpr uint32_t MSM_SUBB = 16;
 
 __global__ __launch_bounds__(64) void k_msm_window_combine(const G1Jac* partials,
                                                            uint32_t c,
+                                                           uint32_t chunk_sz,
                                                            G1Jac* window_partials) {
     __shared__ G1Jac red[64];
     uint32_t w = blockIdx.x / MSM_SUBB;
     uint32_t sb = blockIdx.x % MSM_SUBB;
     uint32_t nb = 1u << (c - 1);
-    uint32_t chunks_per_w = nb / MSM_CHUNK;
+    uint32_t chunks_per_w = nb / chunk_sz;
     uint32_t per_sb = (chunks_per_w + MSM_SUBB - 1) / MSM_SUBB;
     uint32_t lo = sb * per_sb;
     uint32_t hi = lo + per_sb < chunks_per_w ? lo + per_sb : chunks_per_w;
@@ -234,7 +239,7 @@ __global__ __launch_bounds__(64) void k_msm_window_combine(const G1Jac* partials
         uint32_t t = w * chunks_per_w + chunk;
         G1Jac T = partials[2 * t];
         G1Jac S = partials[2 * t + 1];
-        uint32_t base = chunk * MSM_CHUNK;
+        uint32_t base = chunk * chunk_sz;
         G1Jac bT = G1Jac::identity();
         G1Jac addend = T;
         while (base) {

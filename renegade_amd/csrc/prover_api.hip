@@ -266,7 +266,15 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint32_t G = B * W;  // key groups
     uint64_t total = n * G;
     uint64_t nb = (1ull << (c - 1)) * G;  // total buckets
-    uint64_t nchunks = ((1ull << (c - 1)) / MSM_CHUNK) * G;
+    // chunk size adapts to keep the window-sum stages wide: small MSMs (few
+    // windows x small bucket counts) get chunk=2 (4x the lanes, 4x shorter
+    // latency chains), the 2^20 leg keeps chunk=16.
+    uint32_t chunk_sz = MSM_CHUNK;
+    {
+        uint64_t nbw = (1ull << (c - 1)) * G;
+        while (chunk_sz > 2 && nbw / chunk_sz < 32768) chunk_sz >>= 1;
+    }
+    uint64_t nchunks = ((1ull << (c - 1)) / chunk_sz) * G;
 
     if (!tls_msm_scratch) tls_msm_scratch = std::make_unique<MsmScratch>();
     MsmScratch* s = tls_msm_scratch.get();
@@ -382,11 +390,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     }
     et.mark(stream);
     hipLaunchKernelGGL(k_msm_window_chunks, dim3((uint32_t)((nchunks + tb - 1) / tb)),
-                       dim3(tb), 0, stream, s->buckets, c, G, s->partials);
+                       dim3(tb), 0, stream, s->buckets, c, G, chunk_sz, s->partials);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
     hipLaunchKernelGGL(k_msm_window_combine, dim3(G * MSM_SUBB), dim3(64), 0, stream,
-                       s->partials, c, s->window_sums);
+                       s->partials, c, chunk_sz, s->window_sums);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
     // host-side fold: G*SUBB Jacobians; a single-lane dependent EC chain is
@@ -548,7 +556,13 @@ static int commit_dev_batch(RngCtxImpl* ctx, const std::vector<Fr>* const* polys
                        s->stage, s->canon, (uint32_t)(B * m));
     HIP_CHECK(hipGetLastError());
     G1Jac res[13];
-    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, msm_auto_c(m), res, B);
+    // window size: auto by problem size; RNG_MSM_C overrides for tuning
+    static int c_env = [] {
+        const char* e = getenv("RNG_MSM_C");
+        return e ? atoi(e) : 0;
+    }();
+    uint32_t c_used = (c_env >= 8 && c_env <= 16) ? (uint32_t)c_env : msm_auto_c(m);
+    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c_used, res, B);
     if (rc != RNG_OK) return rc;
     for (uint32_t b = 0; b < B; ++b) {
         uint64_t rec[9];
