@@ -33,18 +33,26 @@ def main():
     out_dir = REPO / "tests" / "golden"
     out_dir.mkdir(exist_ok=True)
 
-    for name, builder in [("settlement", "rng_circ_build_settlement"),
-                          ("vbc", "rng_circ_build_vbc")]:
+    # (name, builder, extra builder args); covers every gadget family:
+    # rotation+partial commitments, links, Schnorr, ElGamal, notes
+    targets = [("settlement", "rng_circ_build_settlement", ()),
+               ("vbc", "rng_circ_build_vbc", ()),
+               ("validity", "rng_circ_build_validity", (0,)),
+               ("ff_validity", "rng_circ_build_ff_validity", (0,)),
+               ("nob_validity", "rng_circ_build_nob_validity", ()),
+               ("fee_private_protocol", "rng_circ_build_fee_private_protocol", ()),
+               ("io_validity", "rng_circ_build_io_validity", ())]
+    for name, builder, extra in targets:
         fn = getattr(lib, builder)
         fn.restype = ctypes.c_void_p
-        fn.argtypes = [ctypes.c_uint64]
+        fn.argtypes = [ctypes.c_uint64] * (1 + len(extra))
         lib.rng_circ_n.restype = ctypes.c_uint64
         lib.rng_circ_n.argtypes = [ctypes.c_void_p]
         lib.rng_circ_npub.restype = ctypes.c_uint64
         lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
         lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
         lib.rng_circ_free.argtypes = [ctypes.c_void_p]
-        h = fn(42)
+        h = fn(42, *extra)
         assert h
         n = lib.rng_circ_n(h)
         npub = lib.rng_circ_npub(h)
