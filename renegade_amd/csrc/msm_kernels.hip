@@ -225,13 +225,14 @@ constexpr uint32_t MSM_SUBB = 16;
 __global__ __launch_bounds__(64) void k_msm_window_combine(const G1Jac* partials,
                                                            uint32_t c,
                                                            uint32_t chunk_sz,
+                                                           uint32_t subb,
                                                            G1Jac* window_partials) {
     __shared__ G1Jac red[64];
-    uint32_t w = blockIdx.x / MSM_SUBB;
-    uint32_t sb = blockIdx.x % MSM_SUBB;
+    uint32_t w = blockIdx.x / subb;
+    uint32_t sb = blockIdx.x % subb;
     uint32_t nb = 1u << (c - 1);
     uint32_t chunks_per_w = nb / chunk_sz;
-    uint32_t per_sb = (chunks_per_w + MSM_SUBB - 1) / MSM_SUBB;
+    uint32_t per_sb = (chunks_per_w + subb - 1) / subb;
     uint32_t lo = sb * per_sb;
     uint32_t hi = lo + per_sb < chunks_per_w ? lo + per_sb : chunks_per_w;
     G1Jac acc = G1Jac::identity();
@@ -256,7 +257,7 @@ __global__ __launch_bounds__(64) void k_msm_window_combine(const G1Jac* partials
             red[threadIdx.x] = red[threadIdx.x].add(red[threadIdx.x + stride]);
         __syncthreads();
     }
-    if (threadIdx.x == 0) window_partials[w * MSM_SUBB + sb] = red[0];
+    if (threadIdx.x == 0) window_partials[w * subb + sb] = red[0];
 }
 
 // (The final fold across windows — W*SUBB <= 512 Jacobians, ~1.5 KB — is

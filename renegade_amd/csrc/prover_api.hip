@@ -273,6 +273,12 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     {
         uint64_t nbw = (1ull << (c - 1)) * G;
         while (chunk_sz > 2 && nbw / chunk_sz < 32768) chunk_sz >>= 1;
+        static int chunk_env = [] {
+            const char* e = getenv("RNG_MSM_CHUNK");
+            return e ? atoi(e) : 0;
+        }();
+        if (chunk_env >= 2 && chunk_env <= 16) chunk_sz = (uint32_t)chunk_env;
+        if ((1u << (c - 1)) < chunk_sz) chunk_sz = 1u << (c - 1);
     }
     uint64_t nchunks = ((1ull << (c - 1)) / chunk_sz) * G;
 
@@ -393,14 +399,26 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
                        dim3(tb), 0, stream, s->buckets, c, G, chunk_sz, s->partials);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_window_combine, dim3(G * MSM_SUBB), dim3(64), 0, stream,
-                       s->partials, c, chunk_sz, s->window_sums);
+    // sub-blocks per window: enough blocks to spread chunks, but never more
+    // than chunks (idle blocks); RNG_MSM_SUBB overrides for tuning
+    uint32_t subb = MSM_SUBB;
+    {
+        uint32_t cw = (1u << (c - 1)) / chunk_sz;
+        while (subb > 1 && cw / subb < 64) subb >>= 1;
+        static int subb_env = [] {
+            const char* e = getenv("RNG_MSM_SUBB");
+            return e ? atoi(e) : 0;
+        }();
+        if (subb_env >= 1 && subb_env <= 16) subb = (uint32_t)subb_env;
+    }
+    hipLaunchKernelGGL(k_msm_window_combine, dim3(G * subb), dim3(64), 0, stream,
+                       s->partials, c, chunk_sz, subb, s->window_sums);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
     // host-side fold: G*SUBB Jacobians; a single-lane dependent EC chain is
     // far faster on a host core than on one GPU lane
-    std::vector<G1Jac> wsums(G * MSM_SUBB);
-    HIP_CHECK(hipMemcpyAsync(wsums.data(), s->window_sums, G * MSM_SUBB * sizeof(G1Jac),
+    std::vector<G1Jac> wsums((size_t)G * subb);
+    HIP_CHECK(hipMemcpyAsync(wsums.data(), s->window_sums, G * subb * sizeof(G1Jac),
                              hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     et.collect(tls_msm_times, 5);
@@ -408,9 +426,9 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         G1Jac ws[32];
         for (uint32_t w = 0; w < W; ++w) {
             uint32_t g = b * W + w;
-            G1Jac sum = wsums[(size_t)g * MSM_SUBB];
-            for (uint32_t i = 1; i < MSM_SUBB; ++i)
-                sum = sum.add(wsums[(size_t)g * MSM_SUBB + i]);
+            G1Jac sum = wsums[(size_t)g * subb];
+            for (uint32_t i = 1; i < subb; ++i)
+                sum = sum.add(wsums[(size_t)g * subb + i]);
             ws[w] = sum;
         }
         G1Jac acc = ws[W - 1];
