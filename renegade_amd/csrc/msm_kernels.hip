@@ -89,6 +89,202 @@ __global__ __launch_bounds__(256) void k_msm_digits(const uint64_t* scalars, uin
     // carry out of the top window must be zero for scalars < 2^(W*c-1)
 }
 
+// ---- 1b. GLV endomorphism path (include/bn254_glv.h, self-verified) ----
+// k = k1 + lambda*k2 (mod r), |ki| < 2^127; k*P = k1*P + k2*phi(P) with
+// phi(x,y) = (beta*x, y).  Halves the Pippenger windows (127-bit halves),
+// so the aggregation stages (buckets, window sums, Horner) halve too.
+#include "../../include/bn254_glv.h"
+
+#if GLV_A1_NEG || !GLV_B1_NEG || GLV_A2_NEG || GLV_B2_NEG
+#error "GLV kernels assume sign pattern a1>=0, b1<=0, a2>=0, b2>=0"
+#endif
+
+// (g * k + 2^(SHIFT-1)) >> SHIFT for 4-limb g,k; result < 2^127 (2 limbs)
+__device__ inline void glv_mul_shift(const u64 g[4], const u64 k[4], u64 out[2]) {
+    u64 acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int i = 0; i < 4; ++i) {
+        unsigned __int128 carry = 0;
+        for (int j = 0; j < 4; ++j) {
+            unsigned __int128 t = (unsigned __int128)g[i] * k[j] + acc[i + j] + (u64)carry;
+            acc[i + j] = (u64)t;
+            carry = t >> 64;
+        }
+        int idx = i + 4;
+        while (carry) {
+            unsigned __int128 t = (unsigned __int128)acc[idx] + (u64)carry;
+            acc[idx] = (u64)t;
+            carry = t >> 64;
+            ++idx;
+        }
+    }
+    // + 2^319 (rounding), then >> 320
+    unsigned __int128 t = (unsigned __int128)acc[4] + (1ull << 63);
+    acc[4] = (u64)t;
+    u64 cy = (u64)(t >> 64);
+    for (int idx = 5; cy && idx < 8; ++idx) {
+        t = (unsigned __int128)acc[idx] + cy;
+        acc[idx] = (u64)t;
+        cy = (u64)(t >> 64);
+    }
+    out[0] = acc[5];
+    out[1] = acc[6];
+}
+
+// 2-limb x 2-limb -> 4-limb
+__device__ inline void glv_mul128(const u64 a[2], const u64 b[2], u64 out[4]) {
+    unsigned __int128 t0 = (unsigned __int128)a[0] * b[0];
+    unsigned __int128 t1 = (unsigned __int128)a[0] * b[1];
+    unsigned __int128 t2 = (unsigned __int128)a[1] * b[0];
+    unsigned __int128 t3 = (unsigned __int128)a[1] * b[1];
+    out[0] = (u64)t0;
+    unsigned __int128 m = (t0 >> 64) + (u64)t1 + (u64)t2;
+    out[1] = (u64)m;
+    unsigned __int128 h = (m >> 64) + (t1 >> 64) + (t2 >> 64) + (u64)t3;
+    out[2] = (u64)h;
+    out[3] = (u64)(h >> 64) + (u64)(t3 >> 64);
+}
+
+__device__ inline int glv_cmp4(const u64 a[4], const u64 b[4]) {
+    for (int i = 3; i >= 0; --i) {
+        if (a[i] != b[i]) return a[i] > b[i] ? 1 : -1;
+    }
+    return 0;
+}
+__device__ inline void glv_sub4(const u64 a[4], const u64 b[4], u64 out[4]) {
+    unsigned __int128 borrow = 0;
+    for (int i = 0; i < 4; ++i) {
+        unsigned __int128 t = (unsigned __int128)a[i] - b[i] - (u64)borrow;
+        out[i] = (u64)t;
+        borrow = (t >> 64) ? 1 : 0;
+    }
+}
+__device__ inline void glv_add4(const u64 a[4], const u64 b[4], u64 out[4]) {
+    unsigned __int128 carry = 0;
+    for (int i = 0; i < 4; ++i) {
+        unsigned __int128 t = (unsigned __int128)a[i] + b[i] + (u64)carry;
+        out[i] = (u64)t;
+        carry = t >> 64;
+    }
+}
+
+// canonical 4-limb scalars -> per scalar 4 u64: k1 lo, k1 hi|sign<<63,
+// k2 lo, k2 hi|sign<<63 (magnitudes < 2^127)
+__global__ __launch_bounds__(256) void k_glv_decompose(const u64* canon,
+                                                       uint32_t count, u64* out) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= count) return;
+    const u64 G1C[4] = GLV_G1;
+    const u64 G2C[4] = GLV_G2;
+    const u64 A1[2] = GLV_A1;
+    const u64 B1[2] = GLV_B1;  // |b1|
+    const u64 A2[2] = GLV_A2;
+    const u64 B2[2] = GLV_B2;
+    u64 k[4] = {canon[4 * i], canon[4 * i + 1], canon[4 * i + 2], canon[4 * i + 3]};
+    u64 c1[2], c2[2];
+    glv_mul_shift(G1C, k, c1);
+    glv_mul_shift(G2C, k, c2);
+    // k1 = k - c1*a1 - c2*a2  (signed)
+    u64 s1[4], s2[4], S[4];
+    glv_mul128(c1, A1, s1);
+    glv_mul128(c2, A2, s2);
+    glv_add4(s1, s2, S);
+    u64 k1[4];
+    u64 sign1;
+    if (glv_cmp4(k, S) >= 0) {
+        glv_sub4(k, S, k1);
+        sign1 = 0;
+    } else {
+        glv_sub4(S, k, k1);
+        sign1 = 1;
+    }
+    // k2 = c1*|b1| - c2*b2  (signed)
+    u64 t1[4], t2[4], k2[4];
+    u64 sign2;
+    glv_mul128(c1, B1, t1);
+    glv_mul128(c2, B2, t2);
+    if (glv_cmp4(t1, t2) >= 0) {
+        glv_sub4(t1, t2, k2);
+        sign2 = 0;
+    } else {
+        glv_sub4(t2, t1, k2);
+        sign2 = 1;
+    }
+    out[4 * i] = k1[0];
+    out[4 * i + 1] = k1[1] | (sign1 << 63);
+    out[4 * i + 2] = k2[0];
+    out[4 * i + 3] = k2[1] | (sign2 << 63);
+}
+
+// interleave bases with their endomorphism images: out[2i] = P_i,
+// out[2i+1] = phi(P_i) = (beta*x, y) — adjacency keeps the bucket-phase
+// gathers cache-local (a split phi array costs ~15% on the 2^20 MSM)
+__global__ __launch_bounds__(256) void k_bases_interleave(const G1Aff* in, G1Aff* out,
+                                                          uint32_t count) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= count) return;
+    const u64 BETA[4] = GLV_BETA_MONT;
+    Fq beta;
+    beta.l[0] = BETA[0];
+    beta.l[1] = BETA[1];
+    beta.l[2] = BETA[2];
+    beta.l[3] = BETA[3];
+    G1Aff p = in[i];
+    out[2 * i] = p;
+    p.x = p.x.mul(beta);
+    out[2 * i + 1] = p;
+}
+
+// GLV digit decomposition: each scalar contributes TWO 127-bit halves whose
+// digits share the (poly, window) bucket space; half 2 gathers phi(bases)
+// via bit 30 of the val record.
+__global__ __launch_bounds__(256) void k_msm_digits_glv(const u64* glv, uint32_t n,
+                                                        uint32_t c, uint32_t W,
+                                                        uint32_t B, uint32_t* keys,
+                                                        uint32_t* vals) {
+    uint32_t idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n * B) return;
+    uint32_t b = idx / n, i = idx % n;
+    uint32_t half = 1u << (c - 1);
+    uint64_t cmask = (1ull << c) - 1;
+    for (uint32_t h = 0; h < 2; ++h) {
+        u64 s0 = glv[4 * idx + 2 * h];
+        u64 s1raw = glv[4 * idx + 2 * h + 1];
+        uint32_t hsign = (uint32_t)(s1raw >> 63);
+        u64 s1 = s1raw & ((1ull << 63) - 1);
+        uint32_t carry = 0;
+        for (uint32_t w = 0; w < W; ++w) {
+            uint32_t bit0 = w * c;
+            u64 raw;
+            if (bit0 < 64) {
+                raw = s0 >> bit0;
+                if (bit0 + c > 64) raw |= s1 << (64 - bit0);
+            } else {
+                raw = s1 >> (bit0 - 64);
+            }
+            raw = (raw & cmask) + carry;
+            uint32_t mag, sign;
+            if (raw >= half) {
+                if (raw > half) {
+                    mag = (uint32_t)((1ull << c) - raw);
+                    sign = 1;
+                    carry = 1;
+                } else {
+                    mag = half;
+                    sign = 0;
+                    carry = 0;
+                }
+            } else {
+                mag = (uint32_t)raw;
+                sign = 0;
+                carry = 0;
+            }
+            uint64_t o = ((uint64_t)(b * W + w)) * (2ull * n) + (uint64_t)h * n + i;
+            keys[o] = mag == 0 ? MSM_SENTINEL : (((b * W + w) << 16) | mag);
+            vals[o] = ((sign ^ hsign) << 31) | (2 * i + h);
+        }
+    }
+}
+
 // ---- 3a. segment-head flags (for stream compaction) ----
 __global__ void k_msm_head_flags(const uint32_t* keys, uint32_t total, uint8_t* flags) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;

@@ -101,6 +101,7 @@ struct RngCtxImpl {
     std::vector<uint64_t> srs_g1_host;  // (max_degree+1) * 8 u64 packed affine
     uint64_t srs_count = 0;
     void* srs_dev = nullptr;
+    void* srs_glv_dev = nullptr;  // [P, phi(P)] interleaved for GLV
     uint64_t h_g2[16];       // h: x.c0,x.c1,y.c0,y.c1 Montgomery
     uint64_t beta_h_g2[16];
     std::map<uint32_t, std::unique_ptr<NttPlan>> plans;
@@ -115,6 +116,7 @@ struct RngCtxImpl {
                 if (b) hipFree(b);
         }
         if (srs_dev) hipFree(srs_dev);
+        if (srs_glv_dev) hipFree(srs_glv_dev);
     }
 };
 
@@ -240,6 +242,8 @@ struct MsmScratch {
     G1Jac* partials = nullptr;
     G1Jac* window_sums = nullptr;
     G1Jac* result = nullptr;
+    uint64_t* glv = nullptr;   // per-scalar (k1, k2) magnitudes + signs
+    G1Aff* phi = nullptr;      // phi(bases) scratch for non-SRS base arrays
     uint64_t cap_entries = 0;
     uint64_t cap_nb = 0;
     uint64_t cap_nchunks = 0;
@@ -251,7 +255,7 @@ struct MsmScratch {
                         (void*)sub_len_sorted, (void*)sub_order,
                         (void*)partials2, (void*)head_count, select_temp, scan_temp,
                         (void*)buckets, (void*)partials, (void*)window_sums,
-                        (void*)result})
+                        (void*)result, (void*)glv, (void*)phi})
             if (b) hipFree(b);
     }
 };
@@ -259,12 +263,28 @@ struct MsmScratch {
 static thread_local std::unique_ptr<MsmScratch> tls_msm_scratch;
 
 // B polynomials sharing one base array -> one fused pipeline; results[B].
+// GLV endomorphism path (RNG_MSM_GLV=1): bit-exact (parity-green) but OFF
+// by default — measured on MI355X the halved aggregation does not pay for
+// the decompose + doubled digit work at proof sizes (335 vs 387 proofs/s)
+// or at 2^20 (7.4 vs 6.3 ms); see DESIGN.md §4.1.
+static inline bool msm_glv_enabled() {
+    static int v = [] {
+        const char* e = getenv("RNG_MSM_GLV");
+        return e ? atoi(e) : 0;
+    }();
+    return v != 0;
+}
+
 static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t n,
                        uint32_t c, G1Jac* h_result, uint32_t B = 1,
-                       hipStream_t stream = RNG_STREAM) {
-    uint32_t W = (256 + c - 1) / c;
+                       hipStream_t stream = RNG_STREAM,
+                       const G1Aff* d_glv_bases = nullptr) {
+    const bool glv = msm_glv_enabled();
+    // GLV halves scalar width: two 127-bit halves share each window's buckets
+    uint32_t W = glv ? (128 + c - 1) / c : (256 + c - 1) / c;
     uint32_t G = B * W;  // key groups
-    uint64_t total = n * G;
+    uint64_t per_scalar_entries = glv ? 2 * W : W;
+    uint64_t total = n * B * per_scalar_entries;
     uint64_t nb = (1ull << (c - 1)) * G;  // total buckets
     // chunk size adapts to keep the window-sum stages wide: small MSMs (few
     // windows x small bucket counts) get chunk=2 (4x the lanes, 4x shorter
@@ -326,6 +346,10 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->partials, 2 * cap_nch * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->window_sums, 512 * MSM_SUBB * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->result, sizeof(G1Jac)));
+        // glv: 32 B per scalar = 16*total/W bytes <= 2*total (W >= 8)
+        HIP_CHECK(hipMalloc(&s->glv, 2 * cap_total));
+        // phi: 64 B per base = 32*total/(W*B) bytes <= 4*total (W >= 8, B >= 1)
+        HIP_CHECK(hipMalloc(&s->phi, 4 * cap_total));
         s->cap_entries = cap_total;
         s->cap_nb = cap_nb2;
         s->cap_nchunks = cap_nch;
@@ -334,10 +358,27 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint32_t tb = 256;
     EvtTimer et;
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_digits, dim3((uint32_t)((n * B + tb - 1) / tb)), dim3(tb),
-                       0, stream, d_scalars, (uint32_t)n, c, W, B, s->keys_in,
-                       s->vals_in);
-    HIP_CHECK(hipGetLastError());
+    if (glv) {
+        if (!d_glv_bases) {  // non-SRS bases: interleave into scratch
+            hipLaunchKernelGGL(k_bases_interleave, dim3((uint32_t)((n + tb - 1) / tb)),
+                               dim3(tb), 0, stream, d_bases, s->phi, (uint32_t)n);
+            HIP_CHECK(hipGetLastError());
+            d_glv_bases = s->phi;
+        }
+        d_bases = d_glv_bases;  // bucket gathers index the interleaved array
+        hipLaunchKernelGGL(k_glv_decompose, dim3((uint32_t)((n * B + tb - 1) / tb)),
+                           dim3(tb), 0, stream, d_scalars, (uint32_t)(n * B), s->glv);
+        HIP_CHECK(hipGetLastError());
+        hipLaunchKernelGGL(k_msm_digits_glv, dim3((uint32_t)((n * B + tb - 1) / tb)),
+                           dim3(tb), 0, stream, s->glv, (uint32_t)n, c, W, B,
+                           s->keys_in, s->vals_in);
+        HIP_CHECK(hipGetLastError());
+    } else {
+        hipLaunchKernelGGL(k_msm_digits, dim3((uint32_t)((n * B + tb - 1) / tb)),
+                           dim3(tb), 0, stream, d_scalars, (uint32_t)n, c, W, B,
+                           s->keys_in, s->vals_in);
+        HIP_CHECK(hipGetLastError());
+    }
     et.mark(stream);
     rocprim::radix_sort_pairs(s->sort_temp, s->sort_temp_bytes, s->keys_in, s->keys_out,
                               s->vals_in, s->vals_out, total, 0, 27, stream);
@@ -405,6 +446,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     {
         uint32_t cw = (1u << (c - 1)) / chunk_sz;
         while (subb > 1 && cw / subb < 64) subb >>= 1;
+        // few windows (GLV, big c): grow subb so the combine grid still
+        // spreads across CUs (window_sums is sized for G*subb <= 8192)
+        while (subb < 64 && G * subb < 256 && cw / (2 * subb) >= 32 &&
+               (uint64_t)G * subb * 2 <= 8192)
+            subb <<= 1;
         static int subb_env = [] {
             const char* e = getenv("RNG_MSM_SUBB");
             return e ? atoi(e) : 0;
@@ -580,7 +626,8 @@ static int commit_dev_batch(RngCtxImpl* ctx, const std::vector<Fr>* const* polys
         return e ? atoi(e) : 0;
     }();
     uint32_t c_used = (c_env >= 8 && c_env <= 16) ? (uint32_t)c_env : msm_auto_c(m);
-    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c_used, res, B);
+    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c_used, res, B,
+                         RNG_STREAM, (const G1Aff*)ctx->srs_glv_dev);
     if (rc != RNG_OK) return rc;
     for (uint32_t b = 0; b < B; ++b) {
         uint64_t rec[9];
@@ -1037,6 +1084,13 @@ RngCtx* rng_prover_init(const uint8_t* srs_ptau, size_t len, uint64_t max_degree
         if (hipMemcpy(im->srs_dev, im->srs_g1_host.data(), npoints * 64,
                       hipMemcpyHostToDevice) != hipSuccess)
             return nullptr;
+        // GLV: precompute the interleaved [P, phi(P)] array once (32 MB at 2^17)
+        if (hipMalloc(&im->srs_glv_dev, npoints * 128) != hipSuccess) return nullptr;
+        uint32_t blocks = (uint32_t)((npoints + 255) / 256);
+        hipLaunchKernelGGL(k_bases_interleave, dim3(blocks), dim3(256), 0, 0,
+                           (const G1Aff*)im->srs_dev, (G1Aff*)im->srs_glv_dev,
+                           (uint32_t)npoints);
+        if (hipDeviceSynchronize() != hipSuccess) return nullptr;
     }
     return ctx.release();
 }
