@@ -5,11 +5,10 @@ An unmodified relayer selects an external prover with --prover-service-url /
 proof-manager/src/implementations/external_proof_manager/
 prover_service_client.rs:100-147, request/response types api_types.rs:80-135).
 
-This daemon serves the settlement-path routes on top of the MI355X C ABI:
-
-  POST /prove-valid-balance-create                    -> ProofResponse
-  POST /prove-intent-and-balance-private-settlement   -> PrivateSettlementProofResponse
-  GET  /health
+This daemon serves ALL 20 prover routes of prover_service_client.rs on top
+of the MI355X C ABI (the private-settlement response carries the 4 link
+proofs; validity/settlement responses carry their link hint), plus
+GET /health.
 
 Wire encoding note: the reference's request bodies are serde
 serializations of the Rust witness/statement structs.  This round encodes
@@ -19,7 +18,7 @@ struct field order, and proofs/link-proofs/hints in the rkyv field order
 byte-for-byte is scheduled work.  Authentication mirrors the client's bearer
 password header.
 
-Run:  python -m renegade_amd.prover_service --port 8000 [--srs-power 13]
+Run:  python -m renegade_amd.prover_service --port 8000 [--srs-power 15]
 """
 import argparse
 import ctypes
@@ -59,7 +58,7 @@ class ProverService:
     read-only across request threads like the reference's key cache,
     traits.rs:80-92)."""
 
-    def __init__(self, srs_power=14, srs_seed=42):
+    def __init__(self, srs_power=15, srs_seed=42):
         import sys
         sys.path.insert(0, str(REPO))
         from renegade_amd import load_prover
@@ -285,7 +284,7 @@ def create_app(service=None, password=None):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--port", type=int, default=8000)
-    ap.add_argument("--srs-power", type=int, default=14)
+    ap.add_argument("--srs-power", type=int, default=15)  # covers all 20 circuits (n <= 32768)
     ap.add_argument("--password", default=os.environ.get("PROVER_SERVICE_PASSWORD"))
     args = ap.parse_args()
     import uvicorn
