@@ -58,6 +58,71 @@ class ProverService:
     read-only across request threads like the reference's key cache,
     traits.rs:80-92)."""
 
+    #: (route cache name, builder, builder args) preloaded by --preload;
+    #: route names match prove_kind's `path.lstrip('/')` keys
+    PRELOAD = [
+        ("valid_balance_create", "rng_circ_build_vbc", ()),
+        ("intent_and_balance_private_settlement", "rng_circ_build_settlement", ()),
+        ("prove-valid-deposit", "rng_circ_build_valid_deposit", ()),
+        ("prove-valid-withdrawal", "rng_circ_build_valid_withdrawal", ()),
+        ("prove-valid-order-cancellation", "rng_circ_build_valid_order_cancellation", ()),
+        ("prove-intent-and-balance-validity", "rng_circ_build_validity", (0,)),
+        ("prove-intent-and-balance-first-fill-validity", "rng_circ_build_ff_validity", (0,)),
+        ("prove-intent-only-validity", "rng_circ_build_io_validity", ()),
+        ("prove-intent-only-first-fill-validity", "rng_circ_build_ioff", ()),
+        ("prove-new-output-balance-validity", "rng_circ_build_nob_validity", ()),
+        ("prove-output-balance-validity", "rng_circ_build_ob_validity", (0,)),
+        ("prove-intent-and-balance-public-settlement", "rng_circ_build_public_settlement", ()),
+        ("prove-intent-and-balance-bounded-settlement", "rng_circ_build_ib_bounded_settlement", ()),
+        ("prove-intent-only-public-settlement", "rng_circ_build_io_settlement", ()),
+        ("prove-intent-only-bounded-settlement", "rng_circ_build_io_bounded_settlement", ()),
+        ("prove-valid-note-redemption", "rng_circ_build_note_redemption", ()),
+        ("prove-valid-public-relayer-fee-payment", "rng_circ_build_fee_public_relayer", ()),
+        ("prove-valid-public-protocol-fee-payment", "rng_circ_build_fee_public_protocol", ()),
+        ("prove-valid-private-relayer-fee-payment", "rng_circ_build_fee_private_relayer", ()),
+        ("prove-valid-private-protocol-fee-payment", "rng_circ_build_fee_private_protocol", ()),
+    ]
+
+    def preload_keys(self):
+        """Preprocess every circuit's PK up front under its route's cache
+        name (all 20 in <1 s on the GPU; the reference's CPU startup warmup
+        is minutes-scale)."""
+        import time
+        t0 = time.perf_counter()
+        for route_name, builder, extra in self.PRELOAD:
+            fn = getattr(self.lib, builder)
+            fn.restype = ctypes.c_void_p
+            fn.argtypes = [ctypes.c_uint64] * (1 + len(extra))
+            h = fn(42, *extra)
+            assert h, builder
+            self._preprocess(route_name, h)
+        print(f"preloaded {len(self.PRELOAD)} circuit PKs in "
+              f"{time.perf_counter() - t0:.2f} s", flush=True)
+
+    def _preprocess(self, circuit_name, handle):
+        """Preprocess-or-cache the PK for `handle` under `circuit_name`."""
+        lib = self.lib
+        n = lib.rng_circ_n(handle)
+        npub = lib.rng_circ_npub(handle)
+        nlg = lib.rng_circ_num_link_groups(handle)
+        lg = np.zeros(3 * max(1, nlg), dtype=np.uint64)
+        lib.rng_circ_link_groups(handle, ptr(lg))
+        sel = np.zeros(13 * n * 4, dtype=np.uint64)
+        sigma = np.zeros(5 * n, dtype=np.uint64)
+        wires = np.zeros(5 * n * 4, dtype=np.uint64)
+        pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+        lib.rng_circ_get(handle, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+        lib.rng_circ_free(handle)
+        with self.lock:
+            if circuit_name not in self.pks:
+                desc = self._Desc(n, npub, ptr(sel), ptr(sigma), 0, None)
+                pk = lib.rng_preprocess(self.ctx.h, ctypes.byref(desc))
+                if not pk:
+                    raise RuntimeError("preprocess failed")
+                self.pks[circuit_name] = pk
+                self.pk_meta[circuit_name] = dict(n=n, lg=lg.reshape(-1, 3))
+        return n, npub, wires, pubs
+
     def __init__(self, srs_power=15, srs_seed=42):
         import sys
         sys.path.insert(0, str(REPO))
@@ -112,25 +177,8 @@ class ProverService:
     def _prove_tables(self, circuit_name, handle, want_hint):
         """Preprocess-or-cache, then prove; returns (proof, hint, meta)."""
         lib = self.lib
-        n = lib.rng_circ_n(handle)
-        npub = lib.rng_circ_npub(handle)
-        nlg = lib.rng_circ_num_link_groups(handle)
-        lg = np.zeros(3 * max(1, nlg), dtype=np.uint64)
-        lib.rng_circ_link_groups(handle, ptr(lg))
-        sel = np.zeros(13 * n * 4, dtype=np.uint64)
-        sigma = np.zeros(5 * n, dtype=np.uint64)
-        wires = np.zeros(5 * n * 4, dtype=np.uint64)
-        pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
-        lib.rng_circ_get(handle, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
-        lib.rng_circ_free(handle)
+        n, npub, wires, pubs = self._preprocess(circuit_name, handle)
         with self.lock:
-            if circuit_name not in self.pks:
-                desc = self._Desc(n, npub, ptr(sel), ptr(sigma), 0, None)
-                pk = lib.rng_preprocess(self.ctx.h, ctypes.byref(desc))
-                if not pk:
-                    raise RuntimeError("preprocess failed")
-                self.pks[circuit_name] = pk
-                self.pk_meta[circuit_name] = dict(n=n, lg=lg.reshape(-1, 3))
             self.seed_ctr += 1
             seed = self.seed_ctr
         pk = self.pks[circuit_name]
@@ -286,9 +334,13 @@ def main():
     ap.add_argument("--port", type=int, default=8000)
     ap.add_argument("--srs-power", type=int, default=15)  # covers all 20 circuits (n <= 32768)
     ap.add_argument("--password", default=os.environ.get("PROVER_SERVICE_PASSWORD"))
+    ap.add_argument("--preload", action="store_true",
+                    help="preprocess all 20 circuit PKs at startup (<1 s)")
     args = ap.parse_args()
     import uvicorn
     svc = ProverService(srs_power=args.srs_power)
+    if args.preload:
+        svc.preload_keys()
     app = create_app(svc, args.password)
     uvicorn.run(app, host="0.0.0.0", port=args.port)
 
