@@ -21,6 +21,7 @@ def client(orc):
     from fastapi.testclient import TestClient
     from renegade_amd.prover_service import ProverService, create_app
     svc = ProverService(srs_power=15, srs_seed=42)  # covers n up to 32768
+    svc.preload_keys()  # routes below must then hit the warm PK cache
     app = create_app(svc, password="hunter2")
     return TestClient(app), svc
 
@@ -123,6 +124,7 @@ class TestProverService:
             "/prove-valid-private-protocol-fee-payment": 19,
         }
         hdr = {"authorization": "Bearer hunter2"}
+        pk_count_before = len(svc.pks)
         for path, kind in routes.items():
             nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
             assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
@@ -133,6 +135,9 @@ class TestProverService:
                        json={"witness": _scal_json(w), "statement": _scal_json(s)})
             assert r.status_code == 200, f"{path}: {r.text}"
             assert len(r.json()["proof"]) == 157
+        # every route must have hit the PRELOADED cache (name mapping check)
+        assert len(svc.pks) == pk_count_before, \
+            "a route preprocessed a fresh PK despite --preload"
         # tamper one
         s[0] ^= np.uint64(1)
         r = c.post("/prove-valid-private-protocol-fee-payment", headers=hdr,
