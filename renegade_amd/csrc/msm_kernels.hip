@@ -25,7 +25,11 @@
 
 namespace rng {
 
-constexpr uint32_t MSM_SENTINEL = 0x3FFFFFFu;  // > any (group<<16|mag); 26 bits
+// zero digits get the sentinel key (G << 16) — one past the last real
+// group, so it sorts after every real (group<<16|mag) key while keeping the
+// radix sort to 16 + ceil(log2(G+1)) bits (3 passes instead of 4 for every
+// workload here)
+__host__ __device__ inline uint32_t msm_sentinel(uint32_t G) { return G << 16; }
 constexpr uint32_t MSM_CHUNK = 16;            // buckets per window-sum thread
 
 // window size by problem size.  Besides balancing bucket-phase work
@@ -83,7 +87,7 @@ __global__ __launch_bounds__(256) void k_msm_digits(const uint64_t* scalars, uin
             carry = 0;
         }
         uint64_t o = ((uint64_t)b * W + w) * n + i;
-        keys[o] = mag == 0 ? MSM_SENTINEL : (((b * W + w) << 16) | mag);
+        keys[o] = mag == 0 ? msm_sentinel(B * W) : (((b * W + w) << 16) | mag);
         vals[o] = (sign << 31) | i;
     }
     // carry out of the top window must be zero for scalars < 2^(W*c-1)
@@ -279,18 +283,19 @@ __global__ __launch_bounds__(256) void k_msm_digits_glv(const u64* glv, uint32_t
                 carry = 0;
             }
             uint64_t o = ((uint64_t)(b * W + w)) * (2ull * n) + (uint64_t)h * n + i;
-            keys[o] = mag == 0 ? MSM_SENTINEL : (((b * W + w) << 16) | mag);
+            keys[o] = mag == 0 ? msm_sentinel(B * W) : (((b * W + w) << 16) | mag);
             vals[o] = ((sign ^ hsign) << 31) | (2 * i + h);
         }
     }
 }
 
 // ---- 3a. segment-head flags (for stream compaction) ----
-__global__ void k_msm_head_flags(const uint32_t* keys, uint32_t total, uint8_t* flags) {
+__global__ void k_msm_head_flags(const uint32_t* keys, uint32_t total, uint32_t sentinel,
+                                 uint8_t* flags) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= total) return;
     uint32_t key = keys[t];
-    flags[t] = (key != MSM_SENTINEL && (t == 0 || keys[t - 1] != key)) ? 1 : 0;
+    flags[t] = (key != sentinel && (t == 0 || keys[t - 1] != key)) ? 1 : 0;
 }
 
 // Segment walks are capped at MSM_MAX_SEG entries: skewed digit
