@@ -2694,7 +2694,7 @@ inline void nob_build(uint64_t seed, NobWitness& w, NobStatement& st) {
         return p;
     };
     std::vector<Fr> priv_n = priv_of(w.new_balance.inner, w.new_balance.public_share);
-    Fr full_comm, partial_pub_unused;
+    Fr full_comm;
     {
         // full commitment at the PRE-advance recovery index
         std::vector<Fr> pubs(w.new_balance.public_share, w.new_balance.public_share + 8);
@@ -2707,7 +2707,6 @@ inline void nob_build(uint64_t seed, NobWitness& w, NobStatement& st) {
         priv_n, nb.recovery, nb.share,
         std::vector<Fr>(nb.public_share, nb.public_share + 8),
         BALANCE_PARTIAL_COMMITMENT_SIZE, st.partial_private, st.partial_public);
-    (void)partial_pub_unused;
     for (int k = 0; k < 5; ++k) st.pre_match_shares[k] = w.new_balance.public_share[k];
     std::vector<Fr> priv_e = priv_of(w.existing_balance.inner,
                                      w.existing_balance.public_share);
@@ -2951,13 +2950,11 @@ inline void fee_payment_build(uint64_t seed, int field, VdWitness& w,
     note.blinder = rng.fr();
     // rotate with fee field := 0, re-encrypted
     StateBalance nb = w.old_balance;
-    auto nv = nb.inner.to_scalars();
     Fr pad = nb.share.next();
     st.new_fee_balance_share = Fr::zero().sub(pad);
     nb.public_share[field] = st.new_fee_balance_share;
     std::vector<Fr> new_priv = old_priv;
     new_priv[field] = pad;
-    (void)nv;
     if (field == 5) nb.inner.relayer_fee_balance = Fr::zero();
     else nb.inner.protocol_fee_balance = Fr::zero();
     st.recovery_id = nb.recovery.next();

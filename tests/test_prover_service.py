@@ -143,3 +143,32 @@ class TestProverService:
         r = c.post("/prove-valid-private-protocol-fee-payment", headers=hdr,
                    json={"witness": _scal_json(w), "statement": _scal_json(s)})
         assert r.status_code == 400
+
+    def test_concurrent_requests(self, client):
+        """Parallel requests across different circuits share the context and
+        PK cache safely (the reference proves from a rayon pool)."""
+        from concurrent.futures import ThreadPoolExecutor
+        c, svc = client
+        lib = svc.lib
+        hdr = {"authorization": "Bearer hunter2"}
+        jobs = []
+        for kind, path in [(1, "/prove-valid-deposit"),
+                           (4, "/prove-intent-and-balance-validity"),
+                           (6, "/prove-intent-only-validity"),
+                           (15, "/prove-valid-note-redemption")]:
+            nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+            lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns))
+            w = np.zeros(4 * nw.value, dtype=np.uint64)
+            s = np.zeros(4 * ns.value, dtype=np.uint64)
+            lib.rng_witness_statement(kind, 21, ptr(w), ptr(s))
+            jobs.append((path, _scal_json(w), _scal_json(s)))
+
+        def hit(j):
+            path, w, s = j
+            r = c.post(path, headers=hdr, json={"witness": w, "statement": s})
+            assert r.status_code == 200, f"{path}: {r.text}"
+            return r.json()["proof"]
+
+        with ThreadPoolExecutor(max_workers=8) as ex:
+            proofs = list(ex.map(hit, jobs * 3))  # 12 concurrent requests
+        assert all(len(p) == 157 for p in proofs)
