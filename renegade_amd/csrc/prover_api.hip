@@ -446,18 +446,34 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         et.mark(stream);
     }
     et.mark(stream);
-    hipLaunchKernelGGL(k_msm_window_chunks, dim3((uint32_t)((nchunks + tb - 1) / tb)),
-                       dim3(tb), 0, stream, s->buckets, c, G, chunk_sz, s->partials);
-    HIP_CHECK(hipGetLastError());
+    uint32_t cw = (1u << (c - 1)) / chunk_sz;
+    // fused suffix-scan fold (k_msm_window_fold) measured SLOWER than the
+    // chunk+combine pair at proof sizes (368 vs 402 proofs/s): Hillis-Steele
+    // does a full-width EC add per lane per level and each level carries a
+    // barrier + LDS Jacobian traffic; kept behind RNG_MSM_FOLD=1.
+    static int fold_env = [] {
+        const char* e = getenv("RNG_MSM_FOLD");
+        return e ? atoi(e) : 0;
+    }();
+    bool use_fold = fold_env && cw <= 256;
+    if (!use_fold) {  // separate chunk pass feeding the combine
+        hipLaunchKernelGGL(k_msm_window_chunks,
+                           dim3((uint32_t)((nchunks + tb - 1) / tb)), dim3(tb), 0,
+                           stream, s->buckets, c, G, chunk_sz, s->partials);
+        HIP_CHECK(hipGetLastError());
+    }
     et.mark(stream);
-    // sub-blocks per window: enough blocks to spread chunks, but never more
-    // than chunks (idle blocks); RNG_MSM_SUBB overrides for tuning
-    uint32_t subb = MSM_SUBB;
-    {
-        uint32_t cw = (1u << (c - 1)) / chunk_sz;
+    uint32_t subb = 1;
+    if (use_fold) {
+        hipLaunchKernelGGL(k_msm_window_fold, dim3(G), dim3(cw), 0, stream,
+                           s->buckets, c, chunk_sz, s->window_sums);
+        HIP_CHECK(hipGetLastError());
+        et.mark(stream);
+    } else {
+        // sub-blocks per window: enough blocks to spread chunks, but never
+        // more than chunks (idle blocks); RNG_MSM_SUBB overrides for tuning
+        subb = MSM_SUBB;
         while (subb > 1 && cw / subb < 64) subb >>= 1;
-        // few windows (GLV, big c): grow subb so the combine grid still
-        // spreads across CUs (window_sums is sized for G*subb <= 8192)
         while (subb < 64 && G * subb < 256 && cw / (2 * subb) >= 32 &&
                (uint64_t)G * subb * 2 <= 8192)
             subb <<= 1;
@@ -466,11 +482,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
             return e ? atoi(e) : 0;
         }();
         if (subb_env >= 1 && subb_env <= 16) subb = (uint32_t)subb_env;
+        hipLaunchKernelGGL(k_msm_window_combine, dim3(G * subb), dim3(64), 0, stream,
+                           s->partials, c, chunk_sz, subb, s->window_sums);
+        HIP_CHECK(hipGetLastError());
+        et.mark(stream);
     }
-    hipLaunchKernelGGL(k_msm_window_combine, dim3(G * subb), dim3(64), 0, stream,
-                       s->partials, c, chunk_sz, subb, s->window_sums);
-    HIP_CHECK(hipGetLastError());
-    et.mark(stream);
     // host-side fold: G*SUBB Jacobians; a single-lane dependent EC chain is
     // far faster on a host core than on one GPU lane
     std::vector<G1Jac> wsums((size_t)G * subb);
