@@ -310,7 +310,7 @@ constexpr uint32_t MSM_MAX_SEG = 128;
 // work, so keep within [8, MSM_MAX_SEG]
 __host__ inline uint32_t msm_seg_cap(uint64_t total_entries) {
     uint64_t cap = total_entries / (1u << 16);
-    if (cap < 8) cap = 8;
+    if (cap < 8) cap = 8;  // scratch sizes subs as total/8 — keep min 8
     if (cap > MSM_MAX_SEG) cap = MSM_MAX_SEG;
     return (uint32_t)cap;
 }
@@ -368,6 +368,11 @@ __global__ __launch_bounds__(256) void k_msm_bucket_reduce(
     if (tt >= sub_count) return;
     uint32_t t = sub_order[tt];
     uint32_t start = sub_start[t], len = sub_len[t];
+    // NOTE on ILP: a two-accumulator interleave was tried and reverted —
+    // the v_mad_u64_u32 hazard s_nops it targets come from VCC carry
+    // serialization, which BOTH chains share, so it only added an extra EC
+    // add per sub-segment; the hazard slots are instead covered by running
+    // 16 concurrent proof streams.
     G1Jac acc = G1Jac::identity();
     for (uint32_t j = start; j < start + len; ++j) {
         uint32_t v = vals[j];
