@@ -1588,6 +1588,67 @@ void* rng_circ_build_ioff(uint64_t seed) {
     }
 }
 
+// Baby Jubjub native test shims (tests/test_jubjub.py pins the C++ curve
+// arithmetic + Schnorr/ElGamal against an independent pure-Python bignum
+// implementation).  Scalars: 4 u64 plain LE; points: x,y Montgomery Fr.
+int rng_jj_mul(const uint64_t* scalar4, const uint64_t* px4, const uint64_t* py4,
+               uint64_t* out_xy8) {
+    JjPoint p;
+    memcpy(p.x.l, px4, 32);
+    memcpy(p.y.l, py4, 32);
+    if (!jj_on_curve(p) && !(p.x.is_zero())) return RNG_ERR_BAD_ARG;
+    JjPoint r = jj_mul(scalar4, p);
+    memcpy(out_xy8, r.x.l, 32);
+    memcpy(out_xy8 + 4, r.y.l, 32);
+    return RNG_OK;
+}
+
+void rng_jj_base(uint64_t* out_xy8) {
+    JjPoint b = jj_base();
+    memcpy(out_xy8, b.x.l, 32);
+    memcpy(out_xy8 + 4, b.y.l, 32);
+}
+
+// sign/verify round trip: sk, nonce k as plain scalars; msg = n Fr
+// (Montgomery); out = s (4 u64 plain), R (8 u64 Montgomery xy)
+int rng_jj_sign(const uint64_t* sk4, const uint64_t* k4, const uint64_t* msg,
+                uint64_t n, uint64_t* out_s4, uint64_t* out_r8) {
+    JjScalar sk, k;
+    memcpy(sk.v, sk4, 32);
+    memcpy(k.v, k4, 32);
+    JjSignature sig = jj_sign(sk, k, (const Fr*)msg, n);
+    memcpy(out_s4, sig.s.v, 32);
+    memcpy(out_r8, sig.R.x.l, 32);
+    memcpy(out_r8 + 4, sig.R.y.l, 32);
+    return RNG_OK;
+}
+
+int rng_jj_verify(const uint64_t* vk8, const uint64_t* s4, const uint64_t* r8,
+                  const uint64_t* msg, uint64_t n) {
+    JjPoint vk;
+    memcpy(vk.x.l, vk8, 32);
+    memcpy(vk.y.l, vk8 + 4, 32);
+    JjSignature sig;
+    memcpy(sig.s.v, s4, 32);
+    memcpy(sig.R.x.l, r8, 32);
+    memcpy(sig.R.y.l, r8 + 4, 32);
+    return jj_verify(vk, sig, (const Fr*)msg, n) ? 1 : 0;
+}
+
+int rng_jj_elgamal(const uint64_t* pk8, const uint64_t* k4, const uint64_t* msg12,
+                   uint64_t* out_eph8, uint64_t* out_c12) {
+    JjPoint pk;
+    memcpy(pk.x.l, pk8, 32);
+    memcpy(pk.y.l, pk8 + 4, 32);
+    JjScalar k;
+    memcpy(k.v, k4, 32);
+    JjCiphertext<3> ct = jj_elgamal_encrypt<3>(pk, k, (const Fr*)msg12);
+    memcpy(out_eph8, ct.ephemeral_key.x.l, 32);
+    memcpy(out_eph8 + 4, ct.ephemeral_key.y.l, 32);
+    memcpy(out_c12, ct.ciphertext, 96);
+    return RNG_OK;
+}
+
 // embedded-curve gadget self-tests (Schnorr / ElGamal over Baby Jubjub):
 // native-sign -> in-circuit verify; returns the finalized tables or null if
 // the circuit is unsatisfied (tamper != 0 flips a signature/ciphertext bit
