@@ -104,7 +104,7 @@ __global__ __launch_bounds__(256) void k_msm_digits(const uint64_t* scalars, uin
 #endif
 
 // (g * k + 2^(SHIFT-1)) >> SHIFT for 4-limb g,k; result < 2^127 (2 limbs)
-__device__ inline void glv_mul_shift(const u64 g[4], const u64 k[4], u64 out[2]) {
+__host__ __device__ inline void glv_mul_shift(const u64 g[4], const u64 k[4], u64 out[2]) {
     u64 acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     for (int i = 0; i < 4; ++i) {
         unsigned __int128 carry = 0;
@@ -135,7 +135,7 @@ __device__ inline void glv_mul_shift(const u64 g[4], const u64 k[4], u64 out[2])
 }
 
 // 2-limb x 2-limb -> 4-limb
-__device__ inline void glv_mul128(const u64 a[2], const u64 b[2], u64 out[4]) {
+__host__ __device__ inline void glv_mul128(const u64 a[2], const u64 b[2], u64 out[4]) {
     unsigned __int128 t0 = (unsigned __int128)a[0] * b[0];
     unsigned __int128 t1 = (unsigned __int128)a[0] * b[1];
     unsigned __int128 t2 = (unsigned __int128)a[1] * b[0];
@@ -148,13 +148,13 @@ __device__ inline void glv_mul128(const u64 a[2], const u64 b[2], u64 out[4]) {
     out[3] = (u64)(h >> 64) + (u64)(t3 >> 64);
 }
 
-__device__ inline int glv_cmp4(const u64 a[4], const u64 b[4]) {
+__host__ __device__ inline int glv_cmp4(const u64 a[4], const u64 b[4]) {
     for (int i = 3; i >= 0; --i) {
         if (a[i] != b[i]) return a[i] > b[i] ? 1 : -1;
     }
     return 0;
 }
-__device__ inline void glv_sub4(const u64 a[4], const u64 b[4], u64 out[4]) {
+__host__ __device__ inline void glv_sub4(const u64 a[4], const u64 b[4], u64 out[4]) {
     unsigned __int128 borrow = 0;
     for (int i = 0; i < 4; ++i) {
         unsigned __int128 t = (unsigned __int128)a[i] - b[i] - (u64)borrow;
@@ -162,7 +162,7 @@ __device__ inline void glv_sub4(const u64 a[4], const u64 b[4], u64 out[4]) {
         borrow = (t >> 64) ? 1 : 0;
     }
 }
-__device__ inline void glv_add4(const u64 a[4], const u64 b[4], u64 out[4]) {
+__host__ __device__ inline void glv_add4(const u64 a[4], const u64 b[4], u64 out[4]) {
     unsigned __int128 carry = 0;
     for (int i = 0; i < 4; ++i) {
         unsigned __int128 t = (unsigned __int128)a[i] + b[i] + (u64)carry;

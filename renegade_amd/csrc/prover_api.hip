@@ -1649,6 +1649,51 @@ int rng_jj_elgamal(const uint64_t* pk8, const uint64_t* k4, const uint64_t* msg1
     return RNG_OK;
 }
 
+// GLV decomposition test shim (host mirror of k_glv_decompose's math):
+// canonical scalar -> (|k1|, sign1, |k2|, sign2); tests pin it against the
+// Python derivation in scripts/gen_glv_params.py
+int rng_glv_decompose(const uint64_t* canon4, uint64_t* out_k1, uint64_t* out_sign1,
+                      uint64_t* out_k2, uint64_t* out_sign2) {
+    // reuse the device function's logic via a 1-element host-side emulation:
+    // the kernel math is plain integer C++, so call it through a tiny lambda
+    // duplicating the steps (kept in sync with k_glv_decompose).
+    const u64 G1C[4] = GLV_G1;
+    const u64 G2C[4] = GLV_G2;
+    const u64 A1[2] = GLV_A1;
+    const u64 B1[2] = GLV_B1;
+    const u64 A2[2] = GLV_A2;
+    const u64 B2[2] = GLV_B2;
+    u64 k[4];
+    memcpy(k, canon4, 32);
+    u64 c1[2], c2[2];
+    glv_mul_shift(G1C, k, c1);
+    glv_mul_shift(G2C, k, c2);
+    u64 s1[4], s2[4], S[4], k1[4];
+    glv_mul128(c1, A1, s1);
+    glv_mul128(c2, A2, s2);
+    glv_add4(s1, s2, S);
+    if (glv_cmp4(k, S) >= 0) {
+        glv_sub4(k, S, k1);
+        *out_sign1 = 0;
+    } else {
+        glv_sub4(S, k, k1);
+        *out_sign1 = 1;
+    }
+    u64 t1[4], t2[4], k2[4];
+    glv_mul128(c1, B1, t1);
+    glv_mul128(c2, B2, t2);
+    if (glv_cmp4(t1, t2) >= 0) {
+        glv_sub4(t1, t2, k2);
+        *out_sign2 = 0;
+    } else {
+        glv_sub4(t2, t1, k2);
+        *out_sign2 = 1;
+    }
+    memcpy(out_k1, k1, 32);
+    memcpy(out_k2, k2, 32);
+    return RNG_OK;
+}
+
 // embedded-curve gadget self-tests (Schnorr / ElGamal over Baby Jubjub):
 // native-sign -> in-circuit verify; returns the finalized tables or null if
 // the circuit is unsatisfied (tamper != 0 flips a signature/ciphertext bit

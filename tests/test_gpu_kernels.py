@@ -69,7 +69,9 @@ class TestMsmGpu:
         b8 = np.ascontiguousarray(g1[idx][:, :8]).reshape(-1)
         return b8, b9
 
-    @pytest.mark.parametrize("n", [1, 2, 255, 4096])
+    # 70000 exercises the c=13 auto-window tier (2^16 < n <= 2^18), which no
+    # proof workload hits
+    @pytest.mark.parametrize("n", [1, 2, 255, 4096, 70000])
     def test_parity_sizes(self, ctx, orc, srs10, n):
         _, g1 = srs10
         b8, b9 = self._bases(orc, g1, n)
