@@ -191,6 +191,54 @@ class ProverService:
             raise RuntimeError(f"rng_prove rc={rc}")
         return proof, hint, self.pk_meta[circuit_name]
 
+    # ---- link-proof helpers ----
+    @staticmethod
+    def _hint_n(h):
+        """Domain size encoded by a hint's length (4*(n+2)+9 u64)."""
+        return (h.size - 9) // 4 - 2
+
+    @staticmethod
+    def _ext_hint(h, n_big):
+        out = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+        out[:h.size - 9] = h[:-9]
+        out[-9:] = h[-9:]
+        return out
+
+    def _pk_for_n(self, n):
+        """Any cached proving key whose domain is n (cross-domain links run
+        under the larger domain's key; --preload guarantees coverage)."""
+        for name, meta in self.pk_meta.items():
+            if int(meta["n"]) == n:
+                return self.pks[name]
+        raise ValueError(f"no cached proving key for domain {n}; "
+                         "start the service with --preload")
+
+    def _link(self, my_hint, ext_hint, group, n_mine):
+        """Link an external validity hint against this proof's hint at the
+        given (alignment, offset, count) group — validity first, matching
+        native_proof_manager.rs:746-777."""
+        align, off, cnt = (int(x) for x in group)
+        n_ext = self._hint_n(ext_hint)
+        n_big = max(n_ext, n_mine)
+        pk_big = self._pk_for_n(n_big)
+        ha = self._ext_hint(ext_hint, n_big)
+        hb = self._ext_hint(my_hint, n_big)
+        lp = np.zeros(18, dtype=np.uint64)
+        rc = self.lib.rng_link_proofs(self.ctx.h, ctypes.c_void_p(pk_big), ptr(ha),
+                                      ptr(hb), align, off, cnt, ptr(lp))
+        if rc != 0:
+            raise ValueError(f"rng_link_proofs rc={rc}")
+        return [str(int(x)) for x in lp]
+
+    @staticmethod
+    def _hint_from_body(body, key):
+        if key not in body:
+            raise ValueError(f"missing {key}")
+        h = np.array([int(x) for x in body[key]], dtype=np.uint64)
+        if (h.size - 9) % 4 or h.size < 9 + 4 * 3:
+            raise ValueError(f"bad hint size for {key}")
+        return h
+
     # ---- route handlers ----
     def prove_valid_balance_create(self, body):
         w = json_to_scalars(body["witness"], 12)
@@ -204,7 +252,10 @@ class ProverService:
 
     def prove_kind(self, route_name, kind, body, want_hint):
         """Generic route: witness/statement scalar arrays -> proof
-        (+ link hint for circuits that later enter link proofs)."""
+        (+ link hint for validity circuits; + link proofs for the public/
+        bounded settlement routes whose requests carry validity hints —
+        api_types.rs SettlementProofResponse / PublicSettlementProofResponse).
+        """
         nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
         if self.lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) != 0:
             raise ValueError(f"unknown kind {kind}")
@@ -214,9 +265,22 @@ class ProverService:
         h = self.lib.rng_circ_from_scalars(kind, ptr(w), ptr(s))
         if not h:
             raise ValueError("unsatisfied witness/statement")
-        proof, hint, _ = self._prove_tables(route_name, h, want_hint)
+        proof, hint, meta = self._prove_tables(route_name, h, want_hint)
         out = {"proof": [str(int(x)) for x in proof]}
-        if want_hint:
+        if kind in (11, 12):  # ib public/bounded settlement: 2 link proofs
+            groups = {int(r[2]): r for r in meta["lg"]}
+            n_mine = int(meta["n"])
+            out["validity_link_proof"] = self._link(
+                hint, self._hint_from_body(body, "validity_link_hint"),
+                groups[17], n_mine)
+            out["output_balance_link_proof"] = self._link(
+                hint, self._hint_from_body(body, "output_balance_link_hint"),
+                groups[11], n_mine)
+        elif kind in (13, 14):  # intent-only settlements: 1 link proof
+            out["link_proof"] = self._link(
+                hint, self._hint_from_body(body, "validity_link_hint"),
+                meta["lg"][0], int(meta["n"]))
+        elif want_hint:
             out["link_hint"] = [str(int(x)) for x in hint]
         return out
 
@@ -243,10 +307,11 @@ class ProverService:
                                 dtype=np.uint64)
             assert ext_hint.size == 4 * (n + 2) + 9, "bad hint size"
             lp = np.zeros(18, dtype=np.uint64)
+            # validity hint first, settlement second (native_proof_manager.rs:746-777)
             rc = self.lib.rng_link_proofs(self.ctx.h,
                                           ctypes.c_void_p(
                                               self.pks["intent_and_balance_private_settlement"]),
-                                          ptr(hint), ptr(ext_hint), int(align), int(off),
+                                          ptr(ext_hint), ptr(hint), int(align), int(off),
                                           int(count), ptr(lp))
             if rc != 0:
                 raise RuntimeError(f"rng_link_proofs rc={rc}")

@@ -125,16 +125,50 @@ class TestProverService:
         }
         hdr = {"authorization": "Bearer hunter2"}
         pk_count_before = len(svc.pks)
-        for path, kind in routes.items():
+
+        def vectors(kind, seed=9):
             nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
             assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
             w = np.zeros(4 * nw.value, dtype=np.uint64)
             s = np.zeros(4 * ns.value, dtype=np.uint64)
-            assert lib.rng_witness_statement(kind, 9, ptr(w), ptr(s)) == 0
-            r = c.post(path, headers=hdr,
+            assert lib.rng_witness_statement(kind, seed, ptr(w), ptr(s)) == 0
+            return w, s
+
+        def validity_hint(route, kind, seed=9):
+            w, s = vectors(kind, seed)
+            r = c.post(route, headers=hdr,
                        json={"witness": _scal_json(w), "statement": _scal_json(s)})
+            assert r.status_code == 200, f"{route}: {r.text}"
+            return r.json()["link_hint"]
+
+        # hints for the settlement routes: validity proofs of the SAME seed
+        # (the builders share the seed's bundle, so the link values match)
+        hints = {
+            "ib_validity": validity_hint("/prove-intent-and-balance-validity", 4),
+            "ob_validity": validity_hint("/prove-output-balance-validity", 9),
+            "io_validity": validity_hint("/prove-intent-only-validity", 6),
+        }
+        extra_body = {
+            11: {"validity_link_hint": hints["ib_validity"],
+                 "output_balance_link_hint": hints["ob_validity"]},
+            12: {"validity_link_hint": hints["ib_validity"],
+                 "output_balance_link_hint": hints["ob_validity"]},
+            13: {"validity_link_hint": hints["io_validity"]},
+            14: {"validity_link_hint": hints["io_validity"]},
+        }
+        for path, kind in routes.items():
+            w, s = vectors(kind)
+            body = {"witness": _scal_json(w), "statement": _scal_json(s)}
+            body.update(extra_body.get(kind, {}))
+            r = c.post(path, headers=hdr, json=body)
             assert r.status_code == 200, f"{path}: {r.text}"
-            assert len(r.json()["proof"]) == 157
+            out = r.json()
+            assert len(out["proof"]) == 157
+            if kind in (11, 12):
+                assert len(out["validity_link_proof"]) == 18
+                assert len(out["output_balance_link_proof"]) == 18
+            if kind in (13, 14):
+                assert len(out["link_proof"]) == 18
         # every route must have hit the PRELOADED cache (name mapping check)
         assert len(svc.pks) == pk_count_before, \
             "a route preprocessed a fresh PK despite --preload"
