@@ -123,17 +123,27 @@ class ProverService:
                 self.pk_meta[circuit_name] = dict(n=n, lg=lg.reshape(-1, 3))
         return n, npub, wires, pubs
 
-    def __init__(self, srs_power=15, srs_seed=42):
+    def __init__(self, srs_power=15, srs_seed=42, ptau_path=None):
         import sys
         sys.path.insert(0, str(REPO))
         from renegade_amd import load_prover
-        from tests.orc_bindings import OracleLib
         self.plib = load_prover()
         self.plib.require_gpu()
         self.lib = self.plib.lib
         self._sig()
-        orc = OracleLib(str(REPO / "oracle" / "liborc.so"))
-        ptau = orc.srs_generate_ptau(srs_power, seed=srs_seed)
+        if ptau_path:
+            # production path: the deployment's real ptau bytes (same
+            # snarkjs layout srs.rs:63-214 parses)
+            ptau = open(ptau_path, "rb").read()
+        else:
+            # DEV/TEST fallback: deterministic generated SRS via the CPU
+            # oracle (test infrastructure); production deployments pass
+            # --ptau with the ceremony file
+            from tests.orc_bindings import OracleLib
+            orc = OracleLib(str(REPO / "oracle" / "liborc.so"))
+            ptau = orc.srs_generate_ptau(srs_power, seed=srs_seed)
+            print(f"WARNING: using deterministic TEST SRS (power {srs_power}, "
+                  f"seed {srs_seed}); pass --ptau for production", flush=True)
         self.max_degree = (1 << srs_power) + 2
         self.ctx = self.plib.init(ptau, self.max_degree)
         self.pks = {}
@@ -401,9 +411,12 @@ def main():
     ap.add_argument("--password", default=os.environ.get("PROVER_SERVICE_PASSWORD"))
     ap.add_argument("--preload", action="store_true",
                     help="preprocess all 20 circuit PKs at startup (<1 s)")
+    ap.add_argument("--ptau", default=None,
+                    help="path to the production ptau file (snarkjs layout); "
+                         "omitted = deterministic TEST SRS")
     args = ap.parse_args()
     import uvicorn
-    svc = ProverService(srs_power=args.srs_power)
+    svc = ProverService(srs_power=args.srs_power, ptau_path=args.ptau)
     if args.preload:
         svc.preload_keys()
     app = create_app(svc, args.password)
