@@ -386,10 +386,12 @@ __global__ __launch_bounds__(256) void k_msm_bucket_reduce(
 // buckets: W * 2^(c-1) Jacobian points (zero-init = identity for untouched).
 __global__ __launch_bounds__(256) void k_msm_seg_merge(
     const uint32_t* keys, const uint32_t* heads, const uint32_t* sub_off,
-    const uint32_t* nsub, uint32_t head_count, const G1Jac* partials2,
+    const uint32_t* nsub, const uint32_t* head_order /* by nsub desc */,
+    uint32_t head_count, const G1Jac* partials2,
     G1Jac* buckets, uint32_t c) {
-    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
-    if (t >= head_count) return;
+    uint32_t tt = blockIdx.x * blockDim.x + threadIdx.x;
+    if (tt >= head_count) return;
+    uint32_t t = head_order[tt];
     uint32_t off = sub_off[t], ns = nsub[t];
     G1Jac acc = partials2[off];
     for (uint32_t k = 1; k < ns; ++k) acc = acc.add(partials2[off + k]);

@@ -237,6 +237,8 @@ struct MsmScratch {
     uint32_t* sub_len = nullptr;
     uint32_t* sub_len_sorted = nullptr;
     uint32_t* sub_order = nullptr;
+    uint32_t* nsub_sorted = nullptr;
+    uint32_t* head_order = nullptr;
     G1Jac* partials2 = nullptr;
     uint32_t* head_count = nullptr;  // device u32
     void* select_temp = nullptr;
@@ -258,6 +260,7 @@ struct MsmScratch {
                         sort_temp, (void*)head_flags, (void*)heads, (void*)lens,
                         (void*)nsub, (void*)sub_off, (void*)sub_start, (void*)sub_len,
                         (void*)sub_len_sorted, (void*)sub_order,
+                        (void*)nsub_sorted, (void*)head_order,
                         (void*)partials2, (void*)head_count, select_temp, scan_temp,
                         (void*)buckets, (void*)partials, (void*)window_sums,
                         (void*)result, (void*)glv, (void*)phi})
@@ -335,6 +338,8 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->sub_len, max_subs_cap * 4));
         HIP_CHECK(hipMalloc(&s->sub_len_sorted, max_subs_cap * 4));
         HIP_CHECK(hipMalloc(&s->sub_order, max_subs_cap * 4));
+        HIP_CHECK(hipMalloc(&s->nsub_sorted, max_heads_cap * 4));
+        HIP_CHECK(hipMalloc(&s->head_order, max_heads_cap * 4));
         HIP_CHECK(hipMalloc(&s->partials2, max_subs_cap * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->head_count, 4));
         {
@@ -438,9 +443,17 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
                            0, stream, s->vals_out, s->sub_start, s->sub_len, s->sub_order,
                            sub_total, d_bases, s->partials2);
         HIP_CHECK(hipGetLastError());
+        {
+            // order heads by sub count desc: the handful of long-merge heads
+            // (top-window buckets) otherwise serialize whole waves
+            rocprim::counting_iterator<uint32_t> cit(0);
+            rocprim::radix_sort_pairs_desc(s->sort_temp, s->sort_temp_bytes, s->nsub,
+                                           s->nsub_sorted, cit, s->head_order, hc, 0,
+                                           16, stream);
+        }
         hipLaunchKernelGGL(k_msm_seg_merge, dim3((hc + tb - 1) / tb), dim3(tb), 0,
-                           stream, s->keys_out, s->heads, s->sub_off, s->nsub, hc,
-                           s->partials2, s->buckets, c);
+                           stream, s->keys_out, s->heads, s->sub_off, s->nsub,
+                           s->head_order, hc, s->partials2, s->buckets, c);
         HIP_CHECK(hipGetLastError());
     } else {
         et.mark(stream);
