@@ -178,6 +178,33 @@ class TestProverService:
                    json={"witness": _scal_json(w), "statement": _scal_json(s)})
         assert r.status_code == 400
 
+    def test_python_client_sdk(self, client):
+        """renegade_amd.prover_client against the in-process daemon: a
+        validity proof and a settlement with its link proof."""
+        c, svc = client
+        lib = svc.lib
+        from renegade_amd.prover_client import ProverServiceClient
+        sdk = ProverServiceClient("", password="hunter2", transport=c)
+        assert sdk.health()["status"] == "ok"
+
+        def vectors(kind, seed=13):
+            nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+            assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
+            w = np.zeros(4 * nw.value, dtype=np.uint64)
+            s = np.zeros(4 * ns.value, dtype=np.uint64)
+            assert lib.rng_witness_statement(kind, seed, ptr(w), ptr(s)) == 0
+            return w, s
+
+        w, s = vectors(6)  # intent-only validity
+        vresp = sdk.prove_intent_only_validity(w.reshape(-1, 4), s.reshape(-1, 4))
+        assert len(vresp["proof"]) == 157
+        hint = np.array([int(x) for x in vresp["link_hint"]], dtype=np.uint64)
+        w2, s2 = vectors(13)  # intent-only public settlement, same seed bundle
+        sresp = sdk.prove_intent_only_public_settlement(
+            w2.reshape(-1, 4), s2.reshape(-1, 4), hint)
+        assert len(sresp["proof"]) == 157
+        assert len(sresp["link_proof"]) == 18
+
     def test_concurrent_requests(self, client):
         """Parallel requests across different circuits share the context and
         PK cache safely (the reference proves from a rayon pool)."""
