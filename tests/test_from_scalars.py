@@ -80,3 +80,23 @@ def test_tables_match_native_builder(fs):
     b = tables(lib.rng_circ_build_validity(9, 0))
     for x, y, what in zip(a, b, ["selectors", "sigma", "wires", "pubs"]):
         assert np.array_equal(x, y), f"{what} differ between builders"
+
+
+def test_witness_statement_respects_sizes(fs):
+    """rng_witness_statement must write exactly rng_ws_sizes scalars (a
+    mismatch would silently corrupt the caller's buffers): canary words
+    beyond the declared sizes stay untouched."""
+    lib = fs
+    CANARY = np.uint64(0xDEADBEEFCAFEBABE)
+    for kind in ALL_KINDS:
+        nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+        assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
+        w = np.full(4 * nw.value + 8, CANARY, dtype=np.uint64)
+        s = np.full(4 * ns.value + 8, CANARY, dtype=np.uint64)
+        assert lib.rng_witness_statement(kind, 3, ptr(w), ptr(s)) == 0
+        assert (w[4 * nw.value:] == CANARY).all(), f"kind {kind} overwrote witness"
+        assert (s[4 * ns.value:] == CANARY).all(), f"kind {kind} overwrote statement"
+        # and the declared region was fully written (no canary residue --
+        # scalar values equal to the canary are astronomically unlikely)
+        assert (w[:4 * nw.value] != CANARY).any() or nw.value == 0
+        assert not (s[:4 * ns.value] == CANARY).all()
