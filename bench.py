@@ -343,8 +343,8 @@ def msm_shard_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--jobs", type=int, default=16,
                     help="concurrent prover threads per GPU (the reference proves "
