@@ -3,19 +3,27 @@
 // Replaces: arkworks ark-ec VariableBaseMSM as consumed through the
 // reference's KZG commitments (BASELINE config #2).
 //
-// Structure (sort-based, designed for gfx950):
-//  1. k_msm_digits: signed windowed digit decomposition (c bits, digits in
-//     [-2^(c-1), 2^(c-1)]), one (key = window<<16 | magnitude, val =
-//     sign<<31 | point index) pair per nonzero digit; zero digits get a
-//     sentinel key that sorts last.
-//  2. rocPRIM device radix sort on the 21-bit keys (groups all points of a
-//     bucket together — no atomics or EC critical sections anywhere).
-//  3. k_msm_bucket_reduce: one thread per segment head walks its bucket's
-//     points with Jacobian mixed-adds (bases gathered from HBM).
-//  4. k_msm_window_chunks: per (window, chunk of 2^(c-1)/CHUNK buckets):
-//     suffix running sums -> (sum, weighted-sum) partials.
-//  5. k_msm_final: combine chunk partials, fold windows (Horner with c
-//     doublings), single workgroup; result Jacobian to host.
+// Structure (sort-based, designed for gfx950; B polynomials batch through
+// one pipeline via key group g = poly*W + window):
+//  1. k_msm_digits: signed windowed digit decomposition (c in {8,13,16} so
+//     the TOP window stays populated — see msm_auto_c), one
+//     (key = group<<16 | magnitude, val = sign<<31 | point index) pair per
+//     nonzero digit; zero digits get the dynamic sentinel (G<<16).
+//  2. rocPRIM device radix sort over the live key bits only (3 passes) —
+//     groups every bucket's points together; no atomics or EC critical
+//     sections anywhere.
+//  3. segment heads via flag+select; long segments split into capped
+//     sub-segments (k_msm_seg_lengths / k_msm_make_subs), reduced
+//     longest-first one lane each with Jacobian mixed adds
+//     (k_msm_bucket_reduce), then merged per bucket longest-first
+//     (k_msm_seg_merge).
+//  4. k_msm_window_chunks: per (window, chunk) suffix running sums ->
+//     (sum, weighted-sum) partials; chunk size adapts to keep >=32k lanes.
+//  5. k_msm_window_combine: per-window fold of the chunk partials
+//     (sub-block count adapts); the final Horner across windows runs on
+//     the HOST (a 1-lane dependent EC chain is far faster on a host core).
+// A GLV-endomorphism variant and a fused suffix-scan fold exist behind
+// RNG_MSM_GLV / RNG_MSM_FOLD (both measured slower on MI355X).
 #include <hip/hip_runtime.h>
 #include <rocprim/device/device_radix_sort.hpp>
 #include <rocprim/device/device_select.hpp>
