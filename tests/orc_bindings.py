@@ -132,15 +132,31 @@ class OracleLib:
         return bytes(out)
 
     # SRS ------------------------------------------------------------------
+    # SRS generation/parsing are pure functions of (power, seed) — memoize
+    # per process (many test modules regenerate the same SRS; under a loaded
+    # container the repeats dominated suite time)
+    _srs_gen_cache = {}
+    _srs_parse_cache = {}
+
     def srs_generate_ptau(self, power, seed) -> bytes:
+        key = (power, seed)
+        hit = OracleLib._srs_gen_cache.get(key)
+        if hit is not None:
+            return hit
         size = self.lib.orc_srs_ptau_size(ctypes.c_int(power))
         buf = (ctypes.c_uint8 * size)()
         n = self.lib.orc_srs_generate_ptau(ctypes.c_int(power), ctypes.c_uint64(seed), buf,
                                            ctypes.c_uint64(size))
         assert n != 0, "srs buffer too small"
-        return bytes(buf[:n])
+        out = bytes(buf[:n])
+        OracleLib._srs_gen_cache[key] = out
+        return out
 
     def srs_parse(self, data: bytes, max_degree):
+        key = (id(data), len(data), max_degree)
+        hit = OracleLib._srs_parse_cache.get(key)
+        if hit is not None:
+            return hit
         g1 = arr(9 * (max_degree + 1))
         h = arr(16)
         bh = arr(16)
@@ -148,4 +164,8 @@ class OracleLib:
         rc = self.lib.orc_srs_parse(buf, ctypes.c_uint64(len(data)),
                                     ctypes.c_uint64(max_degree), ptr(g1), ptr(h), ptr(bh))
         assert rc == 0, f"srs parse failed rc={rc}"
-        return g1.reshape(max_degree + 1, 9), h, bh
+        out = (g1.reshape(max_degree + 1, 9), h, bh)
+        # key by object identity of the (cached, immutable) bytes + length:
+        # safe because generate_ptau returns the same interned object per key
+        OracleLib._srs_parse_cache[key] = out
+        return out
