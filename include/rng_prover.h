@@ -195,6 +195,42 @@ int rng_ws_sizes(int kind, uint64_t* out_num_witness, uint64_t* out_num_statemen
 int rng_witness_statement(int kind, uint64_t seed, uint64_t* out_witness,
                           uint64_t* out_statement);
 
+/* Fixed-seed circuit builders (deterministic witness/statement generators
+ * mirroring the reference's test_helpers; returns an opaque CircuitTables
+ * handle or NULL if unsatisfied).  Two-arg builders take (seed, party). */
+void* rng_circ_build_vbc(uint64_t seed);
+void* rng_circ_build_settlement(uint64_t seed);
+void* rng_circ_build_settlement_bundle(uint64_t seed);
+void* rng_circ_build_valid_deposit(uint64_t seed);
+void* rng_circ_build_valid_withdrawal(uint64_t seed);
+void* rng_circ_build_valid_order_cancellation(uint64_t seed);
+void* rng_circ_build_validity(uint64_t seed, uint64_t party);
+void* rng_circ_build_ob_validity(uint64_t seed, uint64_t party);
+void* rng_circ_build_ff_validity(uint64_t seed, uint64_t party);
+void* rng_circ_build_io_validity(uint64_t seed);
+void* rng_circ_build_ioff(uint64_t seed);
+void* rng_circ_build_nob_validity(uint64_t seed);
+void* rng_circ_build_public_settlement(uint64_t seed);
+void* rng_circ_build_ib_bounded_settlement(uint64_t seed);
+void* rng_circ_build_io_settlement(uint64_t seed);
+void* rng_circ_build_io_bounded_settlement(uint64_t seed);
+void* rng_circ_build_note_redemption(uint64_t seed);
+void* rng_circ_build_fee_public_relayer(uint64_t seed);
+void* rng_circ_build_fee_public_protocol(uint64_t seed);
+void* rng_circ_build_fee_private_relayer(uint64_t seed);
+void* rng_circ_build_fee_private_protocol(uint64_t seed);
+
+/* CircuitTables accessors (the finalized arithmetization rng_preprocess /
+ * rng_prove consume; column layouts as in RngCircuitDesc) */
+uint64_t rng_circ_n(void* tables);
+uint64_t rng_circ_npub(void* tables);
+int rng_circ_get(void* tables, uint64_t* selectors, uint64_t* sigma,
+                 uint64_t* wires, uint64_t* public_inputs);
+uint64_t rng_circ_num_link_groups(void* tables);
+/* per group: (alignment, grid offset, count) triples */
+void rng_circ_link_groups(void* tables, uint64_t* out3xN);
+void rng_circ_free(void* tables);
+
 #ifdef __cplusplus
 }
 #endif
