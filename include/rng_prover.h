@@ -100,8 +100,8 @@ int rng_msm_g1(RngCtx* ctx, const uint64_t* bases, const uint64_t* scalars,
 /* Device-resident variants for benchmarking/pipelining (inputs already in
  * HBM; see rng_dbuf_*). */
 int rng_ntt_fr_dev(RngCtx* ctx, void* dev_data, uint64_t n, uint64_t batch, int inverse);
-/* out-of-place: result lands in dev_out (dev_in preserved for n > 4096;
- * for n <= 4096 the transform runs in dev_in then copies to dev_out). */
+/* out-of-place: result lands in dev_out (dev_in preserved for n > 1024;
+ * for n <= 1024 the single-workgroup path runs in dev_in then copies). */
 int rng_ntt_fr_dev_oop(RngCtx* ctx, void* dev_in, void* dev_out, uint64_t n,
                        uint64_t batch, int inverse);
 int rng_msm_g1_dev(RngCtx* ctx, const void* dev_bases, const void* dev_scalars,
@@ -118,7 +118,7 @@ int rng_device_sync(void);
  * returns device pointer (n*8 u64 packed affine) and count. */
 const void* rng_srs_dev_bases(RngCtx* ctx, uint64_t* count);
 
-/* ---- PlonK prover (populated as the plonk layer lands) ---- */
+/* ---- PlonK prover (see DESIGN.md for the round structure) ---- */
 
 /* Circuit description: the output of arithmetization, i.e. exactly what
  * PlonkKzgSnark::preprocess consumes from a finalized PlonkCircuit
