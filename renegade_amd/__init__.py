@@ -6,7 +6,10 @@ reference's circuits crate makes (see SURVEY.md §8b and DESIGN.md).
 
 The HIP extension is REQUIRED on a GPU box: nothing here falls back to a CPU
 or torch implementation.  The CPU oracle under oracle/ is test
-infrastructure only and is never imported from this package.
+infrastructure only and is never used for any computation this package
+serves; the one exception is the service daemon's DEV fallback, which
+synthesizes a deterministic TEST SRS through it when no --ptau file is
+given (clearly logged; production passes --ptau).
 """
 from renegade_amd.prover import ProverLib, load_prover  # noqa: F401
 
