@@ -224,8 +224,8 @@ void* rng_circ_build_fee_private_protocol(uint64_t seed);
  * rng_prove consume; column layouts as in RngCircuitDesc) */
 uint64_t rng_circ_n(void* tables);
 uint64_t rng_circ_npub(void* tables);
-int rng_circ_get(void* tables, uint64_t* selectors, uint64_t* sigma,
-                 uint64_t* wires, uint64_t* public_inputs);
+void rng_circ_get(void* tables, uint64_t* selectors, uint64_t* sigma,
+                  uint64_t* wires, uint64_t* public_inputs);
 uint64_t rng_circ_num_link_groups(void* tables);
 /* per group: (alignment, grid offset, count) triples */
 void rng_circ_link_groups(void* tables, uint64_t* out3xN);
