@@ -128,3 +128,30 @@ class TestCrossDomainLink:
                                      ctypes.c_uint64(ALIGN), ctypes.c_uint64(OFF),
                                      ctypes.c_uint64(COUNT), ptr(xd["tau"]))
         assert ok != 1
+
+
+    def test_wrong_geometry_rejected(self, xd):
+        """A link proof computed at the right placement must NOT verify at a
+        different offset, alignment or count."""
+        o = xd["o"]
+        tA = xd["tables"](7, 2)
+        tB = xd["tables"](7, 30)
+        pkB = xd["setup"](tB)
+        _, hA = xd["prove"](xd["setup"](tA), tA, 3)
+        _, hB = xd["prove"](pkB, tB, 4)
+        nB = tB["n"]
+        hA_ext = np.zeros(4 * (nB + 2) + 9, dtype=np.uint64)
+        hA_ext[:4 * (tA["n"] + 2)] = hA[:4 * (tA["n"] + 2)]
+        hA_ext[-9:] = hA[-9:]
+        lp = np.zeros(18, dtype=np.uint64)
+        assert o.orc_plonk_link(ctypes.c_void_p(pkB), ptr(hA_ext), ptr(hB),
+                                ctypes.c_uint64(ALIGN), ctypes.c_uint64(OFF),
+                                ctypes.c_uint64(COUNT), ptr(lp)) == 0
+        for a, off, cnt in [(ALIGN, OFF + 1, COUNT),   # wrong offset
+                            (ALIGN + 1, OFF, COUNT),   # wrong alignment
+                            (ALIGN, OFF, COUNT - 1)]:  # wrong count
+            ok = o.orc_plonk_link_verify(ctypes.c_void_p(pkB), ptr(hA_ext[-9:].copy()),
+                                         ptr(hB[-9:].copy()), ptr(lp),
+                                         ctypes.c_uint64(a), ctypes.c_uint64(off),
+                                         ctypes.c_uint64(cnt), ptr(xd["tau"]))
+            assert ok != 1, f"verified at wrong geometry {(a, off, cnt)}"
