@@ -105,8 +105,10 @@ class TestPlonkOracle:
     def test_tampered_proof_rejected(self, plonk_setup):
         s = plonk_setup
         base = prove(s)
-        # corrupt each section: a wire comm limb, an eval limb, an opening
-        for idx in [0, 9 * 5, 117, 117 + 39, 9 * 11]:
+        # corrupt EVERY proof element: each of the 13 G1 records (x limb)
+        # and each of the 10 evaluations (limb 0)
+        idxs = [9 * g for g in range(13)] + [117 + 4 * e for e in range(10)]
+        for idx in idxs:
             p = base.copy()
             p[idx] ^= np.uint64(1)
             assert verify(s, p) != 1, f"tampered index {idx} accepted"
