@@ -88,10 +88,12 @@ class TestSettlementCircuit:
         o.orc_derive_tau(42, ptr(tau))
         assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(stl["pubs"]), ptr(proof),
                                   ptr(tau)) == 1
-        # tamper a fee-rate public input -> reject
-        bad = stl["pubs"].copy()
-        bad[14 * 4] ^= np.uint64(1)
-        assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(bad), ptr(proof), ptr(tau)) != 1
+        # tamper EVERY statement scalar -> reject (PI binding is complete)
+        for i in range(17):
+            bad = stl["pubs"].copy()
+            bad[i * 4] ^= np.uint64(1)
+            assert o.orc_plonk_verify(ctypes.c_void_p(pk), ptr(bad), ptr(proof),
+                                      ptr(tau)) != 1, f"statement scalar {i}"
         o.orc_plonk_pk_free(ctypes.c_void_p(pk))
 
     def test_seeds_vary(self, plib):
