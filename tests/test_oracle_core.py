@@ -116,6 +116,35 @@ class TestNTT:
             assert ref.from_mont(got, ref.R) == orig[i]
 
 
+    def test_convolution_theorem(self, orc):
+        """NTT(a) * NTT(b) pointwise == NTT(a conv b): pins the transform as
+        a true DFT over Fr, not just an involution."""
+        import random
+        n = 64
+        rng = random.Random(9)
+        a = [rng.randrange(ref.R) for _ in range(n // 2)] + [0] * (n // 2)
+        b = [rng.randrange(ref.R) for _ in range(n // 2)] + [0] * (n // 2)
+        conv = [0] * n
+        for i in range(n // 2):
+            for j in range(n // 2):
+                conv[i + j] = (conv[i + j] + a[i] * b[j]) % ref.R
+
+        def mont_arr(vals):
+            out = np.zeros(4 * n, dtype=np.uint64)
+            for i, v in enumerate(vals):
+                out[4 * i:4 * i + 4] = ref.int_to_limbs(ref.to_mont(v, ref.R))
+            return out
+
+        fa, fb, fc = mont_arr(a), mont_arr(b), mont_arr(conv)
+        orc.ntt(fa, n)
+        orc.ntt(fb, n)
+        orc.ntt(fc, n)
+        for i in range(n):
+            va = ref.from_mont(ref.limbs_to_int(fa[4 * i:4 * i + 4]), ref.R)
+            vb = ref.from_mont(ref.limbs_to_int(fb[4 * i:4 * i + 4]), ref.R)
+            vc = ref.from_mont(ref.limbs_to_int(fc[4 * i:4 * i + 4]), ref.R)
+            assert va * vb % ref.R == vc, f"slot {i}"
+
 class TestMSM:
     def _mk(self, orc, n, seed):
         rng = random.Random(seed)
