@@ -72,6 +72,14 @@ typedef struct RngProvingKey RngProvingKey;
 RngCtx* rng_prover_init(const uint8_t* srs_ptau, size_t len, uint64_t max_degree);
 void rng_ctx_free(RngCtx* ctx);
 
+/* Deterministic TEST SRS in the snarkjs subset layout parse_ptau_file reads
+ * (srs.rs:63-214): tau = keccak256("renegade-amd-srs-tau" || le64(seed)) mod r.
+ * Dev/test only — production passes real ceremony bytes to rng_prover_init.
+ * rng_srs_test_ptau_size returns the byte size for `power` (0 = bad power);
+ * rng_srs_gen_test_ptau fills a caller buffer of at least that size. */
+uint64_t rng_srs_test_ptau_size(int power);
+int rng_srs_gen_test_ptau(int power, uint64_t seed, uint8_t* out, size_t out_len);
+
 /* Library/device introspection */
 int rng_gpu_available(void);
 const char* rng_version(void);
@@ -194,6 +202,11 @@ void* rng_circ_from_scalars(int kind, const uint64_t* witness64,
 int rng_ws_sizes(int kind, uint64_t* out_num_witness, uint64_t* out_num_statement);
 int rng_witness_statement(int kind, uint64_t seed, uint64_t* out_witness,
                           uint64_t* out_statement);
+/* Party-aware vectors for the per-party validity kinds 4 and 9: party 0/1 of
+ * the seed's consistent production bundle (the private-settlement route needs
+ * all four external hints — native_proof_manager.rs:554-590). */
+int rng_witness_statement_party(int kind, uint64_t seed, uint64_t party,
+                                uint64_t* out_witness, uint64_t* out_statement);
 
 /* Fixed-seed circuit builders (deterministic witness/statement generators
  * mirroring the reference's test_helpers; returns an opaque CircuitTables

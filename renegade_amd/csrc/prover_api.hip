@@ -4,12 +4,16 @@
 // semantics of crates/circuits/circuit-types/src/primitives/srs.rs:63-214),
 // NTT plans, MSM pipeline.  Single translation unit: includes the kernel
 // files directly.
+#include <array>
+#include <atomic>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <memory>
 #include <mutex>
 #include <stdexcept>
+#include <thread>
 #include <vector>
 
 #include "../../include/rng_prover.h"
@@ -38,11 +42,26 @@ namespace rng {
 // concurrent rng_prove calls from a thread pool overlap on the GPU
 #define RNG_STREAM hipStreamPerThread
 
+// Teardown guard: main-thread TLS / static / Python-GC destructors can run
+// AFTER the HIP runtime's own atexit teardown, and a hipFree then faults
+// inside the dead runtime (r01 bench exited rc=139 this way).  We register
+// our atexit handler AFTER the first successful HIP call, so (LIFO order) it
+// runs BEFORE the runtime's teardown and flips this flag; every destructor
+// hipFree is gated on it.  Device memory skipped this way is reclaimed by
+// process exit anyway.
+static std::atomic<bool> g_hip_alive{true};
+static bool hip_alive() { return g_hip_alive.load(std::memory_order_relaxed); }
+static void hip_free_guarded(void* p) {
+    if (p && hip_alive()) hipFree(p);
+}
+
 static bool gpu_ok() {
     static int cached = -1;
     if (cached < 0) {
         int n = 0;
         cached = (hipGetDeviceCount(&n) == hipSuccess && n > 0) ? 1 : 0;
+        if (cached == 1)
+            std::atexit([] { g_hip_alive.store(false, std::memory_order_relaxed); });
     }
     return cached == 1;
 }
@@ -113,10 +132,10 @@ struct RngCtxImpl {
             for (Fr* b : {p->wst1_f, p->wst2_f, p->ta_f, p->tb_f, p->wst1_i,
                           p->wst2_i, p->ta_i, p->tb_i, p->scratch, p->gpow,
                           p->gpow_inv, p->xpow})
-                if (b) hipFree(b);
+                hip_free_guarded(b);
         }
-        if (srs_dev) hipFree(srs_dev);
-        if (srs_glv_dev) hipFree(srs_glv_dev);
+        hip_free_guarded(srs_dev);
+        hip_free_guarded(srs_glv_dev);
     }
 };
 
@@ -264,7 +283,7 @@ struct MsmScratch {
                         (void*)partials2, (void*)head_count, select_temp, scan_temp,
                         (void*)buckets, (void*)partials, (void*)window_sums,
                         (void*)result, (void*)glv, (void*)phi})
-            if (b) hipFree(b);
+            hip_free_guarded(b);
     }
 };
 
@@ -600,7 +619,7 @@ struct ProveScratch {  // per-context device scratch for proving at size n
     ~ProveScratch() {
         for (void* b : {(void*)w_coset, (void*)z_coset, (void*)pi_coset, (void*)q_buf,
                         (void*)tmp, (void*)stage, (void*)canon})
-            if (b) hipFree(b);
+            hip_free_guarded(b);
     }
 };
 
@@ -617,7 +636,7 @@ struct PlonkPkImpl {
     RngCtxImpl* ctx = nullptr;
     ~PlonkPkImpl() {
         for (void* b : {(void*)sel_coset, (void*)sig_coset, (void*)l1_coset})
-            if (b) hipFree(b);
+            hip_free_guarded(b);
     }
 };
 
@@ -1135,6 +1154,180 @@ RngCtx* rng_prover_init(const uint8_t* srs_ptau, size_t len, uint64_t max_degree
 }
 
 void rng_ctx_free(RngCtx* ctx) { delete ctx; }
+
+// ---------------- deterministic TEST SRS (product-side) ----------------
+// Dev/test deployments need a well-formed SRS without shipping ceremony
+// bytes; generating it HERE means the daemon never loads oracle code
+// (tier contract: oracle = test infrastructure only).  Byte-identical to
+// the test oracle's generator (same tau derivation, same snarkjs subset
+// layout parse_ptau_file reads — srs.rs:63-214); production passes real
+// ptau bytes to rng_prover_init instead.
+
+namespace {
+// minimal host G2 Jacobian over Fq2 (EFD a=0 short-Weierstrass formulas,
+// the same shapes as G1Jac) — only used for beta_h = tau*H at generation
+struct HG2Jac {
+    PFq2 X, Y, Z;
+    static HG2Jac from_xy(const PFq2& x, const PFq2& y) { return {x, y, PFq2::one()}; }
+    bool is_identity() const { return Z.eq(PFq2::zero()); }
+    HG2Jac dbl() const {
+        if (is_identity()) return *this;
+        PFq2 A = X.sqr(), B = Y.sqr(), C = B.sqr();
+        PFq2 D = X.add(B).sqr().sub(A).sub(C);
+        D = D.add(D);
+        PFq2 E = A.add(A).add(A);
+        PFq2 F = E.sqr();
+        HG2Jac r;
+        r.X = F.sub(D.add(D));
+        PFq2 C8 = C.add(C); C8 = C8.add(C8); C8 = C8.add(C8);
+        r.Y = E.mul(D.sub(r.X)).sub(C8);
+        r.Z = Y.mul(Z); r.Z = r.Z.add(r.Z);
+        return r;
+    }
+    HG2Jac add(const HG2Jac& o) const {
+        if (is_identity()) return o;
+        if (o.is_identity()) return *this;
+        PFq2 Z1Z1 = Z.sqr(), Z2Z2 = o.Z.sqr();
+        PFq2 U1 = X.mul(Z2Z2), U2 = o.X.mul(Z1Z1);
+        PFq2 S1 = Y.mul(o.Z).mul(Z2Z2), S2 = o.Y.mul(Z).mul(Z1Z1);
+        if (U1.eq(U2)) {
+            if (S1.eq(S2)) return dbl();
+            return {PFq2::one(), PFq2::one(), PFq2::zero()};
+        }
+        PFq2 H = U2.sub(U1);
+        PFq2 I = H.add(H).sqr();
+        PFq2 J = H.mul(I);
+        PFq2 rr = S2.sub(S1); rr = rr.add(rr);
+        PFq2 V = U1.mul(I);
+        HG2Jac r;
+        r.X = rr.sqr().sub(J).sub(V.add(V));
+        PFq2 SJ = S1.mul(J);
+        r.Y = rr.mul(V.sub(r.X)).sub(SJ.add(SJ));
+        r.Z = Z.add(o.Z).sqr().sub(Z1Z1).sub(Z2Z2).mul(H);
+        return r;
+    }
+};
+
+uint64_t srs_test_ptau_npoints(int power) { return (1ull << power) + 3; }
+}  // namespace
+
+uint64_t rng_srs_test_ptau_size(int power) {
+    if (power < 2 || power > 20) return 0;
+    return 12 + (12 + 44) + (12 + srs_test_ptau_npoints(power) * 64) + (12 + 256);
+}
+
+int rng_srs_gen_test_ptau(int power, uint64_t seed, uint8_t* out, size_t out_len) {
+    uint64_t need = rng_srs_test_ptau_size(power);
+    if (!out || !need || out_len < need) return RNG_ERR_BAD_ARG;
+    // tau = keccak256("renegade-amd-srs-tau" || le64(seed)) reduced mod r
+    uint8_t msg[28];
+    memcpy(msg, "renegade-amd-srs-tau", 20);
+    memcpy(msg + 20, &seed, 8);
+    uint8_t h32[32];
+    keccak256_h(msg, 28, h32);
+    u64 limbs[4];
+    memcpy(limbs, h32, 32);
+    Fr tau = Fr::from_canonical(limbs);
+    const uint64_t npoints = srs_test_ptau_npoints(power);
+    std::vector<std::array<u64, 4>> scalars(npoints);
+    Fr acc = Fr::one();
+    for (uint64_t i = 0; i < npoints; ++i) {
+        acc.to_canonical(scalars[i].data());
+        acc = acc.mul(tau);
+    }
+    // fixed-base doubling table G_b = 2^b * G
+    std::vector<G1Jac> table(256);
+    {
+        G1Aff g;
+        static const u64 gx[4] = G1_GEN_X_MONT, gy[4] = G1_GEN_Y_MONT;
+        memcpy(g.x.l, gx, 32);
+        memcpy(g.y.l, gy, 32);
+        table[0] = G1Jac::from_affine(g);
+    }
+    for (int j = 1; j < 256; ++j) table[j] = table[j - 1].dbl();
+    std::vector<G1Aff> pts(npoints);
+    unsigned nthreads = std::thread::hardware_concurrency();
+    if (nthreads == 0) nthreads = 1;
+    if (nthreads > 64) nthreads = 64;
+    std::vector<std::thread> ths;
+    std::atomic<uint64_t> next{0};
+    for (unsigned t = 0; t < nthreads; ++t)
+        ths.emplace_back([&] {
+            for (;;) {
+                uint64_t lo = next.fetch_add(512);
+                if (lo >= npoints) break;
+                uint64_t hi = lo + 512 < npoints ? lo + 512 : npoints;
+                for (uint64_t i = lo; i < hi; ++i) {
+                    G1Jac p = G1Jac::identity();
+                    const u64* s = scalars[i].data();
+                    for (int b = 0; b < 256; ++b)
+                        if ((s[b >> 6] >> (b & 63)) & 1) p = p.add(table[b]);
+                    Fq zi = p.Z.inverse();
+                    Fq zi2 = zi.sqr();
+                    pts[i].x = p.X.mul(zi2);
+                    pts[i].y = p.Y.mul(zi2.mul(zi));
+                }
+            }
+        });
+    for (auto& th : ths) th.join();
+    // beta_h = tau * H (canonical double-and-add, MSB first)
+    PFq2 hx, hy;
+    {
+        static const u64 xc0[4] = G2_GEN_X_C0_MONT, xc1[4] = G2_GEN_X_C1_MONT;
+        static const u64 yc0[4] = G2_GEN_Y_C0_MONT, yc1[4] = G2_GEN_Y_C1_MONT;
+        memcpy(hx.a.l, xc0, 32);
+        memcpy(hx.b.l, xc1, 32);
+        memcpy(hy.a.l, yc0, 32);
+        memcpy(hy.b.l, yc1, 32);
+    }
+    u64 tc[4];
+    tau.to_canonical(tc);
+    HG2Jac hjac = HG2Jac::from_xy(hx, hy);
+    HG2Jac bacc{PFq2::one(), PFq2::one(), PFq2::zero()};
+    for (int i = 255; i >= 0; --i) {
+        bacc = bacc.dbl();
+        if ((tc[i >> 6] >> (i & 63)) & 1) bacc = bacc.add(hjac);
+    }
+    PFq2 bx, by;
+    {
+        PFq2 zi = bacc.Z.inverse();
+        PFq2 zi2 = zi.sqr();
+        bx = bacc.X.mul(zi2);
+        by = bacc.Y.mul(zi2.mul(zi));
+    }
+    // --- serialize (layout of srs_to_ptau / parse_ptau_file) ---
+    uint8_t* w = out;
+    auto put = [&](const void* p, size_t n) { memcpy(w, p, n); w += n; };
+    auto put_u32 = [&](uint32_t v) { put(&v, 4); };
+    auto put_u64v = [&](uint64_t v) { put(&v, 8); };
+    put("ptau", 4);
+    put_u32(1);
+    put_u32(11);
+    put_u32(1);
+    put_u64v(44);
+    put_u32(32);
+    static const u64 qmod[4] = FQ_MODULUS;
+    put(qmod, 32);
+    put_u32((uint32_t)power);
+    put_u32((uint32_t)power);
+    put_u32(2);
+    put_u64v(npoints * 64);
+    for (uint64_t i = 0; i < npoints; ++i) {
+        put(pts[i].x.l, 32);
+        put(pts[i].y.l, 32);
+    }
+    put_u32(3);
+    put_u64v(256);
+    put(hx.a.l, 32);
+    put(hx.b.l, 32);
+    put(hy.a.l, 32);
+    put(hy.b.l, 32);
+    put(bx.a.l, 32);
+    put(bx.b.l, 32);
+    put(by.a.l, 32);
+    put(by.b.l, 32);
+    return RNG_OK;
+}
 
 const void* rng_srs_dev_bases(RngCtx* ctx, uint64_t* count) {
     if (count) *count = ctx->impl.srs_count;
@@ -2405,6 +2598,46 @@ int rng_witness_statement(int kind, uint64_t seed, uint64_t* out_w, uint64_t* ou
     }
 }
 
+// Party-aware test vectors for the two per-party validity kinds (4 = intent-
+// and-balance validity, 9 = output-balance validity): the private-settlement
+// route takes FOUR external hints (party 0/1 of each), all from one seed's
+// consistent bundle (native_proof_manager.rs:554-590).
+int rng_witness_statement_party(int kind, uint64_t seed, uint64_t party,
+                                uint64_t* out_w, uint64_t* out_s) {
+    try {
+        Wt w{(Fr*)out_w};
+        Wt s{(Fr*)out_s};
+        int p = (int)(party & 1);
+        ValidityBundle b;
+        validity_bundle_build(seed, b);
+        if (kind == 4) {
+            const ValidityWitness& vw = b.vw[p];
+            w.sint(vw.old_intent);
+            w.opening(vw.intent_opening_elems, vw.intent_opening_idx);
+            w.in(vw.intent);
+            w.f(vw.new_amount_public_share);
+            w.sbal(vw.old_balance);
+            w.opening(vw.balance_opening_elems, vw.balance_opening_idx);
+            w.bal(vw.balance);
+            w.pms(vw.post_match_balance_shares);
+            s.st(b.vst[p].to_scalars());
+        } else if (kind == 9) {
+            const ObValidityWitness& vw = b.ow[p];
+            w.sbal(vw.old_balance);
+            w.opening(vw.opening_elems, vw.opening_idx);
+            w.bal(vw.balance);
+            w.pms(vw.post_match_balance_shares);
+            s.st(b.ost[p].to_scalars());
+        } else {
+            return RNG_ERR_BAD_ARG;
+        }
+        return RNG_OK;
+    } catch (const std::exception& e) {
+        fprintf(stderr, "rng_witness_statement_party: %s\n", e.what());
+        return RNG_ERR_BAD_ARG;
+    }
+}
+
 // Build any circuit from caller-supplied witness/statement scalars (the
 // prover-service request shape).  `kind` table in include/rng_prover.h;
 // returns null if the witness does not satisfy the circuit.
@@ -3052,6 +3285,25 @@ int rng_verify(RngCtx* ctx, const RngProvingKey* pkw, const uint64_t* public_inp
             memcpy(a.y.l, proof + 9 * i + 4, 32);
             comms[i] = G1Jac::from_affine(a);
         }
+    }
+    // Reject malformed proofs BEFORE transcript replay / pairing: the
+    // reference's arkworks deserialization enforces curve membership and
+    // canonical limb ranges (ark-serialize Validate::Yes on Proof fields);
+    // accepting raw records here would feed off-curve points into the Miller
+    // loop (invalid-point soundness hazard — ADVICE r01).
+    {
+        const Fq b3 = Fq::from_u64(3);  // BN254: y^2 = x^3 + 3
+        for (int i = 0; i < 13; ++i) {
+            if (infs[i]) continue;
+            const uint64_t* rec = proof + 9 * i;
+            if (Fq::geq_mod(rec) || Fq::geq_mod(rec + 4)) return RNG_ERR_VERIFY;
+            Fq x, y;
+            memcpy(x.l, rec, 32);
+            memcpy(y.l, rec + 4, 32);
+            if (!y.sqr().eq(x.sqr().mul(x).add(b3))) return RNG_ERR_VERIFY;
+        }
+        for (int i = 0; i < 10; ++i)
+            if (Fr::geq_mod(proof + 117 + 4 * i)) return RNG_ERR_VERIFY;
     }
     Fr wire_evals[5], sigma_evals[4], z_shift_eval;
     {

@@ -136,12 +136,22 @@ class ProverService:
             # snarkjs layout srs.rs:63-214 parses)
             ptau = open(ptau_path, "rb").read()
         else:
-            # DEV/TEST fallback: deterministic generated SRS via the CPU
-            # oracle (test infrastructure); production deployments pass
-            # --ptau with the ceremony file
-            from tests.orc_bindings import OracleLib
-            orc = OracleLib(str(REPO / "oracle" / "liborc.so"))
-            ptau = orc.srs_generate_ptau(srs_power, seed=srs_seed)
+            # DEV/TEST fallback: deterministic generated SRS from the PRODUCT
+            # library (rng_srs_gen_test_ptau) — the daemon never loads oracle
+            # code; production deployments pass --ptau with the ceremony file
+            self.lib.rng_srs_test_ptau_size.restype = ctypes.c_uint64
+            self.lib.rng_srs_test_ptau_size.argtypes = [ctypes.c_int]
+            self.lib.rng_srs_gen_test_ptau.argtypes = [
+                ctypes.c_int, ctypes.c_uint64, ctypes.POINTER(ctypes.c_uint8),
+                ctypes.c_size_t]
+            size = self.lib.rng_srs_test_ptau_size(srs_power)
+            if not size:
+                raise ValueError(f"bad srs power {srs_power}")
+            buf = (ctypes.c_uint8 * size)()
+            rc = self.lib.rng_srs_gen_test_ptau(srs_power, srs_seed, buf, size)
+            if rc != 0:
+                raise RuntimeError(f"rng_srs_gen_test_ptau rc={rc}")
+            ptau = bytes(buf)
             print(f"WARNING: using deterministic TEST SRS (power {srs_power}, "
                   f"seed {srs_seed}); pass --ptau for production", flush=True)
         self.max_degree = (1 << srs_power) + 2
@@ -149,7 +159,6 @@ class ProverService:
         self.pks = {}
         self.pk_meta = {}
         self.lock = threading.Lock()
-        self.seed_ctr = int.from_bytes(os.urandom(4), "little")
 
     def _sig(self):
         lib = self.lib
@@ -188,9 +197,10 @@ class ProverService:
         """Preprocess-or-cache, then prove; returns (proof, hint, meta)."""
         lib = self.lib
         n, npub, wires, pubs = self._preprocess(circuit_name, handle)
-        with self.lock:
-            self.seed_ctr += 1
-            seed = self.seed_ctr
+        # fresh blinder seed per request from the OS CSPRNG — never derive
+        # successive proofs' blinders from a counter (the reference blinds
+        # with thread_rng at traits.rs:994)
+        seed = int.from_bytes(os.urandom(8), "little")
         pk = self.pks[circuit_name]
         proof = np.zeros(157, dtype=np.uint64)
         hint = np.zeros(4 * (n + 2) + 9, dtype=np.uint64) if want_hint else None
@@ -302,30 +312,21 @@ class ProverService:
             raise ValueError("unsatisfied witness/statement")
         proof, hint, meta = self._prove_tables("intent_and_balance_private_settlement",
                                                h, True)
-        n = meta["n"]
+        n_mine = int(meta["n"])
         # 4 link proofs: settlement hint vs the 4 supplied hints, one per
         # link group (party0, party1 validity; party0, party1 output balance;
-        # group order = creation order in the circuit)
+        # group order = creation order in the circuit).  Real hints come from
+        # DIFFERENT domains (intent-and-balance validity n=16384, output-
+        # balance validity n=8192), so each link runs cross-domain via _link:
+        # zero-extend to the larger domain, prove under that domain's PK
+        # (requires --preload; native_proof_manager.rs:746-777).
         names = ["validity_link_proof_0", "validity_link_proof_1",
                  "output_balance_link_proof_0", "output_balance_link_proof_1"]
         group_order = [0, 2, 1, 3]  # groups created party0(v,o), party1(v,o)
         out = {"proof": [str(int(x)) for x in proof]}
         for name, gidx in zip(names, group_order):
-            align, off, count = meta["lg"][gidx]
-            # external hints arrive as flat limb arrays of the hint layout
-            ext_hint = np.array([int(x) for x in body[name.replace("_proof", "_hint")]],
-                                dtype=np.uint64)
-            assert ext_hint.size == 4 * (n + 2) + 9, "bad hint size"
-            lp = np.zeros(18, dtype=np.uint64)
-            # validity hint first, settlement second (native_proof_manager.rs:746-777)
-            rc = self.lib.rng_link_proofs(self.ctx.h,
-                                          ctypes.c_void_p(
-                                              self.pks["intent_and_balance_private_settlement"]),
-                                          ptr(ext_hint), ptr(hint), int(align), int(off),
-                                          int(count), ptr(lp))
-            if rc != 0:
-                raise RuntimeError(f"rng_link_proofs rc={rc}")
-            out[name] = [str(int(x)) for x in lp]
+            ext_hint = self._hint_from_body(body, name.replace("_proof", "_hint"))
+            out[name] = self._link(hint, ext_hint, meta["lg"][gidx], n_mine)
         return out
 
 
@@ -338,8 +339,9 @@ def create_app(service=None, password=None):
     def auth(request: Request):
         if password is None:
             return
+        import hmac
         hdr = request.headers.get("authorization", "")
-        if hdr != f"Bearer {password}":
+        if not hmac.compare_digest(hdr.encode(), f"Bearer {password}".encode()):
             raise HTTPException(status_code=401, detail="bad password")
 
     @app.get("/health")

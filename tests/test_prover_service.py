@@ -233,3 +233,110 @@ class TestProverService:
         with ThreadPoolExecutor(max_workers=8) as ex:
             proofs = list(ex.map(hit, jobs * 3))  # 12 concurrent requests
         assert all(len(p) == 157 for p in proofs)
+
+    def test_settlement_with_real_validity_hints(self, client, orc):
+        """The production headline flow with REAL cross-domain hints: four
+        validity proofs (intent-and-balance n=16384, output-balance n=8192,
+        party 0 and 1 of one seed's bundle) are proved through their own
+        routes, and their link hints feed the private-settlement route
+        (n=4096).  Each returned link proof is then verified with the oracle
+        link verifier against the two proofs' wire-0 commitments
+        (native_proof_manager.rs:554-590; ADVICE r01 #1 regression test)."""
+        c, svc = client
+        lib = svc.lib
+        hdr = {"authorization": "Bearer hunter2"}
+        seed = 9
+        lib.rng_witness_statement_party.restype = ctypes.c_int
+        lib.rng_witness_statement_party.argtypes = [ctypes.c_int, ctypes.c_uint64,
+                                                    ctypes.c_uint64, U64P, U64P]
+
+        def vectors_party(kind, party):
+            nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+            assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
+            w = np.zeros(4 * nw.value, dtype=np.uint64)
+            s = np.zeros(4 * ns.value, dtype=np.uint64)
+            assert lib.rng_witness_statement_party(kind, seed, party, ptr(w),
+                                                   ptr(s)) == 0
+            return w, s
+
+        def prove_route(route, kind, party):
+            w, s = vectors_party(kind, party)
+            r = c.post(route, headers=hdr,
+                       json={"witness": _scal_json(w), "statement": _scal_json(s)})
+            assert r.status_code == 200, f"{route} p{party}: {r.text}"
+            out = r.json()
+            return (np.array([int(x) for x in out["proof"]], dtype=np.uint64),
+                    np.array([int(x) for x in out["link_hint"]], dtype=np.uint64))
+
+        ib = [prove_route("/prove-intent-and-balance-validity", 4, p) for p in (0, 1)]
+        ob = [prove_route("/prove-output-balance-validity", 9, p) for p in (0, 1)]
+        # real hints come from domains LARGER than the settlement's
+        assert svc._hint_n(ib[0][1]) > 4096 and svc._hint_n(ob[0][1]) > 4096
+
+        w = np.zeros(64 * 4, dtype=np.uint64)
+        s = np.zeros(17 * 4, dtype=np.uint64)
+        lib.rng_settlement_witness_statement.argtypes = [ctypes.c_uint64, U64P, U64P]
+        lib.rng_settlement_witness_statement(seed, ptr(w), ptr(s))
+        body = {"witness": _scal_json(w), "statement": _scal_json(s),
+                "validity_link_hint_0": [str(int(x)) for x in ib[0][1]],
+                "validity_link_hint_1": [str(int(x)) for x in ib[1][1]],
+                "output_balance_link_hint_0": [str(int(x)) for x in ob[0][1]],
+                "output_balance_link_hint_1": [str(int(x)) for x in ob[1][1]]}
+        r = c.post("/prove-intent-and-balance-private-settlement", headers=hdr,
+                   json=body)
+        assert r.status_code == 200, r.text
+        out = r.json()
+        sproof = np.array([int(x) for x in out["proof"]], dtype=np.uint64)
+
+        # --- verify the four link proofs with the oracle link verifier ---
+        meta = svc.pk_meta["intent_and_balance_private_settlement"]
+        legs = [("validity_link_proof_0", 0, ib[0], 16384),
+                ("validity_link_proof_1", 2, ib[1], 16384),
+                ("output_balance_link_proof_0", 1, ob[0], 8192),
+                ("output_balance_link_proof_1", 3, ob[1], 8192)]
+        o = orc.lib
+        o.orc_plonk_preprocess.restype = ctypes.c_void_p
+        o.orc_plonk_preprocess.argtypes = [ctypes.c_uint64, ctypes.c_uint64, U64P,
+                                           U64P, U64P, ctypes.c_uint64]
+        o.orc_plonk_link_verify.argtypes = [ctypes.c_void_p, U64P, U64P, U64P] + \
+            [ctypes.c_uint64] * 3 + [U64P]
+        tau = np.zeros(4, dtype=np.uint64)
+        o.orc_derive_tau.argtypes = [ctypes.c_uint64, U64P]
+        o.orc_derive_tau(42, ptr(tau))
+        # oracle PKs for the two big domains (link runs under the larger pk)
+        power = 15
+        ptau = orc.srs_generate_ptau(power, seed=42)
+        g1, _, _ = orc.srs_parse(ptau, (1 << power) + 2)
+        srs = np.ascontiguousarray(g1).reshape(-1)
+        opks = {}
+        for route, nbig in [("prove-intent-and-balance-validity", 16384),
+                            ("prove-output-balance-validity", 8192)]:
+            builder = {16384: ("rng_circ_build_validity", (0,)),
+                       8192: ("rng_circ_build_ob_validity", (0,))}[nbig]
+            fn = getattr(lib, builder[0])
+            fn.restype = ctypes.c_void_p
+            fn.argtypes = [ctypes.c_uint64] * 2
+            h = fn(seed, *builder[1])
+            assert h
+            n = lib.rng_circ_n(h)
+            npub = lib.rng_circ_npub(h)
+            sel = np.zeros(13 * n * 4, dtype=np.uint64)
+            sigma = np.zeros(5 * n, dtype=np.uint64)
+            wires = np.zeros(5 * n * 4, dtype=np.uint64)
+            pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+            lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+            lib.rng_circ_free(h)
+            opks[nbig] = o.orc_plonk_preprocess(n, npub, ptr(sel), ptr(sigma),
+                                                ptr(srs), (1 << power) + 3)
+            assert opks[nbig]
+        for name, gidx, (vproof, _vhint), nbig in legs:
+            align, off, cnt = (int(x) for x in meta["lg"][gidx])
+            lp = np.array([int(x) for x in out[name]], dtype=np.uint64)
+            assert lp.size == 18
+            comm_a = np.ascontiguousarray(vproof[:9])   # validity wire-0 comm
+            comm_b = np.ascontiguousarray(sproof[:9])   # settlement wire-0 comm
+            ok = o.orc_plonk_link_verify(ctypes.c_void_p(opks[nbig]), ptr(comm_a),
+                                         ptr(comm_b), ptr(lp),
+                                         ctypes.c_uint64(align), ctypes.c_uint64(off),
+                                         ctypes.c_uint64(cnt), ptr(tau))
+            assert ok == 1, f"{name} failed oracle link verification"
