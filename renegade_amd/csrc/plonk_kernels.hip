@@ -83,16 +83,16 @@ __global__ __launch_bounds__(256) void k_bench_frmul(Fr* io, uint32_t iters, int
     Fr b = io[(t + 1) & 1023];
     if (dep) {
         for (uint32_t i = 0; i < iters; ++i)
-            a = (VARIANT == 0) ? a.mul(b) : (VARIANT == 1 ? a.mul32(b) : a.mul_sos(b));
+            a = (VARIANT == 0) ? a.mul_cios(b) : (VARIANT == 1 ? a.mul32(b) : a.mul_sos(b));
     } else {
         Fr c = io[(t + 2) & 1023], d = io[(t + 3) & 1023];
         Fr e = io[(t + 4) & 1023], f = io[(t + 5) & 1023];
         for (uint32_t i = 0; i < iters / 4; ++i) {
             if (VARIANT == 0) {
-                a = a.mul(b);
-                c = c.mul(b);
-                d = d.mul(b);
-                e = e.mul(f);
+                a = a.mul_cios(b);
+                c = c.mul_cios(b);
+                d = d.mul_cios(b);
+                e = e.mul_cios(f);
             } else if (VARIANT == 1) {
                 a = a.mul32(b);
                 c = c.mul32(b);
