@@ -104,20 +104,13 @@ struct Fp4 {
 
     RNG_HD Fp4 dbl() const { return add(*this); }
 
-    // Montgomery multiplication, N = 4.  DEVICE: SOS (full 4x4 product with
-    // independent partial products, then 4-round reduction) — measured ~1.5x
-    // fewer VALU issue slots than CIOS on gfx950 (505 vs 794 incl. mandatory
-    // s_nop hazard slots; the CIOS inner loop serializes on VCC carries).
-    // HOST: CIOS (Acar) — keeping a different formulation on the host side
-    // preserves the host/device cross-check.  All variants are
-    // bit-identical (tests pin them against each other and a Python bignum).
-    RNG_HD Fp4 mul(const Fp4& b) const {
-#ifdef __HIP_DEVICE_COMPILE__
-        return mul_sos(b);
-#else
-        return mul_cios(b);
-#endif
-    }
+    // Montgomery multiplication, N = 4: CIOS (Acar).  Formulation A/B on a
+    // real MI355X (rng_bench_frmul, r02): CIOS-u64 78.6 Gmul/s, SOS-u64 57.7,
+    // CIOS-u32 49.6 — despite SOS having ~1.5x fewer static issue slots, its
+    // dynamically-indexed t[] spills into s_set_gpr_idx sequences and loses;
+    // the u32 variant drowns in pack/unpack movs.  All variants are
+    // bit-identical (pinned against each other and a Python bignum).
+    RNG_HD Fp4 mul(const Fp4& b) const { return mul_cios(b); }
 
     // CIOS Montgomery multiplication (Acar), N = 4.
     RNG_HD Fp4 mul_cios(const Fp4& b) const {

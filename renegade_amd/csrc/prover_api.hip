@@ -2628,6 +2628,33 @@ int rng_witness_statement_party(int kind, uint64_t seed, uint64_t party,
             w.bal(vw.balance);
             w.pms(vw.post_match_balance_shares);
             s.st(b.ost[p].to_scalars());
+        } else if (kind == 10) {
+            // the BUNDLE's settlement witness/statement (64/17 scalars, same
+            // order as rng_settlement_witness_statement).  Needed because
+            // validity_bundle_build mutates the settlement witness after
+            // building it (real Schnorr authorities on the input balances),
+            // so only the bundle's version links against the bundle's
+            // validity proofs.  `party` is ignored.
+            const SettlementWitness& sw = b.sw;
+            std::vector<Fr> ws;
+            for (int i = 0; i < 2; ++i) {
+                auto a = sw.p[i].obligation.to_scalars();
+                auto bi = sw.p[i].intent.to_scalars();
+                auto c = sw.p[i].input_balance.to_scalars();
+                auto d = sw.p[i].pre_in_shares.to_scalars();
+                auto e = sw.p[i].output_balance.to_scalars();
+                auto f = sw.p[i].pre_out_shares.to_scalars();
+                ws.insert(ws.end(), a.begin(), a.end());
+                ws.insert(ws.end(), bi.begin(), bi.end());
+                ws.push_back(sw.p[i].pre_amount_share);
+                ws.insert(ws.end(), c.begin(), c.end());
+                ws.insert(ws.end(), d.begin(), d.end());
+                ws.insert(ws.end(), e.begin(), e.end());
+                ws.insert(ws.end(), f.begin(), f.end());
+            }
+            auto ss = b.sst.to_scalars();
+            memcpy(out_w, ws.data(), ws.size() * sizeof(Fr));
+            memcpy(out_s, ss.data(), ss.size() * sizeof(Fr));
         } else {
             return RNG_ERR_BAD_ARG;
         }

@@ -273,10 +273,12 @@ class TestProverService:
         # real hints come from domains LARGER than the settlement's
         assert svc._hint_n(ib[0][1]) > 4096 and svc._hint_n(ob[0][1]) > 4096
 
+        # kind 10 = the BUNDLE's settlement witness (validity_bundle_build
+        # mutates the settlement witness after building it, so only this
+        # version shares link values with the validity proofs above)
         w = np.zeros(64 * 4, dtype=np.uint64)
         s = np.zeros(17 * 4, dtype=np.uint64)
-        lib.rng_settlement_witness_statement.argtypes = [ctypes.c_uint64, U64P, U64P]
-        lib.rng_settlement_witness_statement(seed, ptr(w), ptr(s))
+        assert lib.rng_witness_statement_party(10, seed, 0, ptr(w), ptr(s)) == 0
         body = {"witness": _scal_json(w), "statement": _scal_json(s),
                 "validity_link_hint_0": [str(int(x)) for x in ib[0][1]],
                 "validity_link_hint_1": [str(int(x)) for x in ib[1][1]],
