@@ -419,53 +419,64 @@ def main():
     desc = Desc(n, npub, ptr(sel), ptr(sigma), 0, None)
     pk = lib.rng_preprocess(ctx.h, ctypes.byref(desc))
     assert pk, "rng_preprocess failed"
+    lib.rng_prove_cohort.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_uint64, U64P, U64P, U64P, U64P, U64P]
     proof = np.zeros(157, dtype=np.uint64)
 
-    def step(seed, buf=proof):
+    def prove_one(seed, buf=proof):
         rc = lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
                            ctypes.c_uint64(seed), ptr(buf), None)
         assert rc == 0, f"rng_prove rc={rc}"
 
-    # determinism spot check
-    step(12345)
-    p1 = proof.copy()
-    step(12345)
-    assert np.array_equal(p1, proof), "nondeterministic proof"
+    # one STEP = one COHORT of `jobs` proofs advanced in lockstep with fused
+    # per-round MSMs (rng_prove_cohort) — the batched shape of the
+    # reference's proof-job pool (native_proof_manager.rs:143-148).  A few
+    # cohorts stay in flight on separate threads/streams so host phases of
+    # one overlap GPU phases of another; value stays robust at any --steps.
+    k = args.jobs
+    wires_all = np.tile(wires, k)
+    pubs_all = np.tile(pubs, k)
 
-    # thread pool: independent proofs in flight per GPU (weak scaling within
-    # the device, mirroring native_proof_manager.rs:143-148)
+    def step(seed_base, buf):
+        seeds = (seed_base + np.arange(k, dtype=np.uint64)).astype(np.uint64)
+        rc = lib.rng_prove_cohort(ctx.h, ctypes.c_void_p(pk), k, ptr(wires_all),
+                                  ptr(pubs_all), ptr(seeds), ptr(buf), None)
+        assert rc == 0, f"rng_prove_cohort rc={rc}"
+
+    # determinism + cohort-vs-single parity spot check
+    cbuf0 = np.zeros(157 * k, dtype=np.uint64)
+    step(12345, cbuf0)
+    prove_one(12345)
+    assert np.array_equal(cbuf0[:157], proof), "cohort[0] differs from single proof"
+    prove_one(12345 + k - 1)
+    assert np.array_equal(cbuf0[157 * (k - 1):], proof), \
+        "cohort[k-1] differs from single proof"
+
     from concurrent.futures import ThreadPoolExecutor
-    pool = ThreadPoolExecutor(max_workers=args.jobs)
-    bufs = [np.zeros(157, dtype=np.uint64) for _ in range(args.jobs)]
-    # W untimed warmup steps (one step = one batch of `jobs` proofs), which
-    # also warms every worker thread's device scratch
-    for w in range(max(1, args.warmup)):
-        list(pool.map(lambda j: step(500 + w * args.jobs + j, bufs[j]),
-                      range(args.jobs)))
+    inflight = max(1, int(os.environ.get("RNG_BENCH_INFLIGHT", "3")))
+    pool = ThreadPoolExecutor(max_workers=inflight)
+    bufs = [np.zeros(157 * k, dtype=np.uint64) for _ in range(inflight)]
+    for w in range(max(1, min(args.warmup, 8))):
+        list(pool.map(lambda j: step(500 + (w * inflight + j) * 10 * k, bufs[j]),
+                      range(inflight)))
     ctx.sync()
 
-    # one STEP = one batch of `jobs` proofs submitted concurrently, so the
-    # stream pool stays saturated whatever --steps the driver passes
-    # (--steps 20 under-filled the pool when a step was
-    # a single proof).  All steps' proofs are in the pool queue at once; the
-    # timed region is steady-state.
-    total_proofs = args.steps * args.jobs
+    total_proofs = args.steps * k
     if dist:
         dist.barrier()
     ctx.sync()
     t0 = time.perf_counter()
-    futs = [pool.submit(step, 10_000 + rank * 1_000_000 + i, bufs[i % args.jobs])
-            for i in range(total_proofs)]
+    futs = [pool.submit(step, 10_000 + rank * 10_000_000 + i * 10 * k,
+                        bufs[i % inflight])
+            for i in range(args.steps)]
     for f in futs:
         f.result()
     ctx.sync()
-    # concurrency correctness: a proof produced under the thread pool must be
-    # bit-identical to the same seed proved alone.  bufs[0] holds the proof of
-    # the LAST step that used it; recompute that seed serially.
-    last0 = total_proofs - 1 - ((total_proofs - 1) % args.jobs)
-    check = np.zeros(157, dtype=np.uint64)
-    step(10_000 + rank * 1_000_000 + last0, check)
-    assert np.array_equal(check, bufs[0]), "threaded proof differs from serial"
+    # concurrency correctness: a cohort proof produced under the thread pool
+    # must be bit-identical to the same seed proved alone
+    last0 = args.steps - 1 - ((args.steps - 1) % inflight)
+    prove_one(10_000 + rank * 10_000_000 + last0 * 10 * k)
+    assert np.array_equal(proof, bufs[0][:157]), "pooled cohort differs from serial"
     if dist:
         import torch
         # NCCL collectives operate on device tensors
@@ -583,10 +594,11 @@ def main():
                 "domain_n": int(n),
                 "num_public": int(npub),
                 "srs_power": power,
-                "prover_threads": args.jobs,
+                "prover_mode": "cohort (lockstep rounds, fused per-round MSMs)",
                 "proofs_per_step": args.jobs,
                 "total_proofs": total_proofs * n_gpus,
-                "parallelism": f"independent proofs x{n_gpus} gpus x{args.jobs} threads",
+                "parallelism": f"x{n_gpus} gpus, cohort={args.jobs}, "
+                               f"{os.environ.get('RNG_BENCH_INFLIGHT', '3')} in flight",
             },
             "roofline": roofline,
             "cpu_baseline": cb,

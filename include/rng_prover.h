@@ -154,6 +154,18 @@ int rng_prove(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* wires,
               const uint64_t* public_inputs, uint64_t seed,
               uint64_t* out_proof, uint64_t* out_link_hint);
 
+/* Cohort prover: k proofs of the SAME circuit advanced in lockstep, each
+ * round's commitments fused into one GPU MSM run — the batched shape of the
+ * reference's proof-job pool (native_proof_manager.rs:143-148,193-198).
+ * wires = k consecutive 5n-scalar wire blocks, public_inputs = k consecutive
+ * statement blocks, seeds = k blinder seeds; out_proofs = k consecutive
+ * 157-u64 proofs; out_link_hints (optional) = k consecutive hint buffers.
+ * Proof p is bit-identical to rng_prove with seeds[p].  1 <= k <= 128. */
+int rng_prove_cohort(RngCtx* ctx, const RngProvingKey* pk, uint64_t k,
+                     const uint64_t* wires, const uint64_t* public_inputs,
+                     const uint64_t* seeds, uint64_t* out_proofs,
+                     uint64_t* out_link_hints);
+
 int rng_verify(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* public_inputs,
                const uint64_t* proof);
 

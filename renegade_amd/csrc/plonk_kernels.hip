@@ -73,6 +73,27 @@ __global__ __launch_bounds__(256) void k_copy_pad(const Fr* src, uint32_t len, F
     dst[i] = (i < len) ? src[i] : Fr::zero();
 }
 
+// batched coset scaling for nb polynomials of `stride` coefficients each
+// stored back to back: data[b*stride + j] *= table[j]  (table >= stride long)
+__global__ __launch_bounds__(256) void k_mul_pointwise_mod(Fr* data, const Fr* table,
+                                                           uint32_t stride,
+                                                           uint64_t total) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < total) data[i] = data[i].mul(table[i % stride]);
+}
+
+// batched zero-extension: nb polys of `len` coefficients at stride `len`
+// in src -> nb polys padded to m in dst
+__global__ __launch_bounds__(256) void k_copy_pad_batch(const Fr* src, uint32_t len,
+                                                        Fr* dst, uint32_t m,
+                                                        uint64_t total /* nb*m */) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    uint64_t b = i / m;
+    uint32_t j = (uint32_t)(i - b * m);
+    dst[i] = (j < len) ? src[b * len + j] : Fr::zero();
+}
+
 // ---- field-mul microbenchmark kernels (A/B the Montgomery formulations) ----
 // Each thread runs `iters` muls: dep = a dependent chain (latency),
 // otherwise 4 independent chains (throughput).

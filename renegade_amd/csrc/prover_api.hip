@@ -6,7 +6,9 @@
 // files directly.
 #include <array>
 #include <atomic>
+#include <condition_variable>
 #include <cstdio>
+#include <functional>
 #include <cstdlib>
 #include <cstring>
 #include <map>
@@ -68,6 +70,88 @@ static bool gpu_ok() {
 
 static thread_local double tls_msm_times[5] = {0, 0, 0, 0, 0};
 static thread_local double tls_ntt_times[2] = {0, 0};
+
+// Persistent host worker pool: parallelizes the per-proof host phases of
+// cohort proving (grand product, evaluations, linearization) and the
+// per-poly window folds of large fused MSM batches.  One job at a time;
+// concurrent callers queue.  Never destroyed (leaked) so there is no
+// thread-join ordering problem at process exit.
+class HostPool {
+  public:
+    static HostPool& inst() {
+        static HostPool* p = new HostPool();
+        return *p;
+    }
+
+    void parallel_for(uint32_t count, const std::function<void(uint32_t)>& fn) {
+        if (count == 0) return;
+        if (count == 1 || workers_.empty()) {
+            for (uint32_t i = 0; i < count; ++i) fn(i);
+            return;
+        }
+        std::unique_lock<std::mutex> job_lk(job_mu_);
+        {
+            std::lock_guard<std::mutex> lk(mu_);
+            fn_ = &fn;
+            count_ = count;
+            next_.store(0, std::memory_order_relaxed);
+            done_.store(0, std::memory_order_relaxed);
+            ++generation_;
+        }
+        cv_.notify_all();
+        work();  // caller participates
+        std::unique_lock<std::mutex> lk(mu_);
+        done_cv_.wait(lk, [&] { return done_.load(std::memory_order_acquire) >= count_; });
+        fn_ = nullptr;
+    }
+
+  private:
+    HostPool() {
+        unsigned n = std::thread::hardware_concurrency();
+        if (n == 0) n = 4;
+        if (n > 32) n = 32;
+        for (unsigned t = 0; t + 1 < n; ++t)
+            workers_.emplace_back([this] { worker_loop(); });
+    }
+    void work() {
+        const std::function<void(uint32_t)>* fn;
+        uint32_t count;
+        {
+            std::lock_guard<std::mutex> lk(mu_);
+            fn = fn_;
+            count = count_;
+        }
+        if (!fn) return;
+        for (;;) {
+            uint32_t i = next_.fetch_add(1, std::memory_order_relaxed);
+            if (i >= count) break;
+            (*fn)(i);
+            if (done_.fetch_add(1, std::memory_order_acq_rel) + 1 >= count) {
+                std::lock_guard<std::mutex> lk(mu_);
+                done_cv_.notify_all();
+            }
+        }
+    }
+    void worker_loop() {
+        uint64_t seen = 0;
+        for (;;) {
+            {
+                std::unique_lock<std::mutex> lk(mu_);
+                cv_.wait(lk, [&] { return generation_ != seen; });
+                seen = generation_;
+            }
+            work();
+        }
+    }
+    std::mutex job_mu_;
+    std::mutex mu_;
+    std::condition_variable cv_, done_cv_;
+    std::vector<std::thread> workers_;
+    const std::function<void(uint32_t)>* fn_ = nullptr;
+    uint32_t count_ = 0;
+    uint64_t generation_ = 0;
+    std::atomic<uint32_t> next_{0}, done_{0};
+};
 
 struct EvtTimer {
     hipEvent_t ev[8];
@@ -208,7 +292,9 @@ static NttPlan* get_plan(RngCtxImpl* ctx, uint32_t n, uint64_t batch) {
 // out-of-place for n > NTT_SMALL_MAX (result in out); in-place for small n.
 static int ntt_dev_run(RngCtxImpl* ctx, Fr* data, Fr* out, uint32_t n, uint64_t batch,
                        bool inverse, hipStream_t stream = 0) {
-    NttPlan* p = get_plan(ctx, n, batch);
+    // batch=0 to get_plan: tables only — ntt_dev_run never touches the
+    // plan's public-API scratch (rng_ntt_fr_dev sizes it itself)
+    NttPlan* p = get_plan(ctx, n, 0);
     if (!p) return RNG_ERR_HIP;
     if (n <= NTT_SMALL_MAX) {
         Fr* wst = inverse ? p->wst1_i : p->wst1_f;
@@ -526,7 +612,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
                              hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     et.collect(tls_msm_times, 5);
-    for (uint32_t b = 0; b < B; ++b) {
+    auto fold_one = [&](uint32_t b) {
         G1Jac ws[32];
         for (uint32_t w = 0; w < W; ++w) {
             uint32_t g = b * W + w;
@@ -541,7 +627,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
             acc = acc.add(ws[w]);
         }
         h_result[b] = acc;
-    }
+    };
+    if (B >= 8)
+        HostPool::inst().parallel_for(B, fold_one);
+    else
+        for (uint32_t b = 0; b < B; ++b) fold_one(b);
     return RNG_OK;
 }
 
@@ -701,6 +791,461 @@ static int commit_dev(RngCtxImpl* ctx, const std::vector<Fr>& coeffs, G1Aff* out
                       bool* out_inf) {
     const std::vector<Fr>* p = &coeffs;
     return commit_dev_batch(ctx, &p, 1, out, out_inf);
+}
+
+// ---------------- cohort proving ----------------
+// The reference proves many jobs concurrently from a rayon pool
+// (native_proof_manager.rs:143-148,193-198).  Per-proof GPU runs at n=4096
+// are aggregation-latency chains (r01 profile: window_combine 30% +
+// seg_merge 22% of proof-workload GPU time), so the cohort prover advances
+// k proofs in LOCKSTEP — R1 for all, R2 for all, … — and fuses each round's
+// commitments into ONE msm_dev_run (key group g = poly*W + window already
+// supports a batch dimension).  Transcripts and blinders stay per-proof
+// (each transcript consumes only its own commitments), so cohort proofs are
+// bit-identical to rng_prove with the same seed.  Host-side phases (grand
+// product, evaluations, linearization) run on a persistent worker pool
+// (HostPool, defined before msm_dev_run so its window fold can use it).
+
+static void h_transcript_init(HostTranscript& tr, const PlonkPkImpl& pk, const Fr* pubs);
+
+struct CohortScratch {  // per-thread device scratch for cohort proving
+    uint64_t n = 0, k = 0;
+    Fr* stage = nullptr;     // max(5k*(n+3), k*m) staging for commits/NTTs
+    uint64_t* canon = nullptr;  // canonical scalars for the fused MSM
+    Fr* coset_in = nullptr;  // 7k*(n+3) coefficient staging for the coset batch
+    Fr* coset_out = nullptr; // 7k*m coset evaluations (5 wires + z + PI per proof)
+    Fr* q_all = nullptr;     // k*m quotient evals
+    Fr* ntt_tmp = nullptr;   // 7k*m NTT ping-pong
+    ~CohortScratch() {
+        for (void* b : {(void*)stage, (void*)canon, (void*)coset_in, (void*)coset_out,
+                        (void*)q_all, (void*)ntt_tmp})
+            hip_free_guarded(b);
+    }
+};
+
+static thread_local std::unique_ptr<CohortScratch> tls_cohort_scratch;
+
+static int cohort_scratch_ensure(uint64_t n, uint64_t k) {
+    if (tls_cohort_scratch && tls_cohort_scratch->n >= n && tls_cohort_scratch->k >= k)
+        return RNG_OK;
+    uint64_t m = 8 * n;
+    tls_cohort_scratch = std::make_unique<CohortScratch>();
+    CohortScratch* s = tls_cohort_scratch.get();
+    uint64_t stage_elems = 5 * k * (n + 3);
+    if (stage_elems < k * m) stage_elems = k * m;
+    HIP_CHECK(hipMalloc(&s->stage, stage_elems * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->canon, stage_elems * 4 * 8));
+    HIP_CHECK(hipMalloc(&s->coset_in, 7 * k * (n + 3) * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->coset_out, 7 * k * m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->q_all, k * m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->ntt_tmp, 7 * k * m * sizeof(Fr)));
+    s->n = n;
+    s->k = k;
+    return RNG_OK;
+}
+
+// Fused commitment of B host polynomials (any B; shorter polys zero-padded
+// to the longest) through ONE msm_dev_run.  Host window fold + affine
+// conversion are parallelized across the pool for large B.
+static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
+                         uint32_t B, G1Aff* out, bool* out_inf) {
+    CohortScratch* s = tls_cohort_scratch.get();
+    uint64_t m = 0;
+    for (uint32_t b = 0; b < B; ++b)
+        if (polys[b]->size() > m) m = polys[b]->size();
+    if (m > ctx->srs_count) return RNG_ERR_BAD_ARG;
+    // pack on the host (pool) and ship ONE H2D copy: B small async copies of
+    // pageable memory each pay a staging round trip
+    static thread_local std::vector<Fr> packed;
+    packed.resize(B * m);
+    HostPool::inst().parallel_for(B, [&](uint32_t b) {
+        memcpy(packed.data() + (size_t)b * m, polys[b]->data(),
+               polys[b]->size() * sizeof(Fr));
+        memset(packed.data() + (size_t)b * m + polys[b]->size(), 0,
+               (m - polys[b]->size()) * sizeof(Fr));
+    });
+    HIP_CHECK(hipMemcpyAsync(s->stage, packed.data(), B * m * sizeof(Fr),
+                             hipMemcpyHostToDevice, RNG_STREAM));
+    uint32_t blocks = (uint32_t)((B * m + 255) / 256);
+    hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, RNG_STREAM,
+                       s->stage, s->canon, (uint32_t)(B * m));
+    HIP_CHECK(hipGetLastError());
+    // bucket space per key group is 2^(c-1); it must stay well below the
+    // per-poly point count n or empty buckets dominate the fused batch
+    uint32_t c = msm_auto_c(m);
+    if ((uint64_t)B * ((256 + c - 1) / c) > 60000) return RNG_ERR_BAD_ARG;  // g<<16 cap
+    std::vector<G1Jac> res(B);
+    int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c, res.data(), B,
+                         RNG_STREAM, (const G1Aff*)ctx->srs_glv_dev);
+    if (rc != RNG_OK) return rc;
+    HostPool::inst().parallel_for(B, [&](uint32_t b) {
+        uint64_t rec[9];
+        jac_to_affine_record(res[b], rec);
+        memcpy(out[b].x.l, rec, 32);
+        memcpy(out[b].y.l, rec + 4, 32);
+        out_inf[b] = rec[8] != 0;
+    });
+    return RNG_OK;
+}
+
+// k proofs in lockstep under ONE proving key (the headline shape: a stream
+// of same-circuit jobs).  Bit-identical to k calls of plonk_prove_impl with
+// the same seeds; 4 fused MSM runs per cohort instead of 5k.
+static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint32_t k,
+                                   const Fr* wires_all, const Fr* pubs_all,
+                                   const uint64_t* seeds, uint64_t* out_proofs,
+                                   uint64_t* out_hints) {
+    const uint64_t n = pk.n;
+    const uint32_t m = (uint32_t)(8 * n);
+    const uint64_t hint_u64s = 4 * (n + 2) + 9;
+    if (!cosets_ok(n)) return RNG_ERR_BAD_ARG;
+    if (cohort_scratch_ensure(n, k) != RNG_OK) return RNG_ERR_HIP;
+    CohortScratch* cs = tls_cohort_scratch.get();
+    HostPool& pool = HostPool::inst();
+    Fr w = h_fr_root_of_unity((uint32_t)n);
+
+    std::vector<HostDrbg> drbg;
+    drbg.reserve(k);
+    std::vector<HostTranscript> tr(k);
+    for (uint32_t p = 0; p < k; ++p) {
+        drbg.emplace_back(seeds[p]);
+        h_transcript_init(tr[p], pk, pubs_all + (size_t)p * pk.npub);
+    }
+    std::vector<std::vector<Fr>> wpoly(5 * (size_t)k), zpoly(k), quot_chunks(5 * (size_t)k);
+    std::vector<G1Aff> comms(13 * (size_t)k);
+    std::vector<uint8_t> cinf(13 * (size_t)k, 0);
+    std::vector<Fr> beta(k), gamma(k), alpha(k), zeta(k), vch(k);
+
+    // scratch for commit outputs (bool arrays for commit_cohort)
+    std::vector<G1Aff> cbuf(5 * (size_t)k);
+    std::unique_ptr<bool[]> ibuf(new bool[5 * (size_t)k]);
+    std::vector<const std::vector<Fr>*> ps(5 * (size_t)k);
+
+    // --- R1: wire polys (one batched IFFT + one fused MSM) ---
+    {
+        HIP_CHECK(hipMemcpyAsync(cs->stage, wires_all, (size_t)5 * k * n * sizeof(Fr),
+                                 hipMemcpyHostToDevice, RNG_STREAM));
+        int rc = ntt_dev_run(ctx, cs->stage, cs->ntt_tmp, (uint32_t)n, 5 * (uint64_t)k,
+                             true, RNG_STREAM);
+        if (rc != RNG_OK) return rc;
+        static thread_local std::vector<Fr> host_flat;
+        host_flat.resize((size_t)5 * k * n);
+        HIP_CHECK(hipMemcpyAsync(host_flat.data(), cs->ntt_tmp,
+                                 (size_t)5 * k * n * sizeof(Fr), hipMemcpyDeviceToHost,
+                                 RNG_STREAM));
+        HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
+        pool.parallel_for(k, [&](uint32_t p) {
+            for (int j = 0; j < 5; ++j) {
+                auto& wp = wpoly[5 * (size_t)p + j];
+                const Fr* src = host_flat.data() + ((size_t)5 * p + j) * n;
+                wp.assign(src, src + n);
+                Fr b0 = drbg[p].next(), b1 = drbg[p].next();
+                wp.resize(n + 2, Fr::zero());
+                wp[0] = wp[0].sub(b0);
+                wp[1] = wp[1].sub(b1);
+                wp[n] = wp[n].add(b0);
+                wp[n + 1] = wp[n + 1].add(b1);
+            }
+        });
+        for (uint32_t p = 0; p < k; ++p)
+            for (int j = 0; j < 5; ++j) ps[5 * (size_t)p + j] = &wpoly[5 * (size_t)p + j];
+        if (commit_cohort(ctx, ps.data(), 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
+            return RNG_ERR_HIP;
+        pool.parallel_for(k, [&](uint32_t p) {
+            for (int j = 0; j < 5; ++j) {
+                comms[13 * (size_t)p + j] = cbuf[5 * (size_t)p + j];
+                cinf[13 * (size_t)p + j] = ibuf[5 * (size_t)p + j] ? 1 : 0;
+                tr[p].append_g1(comms[13 * (size_t)p + j],
+                                cinf[13 * (size_t)p + j] != 0);
+            }
+            if (out_hints) {
+                uint64_t* h = out_hints + (size_t)p * hint_u64s;
+                memcpy(h, wpoly[5 * (size_t)p].data(), (n + 2) * sizeof(Fr));
+                uint64_t* c = h + 4 * (n + 2);
+                memcpy(c, comms[13 * (size_t)p].x.l, 32);
+                memcpy(c + 4, comms[13 * (size_t)p].y.l, 32);
+                c[8] = cinf[13 * (size_t)p] ? 1 : 0;
+            }
+            beta[p] = tr[p].challenge();
+            gamma[p] = tr[p].challenge();
+        });
+    }
+
+    // --- R2: grand products (host pool) + batched IFFT of z and PI ---
+    {
+        static thread_local std::vector<Fr> evals_flat;  // k z-columns + k PI-columns
+        evals_flat.resize(2 * (size_t)k * n);
+        pool.parallel_for(k, [&](uint32_t p) {
+            const Fr* wires = wires_all + (size_t)5 * p * n;
+            std::vector<Fr> znum(n), zden(n);
+            Fr wi = Fr::one();
+            for (uint64_t i = 0; i < n; ++i) {
+                Fr num = Fr::one(), den = Fr::one();
+                for (int j = 0; j < 5; ++j) {
+                    Fr wv = wires[(size_t)j * n + i];
+                    num = num.mul(wv.add(beta[p].mul(pk.k[j]).mul(wi)).add(gamma[p]));
+                    den = den.mul(wv.add(beta[p].mul(pk.sig_evals[j][i])).add(gamma[p]));
+                }
+                znum[i] = num;
+                zden[i] = den;
+                wi = wi.mul(w);
+            }
+            std::vector<Fr> zden_inv = hbatch_inverse(zden);
+            Fr* ze = evals_flat.data() + (size_t)p * n;
+            ze[0] = Fr::one();
+            for (uint64_t i = 1; i < n; ++i)
+                ze[i] = ze[i - 1].mul(znum[i - 1]).mul(zden_inv[i - 1]);
+            Fr* pie = evals_flat.data() + ((size_t)k + p) * n;
+            for (uint64_t i = 0; i < n; ++i) pie[i] = Fr::zero();
+            const Fr* pubs = pubs_all + (size_t)p * pk.npub;
+            for (uint64_t i = 0; i < pk.npub; ++i) pie[i] = pubs[i];
+        });
+        HIP_CHECK(hipMemcpyAsync(cs->stage, evals_flat.data(),
+                                 2 * (size_t)k * n * sizeof(Fr), hipMemcpyHostToDevice,
+                                 RNG_STREAM));
+        int rc = ntt_dev_run(ctx, cs->stage, cs->ntt_tmp, (uint32_t)n, 2 * (uint64_t)k,
+                             true, RNG_STREAM);
+        if (rc != RNG_OK) return rc;
+        HIP_CHECK(hipMemcpyAsync(evals_flat.data(), cs->ntt_tmp,
+                                 2 * (size_t)k * n * sizeof(Fr), hipMemcpyDeviceToHost,
+                                 RNG_STREAM));
+        HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
+        // evals_flat now holds z coefficients (first k) and PI coefficients
+        // (second k); PI coefficients feed the R3 coset batch below
+        pool.parallel_for(k, [&](uint32_t p) {
+            auto& zp = zpoly[p];
+            const Fr* src = evals_flat.data() + (size_t)p * n;
+            zp.assign(src, src + n);
+            Fr b2 = drbg[p].next(), b3 = drbg[p].next(), b4 = drbg[p].next();
+            zp.resize(n + 3, Fr::zero());
+            zp[0] = zp[0].sub(b4);
+            zp[1] = zp[1].sub(b3);
+            zp[2] = zp[2].sub(b2);
+            zp[n] = zp[n].add(b4);
+            zp[n + 1] = zp[n + 1].add(b3);
+            zp[n + 2] = zp[n + 2].add(b2);
+        });
+        for (uint32_t p = 0; p < k; ++p) ps[p] = &zpoly[p];
+        if (commit_cohort(ctx, ps.data(), k, cbuf.data(), ibuf.get()) != RNG_OK)
+            return RNG_ERR_HIP;
+        pool.parallel_for(k, [&](uint32_t p) {
+            comms[13 * (size_t)p + 5] = cbuf[p];
+            cinf[13 * (size_t)p + 5] = ibuf[p] ? 1 : 0;
+            tr[p].append_g1(cbuf[p], ibuf[p]);
+            alpha[p] = tr[p].challenge();
+        });
+
+        // --- R3 staging: 7 polys per proof (5 wires, z, PI) at stride n+3 ---
+        static thread_local std::vector<Fr> coset_host;
+        const uint64_t stride = n + 3;
+        coset_host.resize(7 * (size_t)k * stride);
+        pool.parallel_for(k, [&](uint32_t p) {
+            Fr* base = coset_host.data() + 7 * (size_t)p * stride;
+            for (int j = 0; j < 5; ++j) {
+                memcpy(base + j * stride, wpoly[5 * (size_t)p + j].data(),
+                       (n + 2) * sizeof(Fr));
+                base[j * stride + n + 2] = Fr::zero();
+            }
+            memcpy(base + 5 * stride, zpoly[p].data(), (n + 3) * sizeof(Fr));
+            memcpy(base + 6 * stride, evals_flat.data() + ((size_t)k + p) * n,
+                   n * sizeof(Fr));
+            for (uint64_t j = n; j < stride; ++j) base[6 * stride + j] = Fr::zero();
+        });
+        HIP_CHECK(hipMemcpyAsync(cs->coset_in, coset_host.data(),
+                                 7 * (size_t)k * stride * sizeof(Fr),
+                                 hipMemcpyHostToDevice, RNG_STREAM));
+    }
+
+    // --- R3: coset transforms (batched) + quotient + fused MSM ---
+    {
+        NttPlan* mp = get_plan(ctx, m, 0);  // tables only; batch NTTs below
+        if (!mp || ensure_coset_tables(ctx, mp) != RNG_OK) return RNG_ERR_HIP;
+        const uint64_t stride = n + 3;
+        uint64_t total_in = 7 * (uint64_t)k * stride;
+        hipLaunchKernelGGL(k_mul_pointwise_mod,
+                           dim3((uint32_t)((total_in + 255) / 256)), dim3(256), 0,
+                           RNG_STREAM, cs->coset_in, mp->gpow, (uint32_t)stride,
+                           total_in);
+        HIP_CHECK(hipGetLastError());
+        uint64_t total_m = 7 * (uint64_t)k * m;
+        hipLaunchKernelGGL(k_copy_pad_batch, dim3((uint32_t)((total_m + 255) / 256)),
+                           dim3(256), 0, RNG_STREAM, cs->coset_in, (uint32_t)stride,
+                           cs->ntt_tmp, m, total_m);
+        HIP_CHECK(hipGetLastError());
+        int rc = ntt_dev_run(ctx, cs->ntt_tmp, cs->coset_out, m, 7 * (uint64_t)k, false,
+                             RNG_STREAM);
+        if (rc != RNG_OK) return rc;
+
+        QuotChal ch;
+        for (int j = 0; j < 5; ++j) ch.k[j] = pk.k[j];
+        {
+            Fr g = Fr::from_u64(FR_GENERATOR);
+            Fr gn = g.pow_u64(n);
+            Fr w8 = h_fr_root_of_unity(m).pow_u64(n);
+            std::vector<Fr> zh(8);
+            Fr cur = gn;
+            for (int t = 0; t < 8; ++t) {
+                zh[t] = cur.sub(Fr::one());
+                cur = cur.mul(w8);
+            }
+            zh = hbatch_inverse(zh);
+            for (int t = 0; t < 8; ++t) ch.zh_inv[t] = zh[t];
+        }
+        uint32_t blocks = (m + 255) / 256;
+        for (uint32_t p = 0; p < k; ++p) {
+            QuotChal chp = ch;
+            chp.beta = beta[p];
+            chp.gamma = gamma[p];
+            chp.alpha = alpha[p];
+            chp.alpha2 = alpha[p].sqr();
+            Fr* base = cs->coset_out + 7 * (size_t)p * m;
+            hipLaunchKernelGGL(k_quotient, dim3(blocks), dim3(256), 0, RNG_STREAM,
+                               pk.sel_coset, pk.sig_coset, base /* 5 wires */,
+                               base + 5 * (size_t)m, base + 6 * (size_t)m,
+                               pk.l1_coset, mp->xpow, cs->q_all + (size_t)p * m, m,
+                               chp);
+            HIP_CHECK(hipGetLastError());
+        }
+        // batched inverse coset NTT of the k quotients
+        rc = ntt_dev_run(ctx, cs->q_all, cs->ntt_tmp, m, k, true, RNG_STREAM);
+        if (rc != RNG_OK) return rc;
+        uint64_t total_q = (uint64_t)k * m;
+        hipLaunchKernelGGL(k_mul_pointwise_mod, dim3((uint32_t)((total_q + 255) / 256)),
+                           dim3(256), 0, RNG_STREAM, cs->ntt_tmp, mp->gpow_inv, m,
+                           total_q);
+        HIP_CHECK(hipGetLastError());
+        static thread_local std::vector<Fr> quot_host;
+        quot_host.resize((size_t)k * m);
+        HIP_CHECK(hipMemcpyAsync(quot_host.data(), cs->ntt_tmp,
+                                 (size_t)k * m * sizeof(Fr), hipMemcpyDeviceToHost,
+                                 RNG_STREAM));
+        HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
+        pool.parallel_for(k, [&](uint32_t p) {
+            const Fr* quot = quot_host.data() + (size_t)p * m;
+            Fr prev = Fr::zero();
+            for (int i = 0; i < 5; ++i) {
+                auto& qc = quot_chunks[5 * (size_t)p + i];
+                qc.assign(quot + (size_t)i * (n + 2), quot + (size_t)(i + 1) * (n + 2));
+                Fr bnext = (i < 4) ? drbg[p].next() : Fr::zero();
+                qc[0] = qc[0].sub(prev);
+                if (i < 4) {
+                    qc.resize(n + 3, Fr::zero());
+                    qc[n + 2] = qc[n + 2].add(bnext);
+                }
+                prev = bnext;
+            }
+        });
+        for (uint32_t p = 0; p < k; ++p)
+            for (int i = 0; i < 5; ++i)
+                ps[5 * (size_t)p + i] = &quot_chunks[5 * (size_t)p + i];
+        if (commit_cohort(ctx, ps.data(), 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
+            return RNG_ERR_HIP;
+        pool.parallel_for(k, [&](uint32_t p) {
+            for (int i = 0; i < 5; ++i) {
+                comms[13 * (size_t)p + 6 + i] = cbuf[5 * (size_t)p + i];
+                cinf[13 * (size_t)p + 6 + i] = ibuf[5 * (size_t)p + i] ? 1 : 0;
+                tr[p].append_g1(cbuf[5 * (size_t)p + i], ibuf[5 * (size_t)p + i]);
+            }
+            zeta[p] = tr[p].challenge();
+        });
+    }
+
+    // --- R4 + R5: evaluations, linearization, openings (host pool) ---
+    std::vector<std::vector<Fr>> Wz(k), Wzw(k);
+    std::vector<Fr> wire_evals(5 * (size_t)k), sigma_evals(4 * (size_t)k), zshift(k);
+    pool.parallel_for(k, [&](uint32_t p) {
+        Fr we[5], se[4];
+        for (int j = 0; j < 5; ++j) {
+            we[j] = hpoly_eval(wpoly[5 * (size_t)p + j], zeta[p]);
+            tr[p].append_fr(we[j]);
+            wire_evals[5 * (size_t)p + j] = we[j];
+        }
+        for (int j = 0; j < 4; ++j) {
+            se[j] = hpoly_eval(pk.sigp[j], zeta[p]);
+            tr[p].append_fr(se[j]);
+            sigma_evals[4 * (size_t)p + j] = se[j];
+        }
+        Fr zs = hpoly_eval(zpoly[p], zeta[p].mul(w));
+        zshift[p] = zs;
+        tr[p].append_fr(zs);
+        Fr v = tr[p].challenge();
+        vch[p] = v;
+
+        Fr zeta_n = zeta[p].pow_u64(n);
+        Fr zh_zeta = zeta_n.sub(Fr::one());
+        Fr l1_zeta = zh_zeta.mul(Fr::from_u64(n).mul(zeta[p].sub(Fr::one())).inverse());
+        auto p5f = [](const Fr& x) {
+            Fr x2 = x.sqr();
+            return x2.sqr().mul(x);
+        };
+        const Fr* wb = we;
+        std::vector<Fr> D;
+        hpoly_add_scaled(D, pk.selq[11], Fr::one());
+        for (int j = 0; j < 4; ++j) hpoly_add_scaled(D, pk.selq[j], wb[j]);
+        hpoly_add_scaled(D, pk.selq[4], wb[0].mul(wb[1]));
+        hpoly_add_scaled(D, pk.selq[5], wb[2].mul(wb[3]));
+        for (int j = 0; j < 4; ++j) hpoly_add_scaled(D, pk.selq[6 + j], p5f(wb[j]));
+        hpoly_add_scaled(D, pk.selq[12],
+                         wb[0].mul(wb[1]).mul(wb[2]).mul(wb[3]).mul(wb[4]));
+        hpoly_add_scaled(D, pk.selq[10], wb[4].neg());
+        Fr fbar = Fr::one(), Bbar = Fr::one();
+        for (int j = 0; j < 5; ++j)
+            fbar = fbar.mul(wb[j].add(beta[p].mul(pk.k[j]).mul(zeta[p])).add(gamma[p]));
+        for (int j = 0; j < 4; ++j)
+            Bbar = Bbar.mul(wb[j].add(beta[p].mul(se[j])).add(gamma[p]));
+        hpoly_add_scaled(D, zpoly[p],
+                         alpha[p].mul(fbar).add(alpha[p].sqr().mul(l1_zeta)));
+        hpoly_add_scaled(D, pk.sigp[4],
+                         alpha[p].mul(beta[p]).mul(zs).mul(Bbar).neg());
+        {
+            Fr zpow = zh_zeta.neg();
+            Fr step = zeta[p].pow_u64(n + 2);
+            for (int i = 0; i < 5; ++i) {
+                hpoly_add_scaled(D, quot_chunks[5 * (size_t)p + i], zpow);
+                zpow = zpow.mul(step);
+            }
+        }
+        std::vector<Fr> C = D;
+        Fr vp = Fr::one();
+        for (int j = 0; j < 5; ++j) {
+            vp = vp.mul(v);
+            hpoly_add_scaled(C, wpoly[5 * (size_t)p + j], vp);
+        }
+        for (int j = 0; j < 4; ++j) {
+            vp = vp.mul(v);
+            hpoly_add_scaled(C, pk.sigp[j], vp);
+        }
+        Wz[p] = hpoly_div_linear(C, zeta[p]);
+        Wzw[p] = hpoly_div_linear(zpoly[p], zeta[p].mul(w));
+    });
+    for (uint32_t p = 0; p < k; ++p) {
+        ps[2 * (size_t)p] = &Wz[p];
+        ps[2 * (size_t)p + 1] = &Wzw[p];
+    }
+    if (commit_cohort(ctx, ps.data(), 2 * k, cbuf.data(), ibuf.get()) != RNG_OK)
+        return RNG_ERR_HIP;
+    pool.parallel_for(k, [&](uint32_t p) {
+        comms[13 * (size_t)p + 11] = cbuf[2 * (size_t)p];
+        cinf[13 * (size_t)p + 11] = ibuf[2 * (size_t)p] ? 1 : 0;
+        comms[13 * (size_t)p + 12] = cbuf[2 * (size_t)p + 1];
+        cinf[13 * (size_t)p + 12] = ibuf[2 * (size_t)p + 1] ? 1 : 0;
+        tr[p].append_g1(cbuf[2 * (size_t)p], ibuf[2 * (size_t)p]);
+        tr[p].append_g1(cbuf[2 * (size_t)p + 1], ibuf[2 * (size_t)p + 1]);
+        // --- serialize ---
+        uint64_t* out_proof = out_proofs + (size_t)p * 157;
+        for (int i = 0; i < 13; ++i) {
+            memcpy(out_proof + 9 * i, comms[13 * (size_t)p + i].x.l, 32);
+            memcpy(out_proof + 9 * i + 4, comms[13 * (size_t)p + i].y.l, 32);
+            out_proof[9 * i + 8] = cinf[13 * (size_t)p + i] ? 1 : 0;
+        }
+        uint64_t* e = out_proof + 117;
+        for (int i = 0; i < 5; ++i)
+            memcpy(e + 4 * i, wire_evals[5 * (size_t)p + i].l, 32);
+        for (int i = 0; i < 4; ++i)
+            memcpy(e + 20 + 4 * i, sigma_evals[4 * (size_t)p + i].l, 32);
+        memcpy(e + 36, zshift[p].l, 32);
+    });
+    return RNG_OK;
 }
 
 // upload host coeffs, pad to m, coset-forward into dst (device, m elems)
@@ -3259,6 +3804,19 @@ int rng_prove(RngCtx* ctx, const RngProvingKey* pk, const uint64_t* wires,
     if (pk->impl.npub > 0 && !public_inputs) return RNG_ERR_BAD_ARG;
     return plonk_prove_impl(&ctx->impl, pk->impl, (const Fr*)wires,
                             (const Fr*)public_inputs, seed, out_proof, out_link_hint);
+}
+
+int rng_prove_cohort(RngCtx* ctx, const RngProvingKey* pk, uint64_t k,
+                     const uint64_t* wires, const uint64_t* public_inputs,
+                     const uint64_t* seeds, uint64_t* out_proofs,
+                     uint64_t* out_link_hints) {
+    if (!gpu_ok()) return RNG_ERR_NO_GPU;
+    if (!ctx || !pk || !wires || !seeds || !out_proofs || k == 0 || k > 128)
+        return RNG_ERR_BAD_ARG;
+    if (pk->impl.npub > 0 && !public_inputs) return RNG_ERR_BAD_ARG;
+    return plonk_prove_cohort_impl(&ctx->impl, pk->impl, (uint32_t)k,
+                                   (const Fr*)wires, (const Fr*)public_inputs, seeds,
+                                   out_proofs, out_link_hints);
 }
 
 // PK introspection for tests / the verifier side

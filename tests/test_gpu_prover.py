@@ -114,13 +114,46 @@ class TestGpuProver:
         assert ok == 1
 
     def test_pk_comms_match_oracle(self, setup):
+        """Direct 18-commitment comparison: GPU preprocess vs oracle
+        preprocess on the same tables/SRS (VERDICT r01 missing #4)."""
         s = setup
         comms = np.zeros(18 * 9, dtype=np.uint64)
         s["lib"].rng_pk_comms(ctypes.c_void_p(s["pk"]), ptr(comms))
-        # oracle recomputes the same commitments inside preprocess; compare via
-        # a proof transcript round trip instead: already covered by bit-exact
-        # proof equality (transcript binds the comms). Check non-degenerate:
+        ocomms = np.zeros(18 * 9, dtype=np.uint64)
+        o = s["orc"].lib
+        o.orc_plonk_pk_comms.argtypes = [ctypes.c_void_p, U64P]
+        o.orc_plonk_pk_comms(ctypes.c_void_p(s["opk"]), ptr(ocomms))
+        assert np.array_equal(comms, ocomms), \
+            f"PK commitment mismatch at records {set((np.nonzero(comms != ocomms)[0] // 9).tolist())}"
         assert np.any(comms != 0)
+
+    def test_cohort_bit_exact_vs_single(self, setup):
+        """rng_prove_cohort(k) proofs (and hints) are bit-identical to k
+        independent rng_prove calls with the same seeds."""
+        s = setup
+        lib = s["lib"]
+        n, npub = int(s["n"]), int(s["npub"])
+        k = 5
+        lib.rng_prove_cohort.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                         ctypes.c_uint64, U64P, U64P, U64P, U64P, U64P]
+        wires_all = np.tile(s["wires"], k)
+        pubs_all = np.tile(s["pubs"], k)
+        seeds = np.array([31, 32, 33, 31, 99], dtype=np.uint64)
+        proofs = np.zeros(157 * k, dtype=np.uint64)
+        hint_len = 4 * (n + 2) + 9
+        hints = np.zeros(hint_len * k, dtype=np.uint64)
+        rc = lib.rng_prove_cohort(s["ctx"].h, ctypes.c_void_p(s["pk"]), k,
+                                  ptr(wires_all), ptr(pubs_all), ptr(seeds),
+                                  ptr(proofs), ptr(hints))
+        assert rc == 0, f"rng_prove_cohort rc={rc}"
+        for p in range(k):
+            single, shint = gpu_prove(s, seed=int(seeds[p]), with_hint=True)
+            assert np.array_equal(proofs[157 * p:157 * (p + 1)], single), \
+                f"cohort proof {p} differs from single-proof path"
+            assert np.array_equal(hints[hint_len * p:hint_len * (p + 1)], shint), \
+                f"cohort hint {p} differs"
+        # same seed -> same proof within a cohort too
+        assert np.array_equal(proofs[:157], proofs[157 * 3:157 * 4])
 
     def test_link_hint(self, setup):
         s = setup
