@@ -210,11 +210,20 @@ void* orc_plonk_preprocess(u64 n, u64 num_public, const u64* selectors,
 void orc_plonk_pk_free(void* pk) { delete static_cast<OrcProvingKey*>(pk); }
 
 // 18 affine records: 13 selector commitments then 5 sigma commitments —
-// the same layout as the product's rng_pk_comms, for direct PK parity tests
+// the same layout as the product's rng_pk_comms (canonical zeroed-infinity
+// records), for direct PK parity tests
 void orc_plonk_pk_comms(void* pk_, u64* out18x9) {
     auto* pk = static_cast<OrcProvingKey*>(pk_);
-    for (int s = 0; s < 13; ++s) store_affine(pk->sel_comms[s], out18x9 + 9 * s);
-    for (int j = 0; j < 5; ++j) store_affine(pk->sig_comms[j], out18x9 + 9 * (13 + j));
+    auto store = [&](const G1Affine& p, u64* rec) {
+        if (p.infinity) {
+            memset(rec, 0, 8 * 8);
+            rec[8] = 1;
+            return;
+        }
+        store_affine(p, rec);
+    };
+    for (int s = 0; s < 13; ++s) store(pk->sel_comms[s], out18x9 + 9 * s);
+    for (int j = 0; j < 5; ++j) store(pk->sig_comms[j], out18x9 + 9 * (13 + j));
 }
 
 int orc_plonk_prove(void* pk_, const u64* wires, const u64* pubs, u64 seed,

@@ -77,9 +77,19 @@ static thread_local double tls_ntt_times[2] = {0, 0};
 // concurrent callers queue.  Never destroyed (leaked) so there is no
 // thread-join ordering problem at process exit.
 class HostPool {
+    // Each job is a heap object shared by every thread that touches it: a
+    // straggler from job A holds A's shared_ptr, sees A's exhausted `next`
+    // counter and exits — it can never execute job B's indices with A's
+    // (destroyed) callable.
+    struct Job {
+        std::function<void(uint32_t)> fn;
+        uint32_t count;
+        std::atomic<uint32_t> next{0}, done{0};
+    };
+
   public:
     static HostPool& inst() {
-        static HostPool* p = new HostPool();
+        static HostPool* p = new HostPool();  // leaked: no join-at-exit problem
         return *p;
     }
 
@@ -89,20 +99,24 @@ class HostPool {
             for (uint32_t i = 0; i < count; ++i) fn(i);
             return;
         }
-        std::unique_lock<std::mutex> job_lk(job_mu_);
+        std::unique_lock<std::mutex> job_lk(job_mu_);  // one job at a time
+        auto job = std::make_shared<Job>();
+        job->fn = fn;
+        job->count = count;
         {
             std::lock_guard<std::mutex> lk(mu_);
-            fn_ = &fn;
-            count_ = count;
-            next_.store(0, std::memory_order_relaxed);
-            done_.store(0, std::memory_order_relaxed);
+            job_ = job;
             ++generation_;
         }
         cv_.notify_all();
-        work();  // caller participates
-        std::unique_lock<std::mutex> lk(mu_);
-        done_cv_.wait(lk, [&] { return done_.load(std::memory_order_acquire) >= count_; });
-        fn_ = nullptr;
+        run_job(*job);  // caller participates
+        {
+            std::unique_lock<std::mutex> lk(mu_);
+            done_cv_.wait(lk, [&] {
+                return job->done.load(std::memory_order_acquire) >= count;
+            });
+            job_.reset();
+        }
     }
 
   private:
@@ -113,20 +127,12 @@ class HostPool {
         for (unsigned t = 0; t + 1 < n; ++t)
             workers_.emplace_back([this] { worker_loop(); });
     }
-    void work() {
-        const std::function<void(uint32_t)>* fn;
-        uint32_t count;
-        {
-            std::lock_guard<std::mutex> lk(mu_);
-            fn = fn_;
-            count = count_;
-        }
-        if (!fn) return;
+    void run_job(Job& job) {
         for (;;) {
-            uint32_t i = next_.fetch_add(1, std::memory_order_relaxed);
-            if (i >= count) break;
-            (*fn)(i);
-            if (done_.fetch_add(1, std::memory_order_acq_rel) + 1 >= count) {
+            uint32_t i = job.next.fetch_add(1, std::memory_order_relaxed);
+            if (i >= job.count) break;
+            job.fn(i);
+            if (job.done.fetch_add(1, std::memory_order_acq_rel) + 1 >= job.count) {
                 std::lock_guard<std::mutex> lk(mu_);
                 done_cv_.notify_all();
             }
@@ -135,22 +141,22 @@ class HostPool {
     void worker_loop() {
         uint64_t seen = 0;
         for (;;) {
+            std::shared_ptr<Job> job;
             {
                 std::unique_lock<std::mutex> lk(mu_);
                 cv_.wait(lk, [&] { return generation_ != seen; });
                 seen = generation_;
+                job = job_;
             }
-            work();
+            if (job) run_job(*job);
         }
     }
     std::mutex job_mu_;
     std::mutex mu_;
     std::condition_variable cv_, done_cv_;
     std::vector<std::thread> workers_;
-    const std::function<void(uint32_t)>* fn_ = nullptr;
-    uint32_t count_ = 0;
+    std::shared_ptr<Job> job_;
     uint64_t generation_ = 0;
-    std::atomic<uint32_t> next_{0}, done_{0};
 };
 
 struct EvtTimer {
@@ -849,6 +855,17 @@ static int cohort_scratch_ensure(uint64_t n, uint64_t k) {
 // conversion are parallelized across the pool for large B.
 static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
                          uint32_t B, G1Aff* out, bool* out_inf) {
+    static int trace = [] {
+        const char* e = getenv("RNG_COHORT_TRACE");
+        return e ? atoi(e) : 0;
+    }();
+    auto tr = [&](const char* tag) {
+        if (trace) {
+            fprintf(stderr, "[commit_cohort B=%u] %s\n", B, tag);
+            fflush(stderr);
+        }
+    };
+    tr("enter");
     CohortScratch* s = tls_cohort_scratch.get();
     uint64_t m = 0;
     for (uint32_t b = 0; b < B; ++b)
@@ -875,8 +892,10 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
     uint32_t c = msm_auto_c(m);
     if ((uint64_t)B * ((256 + c - 1) / c) > 60000) return RNG_ERR_BAD_ARG;  // g<<16 cap
     std::vector<G1Jac> res(B);
+    tr("msm");
     int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c, res.data(), B,
                          RNG_STREAM, (const G1Aff*)ctx->srs_glv_dev);
+    tr("msm-done");
     if (rc != RNG_OK) return rc;
     HostPool::inst().parallel_for(B, [&](uint32_t b) {
         uint64_t rec[9];
@@ -898,6 +917,18 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
     const uint64_t n = pk.n;
     const uint32_t m = (uint32_t)(8 * n);
     const uint64_t hint_u64s = 4 * (n + 2) + 9;
+    static int trace = [] {
+        const char* e = getenv("RNG_COHORT_TRACE");
+        return e ? atoi(e) : 0;
+    }();
+#define COHORT_TRACE(tag)                                              \
+    do {                                                               \
+        if (trace) {                                                   \
+            fprintf(stderr, "[cohort] %s\n", tag);                     \
+            fflush(stderr);                                            \
+        }                                                              \
+    } while (0)
+    COHORT_TRACE("enter");
     if (!cosets_ok(n)) return RNG_ERR_BAD_ARG;
     if (cohort_scratch_ensure(n, k) != RNG_OK) return RNG_ERR_HIP;
     CohortScratch* cs = tls_cohort_scratch.get();
@@ -921,10 +952,12 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
     std::unique_ptr<bool[]> ibuf(new bool[5 * (size_t)k]);
     std::vector<const std::vector<Fr>*> ps(5 * (size_t)k);
 
+    COHORT_TRACE("r1");
     // --- R1: wire polys (one batched IFFT + one fused MSM) ---
     {
         HIP_CHECK(hipMemcpyAsync(cs->stage, wires_all, (size_t)5 * k * n * sizeof(Fr),
                                  hipMemcpyHostToDevice, RNG_STREAM));
+        COHORT_TRACE("r1-ifft");
         int rc = ntt_dev_run(ctx, cs->stage, cs->ntt_tmp, (uint32_t)n, 5 * (uint64_t)k,
                              true, RNG_STREAM);
         if (rc != RNG_OK) return rc;
@@ -971,6 +1004,7 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         });
     }
 
+    COHORT_TRACE("r2");
     // --- R2: grand products (host pool) + batched IFFT of z and PI ---
     {
         static thread_local std::vector<Fr> evals_flat;  // k z-columns + k PI-columns
@@ -1056,6 +1090,7 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
                                  hipMemcpyHostToDevice, RNG_STREAM));
     }
 
+    COHORT_TRACE("r3");
     // --- R3: coset transforms (batched) + quotient + fused MSM ---
     {
         NttPlan* mp = get_plan(ctx, m, 0);  // tables only; batch NTTs below
@@ -1150,6 +1185,7 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         });
     }
 
+    COHORT_TRACE("r4r5");
     // --- R4 + R5: evaluations, linearization, openings (host pool) ---
     std::vector<std::vector<Fr>> Wz(k), Wzw(k);
     std::vector<Fr> wire_evals(5 * (size_t)k), sigma_evals(4 * (size_t)k), zshift(k);
@@ -3823,16 +3859,20 @@ int rng_prove_cohort(RngCtx* ctx, const RngProvingKey* pk, uint64_t k,
 uint64_t rng_pk_n(RngProvingKey* pk) { return pk->impl.n; }
 // out: 18 affine records (13 selector comms, 5 sigma comms)
 void rng_pk_comms(RngProvingKey* pk, uint64_t* out) {
-    for (int s = 0; s < 13; ++s) {
-        memcpy(out + 9 * s, pk->impl.sel_comms[s].x.l, 32);
-        memcpy(out + 9 * s + 4, pk->impl.sel_comms[s].y.l, 32);
-        out[9 * s + 8] = 0;
-    }
-    for (int j = 0; j < 5; ++j) {
-        memcpy(out + 9 * (13 + j), pk->impl.sig_comms[j].x.l, 32);
-        memcpy(out + 9 * (13 + j) + 4, pk->impl.sig_comms[j].y.l, 32);
-        out[9 * (13 + j) + 8] = 0;
-    }
+    auto store = [&](const G1Aff& a, bool inf, uint64_t* rec) {
+        if (inf) {  // canonical infinity record: zeroed coords + flag
+            memset(rec, 0, 8 * 8);
+            rec[8] = 1;
+            return;
+        }
+        memcpy(rec, a.x.l, 32);
+        memcpy(rec + 4, a.y.l, 32);
+        rec[8] = 0;
+    };
+    for (int s = 0; s < 13; ++s)
+        store(pk->impl.sel_comms[s], pk->impl.sel_inf[s], out + 9 * s);
+    for (int j = 0; j < 5; ++j)
+        store(pk->impl.sig_comms[j], pk->impl.sig_inf[j], out + 9 * (13 + j));
 }
 
 // Full PlonK verifier with the real BN254 pairing (replaces
