@@ -872,9 +872,9 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
         if (polys[b]->size() > m) m = polys[b]->size();
     if (m > ctx->srs_count) return RNG_ERR_BAD_ARG;
     // pack on the host (pool) and ship ONE H2D copy: B small async copies of
-    // pageable memory each pay a staging round trip
-    static thread_local std::vector<Fr> packed;
-    packed.resize(B * m);
+    // pageable memory each pay a staging round trip.  Plain local (NOT
+    // thread_local): pool workers reference it from their own threads.
+    std::vector<Fr> packed(B * m);
     HostPool::inst().parallel_for(B, [&](uint32_t b) {
         memcpy(packed.data() + (size_t)b * m, polys[b]->data(),
                polys[b]->size() * sizeof(Fr));
@@ -961,12 +961,13 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         int rc = ntt_dev_run(ctx, cs->stage, cs->ntt_tmp, (uint32_t)n, 5 * (uint64_t)k,
                              true, RNG_STREAM);
         if (rc != RNG_OK) return rc;
-        static thread_local std::vector<Fr> host_flat;
-        host_flat.resize((size_t)5 * k * n);
+        COHORT_TRACE("r1-ifft-done");
+        std::vector<Fr> host_flat((size_t)5 * k * n);
         HIP_CHECK(hipMemcpyAsync(host_flat.data(), cs->ntt_tmp,
                                  (size_t)5 * k * n * sizeof(Fr), hipMemcpyDeviceToHost,
                                  RNG_STREAM));
         HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
+        COHORT_TRACE("r1-sync-done");
         pool.parallel_for(k, [&](uint32_t p) {
             for (int j = 0; j < 5; ++j) {
                 auto& wp = wpoly[5 * (size_t)p + j];
@@ -980,6 +981,7 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
                 wp[n + 1] = wp[n + 1].add(b1);
             }
         });
+        COHORT_TRACE("r1-blind-done");
         for (uint32_t p = 0; p < k; ++p)
             for (int j = 0; j < 5; ++j) ps[5 * (size_t)p + j] = &wpoly[5 * (size_t)p + j];
         if (commit_cohort(ctx, ps.data(), 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
@@ -1007,8 +1009,7 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
     COHORT_TRACE("r2");
     // --- R2: grand products (host pool) + batched IFFT of z and PI ---
     {
-        static thread_local std::vector<Fr> evals_flat;  // k z-columns + k PI-columns
-        evals_flat.resize(2 * (size_t)k * n);
+        std::vector<Fr> evals_flat(2 * (size_t)k * n);  // k z-cols + k PI-cols
         pool.parallel_for(k, [&](uint32_t p) {
             const Fr* wires = wires_all + (size_t)5 * p * n;
             std::vector<Fr> znum(n), zden(n);
@@ -1070,9 +1071,8 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         });
 
         // --- R3 staging: 7 polys per proof (5 wires, z, PI) at stride n+3 ---
-        static thread_local std::vector<Fr> coset_host;
         const uint64_t stride = n + 3;
-        coset_host.resize(7 * (size_t)k * stride);
+        std::vector<Fr> coset_host(7 * (size_t)k * stride);
         pool.parallel_for(k, [&](uint32_t p) {
             Fr* base = coset_host.data() + 7 * (size_t)p * stride;
             for (int j = 0; j < 5; ++j) {
@@ -1149,8 +1149,7 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
                            dim3(256), 0, RNG_STREAM, cs->ntt_tmp, mp->gpow_inv, m,
                            total_q);
         HIP_CHECK(hipGetLastError());
-        static thread_local std::vector<Fr> quot_host;
-        quot_host.resize((size_t)k * m);
+        std::vector<Fr> quot_host((size_t)k * m);
         HIP_CHECK(hipMemcpyAsync(quot_host.data(), cs->ntt_tmp,
                                  (size_t)k * m * sizeof(Fr), hipMemcpyDeviceToHost,
                                  RNG_STREAM));
