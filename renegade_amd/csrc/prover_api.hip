@@ -6,6 +6,7 @@
 // files directly.
 #include <array>
 #include <atomic>
+#include <chrono>
 #include <condition_variable>
 #include <cstdio>
 #include <functional>
@@ -889,7 +890,11 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
     HIP_CHECK(hipGetLastError());
     // bucket space per key group is 2^(c-1); it must stay well below the
     // per-poly point count n or empty buckets dominate the fused batch
-    uint32_t c = msm_auto_c(m);
+    static int c_env = [] {
+        const char* e = getenv("RNG_MSM_C");
+        return e ? atoi(e) : 0;
+    }();
+    uint32_t c = (c_env >= 8 && c_env <= 16) ? (uint32_t)c_env : msm_auto_c(m);
     if ((uint64_t)B * ((256 + c - 1) / c) > 60000) return RNG_ERR_BAD_ARG;  // g<<16 cap
     std::vector<G1Jac> res(B);
     tr("msm");
@@ -921,12 +926,16 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         const char* e = getenv("RNG_COHORT_TRACE");
         return e ? atoi(e) : 0;
     }();
-#define COHORT_TRACE(tag)                                              \
-    do {                                                               \
-        if (trace) {                                                   \
-            fprintf(stderr, "[cohort] %s\n", tag);                     \
-            fflush(stderr);                                            \
-        }                                                              \
+    auto t_enter = std::chrono::steady_clock::now();
+#define COHORT_TRACE(tag)                                                     \
+    do {                                                                      \
+        if (trace) {                                                          \
+            double ms_ = std::chrono::duration<double, std::milli>(           \
+                             std::chrono::steady_clock::now() - t_enter)      \
+                             .count();                                        \
+            fprintf(stderr, "[cohort +%7.2fms] %s\n", ms_, tag);              \
+            fflush(stderr);                                                   \
+        }                                                                     \
     } while (0)
     COHORT_TRACE("enter");
     if (!cosets_ok(n)) return RNG_ERR_BAD_ARG;
