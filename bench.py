@@ -613,10 +613,14 @@ def main():
 
 if __name__ == "__main__":
     main()
+    sys.stdout.flush()
+    sys.stderr.flush()
     # driver-proof exit: the JSON is printed and all pools are joined; skip
     # interpreter teardown entirely so no late TLS/GC destructor can touch a
     # torn-down HIP runtime (r01 headline run exited rc=139 that way — the
-    # library also guards its destructors now, this is belt and braces)
-    sys.stdout.flush()
-    sys.stderr.flush()
-    os._exit(0)
+    # library also guards its destructors now, this is belt and braces).
+    # Under rocprof, exit normally instead: _exit would skip the profiler's
+    # finalization and no stats files would be written.
+    profiled = any(k.startswith(("ROCPROF", "ROCP_", "HSA_TOOLS")) for k in os.environ)
+    if not profiled:
+        os._exit(0)

@@ -866,6 +866,14 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
             fflush(stderr);
         }
     };
+    auto tr_msm_stages = [&] {
+        if (trace)
+            fprintf(stderr,
+                    "[commit_cohort B=%u] msm stages ms: digits %.2f sort %.2f "
+                    "bucket %.2f chunks %.2f final %.2f\n",
+                    B, tls_msm_times[0], tls_msm_times[1], tls_msm_times[2],
+                    tls_msm_times[3], tls_msm_times[4]);
+    };
     tr("enter");
     CohortScratch* s = tls_cohort_scratch.get();
     uint64_t m = 0;
@@ -901,6 +909,7 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
     int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c, res.data(), B,
                          RNG_STREAM, (const G1Aff*)ctx->srs_glv_dev);
     tr("msm-done");
+    tr_msm_stages();
     if (rc != RNG_OK) return rc;
     HostPool::inst().parallel_for(B, [&](uint32_t b) {
         uint64_t rec[9];
