@@ -357,6 +357,11 @@ struct MsmScratch {
     size_t select_temp_bytes = 0;
     void* scan_temp = nullptr;
     size_t scan_temp_bytes = 0;
+    // binned (counting-scatter) pipeline
+    uint32_t* counts = nullptr;   // nb
+    uint32_t* offsets = nullptr;  // nb
+    uint32_t* cursor = nullptr;   // nb
+    uint8_t* bflags = nullptr;    // nb
     G1Jac* buckets = nullptr;
     G1Jac* partials = nullptr;
     G1Jac* window_sums = nullptr;
@@ -375,7 +380,8 @@ struct MsmScratch {
                         (void*)nsub_sorted, (void*)head_order,
                         (void*)partials2, (void*)head_count, select_temp, scan_temp,
                         (void*)buckets, (void*)partials, (void*)window_sums,
-                        (void*)result, (void*)glv, (void*)phi})
+                        (void*)result, (void*)glv, (void*)phi, (void*)counts,
+                        (void*)offsets, (void*)cursor, (void*)bflags})
             hip_free_guarded(b);
     }
 };
@@ -456,14 +462,20 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->head_count, 4));
         {
             rocprim::counting_iterator<uint32_t> cit(0);
+            uint64_t sel_cap = cap_total > cap_nb2 ? cap_total : cap_nb2;
             (void)rocprim::select(nullptr, s->select_temp_bytes, cit, s->head_flags,
-                                  s->heads, s->head_count, cap_total, stream);
+                                  s->heads, s->head_count, sel_cap, stream);
             HIP_CHECK(hipMalloc(&s->select_temp, s->select_temp_bytes));
+            uint64_t scan_cap = max_heads_cap > cap_nb2 ? max_heads_cap : cap_nb2;
             (void)rocprim::exclusive_scan(nullptr, s->scan_temp_bytes, s->nsub,
-                                          s->sub_off, 0u, max_heads_cap,
+                                          s->sub_off, 0u, scan_cap,
                                           rocprim::plus<uint32_t>(), stream);
             HIP_CHECK(hipMalloc(&s->scan_temp, s->scan_temp_bytes));
         }
+        HIP_CHECK(hipMalloc(&s->counts, cap_nb2 * 4));
+        HIP_CHECK(hipMalloc(&s->offsets, cap_nb2 * 4));
+        HIP_CHECK(hipMalloc(&s->cursor, cap_nb2 * 4));
+        HIP_CHECK(hipMalloc(&s->bflags, cap_nb2));
         HIP_CHECK(hipMalloc(&s->buckets, cap_nb2 * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->partials, 2 * cap_nch * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->window_sums, 512 * MSM_SUBB * sizeof(G1Jac)));
@@ -477,9 +489,92 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         s->cap_nchunks = cap_nch;
     }
 
+    // binned (counting-scatter) pipeline: default ON for the non-GLV path;
+    // RNG_MSM_BINNED=0 falls back to the radix sort for A/B
+    static int binned_env = [] {
+        const char* e = getenv("RNG_MSM_BINNED");
+        return e ? atoi(e) : 1;
+    }();
+    const bool binned = binned_env && !glv;
+
     uint32_t tb = 256;
     EvtTimer et;
     et.mark(stream);
+    if (binned) {
+        HIP_CHECK(hipMemsetAsync(s->counts, 0, nb * 4, stream));
+        hipLaunchKernelGGL(k_msm_digits_hist, dim3((uint32_t)((n * B + tb - 1) / tb)),
+                           dim3(tb), 0, stream, d_scalars, (uint32_t)n, c, W, B,
+                           s->keys_in, s->vals_in, s->counts);
+        HIP_CHECK(hipGetLastError());
+        et.mark(stream);
+        (void)rocprim::exclusive_scan(s->scan_temp, s->scan_temp_bytes, s->counts,
+                                      s->offsets, 0u, nb, rocprim::plus<uint32_t>(),
+                                      stream);
+        HIP_CHECK(hipMemcpyAsync(s->cursor, s->offsets, nb * 4,
+                                 hipMemcpyDeviceToDevice, stream));
+        hipLaunchKernelGGL(k_msm_scatter, dim3((uint32_t)((total + tb - 1) / tb)),
+                           dim3(tb), 0, stream, s->keys_in, s->vals_in, total,
+                           s->cursor, s->vals_out);
+        HIP_CHECK(hipGetLastError());
+        HIP_CHECK(hipMemsetAsync(s->buckets, 0, nb * sizeof(G1Jac), stream));
+        hipLaunchKernelGGL(k_nonzero_flags, dim3((uint32_t)((nb + tb - 1) / tb)),
+                           dim3(tb), 0, stream, s->counts, (uint32_t)nb, s->bflags);
+        HIP_CHECK(hipGetLastError());
+        {
+            // keys_out holds the compacted non-empty bucket ids
+            rocprim::counting_iterator<uint32_t> cit(0);
+            (void)rocprim::select(s->select_temp, s->select_temp_bytes, cit, s->bflags,
+                                  s->keys_out, s->head_count, nb, stream);
+        }
+        uint32_t seg_cap = msm_seg_cap(total);
+        hipLaunchKernelGGL(k_msm_seg_from_counts, dim3((uint32_t)((nb + tb - 1) / tb)),
+                           dim3(tb), 0, stream, s->keys_out, s->offsets, s->counts,
+                           s->head_count, s->heads, s->lens, s->nsub, seg_cap);
+        HIP_CHECK(hipGetLastError());
+        uint32_t hc = 0;
+        HIP_CHECK(hipMemcpyAsync(&hc, s->head_count, 4, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        if (hc > 0) {
+            (void)rocprim::exclusive_scan(s->scan_temp, s->scan_temp_bytes, s->nsub,
+                                          s->sub_off, 0u, hc, rocprim::plus<uint32_t>(),
+                                          stream);
+            uint32_t last_off = 0, last_n = 0;
+            HIP_CHECK(hipMemcpyAsync(&last_off, s->sub_off + hc - 1, 4,
+                                     hipMemcpyDeviceToHost, stream));
+            HIP_CHECK(hipMemcpyAsync(&last_n, s->nsub + hc - 1, 4,
+                                     hipMemcpyDeviceToHost, stream));
+            HIP_CHECK(hipStreamSynchronize(stream));
+            uint32_t sub_total = last_off + last_n;
+            hipLaunchKernelGGL(k_msm_make_subs, dim3((hc + tb - 1) / tb), dim3(tb), 0,
+                               stream, s->heads, s->lens, s->sub_off, s->head_count,
+                               s->sub_start, s->sub_len, seg_cap);
+            HIP_CHECK(hipGetLastError());
+            {
+                rocprim::counting_iterator<uint32_t> cit(0);
+                rocprim::radix_sort_pairs_desc(s->sort_temp, s->sort_temp_bytes,
+                                               s->sub_len, s->sub_len_sorted, cit,
+                                               s->sub_order, sub_total, 0, 8, stream);
+            }
+            et.mark(stream);
+            hipLaunchKernelGGL(k_msm_bucket_reduce, dim3((sub_total + tb - 1) / tb),
+                               dim3(tb), 0, stream, s->vals_out, s->sub_start,
+                               s->sub_len, s->sub_order, sub_total, d_bases,
+                               s->partials2);
+            HIP_CHECK(hipGetLastError());
+            {
+                rocprim::counting_iterator<uint32_t> cit(0);
+                rocprim::radix_sort_pairs_desc(s->sort_temp, s->sort_temp_bytes,
+                                               s->nsub, s->nsub_sorted, cit,
+                                               s->head_order, hc, 0, 16, stream);
+            }
+            hipLaunchKernelGGL(k_msm_seg_merge2, dim3((hc + tb - 1) / tb), dim3(tb), 0,
+                               stream, s->keys_out, s->sub_off, s->nsub, s->head_order,
+                               hc, s->partials2, s->buckets);
+            HIP_CHECK(hipGetLastError());
+        } else {
+            et.mark(stream);
+        }
+    } else {  // radix-sort path (GLV, or RNG_MSM_BINNED=0)
     if (glv) {
         if (!d_glv_bases) {  // non-SRS bases: interleave into scratch
             hipLaunchKernelGGL(k_bases_interleave, dim3((uint32_t)((n + tb - 1) / tb)),
@@ -570,6 +665,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     } else {
         et.mark(stream);
     }
+    }  // end radix-sort path
     et.mark(stream);
     uint32_t cw = (1u << (c - 1)) / chunk_sz;
     // fused suffix-scan fold (k_msm_window_fold) measured SLOWER than the
