@@ -198,6 +198,125 @@ struct Fp4 {
         return r;
     }
 
+    // 32-bit CIOS variant B: t[] and carry kept as u64 values with a
+    // maintained hi=0 invariant, so each inner step is one v_mad_u64_u32
+    // (32x32 + 64-bit addend) plus one 64-bit add — no zero-extension mov
+    // dance.  (a*b + t + c <= (2^32-1)^2 + 2*(2^32-1) = 2^64-1: no overflow.)
+    RNG_HD Fp4 mul_cios32b(const Fp4& o) const {
+        uint32_t a[8], b[8], p[8];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            a[2 * i] = (uint32_t)l[i];
+            a[2 * i + 1] = (uint32_t)(l[i] >> 32);
+            b[2 * i] = (uint32_t)o.l[i];
+            b[2 * i + 1] = (uint32_t)(o.l[i] >> 32);
+            p[2 * i] = (uint32_t)P::mod[i];
+            p[2 * i + 1] = (uint32_t)(P::mod[i] >> 32);
+        }
+        const uint32_t inv32 = (uint32_t)P::inv;
+        u64 t[9];
+#pragma unroll
+        for (int i = 0; i < 9; ++i) t[i] = 0;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const u64 bi = b[i];
+            // multiply pipeline: t += a * b[i]
+            u64 cm = 0;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                u64 r = (u64)a[j] * bi + t[j] + cm;
+                t[j] = (uint32_t)r;
+                cm = r >> 32;
+            }
+            u64 top = t[8] + cm;
+            // reduce pipeline: one limb of Montgomery reduction
+            const u64 m = (uint32_t)((uint32_t)t[0] * inv32);
+            u64 cr = ((u64)(uint32_t)m * p[0] + t[0]) >> 32;
+#pragma unroll
+            for (int j = 1; j < 8; ++j) {
+                u64 r = (u64)(uint32_t)m * p[j] + t[j] + cr;
+                t[j - 1] = (uint32_t)r;
+                cr = r >> 32;
+            }
+            u64 r = top + cr;
+            t[7] = (uint32_t)r;
+            t[8] = r >> 32;
+        }
+        Fp4 res{{t[0] | (t[1] << 32), t[2] | (t[3] << 32), t[4] | (t[5] << 32),
+                 t[6] | (t[7] << 32)}};
+        if (t[8] || geq_mod(res.l)) {
+            u128 bw = 0;
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                u128 d = (u128)res.l[i] - P::mod[i] - (u64)bw;
+                res.l[i] = (u64)d;
+                bw = (d >> 64) & 1;
+            }
+        }
+        return res;
+    }
+
+    // 32-bit CIOS variant C: per row, all 8 multiply mads are issued into
+    // INDEPENDENT 64-bit accumulators (no serial carry through the mads),
+    // then one 32-bit carry-resolve chain folds the highs forward.
+    RNG_HD Fp4 mul_cios32c(const Fp4& o) const {
+        uint32_t a[8], b[8], p[8];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            a[2 * i] = (uint32_t)l[i];
+            a[2 * i + 1] = (uint32_t)(l[i] >> 32);
+            b[2 * i] = (uint32_t)o.l[i];
+            b[2 * i + 1] = (uint32_t)(o.l[i] >> 32);
+            p[2 * i] = (uint32_t)P::mod[i];
+            p[2 * i + 1] = (uint32_t)(P::mod[i] >> 32);
+        }
+        const uint32_t inv32 = (uint32_t)P::inv;
+        u64 t[9];
+#pragma unroll
+        for (int i = 0; i < 9; ++i) t[i] = 0;
+        u64 r[8];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            const u64 bi = b[i];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) r[j] = (u64)a[j] * bi + t[j];  // independent
+            // resolve: r[j] < 2^64 - 2^32; adding a 32-bit carry cannot overflow
+            u64 c = 0;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                r[j] += c;
+                t[j] = (uint32_t)r[j];
+                c = r[j] >> 32;
+            }
+            u64 top = t[8] + c;
+            const u64 m = (uint32_t)((uint32_t)t[0] * inv32);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) r[j] = (u64)(uint32_t)m * p[j] + t[j];
+            c = r[0] >> 32;
+#pragma unroll
+            for (int j = 1; j < 8; ++j) {
+                r[j] += c;
+                t[j - 1] = (uint32_t)r[j];
+                c = r[j] >> 32;
+            }
+            u64 rr = top + c;
+            t[7] = (uint32_t)rr;
+            t[8] = rr >> 32;
+        }
+        Fp4 res{{t[0] | (t[1] << 32), t[2] | (t[3] << 32), t[4] | (t[5] << 32),
+                 t[6] | (t[7] << 32)}};
+        if (t[8] || geq_mod(res.l)) {
+            u128 bw = 0;
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                u128 d = (u128)res.l[i] - P::mod[i] - (u64)bw;
+                res.l[i] = (u64)d;
+                bw = (d >> 64) & 1;
+            }
+        }
+        return res;
+    }
+
     // SOS: full 4x4 product (independent partial products -> ILP), then a
     // separate 4-round Montgomery reduction.
     RNG_HD Fp4 mul_sos(const Fp4& b) const {

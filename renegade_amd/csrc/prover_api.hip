@@ -489,11 +489,14 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         s->cap_nchunks = cap_nch;
     }
 
-    // binned (counting-scatter) pipeline: default ON for the non-GLV path;
-    // RNG_MSM_BINNED=0 falls back to the radix sort for A/B
+    // binned (counting-scatter) pipeline, RNG_MSM_BINNED=1 to enable.
+    // Measured SLOWER than the rocPRIM onesweep sort on MI355X (2^20 MSM:
+    // digits+hist 0.63 ms with hot-counter atomics + scatter 1.60 ms of
+    // uncoalesced atomic-cursor writes vs 0.04 + 0.86 for digits+sort;
+    // cohort headline 1026 vs 1071 proofs/s) — kept for A/B re-checks.
     static int binned_env = [] {
         const char* e = getenv("RNG_MSM_BINNED");
-        return e ? atoi(e) : 1;
+        return e ? atoi(e) : 0;
     }();
     const bool binned = binned_env && !glv;
 
