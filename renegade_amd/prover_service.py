@@ -45,12 +45,40 @@ def scalars_to_json(arr: np.ndarray):
 
 def json_to_scalars(vals, count):
     out = np.zeros(4 * count, dtype=np.uint64)
-    assert len(vals) == count, f"expected {count} scalars, got {len(vals)}"
+    if len(vals) != count:
+        raise ValueError(f"expected {count} scalars, got {len(vals)}")
     for i, v in enumerate(vals):
         x = int(v)
         for k in range(4):
             out[4 * i + k] = (x >> (64 * k)) & 0xFFFFFFFFFFFFFFFF
     return out
+
+
+def body_scalars(body, key, kind, count, which):
+    """Decode body[key] into the flat 4-limb scalar array.
+
+    Two accepted encodings:
+    - nested JSON object in the reference's api_types.rs struct shape
+      (field names/nesting per renegade_amd.api_schema — the serde layout of
+      the in-repo Rust witness/statement structs), scalars as decimal strings;
+    - flat decimal-scalar array in struct field order (legacy/debug form).
+    """
+    from renegade_amd import api_schema
+    v = body.get(key)
+    if v is None:
+        raise ValueError(f"missing {key}")
+    if isinstance(v, dict):
+        schema = api_schema.SCHEMAS[kind][0 if which == "witness" else 1]
+        try:
+            flat = api_schema.flatten(v, schema, [])
+        except (ValueError, TypeError, KeyError) as e:
+            raise ValueError(f"bad {key} shape: {e}")
+        if len(flat) != count:
+            raise ValueError(f"{key}: schema flattened to {len(flat)}, "
+                             f"expected {count}")
+        # nested bodies carry canonical values; the ABI takes Montgomery form
+        return json_to_scalars([api_schema.to_mont(x) for x in flat], count)
+    return json_to_scalars(v, count)
 
 
 class ProverService:
@@ -261,8 +289,8 @@ class ProverService:
 
     # ---- route handlers ----
     def prove_valid_balance_create(self, body):
-        w = json_to_scalars(body["witness"], 12)
-        s = json_to_scalars(body["statement"], 13)
+        w = body_scalars(body, "witness", 0, 12, "witness")
+        s = body_scalars(body, "statement", 0, 13, "statement")
         h = self.lib.rng_circ_vbc_from_scalars(ptr(w), ptr(s))
         if not h:
             raise ValueError("unsatisfied witness/statement")
@@ -279,9 +307,9 @@ class ProverService:
         nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
         if self.lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) != 0:
             raise ValueError(f"unknown kind {kind}")
-        w = json_to_scalars(body["witness"], nw.value) if nw.value else \
-            np.zeros(0, dtype=np.uint64)
-        s = json_to_scalars(body["statement"], ns.value)
+        w = body_scalars(body, "witness", kind, nw.value, "witness") if nw.value \
+            else np.zeros(0, dtype=np.uint64)
+        s = body_scalars(body, "statement", kind, ns.value, "statement")
         h = self.lib.rng_circ_from_scalars(kind, ptr(w), ptr(s))
         if not h:
             raise ValueError("unsatisfied witness/statement")
@@ -305,8 +333,8 @@ class ProverService:
         return out
 
     def prove_private_settlement(self, body):
-        w = json_to_scalars(body["witness"], 64)
-        s = json_to_scalars(body["statement"], 17)
+        w = body_scalars(body, "witness", 10, 64, "witness")
+        s = body_scalars(body, "statement", 10, 17, "statement")
         h = self.lib.rng_circ_settlement_from_scalars(ptr(w), ptr(s))
         if not h:
             raise ValueError("unsatisfied witness/statement")

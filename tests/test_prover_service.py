@@ -342,3 +342,56 @@ class TestProverService:
                                          ctypes.c_uint64(align), ctypes.c_uint64(off),
                                          ctypes.c_uint64(cnt), ptr(tau))
             assert ok == 1, f"{name} failed oracle link verification"
+
+    def test_nested_api_types_body(self, client):
+        """A request body in the api_types.rs struct shape (nested field
+        names, canonical values) proves end-to-end, and a tampered nested
+        statement is rejected (VERDICT r01 next #5)."""
+        from renegade_amd import api_schema
+        c, svc = client
+        lib = svc.lib
+        hdr = {"authorization": "Bearer hunter2"}
+
+        def nested_vectors(kind, seed=17):
+            nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+            assert lib.rng_ws_sizes(kind, ctypes.byref(nw), ctypes.byref(ns)) == 0
+            w = np.zeros(4 * nw.value, dtype=np.uint64)
+            s = np.zeros(4 * ns.value, dtype=np.uint64)
+            assert lib.rng_witness_statement(kind, seed, ptr(w), ptr(s)) == 0
+
+            def to_nested(limbs, schema):
+                rows = limbs.reshape(-1, 4)
+                ints = [int(r[0]) | int(r[1]) << 64 | int(r[2]) << 128 |
+                        int(r[3]) << 192 for r in rows]
+                nested, pos = api_schema.unflatten(
+                    [api_schema.from_mont(x) for x in ints], schema)
+                assert pos == len(ints)
+                return nested
+
+            return (to_nested(w, api_schema.SCHEMAS[kind][0]),
+                    to_nested(s, api_schema.SCHEMAS[kind][1]))
+
+        # intent-and-balance validity via nested body
+        wn, sn = nested_vectors(4)
+        assert isinstance(wn["old_intent"]["inner"]["amount_in"], int)
+        r = c.post("/prove-intent-and-balance-validity", headers=hdr,
+                   json={"witness": wn, "statement": sn})
+        assert r.status_code == 200, r.text
+        assert len(r.json()["proof"]) == 157
+        # a fee route with ElGamal material, nested
+        wn19, sn19 = nested_vectors(19)
+        assert "note_ciphertext" in sn19 and "ephemeral_key" in sn19["note_ciphertext"]
+        r = c.post("/prove-valid-private-protocol-fee-payment", headers=hdr,
+                   json={"witness": wn19, "statement": sn19})
+        assert r.status_code == 200, r.text
+        # tampered nested statement -> 400 (circuit unsatisfied)
+        sn["old_intent_nullifier"] = str(int(sn["old_intent_nullifier"]) + 1)
+        r = c.post("/prove-intent-and-balance-validity", headers=hdr,
+                   json={"witness": wn, "statement": sn})
+        assert r.status_code == 400
+        # malformed nested body (missing field) -> 400
+        bad = dict(wn)
+        del bad["intent"]
+        r = c.post("/prove-intent-and-balance-validity", headers=hdr,
+                   json={"witness": bad, "statement": sn})
+        assert r.status_code == 400
