@@ -624,3 +624,10 @@ if __name__ == "__main__":
     profiled = any(k.startswith(("ROCPROF", "ROCP_", "HSA_TOOLS")) for k in os.environ)
     if not profiled:
         os._exit(0)
+    # under a profiler: join the library's worker-pool threads so the
+    # profiler's finalizer can't stall on them, then exit normally
+    try:
+        import ctypes as _ct
+        _ct.CDLL(str(REPO / "renegade_amd" / "librenegade_prover.so")).rng_shutdown_pool()
+    except Exception:
+        pass

@@ -83,6 +83,9 @@ int rng_srs_gen_test_ptau(int power, uint64_t seed, uint8_t* out, size_t out_len
 /* Library/device introspection */
 int rng_gpu_available(void);
 const char* rng_version(void);
+/* Join the library's host worker-pool threads (profiler-finalizer hygiene;
+ * pool work runs serially afterwards).  Safe to call at any time. */
+void rng_shutdown_pool(void);
 int rng_set_device(int device);
 
 /* Per-kernel HIP-event times (ms) of the LAST rng_msm_* / rng_ntt_* call on

@@ -145,7 +145,8 @@ class HostPool {
             std::shared_ptr<Job> job;
             {
                 std::unique_lock<std::mutex> lk(mu_);
-                cv_.wait(lk, [&] { return generation_ != seen; });
+                cv_.wait(lk, [&] { return generation_ != seen || stopping_; });
+                if (stopping_) return;
                 seen = generation_;
                 job = job_;
             }
@@ -158,6 +159,24 @@ class HostPool {
     std::vector<std::thread> workers_;
     std::shared_ptr<Job> job_;
     uint64_t generation_ = 0;
+    bool stopping_ = false;
+
+  public:
+    // Join every worker (profilers' finalizers can stall on foreign live
+    // threads at process exit; bench/debug scripts call rng_shutdown_pool
+    // before exiting under rocprof).  parallel_for falls back to serial
+    // execution afterwards.
+    void stop() {
+        std::unique_lock<std::mutex> job_lk(job_mu_);
+        {
+            std::lock_guard<std::mutex> lk(mu_);
+            if (stopping_) return;
+            stopping_ = true;
+        }
+        cv_.notify_all();
+        for (auto& t : workers_) t.join();
+        workers_.clear();
+    }
 };
 
 struct EvtTimer {
@@ -1755,6 +1774,11 @@ extern "C" {
 const char* rng_version(void) { return "renegade_amd 0.1 (gfx950)"; }
 
 int rng_gpu_available(void) { return gpu_ok() ? 1 : 0; }
+
+/* Join the host worker pool's threads (harmless if never started).  Call
+ * before process exit under a profiler: rocprofv3's finalizer can stall on
+ * live foreign threads.  Subsequent pool work runs serially. */
+void rng_shutdown_pool(void) { HostPool::inst().stop(); }
 
 int rng_set_device(int device) {
     if (!gpu_ok()) return RNG_ERR_NO_GPU;

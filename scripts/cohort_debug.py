@@ -77,3 +77,6 @@ def main():
 
 if __name__ == "__main__":
     main()
+# (cohort_debug may run under rocprofv3: join pool threads before exit)
+import atexit, ctypes as _ct
+atexit.register(lambda: _ct.CDLL(str(REPO / "renegade_amd" / "librenegade_prover.so")).rng_shutdown_pool())
