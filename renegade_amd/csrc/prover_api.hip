@@ -437,7 +437,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint32_t chunk_sz = MSM_CHUNK;
     {
         uint64_t nbw = (1ull << (c - 1)) * G;
-        while (chunk_sz > 2 && nbw / chunk_sz < 32768) chunk_sz >>= 1;
+        // target >=128k lanes: the per-chunk suffix walk is a serial EC
+        // chain, so the chunk kernel is latency-bound until the chip is
+        // several waves deep per SIMD (r02: chunk 16 -> 4 at 2^20 cut
+        // window_chunks ~3x)
+        while (chunk_sz > 2 && nbw / chunk_sz < 131072) chunk_sz >>= 1;
         static int chunk_env = [] {
             const char* e = getenv("RNG_MSM_CHUNK");
             return e ? atoi(e) : 0;
@@ -717,7 +721,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         // more than chunks (idle blocks); RNG_MSM_SUBB overrides for tuning
         subb = MSM_SUBB;
         while (subb > 1 && cw / subb < 64) subb >>= 1;
-        while (subb < 64 && G * subb < 256 && cw / (2 * subb) >= 32 &&
+        while (subb < 64 && G * subb < 2048 && cw / (2 * subb) >= 32 &&
                (uint64_t)G * subb * 2 <= 8192)
             subb <<= 1;
         static int subb_env = [] {
