@@ -721,7 +721,9 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         // more than chunks (idle blocks); RNG_MSM_SUBB overrides for tuning
         subb = MSM_SUBB;
         while (subb > 1 && cw / subb < 64) subb >>= 1;
-        while (subb < 64 && G * subb < 2048 && cw / (2 * subb) >= 32 &&
+        // the combine blocks are serial EC chains: push toward >=4096 waves
+        // (1 block = 1 wave of 64 lanes) before letting per-thread work grow
+        while (subb < 256 && G * subb < 4096 && cw / (2 * subb) >= 32 &&
                (uint64_t)G * subb * 2 <= 8192)
             subb <<= 1;
         static int subb_env = [] {
@@ -741,22 +743,27 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
                              hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     et.collect(tls_msm_times, 5);
+    // two pooled stages: per-(poly, window) sub-block reduction, then the
+    // per-poly Horner across windows (the window count is small; a serial
+    // 1024-add fold on one core showed up at B=1 with wide subb)
+    std::vector<G1Jac> wred((size_t)G);
+    auto red_one = [&](uint32_t g) {
+        G1Jac sum = wsums[(size_t)g * subb];
+        for (uint32_t i = 1; i < subb; ++i) sum = sum.add(wsums[(size_t)g * subb + i]);
+        wred[g] = sum;
+    };
     auto fold_one = [&](uint32_t b) {
-        G1Jac ws[32];
-        for (uint32_t w = 0; w < W; ++w) {
-            uint32_t g = b * W + w;
-            G1Jac sum = wsums[(size_t)g * subb];
-            for (uint32_t i = 1; i < subb; ++i)
-                sum = sum.add(wsums[(size_t)g * subb + i]);
-            ws[w] = sum;
-        }
-        G1Jac acc = ws[W - 1];
+        G1Jac acc = wred[(size_t)b * W + W - 1];
         for (int w = (int)W - 2; w >= 0; --w) {
             for (uint32_t k = 0; k < c; ++k) acc = acc.dbl();
-            acc = acc.add(ws[w]);
+            acc = acc.add(wred[(size_t)b * W + w]);
         }
         h_result[b] = acc;
     };
+    if ((uint64_t)G * subb >= 64)
+        HostPool::inst().parallel_for(G, red_one);
+    else
+        for (uint32_t g = 0; g < G; ++g) red_one(g);
     if (B >= 8)
         HostPool::inst().parallel_for(B, fold_one);
     else
