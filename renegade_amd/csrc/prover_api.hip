@@ -730,7 +730,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
             const char* e = getenv("RNG_MSM_SUBB");
             return e ? atoi(e) : 0;
         }();
-        if (subb_env >= 1 && subb_env <= 16) subb = (uint32_t)subb_env;
+        if (subb_env >= 1 && subb_env <= 256) subb = (uint32_t)subb_env;
         hipLaunchKernelGGL(k_msm_window_combine, dim3(G * subb), dim3(64), 0, stream,
                            s->partials, c, chunk_sz, subb, s->window_sums);
         HIP_CHECK(hipGetLastError());

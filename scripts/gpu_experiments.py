@@ -105,3 +105,7 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+import atexit as _atexit
+import ctypes as _ct
+_atexit.register(lambda: _ct.CDLL(str(REPO / "renegade_amd" / "librenegade_prover.so")).rng_shutdown_pool())
