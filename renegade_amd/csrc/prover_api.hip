@@ -721,9 +721,10 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         // more than chunks (idle blocks); RNG_MSM_SUBB overrides for tuning
         subb = MSM_SUBB;
         while (subb > 1 && cw / subb < 64) subb >>= 1;
-        // the combine blocks are serial EC chains: push toward >=4096 waves
-        // (1 block = 1 wave of 64 lanes) before letting per-thread work grow
-        while (subb < 256 && G * subb < 4096 && cw / (2 * subb) >= 32 &&
+        // the combine blocks are serial EC chains: push toward >=2048 waves
+        // (1 block = 1 wave of 64 lanes) while keeping >=64 chunks per block
+        // so the scan-based combine2 applies
+        while (subb < 256 && G * subb < 4096 && cw / (2 * subb) >= 64 &&
                (uint64_t)G * subb * 2 <= 8192)
             subb <<= 1;
         static int subb_env = [] {
@@ -731,8 +732,16 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
             return e ? atoi(e) : 0;
         }();
         if (subb_env >= 1 && subb_env <= 256) subb = (uint32_t)subb_env;
-        hipLaunchKernelGGL(k_msm_window_combine, dim3(G * subb), dim3(64), 0, stream,
-                           s->partials, c, chunk_sz, subb, s->window_sums);
+        if (cw % subb == 0 && (cw / subb) % 64 == 0) {
+            // scan-based combine: no per-chunk scalar weights
+            hipLaunchKernelGGL(k_msm_window_combine2, dim3(G * subb), dim3(64), 0,
+                               stream, s->partials, c, chunk_sz, subb,
+                               s->window_sums);
+        } else {
+            hipLaunchKernelGGL(k_msm_window_combine, dim3(G * subb), dim3(64), 0,
+                               stream, s->partials, c, chunk_sz, subb,
+                               s->window_sums);
+        }
         HIP_CHECK(hipGetLastError());
         et.mark(stream);
     }
