@@ -732,6 +732,14 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
             return e ? atoi(e) : 0;
         }();
         if (subb_env >= 1 && subb_env <= 256) subb = (uint32_t)subb_env;
+        static int msm_dbg = [] {
+            const char* e = getenv("RNG_MSM_DEBUG");
+            return e ? atoi(e) : 0;
+        }();
+        if (msm_dbg)
+            fprintf(stderr, "[msm] n=%lu B=%u c=%u chunk=%u cw=%u subb=%u combine=%s\n",
+                    (unsigned long)n, B, c, chunk_sz, cw, subb,
+                    (cw % subb == 0 && (cw / subb) % 64 == 0) ? "scan2" : "v1");
         if (cw % subb == 0 && (cw / subb) % 64 == 0) {
             // scan-based combine: no per-chunk scalar weights
             hipLaunchKernelGGL(k_msm_window_combine2, dim3(G * subb), dim3(64), 0,
