@@ -52,6 +52,50 @@ __global__ __launch_bounds__(256) void k_quotient(
     out[t] = gate.add(perm).add(l1term).mul(ch.zh_inv[t & 7]);
 }
 
+// cohort-batched quotient: K proofs in one launch (K*m threads) — a single
+// proof's m=8n grid is only ~0.5 waves/SIMD at n=4096, pure latency; the
+// cohort profile showed 483 small k_quotient launches at 14% of GPU time.
+// Wire/z/PI cosets are laid out per proof at stride 7m (slots 0-4 wires,
+// 5 z, 6 PI — plonk_prove_cohort_impl); challenges per proof from HBM.
+__global__ __launch_bounds__(256) void k_quotient_batch(
+    const Fr* sel, const Fr* sig, const Fr* coset_all, const Fr* l1,
+    const Fr* xpow, Fr* out_all, uint32_t m, uint32_t K, const QuotChal* chs) {
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (tid >= (uint64_t)K * m) return;
+    uint32_t p = (uint32_t)(tid / m);
+    uint32_t t = (uint32_t)(tid - (uint64_t)p * m);
+    const QuotChal& ch = chs[p];
+    const Fr* w = coset_all + 7 * (uint64_t)p * m;
+    const Fr* z = w + 5 * (uint64_t)m;
+    const Fr* pi = w + 6 * (uint64_t)m;
+    Fr* out = out_all + (uint64_t)p * m;
+    auto p5 = [](const Fr& v) {
+        Fr v2 = v.sqr();
+        return v2.sqr().mul(v);
+    };
+    Fr w0 = w[t], w1 = w[m + t], w2 = w[2 * m + t], w3 = w[3 * m + t],
+       w4 = w[4 * m + t];
+    Fr gate = sel[11 * m + t].add(pi[t]);
+    gate = gate.add(sel[0 * m + t].mul(w0)).add(sel[1 * m + t].mul(w1));
+    gate = gate.add(sel[2 * m + t].mul(w2)).add(sel[3 * m + t].mul(w3));
+    gate = gate.add(sel[4 * m + t].mul(w0.mul(w1))).add(sel[5 * m + t].mul(w2.mul(w3)));
+    gate = gate.add(sel[6 * m + t].mul(p5(w0))).add(sel[7 * m + t].mul(p5(w1)));
+    gate = gate.add(sel[8 * m + t].mul(p5(w2))).add(sel[9 * m + t].mul(p5(w3)));
+    gate = gate.add(sel[12 * m + t].mul(w0.mul(w1).mul(w2).mul(w3).mul(w4)));
+    gate = gate.sub(sel[10 * m + t].mul(w4));
+    Fr x = xpow[t];
+    Fr f = Fr::one(), g = Fr::one();
+    const Fr* ws[5] = {&w0, &w1, &w2, &w3, &w4};
+    for (int j = 0; j < 5; ++j) {
+        f = f.mul(ws[j]->add(ch.beta.mul(ch.k[j]).mul(x)).add(ch.gamma));
+        g = g.mul(ws[j]->add(ch.beta.mul(sig[j * m + t])).add(ch.gamma));
+    }
+    Fr zshift = z[(t + 8) % m];
+    Fr perm = ch.alpha.mul(z[t].mul(f).sub(zshift.mul(g)));
+    Fr l1term = ch.alpha2.mul(z[t].sub(Fr::one())).mul(l1[t]);
+    out[t] = gate.add(perm).add(l1term).mul(ch.zh_inv[t & 7]);
+}
+
 // elementwise multiply: data[i] *= table[i] (coset scaling)
 __global__ __launch_bounds__(256) void k_mul_pointwise(Fr* data, const Fr* table,
                                                        uint32_t count) {

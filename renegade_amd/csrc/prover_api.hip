@@ -969,9 +969,10 @@ struct CohortScratch {  // per-thread device scratch for cohort proving
     Fr* coset_out = nullptr; // 7k*m coset evaluations (5 wires + z + PI per proof)
     Fr* q_all = nullptr;     // k*m quotient evals
     Fr* ntt_tmp = nullptr;   // 7k*m NTT ping-pong
+    QuotChal* chs = nullptr;  // k challenge packs (batched quotient)
     ~CohortScratch() {
         for (void* b : {(void*)stage, (void*)canon, (void*)coset_in, (void*)coset_out,
-                        (void*)q_all, (void*)ntt_tmp})
+                        (void*)q_all, (void*)ntt_tmp, (void*)chs})
             hip_free_guarded(b);
     }
 };
@@ -992,6 +993,7 @@ static int cohort_scratch_ensure(uint64_t n, uint64_t k) {
     HIP_CHECK(hipMalloc(&s->coset_out, 7 * k * m * sizeof(Fr)));
     HIP_CHECK(hipMalloc(&s->q_all, k * m * sizeof(Fr)));
     HIP_CHECK(hipMalloc(&s->ntt_tmp, 7 * k * m * sizeof(Fr)));
+    HIP_CHECK(hipMalloc(&s->chs, k * sizeof(QuotChal)));
     s->n = n;
     s->k = k;
     return RNG_OK;
@@ -1290,20 +1292,26 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
             zh = hbatch_inverse(zh);
             for (int t = 0; t < 8; ++t) ch.zh_inv[t] = zh[t];
         }
-        uint32_t blocks = (m + 255) / 256;
-        for (uint32_t p = 0; p < k; ++p) {
-            QuotChal chp = ch;
-            chp.beta = beta[p];
-            chp.gamma = gamma[p];
-            chp.alpha = alpha[p];
-            chp.alpha2 = alpha[p].sqr();
-            Fr* base = cs->coset_out + 7 * (size_t)p * m;
-            hipLaunchKernelGGL(k_quotient, dim3(blocks), dim3(256), 0, RNG_STREAM,
-                               pk.sel_coset, pk.sig_coset, base /* 5 wires */,
-                               base + 5 * (size_t)m, base + 6 * (size_t)m,
-                               pk.l1_coset, mp->xpow, cs->q_all + (size_t)p * m, m,
-                               chp);
+        // ONE batched quotient launch for all k proofs (a single proof's m
+        // grid is latency-starved; 483 small launches were 14% of cohort GPU
+        // time in the r02 profile)
+        {
+            std::vector<QuotChal> chs_h(k, ch);
+            for (uint32_t p = 0; p < k; ++p) {
+                chs_h[p].beta = beta[p];
+                chs_h[p].gamma = gamma[p];
+                chs_h[p].alpha = alpha[p];
+                chs_h[p].alpha2 = alpha[p].sqr();
+            }
+            HIP_CHECK(hipMemcpyAsync(cs->chs, chs_h.data(), k * sizeof(QuotChal),
+                                     hipMemcpyHostToDevice, RNG_STREAM));
+            uint64_t total_kq = (uint64_t)k * m;
+            hipLaunchKernelGGL(k_quotient_batch,
+                               dim3((uint32_t)((total_kq + 255) / 256)), dim3(256), 0,
+                               RNG_STREAM, pk.sel_coset, pk.sig_coset, cs->coset_out,
+                               pk.l1_coset, mp->xpow, cs->q_all, m, k, cs->chs);
             HIP_CHECK(hipGetLastError());
+            HIP_CHECK(hipStreamSynchronize(RNG_STREAM));  // chs_h lifetime
         }
         // batched inverse coset NTT of the k quotients
         rc = ntt_dev_run(ctx, cs->q_all, cs->ntt_tmp, m, k, true, RNG_STREAM);
