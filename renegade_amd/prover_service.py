@@ -24,6 +24,7 @@ import argparse
 import ctypes
 import os
 import threading
+import time
 from pathlib import Path
 
 import numpy as np
@@ -79,6 +80,88 @@ def body_scalars(body, key, kind, count, which):
         # nested bodies carry canonical values; the ABI takes Montgomery form
         return json_to_scalars([api_schema.to_mont(x) for x in flat], count)
     return json_to_scalars(v, count)
+
+
+class CohortBatcher:
+    """Groups concurrent same-circuit prove requests into one
+    `rng_prove_cohort` call (the reference's proof manager drains its job
+    queue into a rayon pool — native_proof_manager.rs:193-198; the cohort
+    prover is that shape with fused per-round GPU batches).
+
+    Requests wait up to `window_ms` for co-batching; a burst of k same-kind
+    requests is proven as one k-cohort.  Single requests degrade to k=1
+    (identical bytes to rng_prove)."""
+
+    def __init__(self, svc, window_ms=3.0, max_k=32):
+        self.svc = svc
+        self.window = window_ms / 1e3
+        self.max_k = max_k
+        self._pending = {}
+        self._lock = threading.Lock()
+        self._wake = threading.Event()
+        #: telemetry: cohort calls vs proofs served (tests assert batching)
+        self.cohort_calls = 0
+        self.proofs_served = 0
+        svc.lib.rng_prove_cohort.restype = ctypes.c_int
+        svc.lib.rng_prove_cohort.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                             ctypes.c_uint64, U64P, U64P, U64P,
+                                             U64P, U64P]
+        t = threading.Thread(target=self._loop, daemon=True,
+                             name="cohort-batcher")
+        t.start()
+
+    def prove(self, circuit_name, wires, pubs, seed, n):
+        item = {"wires": wires, "pubs": pubs, "seed": seed,
+                "done": threading.Event(), "proof": None, "hint": None,
+                "err": None}
+        with self._lock:
+            self._pending.setdefault(circuit_name, []).append(item)
+        self._wake.set()
+        item["done"].wait()
+        if item["err"] is not None:
+            raise RuntimeError(item["err"])
+        return item["proof"], item["hint"]
+
+    def _loop(self):
+        while True:
+            self._wake.wait()
+            time.sleep(self.window)  # let a burst accumulate
+            with self._lock:
+                drained = self._pending
+                self._pending = {}
+                self._wake.clear()
+            for name, items in drained.items():
+                for lo in range(0, len(items), self.max_k):
+                    self._run(name, items[lo:lo + self.max_k])
+
+    def _run(self, name, items):
+        try:
+            svc = self.svc
+            pk = svc.pks[name]
+            n = int(svc.pk_meta[name]["n"])
+            k = len(items)
+            wires_all = np.concatenate([it["wires"] for it in items])
+            pubs_all = np.concatenate([it["pubs"] for it in items])
+            seeds = np.array([it["seed"] for it in items], dtype=np.uint64)
+            proofs = np.zeros(157 * k, dtype=np.uint64)
+            hlen = 4 * (n + 2) + 9
+            hints = np.zeros(hlen * k, dtype=np.uint64)
+            rc = svc.lib.rng_prove_cohort(svc.ctx.h, ctypes.c_void_p(pk), k,
+                                          ptr(wires_all), ptr(pubs_all),
+                                          ptr(seeds), ptr(proofs), ptr(hints))
+            if rc != 0:
+                raise RuntimeError(f"rng_prove_cohort rc={rc}")
+            self.cohort_calls += 1
+            self.proofs_served += k
+            for i, it in enumerate(items):
+                it["proof"] = proofs[157 * i:157 * (i + 1)].copy()
+                it["hint"] = hints[hlen * i:hlen * (i + 1)].copy()
+        except Exception as e:  # noqa: BLE001 — fail every waiter loudly
+            for it in items:
+                it["err"] = str(e)
+        finally:
+            for it in items:
+                it["done"].set()
 
 
 class ProverService:
@@ -151,7 +234,8 @@ class ProverService:
                 self.pk_meta[circuit_name] = dict(n=n, lg=lg.reshape(-1, 3))
         return n, npub, wires, pubs
 
-    def __init__(self, srs_power=15, srs_seed=42, ptau_path=None):
+    def __init__(self, srs_power=15, srs_seed=42, ptau_path=None,
+                 batch_window_ms=None):
         import sys
         sys.path.insert(0, str(REPO))
         from renegade_amd import load_prover
@@ -187,6 +271,10 @@ class ProverService:
         self.pks = {}
         self.pk_meta = {}
         self.lock = threading.Lock()
+        # service-level cohort batching: co-arriving
+        # same-circuit requests prove as one fused cohort
+        self.batcher = (CohortBatcher(self, window_ms=batch_window_ms)
+                        if batch_window_ms else None)
 
     def _sig(self):
         lib = self.lib
@@ -229,6 +317,12 @@ class ProverService:
         # successive proofs' blinders from a counter (the reference blinds
         # with thread_rng at traits.rs:994)
         seed = int.from_bytes(os.urandom(8), "little")
+        if self.batcher is not None:
+            proof, hint = self.batcher.prove(circuit_name, wires, pubs, seed,
+                                             int(n))
+            if not want_hint:
+                hint = None
+            return proof, hint, self.pk_meta[circuit_name]
         pk = self.pks[circuit_name]
         proof = np.zeros(157, dtype=np.uint64)
         hint = np.zeros(4 * (n + 2) + 9, dtype=np.uint64) if want_hint else None
@@ -359,8 +453,11 @@ class ProverService:
 
 
 def create_app(service=None, password=None):
-    from fastapi import FastAPI, HTTPException, Request
+    from fastapi import Body, FastAPI, HTTPException, Request
 
+    # handlers are SYNC (def, not async def): FastAPI then runs them in its
+    # thread pool, so concurrent requests actually overlap (ctypes releases
+    # the GIL) and co-arriving requests can batch into one cohort
     app = FastAPI(title="renegade_amd prover service")
     svc = service
 
@@ -377,18 +474,16 @@ def create_app(service=None, password=None):
         return {"status": "ok", "backend": svc.plib.version}
 
     @app.post("/prove-valid-balance-create")
-    async def vbc(request: Request):
+    def vbc(request: Request, body: dict = Body(...)):
         auth(request)
-        body = await request.json()
         try:
             return svc.prove_valid_balance_create(body)
         except ValueError as e:
             raise HTTPException(status_code=400, detail=str(e))
 
     @app.post("/prove-intent-and-balance-private-settlement")
-    async def settle(request: Request):
+    def settle(request: Request, body: dict = Body(...)):
         auth(request)
-        body = await request.json()
         try:
             return svc.prove_private_settlement(body)
         except ValueError as e:
@@ -419,9 +514,8 @@ def create_app(service=None, password=None):
     }
 
     def make_route(path, kind, want_hint):
-        async def handler(request: Request):
+        def handler(request: Request, body: dict = Body(...)):
             auth(request)
-            body = await request.json()
             try:
                 return svc.prove_kind(path.lstrip("/"), kind, body, want_hint)
             except ValueError as e:
@@ -444,9 +538,13 @@ def main():
     ap.add_argument("--ptau", default=None,
                     help="path to the production ptau file (snarkjs layout); "
                          "omitted = deterministic TEST SRS")
+    ap.add_argument("--batch-window-ms", type=float, default=3.0,
+                    help="co-batch same-circuit requests arriving within this "
+                         "window into one fused proof cohort (0 = off)")
     args = ap.parse_args()
     import uvicorn
-    svc = ProverService(srs_power=args.srs_power, ptau_path=args.ptau)
+    svc = ProverService(srs_power=args.srs_power, ptau_path=args.ptau,
+                        batch_window_ms=args.batch_window_ms or None)
     if args.preload:
         svc.preload_keys()
     app = create_app(svc, args.password)

@@ -395,3 +395,54 @@ class TestProverService:
         r = c.post("/prove-intent-and-balance-validity", headers=hdr,
                    json={"witness": bad, "statement": sn})
         assert r.status_code == 400
+
+
+class TestCohortBatching:
+    def test_concurrent_requests_batch_into_cohorts(self, orc):
+        """A burst of same-circuit requests is proven as fused cohorts
+        (service-level shape of native_proof_manager.rs:193-198): all
+        succeed, and the batcher used fewer rng_prove_cohort calls than
+        there were requests."""
+        from concurrent.futures import ThreadPoolExecutor
+        from fastapi.testclient import TestClient
+        from renegade_amd import load_prover
+        from renegade_amd.prover_service import ProverService, create_app
+        plib = load_prover()
+        if not plib.gpu_available:
+            pytest.skip("no GPU")
+        svc = ProverService(srs_power=14, batch_window_ms=10.0)
+        app = create_app(svc, password=None)
+        c = TestClient(app)
+        lib = svc.lib
+        lib.rng_ws_sizes.restype = ctypes.c_int
+        lib.rng_ws_sizes.argtypes = [ctypes.c_int, ctypes.POINTER(ctypes.c_uint64),
+                                     ctypes.POINTER(ctypes.c_uint64)]
+        lib.rng_witness_statement.restype = ctypes.c_int
+        lib.rng_witness_statement.argtypes = [ctypes.c_int, ctypes.c_uint64,
+                                              U64P, U64P]
+        nw, ns = ctypes.c_uint64(), ctypes.c_uint64()
+        assert lib.rng_ws_sizes(3, ctypes.byref(nw), ctypes.byref(ns)) == 0
+        w = np.zeros(4 * nw.value, dtype=np.uint64)
+        s = np.zeros(4 * ns.value, dtype=np.uint64)
+        assert lib.rng_witness_statement(3, 7, ptr(w), ptr(s)) == 0
+        from renegade_amd.prover_service import scalars_to_json
+        body = {"witness": scalars_to_json(w), "statement": scalars_to_json(s)}
+        # warm the PK cache so the burst hits the batcher together
+        r = c.post("/prove-valid-order-cancellation", json=body)
+        assert r.status_code == 200, r.text
+        calls_before = svc.batcher.cohort_calls
+
+        def hit(_):
+            rr = c.post("/prove-valid-order-cancellation", json=body)
+            assert rr.status_code == 200, rr.text
+            return rr.json()["proof"]
+
+        N = 12
+        with ThreadPoolExecutor(max_workers=N) as ex:
+            proofs = list(ex.map(hit, range(N)))
+        assert all(len(p) == 157 for p in proofs)
+        used = svc.batcher.cohort_calls - calls_before
+        assert used < N, f"no batching happened ({used} cohorts for {N} requests)"
+        assert svc.batcher.proofs_served >= N + 1
+        # blinder seeds are per-request: proofs in one cohort must differ
+        assert len({tuple(p) for p in proofs}) == len(proofs)
