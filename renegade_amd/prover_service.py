@@ -127,9 +127,12 @@ class CohortBatcher:
             self._wake.wait()
             time.sleep(self.window)  # let a burst accumulate
             with self._lock:
+                # clear BEFORE swapping: an enqueue that lands after the swap
+                # re-sets the event and is drained next loop (clearing after
+                # could strand an item whose set() raced the clear)
+                self._wake.clear()
                 drained = self._pending
                 self._pending = {}
-                self._wake.clear()
             for name, items in drained.items():
                 for lo in range(0, len(items), self.max_k):
                     self._run(name, items[lo:lo + self.max_k])
