@@ -117,6 +117,47 @@ __global__ __launch_bounds__(256) void k_copy_pad(const Fr* src, uint32_t len, F
     dst[i] = (i < len) ? src[i] : Fr::zero();
 }
 
+// cohort R1: blind + pad 5k wire polynomials on device.  in: coeffs
+// (5k polys x n contiguous), blinders (2 per poly); out: 5k polys at
+// stride n+2 with wp[0]-=b0, wp[1]-=b1, wp[n]+=b0, wp[n+1]+=b1
+// (traits.rs:994 blinding shape; DRBG order preserved host-side).
+__global__ __launch_bounds__(256) void k_blind_wires_batch(
+    const Fr* coeffs, const Fr* blinders, Fr* out, uint32_t n, uint64_t total) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    uint64_t stride = n + 2;
+    uint64_t g = i / stride;
+    uint32_t j = (uint32_t)(i - g * stride);
+    Fr v = (j < n) ? coeffs[g * n + j] : Fr::zero();
+    Fr b0 = blinders[2 * g], b1 = blinders[2 * g + 1];
+    if (j == 0) v = v.sub(b0);
+    if (j == 1) v = v.sub(b1);
+    if (j == n) v = v.add(b0);
+    if (j == n + 1) v = v.add(b1);
+    out[i] = v;
+}
+
+// cohort R3: split each proof's quotient (m coeffs) into 5 chunks of
+// degree <= n+2 with the linking blinders applied on device: chunk i gets
+// qc[0] -= b_{i-1} and (i<4) qc[n+2] += b_i.  out: 5k polys at stride n+3
+// (chunk 4's top coefficient is zero).
+__global__ __launch_bounds__(256) void k_quot_chunks_batch(
+    const Fr* quot, const Fr* blinders /* 4 per proof */, Fr* out, uint32_t n,
+    uint32_t m, uint64_t total /* 5k*(n+3) */) {
+    uint64_t idx = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;
+    uint64_t stride = (uint64_t)n + 3;
+    uint64_t g = idx / stride;           // p*5 + i
+    uint32_t j = (uint32_t)(idx - g * stride);
+    uint64_t p = g / 5;
+    uint32_t i = (uint32_t)(g - p * 5);
+    Fr v = Fr::zero();
+    if (j < n + 2) v = quot[p * m + (uint64_t)i * (n + 2) + j];
+    if (j == 0 && i > 0) v = v.sub(blinders[4 * p + i - 1]);
+    if (j == n + 2 && i < 4) v = v.add(blinders[4 * p + i]);
+    out[idx] = v;
+}
+
 // batched coset scaling for nb polynomials of `stride` coefficients each
 // stored back to back: data[b*stride + j] *= table[j]  (table >= stride long)
 __global__ __launch_bounds__(256) void k_mul_pointwise_mod(Fr* data, const Fr* table,

@@ -970,9 +970,10 @@ struct CohortScratch {  // per-thread device scratch for cohort proving
     Fr* q_all = nullptr;     // k*m quotient evals
     Fr* ntt_tmp = nullptr;   // 7k*m NTT ping-pong
     QuotChal* chs = nullptr;  // k challenge packs (batched quotient)
+    Fr* blinders = nullptr;   // 10k wire + 4k chunk blinders (device)
     ~CohortScratch() {
         for (void* b : {(void*)stage, (void*)canon, (void*)coset_in, (void*)coset_out,
-                        (void*)q_all, (void*)ntt_tmp, (void*)chs})
+                        (void*)q_all, (void*)ntt_tmp, (void*)chs, (void*)blinders})
             hip_free_guarded(b);
     }
 };
@@ -994,10 +995,14 @@ static int cohort_scratch_ensure(uint64_t n, uint64_t k) {
     HIP_CHECK(hipMalloc(&s->q_all, k * m * sizeof(Fr)));
     HIP_CHECK(hipMalloc(&s->ntt_tmp, 7 * k * m * sizeof(Fr)));
     HIP_CHECK(hipMalloc(&s->chs, k * sizeof(QuotChal)));
+    HIP_CHECK(hipMalloc(&s->blinders, 14 * k * sizeof(Fr)));
     s->n = n;
     s->k = k;
     return RNG_OK;
 }
+
+static int commit_staged(RngCtxImpl* ctx, uint64_t m, uint32_t B, G1Aff* out,
+                         bool* out_inf);
 
 // Fused commitment of B host polynomials (any B; shorter polys zero-padded
 // to the longest) through ONE msm_dev_run.  Host window fold + affine
@@ -1040,6 +1045,19 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
     });
     HIP_CHECK(hipMemcpyAsync(s->stage, packed.data(), B * m * sizeof(Fr),
                              hipMemcpyHostToDevice, RNG_STREAM));
+    tr("msm");
+    int rc = commit_staged(ctx, m, B, out, out_inf);
+    tr("msm-done");
+    tr_msm_stages();
+    return rc;
+}
+
+// commit B polynomials that are ALREADY staged on device in s->stage
+// (B consecutive m-coefficient blocks, Montgomery form)
+static int commit_staged(RngCtxImpl* ctx, uint64_t m, uint32_t B, G1Aff* out,
+                         bool* out_inf) {
+    CohortScratch* s = tls_cohort_scratch.get();
+    if (m > ctx->srs_count) return RNG_ERR_BAD_ARG;
     uint32_t blocks = (uint32_t)((B * m + 255) / 256);
     hipLaunchKernelGGL(k_fr_to_canonical, dim3(blocks), dim3(256), 0, RNG_STREAM,
                        s->stage, s->canon, (uint32_t)(B * m));
@@ -1053,11 +1071,8 @@ static int commit_cohort(RngCtxImpl* ctx, const std::vector<Fr>* const* polys,
     uint32_t c = (c_env >= 8 && c_env <= 16) ? (uint32_t)c_env : msm_auto_c(m);
     if ((uint64_t)B * ((256 + c - 1) / c) > 60000) return RNG_ERR_BAD_ARG;  // g<<16 cap
     std::vector<G1Jac> res(B);
-    tr("msm");
     int rc = msm_dev_run((const G1Aff*)ctx->srs_dev, s->canon, m, c, res.data(), B,
                          RNG_STREAM, (const G1Aff*)ctx->srs_glv_dev);
-    tr("msm-done");
-    tr_msm_stages();
     if (rc != RNG_OK) return rc;
     HostPool::inst().parallel_for(B, [&](uint32_t b) {
         uint64_t rec[9];
@@ -1128,30 +1143,39 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
                              true, RNG_STREAM);
         if (rc != RNG_OK) return rc;
         COHORT_TRACE("r1-ifft-done");
-        std::vector<Fr> host_flat((size_t)5 * k * n);
-        HIP_CHECK(hipMemcpyAsync(host_flat.data(), cs->ntt_tmp,
-                                 (size_t)5 * k * n * sizeof(Fr), hipMemcpyDeviceToHost,
+        // blind + pad ON DEVICE (blinders drawn host-side in the single-
+        // proof DRBG order), commit straight from the staged device buffer;
+        // the blinded coefficients stream back D2H on the same stream and
+        // are complete once the commit's internal sync returns
+        std::vector<Fr> blind_h(10 * (size_t)k);
+        for (uint32_t p = 0; p < k; ++p)
+            for (int j = 0; j < 5; ++j) {
+                blind_h[10 * (size_t)p + 2 * j] = drbg[p].next();
+                blind_h[10 * (size_t)p + 2 * j + 1] = drbg[p].next();
+            }
+        HIP_CHECK(hipMemcpyAsync(cs->blinders, blind_h.data(),
+                                 blind_h.size() * sizeof(Fr), hipMemcpyHostToDevice,
                                  RNG_STREAM));
-        HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
+        uint64_t total1 = 5 * (uint64_t)k * (n + 2);
+        hipLaunchKernelGGL(k_blind_wires_batch,
+                           dim3((uint32_t)((total1 + 255) / 256)), dim3(256), 0,
+                           RNG_STREAM, cs->ntt_tmp, cs->blinders, cs->stage,
+                           (uint32_t)n, total1);
+        HIP_CHECK(hipGetLastError());
+        std::vector<Fr> host_flat(total1);
+        HIP_CHECK(hipMemcpyAsync(host_flat.data(), cs->stage, total1 * sizeof(Fr),
+                                 hipMemcpyDeviceToHost, RNG_STREAM));
         COHORT_TRACE("r1-sync-done");
+        if (commit_staged(ctx, n + 2, 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
+            return RNG_ERR_HIP;
         pool.parallel_for(k, [&](uint32_t p) {
             for (int j = 0; j < 5; ++j) {
                 auto& wp = wpoly[5 * (size_t)p + j];
-                const Fr* src = host_flat.data() + ((size_t)5 * p + j) * n;
-                wp.assign(src, src + n);
-                Fr b0 = drbg[p].next(), b1 = drbg[p].next();
-                wp.resize(n + 2, Fr::zero());
-                wp[0] = wp[0].sub(b0);
-                wp[1] = wp[1].sub(b1);
-                wp[n] = wp[n].add(b0);
-                wp[n + 1] = wp[n + 1].add(b1);
+                const Fr* src = host_flat.data() + ((size_t)5 * p + j) * (n + 2);
+                wp.assign(src, src + (n + 2));
             }
         });
         COHORT_TRACE("r1-blind-done");
-        for (uint32_t p = 0; p < k; ++p)
-            for (int j = 0; j < 5; ++j) ps[5 * (size_t)p + j] = &wpoly[5 * (size_t)p + j];
-        if (commit_cohort(ctx, ps.data(), 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
-            return RNG_ERR_HIP;
         pool.parallel_for(k, [&](uint32_t p) {
             for (int j = 0; j < 5; ++j) {
                 comms[13 * (size_t)p + j] = cbuf[5 * (size_t)p + j];
@@ -1321,31 +1345,34 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
                            dim3(256), 0, RNG_STREAM, cs->ntt_tmp, mp->gpow_inv, m,
                            total_q);
         HIP_CHECK(hipGetLastError());
-        std::vector<Fr> quot_host((size_t)k * m);
-        HIP_CHECK(hipMemcpyAsync(quot_host.data(), cs->ntt_tmp,
-                                 (size_t)k * m * sizeof(Fr), hipMemcpyDeviceToHost,
+        // chunk-split + linking blinders ON DEVICE, commit from the staged
+        // buffer; the blinded chunks stream back D2H for the host R5
+        // linearization (complete after the commit's internal sync)
+        std::vector<Fr> qblind_h(4 * (size_t)k);
+        for (uint32_t p = 0; p < k; ++p)
+            for (int i = 0; i < 4; ++i) qblind_h[4 * (size_t)p + i] = drbg[p].next();
+        HIP_CHECK(hipMemcpyAsync(cs->blinders, qblind_h.data(),
+                                 qblind_h.size() * sizeof(Fr),
+                                 hipMemcpyHostToDevice, RNG_STREAM));
+        uint64_t total_c = 5 * (uint64_t)k * (n + 3);
+        hipLaunchKernelGGL(k_quot_chunks_batch,
+                           dim3((uint32_t)((total_c + 255) / 256)), dim3(256), 0,
+                           RNG_STREAM, cs->ntt_tmp, cs->blinders, cs->stage,
+                           (uint32_t)n, m, total_c);
+        HIP_CHECK(hipGetLastError());
+        std::vector<Fr> chunks_host(total_c);
+        HIP_CHECK(hipMemcpyAsync(chunks_host.data(), cs->stage,
+                                 total_c * sizeof(Fr), hipMemcpyDeviceToHost,
                                  RNG_STREAM));
-        HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
+        if (commit_staged(ctx, n + 3, 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
+            return RNG_ERR_HIP;
         pool.parallel_for(k, [&](uint32_t p) {
-            const Fr* quot = quot_host.data() + (size_t)p * m;
-            Fr prev = Fr::zero();
             for (int i = 0; i < 5; ++i) {
                 auto& qc = quot_chunks[5 * (size_t)p + i];
-                qc.assign(quot + (size_t)i * (n + 2), quot + (size_t)(i + 1) * (n + 2));
-                Fr bnext = (i < 4) ? drbg[p].next() : Fr::zero();
-                qc[0] = qc[0].sub(prev);
-                if (i < 4) {
-                    qc.resize(n + 3, Fr::zero());
-                    qc[n + 2] = qc[n + 2].add(bnext);
-                }
-                prev = bnext;
+                const Fr* src = chunks_host.data() + (5 * (size_t)p + i) * (n + 3);
+                qc.assign(src, src + (i < 4 ? n + 3 : n + 2));
             }
         });
-        for (uint32_t p = 0; p < k; ++p)
-            for (int i = 0; i < 5; ++i)
-                ps[5 * (size_t)p + i] = &quot_chunks[5 * (size_t)p + i];
-        if (commit_cohort(ctx, ps.data(), 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
-            return RNG_ERR_HIP;
         pool.parallel_for(k, [&](uint32_t p) {
             for (int i = 0; i < 5; ++i) {
                 comms[13 * (size_t)p + 6 + i] = cbuf[5 * (size_t)p + i];
