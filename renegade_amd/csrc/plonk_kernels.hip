@@ -121,11 +121,15 @@ __global__ __launch_bounds__(256) void k_copy_pad(const Fr* src, uint32_t len, F
 // (5k polys x n contiguous), blinders (2 per poly); out: 5k polys at
 // stride n+2 with wp[0]-=b0, wp[1]-=b1, wp[n]+=b0, wp[n+1]+=b1
 // (traits.rs:994 blinding shape; DRBG order preserved host-side).
+// Writes TWO copies: `out` (contiguous stride n+2, the commit staging) and
+// `coset_out` slot (p*7 + wire) at stride n+3 zero-padded — the R3 coset
+// batch reads wires/z/PI from that layout with no host round trip.
 __global__ __launch_bounds__(256) void k_blind_wires_batch(
-    const Fr* coeffs, const Fr* blinders, Fr* out, uint32_t n, uint64_t total) {
+    const Fr* coeffs, const Fr* blinders, Fr* out, Fr* coset_out, uint32_t n,
+    uint64_t total /* 5k*(n+3) */) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= total) return;
-    uint64_t stride = n + 2;
+    uint64_t stride = (uint64_t)n + 3;
     uint64_t g = i / stride;
     uint32_t j = (uint32_t)(i - g * stride);
     Fr v = (j < n) ? coeffs[g * n + j] : Fr::zero();
@@ -134,7 +138,46 @@ __global__ __launch_bounds__(256) void k_blind_wires_batch(
     if (j == 1) v = v.sub(b1);
     if (j == n) v = v.add(b0);
     if (j == n + 1) v = v.add(b1);
+    uint64_t p = g / 5;
+    coset_out[(p * 7 + (g - p * 5)) * stride + j] = v;
+    if (j < n + 2) out[g * (uint64_t)(n + 2) + j] = v;
+}
+
+// cohort R2: blind the k grand-product polynomials on device; same double
+// write (commit staging at stride n+3, coset slot p*7+5).  Blinder order
+// mirrors plonk_prove_impl: draws (b2,b3,b4) apply as zp[0]-=b4, zp[1]-=b3,
+// zp[2]-=b2, zp[n]+=b4, zp[n+1]+=b3, zp[n+2]+=b2.
+__global__ __launch_bounds__(256) void k_blind_z_batch(
+    const Fr* coeffs, const Fr* blinders /* 3 per proof: b2,b3,b4 */, Fr* out,
+    Fr* coset_out, uint32_t n, uint64_t total /* k*(n+3) */) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    uint64_t stride = (uint64_t)n + 3;
+    uint64_t p = i / stride;
+    uint32_t j = (uint32_t)(i - p * stride);
+    Fr v = (j < n) ? coeffs[p * n + j] : Fr::zero();
+    Fr b2 = blinders[3 * p], b3 = blinders[3 * p + 1], b4 = blinders[3 * p + 2];
+    if (j == 0) v = v.sub(b4);
+    if (j == 1) v = v.sub(b3);
+    if (j == 2) v = v.sub(b2);
+    if (j == n) v = v.add(b4);
+    if (j == n + 1) v = v.add(b3);
+    if (j == n + 2) v = v.add(b2);
     out[i] = v;
+    coset_out[(p * 7 + 5) * stride + j] = v;
+}
+
+// cohort R2: stage the k public-input polynomials (n coeffs each) into
+// their coset slots (p*7+6) zero-padded to stride n+3
+__global__ __launch_bounds__(256) void k_pi_to_coset(const Fr* coeffs,
+                                                     Fr* coset_out, uint32_t n,
+                                                     uint64_t total /* k*(n+3) */) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    uint64_t stride = (uint64_t)n + 3;
+    uint64_t p = i / stride;
+    uint32_t j = (uint32_t)(i - p * stride);
+    coset_out[(p * 7 + 6) * stride + j] = (j < n) ? coeffs[p * n + j] : Fr::zero();
 }
 
 // cohort R3: split each proof's quotient (m coeffs) into 5 chunks of

@@ -1156,14 +1156,15 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         HIP_CHECK(hipMemcpyAsync(cs->blinders, blind_h.data(),
                                  blind_h.size() * sizeof(Fr), hipMemcpyHostToDevice,
                                  RNG_STREAM));
-        uint64_t total1 = 5 * (uint64_t)k * (n + 2);
+        uint64_t total1 = 5 * (uint64_t)k * (n + 3);
         hipLaunchKernelGGL(k_blind_wires_batch,
                            dim3((uint32_t)((total1 + 255) / 256)), dim3(256), 0,
                            RNG_STREAM, cs->ntt_tmp, cs->blinders, cs->stage,
-                           (uint32_t)n, total1);
+                           cs->coset_in, (uint32_t)n, total1);
         HIP_CHECK(hipGetLastError());
-        std::vector<Fr> host_flat(total1);
-        HIP_CHECK(hipMemcpyAsync(host_flat.data(), cs->stage, total1 * sizeof(Fr),
+        uint64_t wlen = 5 * (uint64_t)k * (n + 2);
+        std::vector<Fr> host_flat(wlen);
+        HIP_CHECK(hipMemcpyAsync(host_flat.data(), cs->stage, wlen * sizeof(Fr),
                                  hipMemcpyDeviceToHost, RNG_STREAM));
         COHORT_TRACE("r1-sync-done");
         if (commit_staged(ctx, n + 2, 5 * k, cbuf.data(), ibuf.get()) != RNG_OK)
@@ -1231,53 +1232,42 @@ static int plonk_prove_cohort_impl(RngCtxImpl* ctx, const PlonkPkImpl& pk, uint3
         int rc = ntt_dev_run(ctx, cs->stage, cs->ntt_tmp, (uint32_t)n, 2 * (uint64_t)k,
                              true, RNG_STREAM);
         if (rc != RNG_OK) return rc;
-        HIP_CHECK(hipMemcpyAsync(evals_flat.data(), cs->ntt_tmp,
-                                 2 * (size_t)k * n * sizeof(Fr), hipMemcpyDeviceToHost,
-                                 RNG_STREAM));
-        HIP_CHECK(hipStreamSynchronize(RNG_STREAM));
-        // evals_flat now holds z coefficients (first k) and PI coefficients
-        // (second k); PI coefficients feed the R3 coset batch below
-        pool.parallel_for(k, [&](uint32_t p) {
-            auto& zp = zpoly[p];
-            const Fr* src = evals_flat.data() + (size_t)p * n;
-            zp.assign(src, src + n);
-            Fr b2 = drbg[p].next(), b3 = drbg[p].next(), b4 = drbg[p].next();
-            zp.resize(n + 3, Fr::zero());
-            zp[0] = zp[0].sub(b4);
-            zp[1] = zp[1].sub(b3);
-            zp[2] = zp[2].sub(b2);
-            zp[n] = zp[n].add(b4);
-            zp[n + 1] = zp[n + 1].add(b3);
-            zp[n + 2] = zp[n + 2].add(b2);
-        });
-        for (uint32_t p = 0; p < k; ++p) ps[p] = &zpoly[p];
-        if (commit_cohort(ctx, ps.data(), k, cbuf.data(), ibuf.get()) != RNG_OK)
+        // device-side z blinding + PI staging straight into the R3 coset
+        // slots (the wires landed there in R1); blinders in the single-proof
+        // DRBG order (b2, b3, b4)
+        std::vector<Fr> zblind_h(3 * (size_t)k);
+        for (uint32_t p = 0; p < k; ++p) {
+            zblind_h[3 * (size_t)p] = drbg[p].next();
+            zblind_h[3 * (size_t)p + 1] = drbg[p].next();
+            zblind_h[3 * (size_t)p + 2] = drbg[p].next();
+        }
+        HIP_CHECK(hipMemcpyAsync(cs->blinders, zblind_h.data(),
+                                 zblind_h.size() * sizeof(Fr),
+                                 hipMemcpyHostToDevice, RNG_STREAM));
+        uint64_t totz = (uint64_t)k * (n + 3);
+        hipLaunchKernelGGL(k_blind_z_batch, dim3((uint32_t)((totz + 255) / 256)),
+                           dim3(256), 0, RNG_STREAM, cs->ntt_tmp, cs->blinders,
+                           cs->stage, cs->coset_in, (uint32_t)n, totz);
+        HIP_CHECK(hipGetLastError());
+        hipLaunchKernelGGL(k_pi_to_coset, dim3((uint32_t)((totz + 255) / 256)),
+                           dim3(256), 0, RNG_STREAM,
+                           cs->ntt_tmp + (size_t)k * n, cs->coset_in, (uint32_t)n,
+                           totz);
+        HIP_CHECK(hipGetLastError());
+        std::vector<Fr> z_host(totz);
+        HIP_CHECK(hipMemcpyAsync(z_host.data(), cs->stage, totz * sizeof(Fr),
+                                 hipMemcpyDeviceToHost, RNG_STREAM));
+        if (commit_staged(ctx, n + 3, k, cbuf.data(), ibuf.get()) != RNG_OK)
             return RNG_ERR_HIP;
+        // commit_staged synchronized: z_host is complete
         pool.parallel_for(k, [&](uint32_t p) {
+            const Fr* src = z_host.data() + (size_t)p * (n + 3);
+            zpoly[p].assign(src, src + (n + 3));
             comms[13 * (size_t)p + 5] = cbuf[p];
             cinf[13 * (size_t)p + 5] = ibuf[p] ? 1 : 0;
             tr[p].append_g1(cbuf[p], ibuf[p]);
             alpha[p] = tr[p].challenge();
         });
-
-        // --- R3 staging: 7 polys per proof (5 wires, z, PI) at stride n+3 ---
-        const uint64_t stride = n + 3;
-        std::vector<Fr> coset_host(7 * (size_t)k * stride);
-        pool.parallel_for(k, [&](uint32_t p) {
-            Fr* base = coset_host.data() + 7 * (size_t)p * stride;
-            for (int j = 0; j < 5; ++j) {
-                memcpy(base + j * stride, wpoly[5 * (size_t)p + j].data(),
-                       (n + 2) * sizeof(Fr));
-                base[j * stride + n + 2] = Fr::zero();
-            }
-            memcpy(base + 5 * stride, zpoly[p].data(), (n + 3) * sizeof(Fr));
-            memcpy(base + 6 * stride, evals_flat.data() + ((size_t)k + p) * n,
-                   n * sizeof(Fr));
-            for (uint64_t j = n; j < stride; ++j) base[6 * stride + j] = Fr::zero();
-        });
-        HIP_CHECK(hipMemcpyAsync(cs->coset_in, coset_host.data(),
-                                 7 * (size_t)k * stride * sizeof(Fr),
-                                 hipMemcpyHostToDevice, RNG_STREAM));
     }
 
     COHORT_TRACE("r3");
