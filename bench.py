@@ -6,7 +6,7 @@ per GPU (torchrun for N>1; invoked directly with --gpus N>1 it self-launches
 torchrun), rank 0 prints ONE JSON line.
 
 Workload (BASELINE.json configs[3], the config the proofs/sec leg of the
-metric is quoted on): one step = one batch of `--jobs` (default 16)
+metric is quoted on): one step = one cohort of `--jobs` (default 32)
 concurrent `Intent And Balance Private Settlement` TurboPlonk proofs (the
 VALID MATCH MPC successor) on the GPU prover — synthetic
 fixed-seed witnesses, deterministic generated SRS, measured domain n reported
@@ -189,39 +189,66 @@ def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
         out[4 * (n_small + 2):-9] = 0
         out[-9:] = h[-9:]
 
-    def bundle(seed):
-        proofs = np.zeros((5, 157), dtype=np.uint64)
-        hs = np.zeros(4 * (int(ts["n"]) + 2) + 9, dtype=np.uint64)
-        assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk_s), ptr(ts["wires"]),
-                             ptr(ts["pubs"]), seed, ptr(proofs[0]), ptr(hs)) == 0
-        hs_e = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
-        ext(hs, int(ts["n"]), hs_e)
-        hv_e = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
-        lps = np.zeros((4, 18), dtype=np.uint64)
-        for i, ((off, align, cnt), pk, t) in enumerate(legs):
-            hv = np.zeros(4 * (int(t["n"]) + 2) + 9, dtype=np.uint64)
-            assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(t["wires"]),
-                                 ptr(t["pubs"]), seed + 1 + i, ptr(proofs[1 + i]),
-                                 ptr(hv)) == 0
-            ext(hv, int(t["n"]), hv_e)
-            assert lib.rng_link_proofs(ctx.h, ctypes.c_void_p(pk_big), ptr(hv_e),
-                                       ptr(hs_e), align, off, cnt, ptr(lps[i])) == 0
-        return proofs, lps
-
-    for i in range(args.warmup):
-        bundle(100 + i)
-    ctx.sync()
+    lib.rng_prove_cohort.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_uint64, U64P, U64P, U64P, U64P, U64P]
     from concurrent.futures import ThreadPoolExecutor
-    pool = ThreadPoolExecutor(max_workers=args.jobs)
-    list(pool.map(lambda j: bundle(500 + j), range(args.jobs)))
+    link_pool = ThreadPoolExecutor(max_workers=16)
+    kb = args.jobs  # bundles per step
+
+    def cohort(pk, tables, kk_each, seeds):
+        """Prove len(tables)*kk_each proofs under one PK: kk_each copies of
+        each table's witness, concatenated table-major."""
+        n = int(tables[0]["n"])
+        kk = kk_each * len(tables)
+        wires = np.concatenate([np.tile(t["wires"], kk_each) for t in tables])
+        pubs = np.concatenate([np.tile(t["pubs"], kk_each) for t in tables])
+        proofs = np.zeros(157 * kk, dtype=np.uint64)
+        hints = np.zeros((4 * (n + 2) + 9) * kk, dtype=np.uint64)
+        rc = lib.rng_prove_cohort(ctx.h, ctypes.c_void_p(pk), kk, ptr(wires),
+                                  ptr(pubs), ptr(seeds), ptr(proofs), ptr(hints))
+        assert rc == 0, f"rng_prove_cohort rc={rc}"
+        return proofs, hints.reshape(kk, -1)
+
+    def step(seed_base):
+        """One step = `jobs` production bundles: the same-position proofs of
+        all bundles prove as cohorts (settlement xk, ib-validity x2k covering
+        both parties, ob-validity x2k), then the 4k cross-domain link proofs
+        run on a thread pool — the batched shape of
+        native_proof_manager.rs:554-590."""
+        sd = lambda off: (np.uint64(seed_base + off * 100000) +
+                          np.arange(2 * kb, dtype=np.uint64))
+        _, sh = cohort(pk_s, [ts], kb, sd(0)[:kb])
+        _, vh = cohort(pk_v, [tv[0], tv[1]], kb, sd(1))
+        _, oh = cohort(pk_o, [to[0], to[1]], kb, sd(2))
+        # legs (offset-ascending): validity p0, ob p0, validity p1, ob p1;
+        # party p's hints are rows [p*kb, (p+1)*kb) of the 2k cohorts
+        leg_hints = [(vh, int(tv[0]["n"]), 0), (oh, int(to[0]["n"]), 0),
+                     (vh, int(tv[0]["n"]), 1), (oh, int(to[0]["n"]), 1)]
+
+        def links_for(b):
+            hs_e = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+            ext(sh[b], int(ts["n"]), hs_e)
+            lps = np.zeros((4, 18), dtype=np.uint64)
+            for i, ((off, align, cnt), _pk, _t) in enumerate(legs):
+                rows, n_leg, party = leg_hints[i]
+                hv_e = np.zeros(4 * (n_big + 2) + 9, dtype=np.uint64)
+                ext(rows[party * kb + b], n_leg, hv_e)
+                assert lib.rng_link_proofs(ctx.h, ctypes.c_void_p(pk_big),
+                                           ptr(hv_e), ptr(hs_e), align, off, cnt,
+                                           ptr(lps[i])) == 0
+            return lps
+
+        list(link_pool.map(links_for, range(kb)))
+
+    for i in range(max(1, min(args.warmup, 4))):
+        step(100 + i * 100_000)
+    ctx.sync()
     if dist:
         dist.barrier()
     ctx.sync()
     t0 = time.perf_counter()
-    futs = [pool.submit(bundle, 10_000 + rank * 100_000 + 10 * i)
-            for i in range(args.steps)]
-    for f in futs:
-        f.result()
+    for i in range(args.steps):
+        step(10_000_000 + rank * 100_000_000 + i * 100_000)
     ctx.sync()
     if dist:
         import torch
@@ -234,7 +261,7 @@ def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
     if rank == 0:
         out = {
             "metric": "settlement_bundles_per_s",
-            "value": round(args.steps * n_gpus / elapsed, 3),
+            "value": round(args.steps * args.jobs * n_gpus / elapsed, 3),
             "unit": "bundles/s",
             "n_gpus": n_gpus,
             "steps": args.steps,
@@ -249,14 +276,15 @@ def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
                 "workload": "production proof bundle: 1 settlement (n=4096) + 2 "
                             "intent-and-balance validity (n=16384) + 2 output-"
                             "balance validity (n=8192) + 4 cross-domain link "
-                            "proofs (native_proof_manager.rs:554-590)",
-                "jobs": args.jobs,
+                            "proofs (native_proof_manager.rs:554-590); one step "
+                            "= `jobs` bundles, same-position proofs cohorted",
+                "bundles_per_step": args.jobs,
             },
         }
         print(json.dumps(out), flush=True)
     # join worker threads BEFORE interpreter teardown so their TLS HIP
     # scratch frees while the runtime is still alive
-    pool.shutdown(wait=True)
+    link_pool.shutdown(wait=True)
     if dist:
         dist.destroy_process_group()
 
