@@ -240,6 +240,8 @@ def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
 
         list(link_pool.map(links_for, range(kb)))
 
+    step_pool = ThreadPoolExecutor(
+        max_workers=max(1, int(os.environ.get("RNG_BENCH_INFLIGHT", "2"))))
     for i in range(max(1, min(args.warmup, 4))):
         step(100 + i * 100_000)
     ctx.sync()
@@ -247,8 +249,10 @@ def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
         dist.barrier()
     ctx.sync()
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        step(10_000_000 + rank * 100_000_000 + i * 100_000)
+    futs = [step_pool.submit(step, 10_000_000 + rank * 100_000_000 + i * 100_000)
+            for i in range(args.steps)]
+    for f in futs:
+        f.result()
     ctx.sync()
     if dist:
         import torch
@@ -285,6 +289,7 @@ def bundle_mode(args, plib, orc, dist, rank, local_rank, n_gpus):
     # join worker threads BEFORE interpreter teardown so their TLS HIP
     # scratch frees while the runtime is still alive
     link_pool.shutdown(wait=True)
+    step_pool.shutdown(wait=True)
     if dist:
         dist.destroy_process_group()
 
