@@ -269,3 +269,67 @@ def test_real_circuit_gpu_parity(orc, builder, seed):
     assert lib.rng_verify(ctx.h, ctypes.c_void_p(pk), ptr(badp), ptr(proof)) != 0
     lib.rng_pk_free(ctypes.c_void_p(pk))
     ctx.close()
+
+
+def test_cohort_large_domain_bit_exact(orc):
+    """Cohort parity at a production validity domain (n=16384): the bundle
+    bench cohorts these tables, so pin them against the single-proof path
+    (the small-n cohort test exercises a different NTT/MSM shape)."""
+    from renegade_amd import load_prover
+    plib = load_prover()
+    if not plib.gpu_available:
+        pytest.skip("no GPU")
+    lib = plib.lib
+    lib.rng_circ_build_validity.restype = ctypes.c_void_p
+    lib.rng_circ_build_validity.argtypes = [ctypes.c_uint64] * 2
+    lib.rng_circ_n.restype = ctypes.c_uint64
+    lib.rng_circ_n.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_npub.restype = ctypes.c_uint64
+    lib.rng_circ_npub.argtypes = [ctypes.c_void_p]
+    lib.rng_circ_get.argtypes = [ctypes.c_void_p, U64P, U64P, U64P, U64P]
+    lib.rng_circ_free.argtypes = [ctypes.c_void_p]
+    lib.rng_preprocess.restype = ctypes.c_void_p
+    lib.rng_preprocess.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.rng_prove.argtypes = [ctypes.c_void_p, ctypes.c_void_p, U64P, U64P,
+                              ctypes.c_uint64, U64P, U64P]
+    lib.rng_prove_cohort.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_uint64, U64P, U64P, U64P, U64P, U64P]
+    h = lib.rng_circ_build_validity(11, 0)
+    assert h
+    n = lib.rng_circ_n(h)
+    assert int(n) >= 16384
+    npub = lib.rng_circ_npub(h)
+    sel = np.zeros(13 * n * 4, dtype=np.uint64)
+    sigma = np.zeros(5 * n, dtype=np.uint64)
+    wires = np.zeros(5 * n * 4, dtype=np.uint64)
+    pubs = np.zeros(max(1, npub * 4), dtype=np.uint64)
+    lib.rng_circ_get(h, ptr(sel), ptr(sigma), ptr(wires), ptr(pubs))
+    lib.rng_circ_free(h)
+    power = max(4, int(n).bit_length())
+    from tests.orc_bindings import OracleLib  # noqa: F811 — reuse orc fixture lib
+    ptau = orc.srs_generate_ptau(power, seed=42)
+    ctx = plib.init(ptau, (1 << power) + 2)
+
+    class Desc(ctypes.Structure):
+        _fields_ = [("n", ctypes.c_uint64), ("num_public", ctypes.c_uint64),
+                    ("selectors", U64P), ("sigma", U64P),
+                    ("num_link_groups", ctypes.c_uint64), ("link_offsets", U64P)]
+
+    pk = lib.rng_preprocess(ctx.h, ctypes.byref(Desc(n, npub, ptr(sel), ptr(sigma),
+                                                     0, None)))
+    assert pk
+    k = 3
+    seeds = np.array([71, 72, 71], dtype=np.uint64)
+    proofs = np.zeros(157 * k, dtype=np.uint64)
+    hints = np.zeros((4 * (int(n) + 2) + 9) * k, dtype=np.uint64)
+    rc = lib.rng_prove_cohort(ctx.h, ctypes.c_void_p(pk), k, ptr(np.tile(wires, k)),
+                              ptr(np.tile(pubs, k)), ptr(seeds), ptr(proofs),
+                              ptr(hints))
+    assert rc == 0
+    single = np.zeros(157, dtype=np.uint64)
+    for p in range(k):
+        assert lib.rng_prove(ctx.h, ctypes.c_void_p(pk), ptr(wires), ptr(pubs),
+                             ctypes.c_uint64(int(seeds[p])), ptr(single), None) == 0
+        assert np.array_equal(single, proofs[157 * p:157 * (p + 1)]), \
+            f"large-domain cohort proof {p} differs"
+    ctx.close()
