@@ -983,6 +983,13 @@ static thread_local std::unique_ptr<CohortScratch> tls_cohort_scratch;
 static int cohort_scratch_ensure(uint64_t n, uint64_t k) {
     if (tls_cohort_scratch && tls_cohort_scratch->n >= n && tls_cohort_scratch->k >= k)
         return RNG_OK;
+    // grow to the max of old/new in BOTH dimensions: a mixed-circuit stream
+    // (service batcher: n=4096 settlements between n=16384 validities)
+    // would otherwise thrash reallocations between shapes
+    if (tls_cohort_scratch) {
+        if (tls_cohort_scratch->n > n) n = tls_cohort_scratch->n;
+        if (tls_cohort_scratch->k > k) k = tls_cohort_scratch->k;
+    }
     uint64_t m = 8 * n;
     tls_cohort_scratch = std::make_unique<CohortScratch>();
     CohortScratch* s = tls_cohort_scratch.get();
