@@ -712,6 +712,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     et.mark(stream);
     uint32_t subb = 1;
     if (use_fold) {
+        if ((uint64_t)G > s->window_sums_cap) {
+            hipFree(s->window_sums);
+            s->window_sums_cap = (uint64_t)G * 2;
+            HIP_CHECK(hipMalloc(&s->window_sums, s->window_sums_cap * sizeof(G1Jac)));
+        }
         hipLaunchKernelGGL(k_msm_window_fold, dim3(G), dim3(cw), 0, stream,
                            s->buckets, c, chunk_sz, s->window_sums);
         HIP_CHECK(hipGetLastError());
@@ -732,6 +737,13 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
             return e ? atoi(e) : 0;
         }();
         if (subb_env >= 1 && subb_env <= 256) subb = (uint32_t)subb_env;
+        if ((uint64_t)G * subb > s->window_sums_cap) {
+            // big fused batches (cohort k=64: G=10240 groups) outgrow the
+            // default window_sums allocation
+            hipFree(s->window_sums);
+            s->window_sums_cap = (uint64_t)G * subb * 2;
+            HIP_CHECK(hipMalloc(&s->window_sums, s->window_sums_cap * sizeof(G1Jac)));
+        }
         static int msm_dbg = [] {
             const char* e = getenv("RNG_MSM_DEBUG");
             return e ? atoi(e) : 0;
