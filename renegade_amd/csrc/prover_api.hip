@@ -384,6 +384,7 @@ struct MsmScratch {
     G1Jac* buckets = nullptr;
     G1Jac* partials = nullptr;
     G1Jac* window_sums = nullptr;
+    uint64_t window_sums_cap = 0;  // entries (G*subb grows with fused B)
     G1Jac* result = nullptr;
     uint64_t* glv = nullptr;   // per-scalar (k1, k2) magnitudes + signs
     G1Aff* phi = nullptr;      // phi(bases) scratch for non-SRS base arrays
@@ -502,6 +503,7 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
         HIP_CHECK(hipMalloc(&s->buckets, cap_nb2 * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->partials, 2 * cap_nch * sizeof(G1Jac)));
         HIP_CHECK(hipMalloc(&s->window_sums, 512 * MSM_SUBB * sizeof(G1Jac)));
+        s->window_sums_cap = 512 * MSM_SUBB;
         HIP_CHECK(hipMalloc(&s->result, sizeof(G1Jac)));
         // glv: 32 B per scalar = 16*total/W bytes <= 2*total (W >= 8)
         HIP_CHECK(hipMalloc(&s->glv, 2 * cap_total));
