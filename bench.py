@@ -6,7 +6,7 @@ per GPU (torchrun for N>1; invoked directly with --gpus N>1 it self-launches
 torchrun), rank 0 prints ONE JSON line.
 
 Workload (BASELINE.json configs[3], the config the proofs/sec leg of the
-metric is quoted on): one step = one cohort of `--jobs` (default 32)
+metric is quoted on): one step = one cohort of `--jobs` (default 48)
 concurrent `Intent And Balance Private Settlement` TurboPlonk proofs (the
 VALID MATCH MPC successor) on the GPU prover — synthetic
 fixed-seed witnesses, deterministic generated SRS, measured domain n reported
@@ -383,7 +383,7 @@ def main():
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--jobs", type=int, default=32,
+    ap.add_argument("--jobs", type=int, default=48,
                     help="cohort size: proofs advanced in lockstep per step "
                          "(the reference proves from a rayon pool; measured "
                          "optimum 32 with 4 cohorts in flight)")
