@@ -386,7 +386,7 @@ def main():
     ap.add_argument("--jobs", type=int, default=48,
                     help="cohort size: proofs advanced in lockstep per step "
                          "(the reference proves from a rayon pool; measured "
-                         "optimum 32 with 4 cohorts in flight)")
+                         "optimum 48 with 4 cohorts in flight)")
     ap.add_argument("--no-kernel-legs", action="store_true",
                     help="skip the MSM/NTT kernel side-measurements")
     ap.add_argument("--mode", choices=["proofs", "msm-shard", "bundle"],
