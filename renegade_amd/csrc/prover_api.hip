@@ -4064,6 +4064,11 @@ int rng_prove_cohort(RngCtx* ctx, const RngProvingKey* pk, uint64_t k,
     if (!ctx || !pk || !wires || !seeds || !out_proofs || k == 0 || k > 128)
         return RNG_ERR_BAD_ARG;
     if (pk->impl.npub > 0 && !public_inputs) return RNG_ERR_BAD_ARG;
+    if (k == 1)  // same bytes by construction; the per-proof path has lower
+                 // latency (fewer phase syncs) when there is nothing to fuse
+        return plonk_prove_impl(&ctx->impl, pk->impl, (const Fr*)wires,
+                                (const Fr*)public_inputs, seeds[0], out_proofs,
+                                out_link_hints);
     return plonk_prove_cohort_impl(&ctx->impl, pk->impl, (uint32_t)k,
                                    (const Fr*)wires, (const Fr*)public_inputs, seeds,
                                    out_proofs, out_link_hints);
