@@ -72,27 +72,33 @@ __global__ __launch_bounds__(512) void k_ntt_small(Fr* data, const Fr* wst,
     }
 }
 
-// ---- pass 1: column-pair DFT of length N2 (stride N1) + outer twiddle ----
-// grid.x = N1/2 * batch; data viewed as [N2 rows][N1 cols].
+// ---- pass 1: column DFT of length N2 (stride N1) + outer twiddle ----
+// grid.x = N1/PW * batch; data viewed as [N2 rows][N1 cols].  PW = columns
+// per block: 2 makes every global access a 64-byte segment (2 adjacent Fr),
+// 1 halves the LDS footprint.  At 2^22 the pair form needs 128 KiB LDS =
+// ONE block per CU (2 waves/SIMD) and the dependent butterfly muls run
+// latency-exposed; singles fit 2 blocks/CU and trade half-sector global
+// access (~0.05 ms of traffic) for doubled occupancy.
+template <int PW>
 __global__ __launch_bounds__(512) void k_ntt_col(Fr* data, const Fr* wst2,
                                                  const Fr* ta, const Fr* tb,
                                                  uint32_t N1, uint32_t N2,
                                                  uint32_t logN2, uint32_t split_log) {
     extern __shared__ Fr lds[];
-    uint32_t wg = blockIdx.x % (N1 / 2);
-    Fr* base = data + (uint64_t)(blockIdx.x / (N1 / 2)) * N1 * N2;
+    uint32_t wg = blockIdx.x % (N1 / PW);
+    Fr* base = data + (uint64_t)(blockIdx.x / (N1 / PW)) * N1 * N2;
     uint32_t split_mask = (1u << split_log) - 1;
 
-    for (uint32_t q = threadIdx.x; q < 2 * N2; q += blockDim.x) {
-        uint32_t m = q >> 1, c = q & 1;
-        Fr v = base[(2 * wg + c) + (uint64_t)N1 * m];
+    for (uint32_t q = threadIdx.x; q < PW * N2; q += blockDim.x) {
+        uint32_t m = q / PW, c = q % PW;
+        Fr v = base[(PW * wg + c) + (uint64_t)N1 * m];
         lds[lds_slot(c * N2 + brev_n(m, logN2))] = v;
     }
     __syncthreads();
     for (uint32_t s = 1; s <= logN2; ++s) {
         uint32_t half = 1u << (s - 1);
-        for (uint32_t q = threadIdx.x; q < N2; q += blockDim.x) {
-            uint32_t c = q >> (logN2 - 1);          // N2/2 butterflies per column
+        for (uint32_t q = threadIdx.x; q < PW * (N2 / 2); q += blockDim.x) {
+            uint32_t c = q / (N2 / 2);
             uint32_t bf = q & (N2 / 2 - 1);
             uint32_t blk = bf >> (s - 1);
             uint32_t k = bf & (half - 1);
@@ -106,36 +112,37 @@ __global__ __launch_bounds__(512) void k_ntt_col(Fr* data, const Fr* wst2,
         }
         __syncthreads();
     }
-    for (uint32_t q = threadIdx.x; q < 2 * N2; q += blockDim.x) {
-        uint32_t k2 = q >> 1, c = q & 1;
-        uint32_t j1 = 2 * wg + c;
+    for (uint32_t q = threadIdx.x; q < PW * N2; q += blockDim.x) {
+        uint32_t k2 = q / PW, c = q % PW;
+        uint32_t j1 = PW * wg + c;
         Fr tw = ta[(uint64_t)j1 * (k2 >> split_log)].mul(tb[(uint64_t)j1 * (k2 & split_mask)]);
         base[j1 + (uint64_t)N1 * k2] = lds[lds_slot(c * N2 + k2)].mul(tw);
     }
 }
 
-// ---- pass 2: row-pair DFT of length N1 (contiguous) + strided store ----
-// grid.x = N2/2 * batch.
+// ---- pass 2: row DFT of length N1 (contiguous) + strided store ----
+// grid.x = N2/PW * batch.
+template <int PW>
 __global__ __launch_bounds__(512) void k_ntt_row(const Fr* in, Fr* out, const Fr* wst1,
                                                  uint32_t N1, uint32_t N2,
                                                  uint32_t logN1, Fr scale, int do_scale) {
     extern __shared__ Fr lds[];
-    uint32_t wg = blockIdx.x % (N2 / 2);
-    uint64_t boff = (uint64_t)(blockIdx.x / (N2 / 2)) * N1 * N2;
+    uint32_t wg = blockIdx.x % (N2 / PW);
+    uint64_t boff = (uint64_t)(blockIdx.x / (N2 / PW)) * N1 * N2;
     const Fr* ibase = in + boff;
     Fr* obase = out + boff;
 
-    for (uint32_t q = threadIdx.x; q < 2 * N1; q += blockDim.x) {
+    for (uint32_t q = threadIdx.x; q < PW * N1; q += blockDim.x) {
         uint32_t c = q >> logN1;
         uint32_t j1 = q & (N1 - 1);
-        Fr v = ibase[(uint64_t)(2 * wg) * N1 + q];
+        Fr v = ibase[(uint64_t)(PW * wg) * N1 + q];
         lds[lds_slot(c * N1 + brev_n(j1, logN1))] = v;
     }
     __syncthreads();
     for (uint32_t s = 1; s <= logN1; ++s) {
         uint32_t half = 1u << (s - 1);
-        for (uint32_t q = threadIdx.x; q < N1; q += blockDim.x) {
-            uint32_t c = q >> (logN1 - 1);
+        for (uint32_t q = threadIdx.x; q < PW * (N1 / 2); q += blockDim.x) {
+            uint32_t c = q / (N1 / 2);
             uint32_t bf = q & (N1 / 2 - 1);
             uint32_t blk = bf >> (s - 1);
             uint32_t k = bf & (half - 1);
@@ -149,12 +156,12 @@ __global__ __launch_bounds__(512) void k_ntt_row(const Fr* in, Fr* out, const Fr
         }
         __syncthreads();
     }
-    // element [c][k1] -> global position k2 + N2*k1, k2 = 2*wg + c
-    for (uint32_t q = threadIdx.x; q < 2 * N1; q += blockDim.x) {
-        uint32_t k1 = q >> 1, c = q & 1;
+    // element [c][k1] -> global position k2 + N2*k1, k2 = PW*wg + c
+    for (uint32_t q = threadIdx.x; q < PW * N1; q += blockDim.x) {
+        uint32_t k1 = q / PW, c = q % PW;
         Fr v = lds[lds_slot(c * N1 + k1)];
         if (do_scale) v = v.mul(scale);
-        obase[(2 * wg + c) + (uint64_t)N2 * k1] = v;
+        obase[(PW * wg + c) + (uint64_t)N2 * k1] = v;
     }
 }
 

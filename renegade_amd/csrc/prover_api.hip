@@ -336,17 +336,34 @@ static int ntt_dev_run(RngCtxImpl* ctx, Fr* data, Fr* out, uint32_t n, uint64_t 
     Fr* wst2 = inverse ? p->wst2_i : p->wst2_f;
     Fr* ta = inverse ? p->ta_i : p->ta_f;
     Fr* tb = inverse ? p->tb_i : p->tb_f;
-    uint32_t lds1 = 2 * p->N2 * sizeof(Fr);
-    uint32_t lds2 = 2 * p->N1 * sizeof(Fr);
+    // pair width per pass: 2 (64-byte-coalesced global access) while the
+    // pair fits 64 KiB LDS (>=2 blocks/CU); above that, singles — at 2^22
+    // the 128 KiB pair form pinned the DFT at 1 block/CU and the dependent
+    // butterfly muls ran latency-exposed
+    const uint32_t pw1 = (2u * p->N2 * sizeof(Fr) <= 65536) ? 2 : 1;
+    const uint32_t pw2 = (2u * p->N1 * sizeof(Fr) <= 65536) ? 2 : 1;
+    uint32_t lds1 = pw1 * p->N2 * sizeof(Fr);
+    uint32_t lds2 = pw2 * p->N1 * sizeof(Fr);
     EvtTimer et;
     et.mark(stream);
-    hipLaunchKernelGGL(k_ntt_col, dim3((uint32_t)(p->N1 / 2 * batch)), dim3(512), lds1,
-                       stream, data, wst2, ta, tb, p->N1, p->N2, p->logN2, p->split_log);
+    if (pw1 == 2)
+        hipLaunchKernelGGL(k_ntt_col<2>, dim3((uint32_t)(p->N1 / 2 * batch)), dim3(512),
+                           lds1, stream, data, wst2, ta, tb, p->N1, p->N2, p->logN2,
+                           p->split_log);
+    else
+        hipLaunchKernelGGL(k_ntt_col<1>, dim3((uint32_t)(p->N1 * batch)), dim3(512),
+                           lds1, stream, data, wst2, ta, tb, p->N1, p->N2, p->logN2,
+                           p->split_log);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
-    hipLaunchKernelGGL(k_ntt_row, dim3((uint32_t)(p->N2 / 2 * batch)), dim3(512), lds2,
-                       stream, data, out, wst1, p->N1, p->N2, p->logN1, p->ninv,
-                       inverse ? 1 : 0);
+    if (pw2 == 2)
+        hipLaunchKernelGGL(k_ntt_row<2>, dim3((uint32_t)(p->N2 / 2 * batch)), dim3(512),
+                           lds2, stream, data, out, wst1, p->N1, p->N2, p->logN1,
+                           p->ninv, inverse ? 1 : 0);
+    else
+        hipLaunchKernelGGL(k_ntt_row<1>, dim3((uint32_t)(p->N2 * batch)), dim3(512),
+                           lds2, stream, data, out, wst1, p->N1, p->N2, p->logN1,
+                           p->ninv, inverse ? 1 : 0);
     HIP_CHECK(hipGetLastError());
     et.mark(stream);
     et.collect(tls_ntt_times, 2);
