@@ -340,8 +340,12 @@ static int ntt_dev_run(RngCtxImpl* ctx, Fr* data, Fr* out, uint32_t n, uint64_t 
     // pair fits 64 KiB LDS (>=2 blocks/CU); above that, singles — at 2^22
     // the 128 KiB pair form pinned the DFT at 1 block/CU and the dependent
     // butterfly muls ran latency-exposed
-    const uint32_t pw1 = (2u * p->N2 * sizeof(Fr) <= 65536) ? 2 : 1;
-    const uint32_t pw2 = (2u * p->N1 * sizeof(Fr) <= 65536) ? 2 : 1;
+    static uint32_t pw_lds_cap = [] {
+        const char* e = getenv("RNG_NTT_PAIR_LDS");  // A/B knob, bytes
+        return e ? (uint32_t)atoi(e) : 65536u;
+    }();
+    const uint32_t pw1 = (2u * p->N2 * sizeof(Fr) <= pw_lds_cap) ? 2 : 1;
+    const uint32_t pw2 = (2u * p->N1 * sizeof(Fr) <= pw_lds_cap) ? 2 : 1;
     uint32_t lds1 = pw1 * p->N2 * sizeof(Fr);
     uint32_t lds2 = pw2 * p->N1 * sizeof(Fr);
     EvtTimer et;
