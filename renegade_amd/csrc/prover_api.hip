@@ -459,12 +459,11 @@ static int msm_dev_run(const G1Aff* d_bases, const uint64_t* d_scalars, uint64_t
     uint32_t chunk_sz = MSM_CHUNK;
     {
         uint64_t nbw = (1ull << (c - 1)) * G;
-        // target >=256k lanes: the per-chunk suffix walk is a serial EC
+        // target >=128k lanes: the per-chunk suffix walk is a serial EC
         // chain, so the chunk kernel is latency-bound until the chip is
         // several waves deep per SIMD (r02: chunk 16 -> 4 at 2^20 cut
-        // window_chunks ~3x; the deeper target also keeps cw >= 64 at the
-        // big cohort rounds so the scan combine applies)
-        while (chunk_sz > 2 && nbw / chunk_sz < 262144) chunk_sz >>= 1;
+        // window_chunks ~3x)
+        while (chunk_sz > 2 && nbw / chunk_sz < 131072) chunk_sz >>= 1;
         static int chunk_env = [] {
             const char* e = getenv("RNG_MSM_CHUNK");
             return e ? atoi(e) : 0;
